@@ -1,0 +1,106 @@
+"""Provider configuration.
+
+Counterpart of the reference's ``pkg/config/config.go`` (config.go:8-26) and the
+CLI flag surface (cmd/virtual_kubelet/main.go:59-73) — with the reference's dead
+knobs actually wired:
+
+- ``max_gpu_price`` is honored by the GPU selector (the reference defines the
+  flag at main.go:62 but always passes the constant ``DefaultMaxPrice`` at
+  runpod_client.go:1281).
+- ``reconcile_interval`` drives the real reconcile ticker (the reference only
+  feeds it to informer resync; its loops are hardcoded 10 s/30 s/5 min).
+- ``pending_pod_timeout`` replaces the hardcoded 15 min cutoff
+  (reference kubelet.go:788); the stuck-terminating ladder thresholds
+  (reference kubelet.go:1333/:1285/:1350) are configurable too.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import os
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import yaml
+
+
+@dataclass
+class Config:
+    # Node identity (reference main.go:59-73 flags)
+    node_name: str = "virtual-runpod"
+    operating_system: str = "Linux"
+    internal_ip: str = "127.0.0.1"
+    listen_port: int = 10250
+    namespace: str = "kube-system"
+
+    # K8s access
+    kubeconfig: str = ""
+
+    # Reconcile cadence. The event-driven runtime makes status sync push-based;
+    # these are the fallback/periodic cadences (reference: 30 s periodic loop
+    # kubelet.go:293, 10 s NotifyPods loop kubelet.go:719, 5 min cleanup
+    # kubelet.go:307, 30 s pending retry kubelet.go:735).
+    reconcile_interval_s: float = 30.0
+    notify_interval_s: float = 10.0
+    cleanup_interval_s: float = 300.0
+    pending_retry_interval_s: float = 5.0
+
+    # Failure ladders (reference kubelet.go:788, :1333, :1285, :1350)
+    pending_pod_timeout_s: float = 900.0
+    stuck_reterminate_after_s: float = 300.0
+    stuck_statuserr_force_after_s: float = 600.0
+    stuck_force_after_s: float = 900.0
+
+    # GPU policy. max_gpu_price is kept for Helm-values compatibility and
+    # remapped to a max fractional "cost" of a GPU (its current busy/HBM
+    # occupancy score in [0,1]); selector skips GPUs costlier than this.
+    max_gpu_price: float = 0.5
+    gpu_memory_default_gb: int = 16  # reference runpod_client.go:1189 default
+    datacenter_ids: List[str] = field(default_factory=list)
+
+    # Ops
+    health_server_address: str = ":8080"
+    log_level: str = "info"
+    heartbeat_interval_s: float = 300.0  # 0 disables (reference kubelet.go:73)
+    registration_endpoint: str = ""  # optional registration hook, off by default
+
+    # Local node backend (no cloud)
+    state_dir: str = "/var/lib/amd-virtual-kubelet"
+    sysfs_root: str = "/sys"
+    cgroup_root: str = "/sys/fs/cgroup"
+    cgroup_parent: str = "amdvk.slice"
+    runtime: str = "process"  # process | fake
+    pod_log_dir: str = ""  # defaults to <state_dir>/logs
+    pod_controller_workers: int = 4  # reference uses 1 (main.go:263)
+
+    # GPU inventory overrides (mostly for tests / CPU-only dev)
+    gpu_count_override: int = -1
+    gpu_vram_gb_override: int = -1
+
+    def resolved_pod_log_dir(self) -> str:
+        return self.pod_log_dir or os.path.join(self.state_dir, "logs")
+
+
+def load_config(path: Optional[str]) -> Config:
+    """Load YAML config overlaid on defaults (reference main.go:76-90).
+
+    Unknown keys are rejected loudly instead of silently ignored.
+    """
+    cfg = Config()
+    if not path:
+        return cfg
+    with open(path, "r", encoding="utf-8") as fh:
+        data = yaml.safe_load(fh) or {}
+    if not isinstance(data, dict):
+        raise ValueError(f"provider config {path!r}: expected a mapping")
+    names = {f.name for f in dataclasses.fields(Config)}
+    for key, value in data.items():
+        norm = key.replace("-", "_")
+        if norm not in names:
+            raise ValueError(f"provider config {path!r}: unknown key {key!r}")
+        setattr(cfg, norm, value)
+    if cfg.datacenter_ids is None:
+        cfg.datacenter_ids = []
+    if isinstance(cfg.datacenter_ids, str):
+        cfg.datacenter_ids = [s.strip() for s in cfg.datacenter_ids.split(",") if s.strip()]
+    return cfg

@@ -1,0 +1,93 @@
+"""In-tree build driver for the native components.
+
+Builds (next to this file, so the .so/binary travel with the repo snapshot):
+- ``_native.*.so``      — pybind11 extension (probe + launcher), plain C++.
+- ``podworker/podworker`` — HIP workload binary for gfx950 (hipcc
+  cross-compiles without a GPU; built with --offload-arch=gfx950).
+
+Run: ``python -m k8s_runpod_kubelet_amd.ops.build`` (or via
+``__graft_entry__.build()``).
+"""
+
+from __future__ import annotations
+
+import os
+import shlex
+import subprocess
+import sys
+import sysconfig
+from pathlib import Path
+
+OPS_DIR = Path(__file__).resolve().parent
+CSRC = OPS_DIR / "csrc"
+PODWORKER_DIR = OPS_DIR / "podworker"
+
+GFX_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+CXX = os.environ.get("CXX", "g++")
+
+
+def _run(cmd: list[str]) -> None:
+    print("+", " ".join(shlex.quote(c) for c in cmd), flush=True)
+    subprocess.run(cmd, check=True)
+
+
+def ext_suffix() -> str:
+    return sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+
+
+def native_ext_path() -> Path:
+    return OPS_DIR / f"_native{ext_suffix()}"
+
+
+def podworker_path() -> Path:
+    return PODWORKER_DIR / "podworker"
+
+
+def _newer_than(target: Path, sources: list[Path]) -> bool:
+    if not target.exists():
+        return False
+    t = target.stat().st_mtime
+    return all(t >= s.stat().st_mtime for s in sources)
+
+
+def build_native(force: bool = False) -> Path:
+    import pybind11
+
+    out = native_ext_path()
+    sources = [CSRC / "module.cpp", CSRC / "probe.cpp", CSRC / "launcher.cpp"]
+    headers = [CSRC / "probe.h", CSRC / "launcher.h"]
+    if not force and _newer_than(out, sources + headers):
+        return out
+    cmd = [
+        CXX, "-O2", "-g", "-std=c++17", "-shared", "-fPIC",
+        "-fvisibility=hidden",
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_paths()['include']}",
+        *[str(s) for s in sources],
+        "-o", str(out),
+    ]
+    _run(cmd)
+    return out
+
+
+def build_podworker(force: bool = False) -> Path:
+    out = podworker_path()
+    src = PODWORKER_DIR / "podworker.hip"
+    if not force and _newer_than(out, [src]):
+        return out
+    cmd = [
+        HIPCC, f"--offload-arch={GFX_ARCH}", "-O2", "-std=c++17",
+        str(src), "-o", str(out),
+    ]
+    _run(cmd)
+    return out
+
+
+def build_all(force: bool = False) -> None:
+    build_native(force=force)
+    build_podworker(force=force)
+
+
+if __name__ == "__main__":
+    build_all(force="--force" in sys.argv)

@@ -1,0 +1,85 @@
+#pragma once
+
+#include <cstdint>
+#include <map>
+#include <mutex>
+#include <string>
+#include <vector>
+
+namespace amdvk {
+
+struct LaunchSpec {
+  std::vector<std::string> argv;
+  std::vector<std::string> env;  // "KEY=VALUE" entries (full environment)
+  std::string cwd;
+  std::string stdout_path;
+  std::string stderr_path;
+  std::string cgroup_dir;  // if non-empty, child is placed here before exec
+  bool new_session = true;
+  bool ready_pipe = true;  // create readiness pipe, exported as AMDVK_READY_FD
+};
+
+struct LaunchResult {
+  int64_t pid = -1;
+  int pidfd = -1;
+  int ready_fd = -1;  // parent read end of the readiness pipe (-1 if disabled)
+  std::string error;
+};
+
+LaunchResult LaunchProcess(const LaunchSpec& spec);
+
+// pidfd for an already-running process (adoption-on-restart path).
+int OpenPidfd(int64_t pid);
+
+// Send a signal to a pid or (negative) to its process group.
+int SignalProcess(int64_t pid, int sig, bool whole_group);
+
+// cgroup v2 helpers (best-effort: return false without throwing when the
+// controller files are absent/unwritable — e.g. unprivileged test runs).
+bool CgroupCreate(const std::string& path, const std::string& cpu_max,
+                  const std::string& memory_max);
+bool CgroupRemove(const std::string& path);
+long CgroupProcCount(const std::string& path);
+
+struct Event {
+  enum Type { kExited = 0, kReady = 1, kReadyClosed = 2 };
+  Type type;
+  int64_t pid = -1;
+  uint64_t token = 0;
+  int exit_code = -1;    // kExited: exit status, or 128+signal
+  std::string data;      // kReady: bytes read from the readiness pipe
+};
+
+// epoll loop over pidfds + readiness pipes. One instance per runtime; a
+// Python watcher thread calls Poll() (GIL released while blocked) and turns
+// events into provider status updates — this is what replaces the
+// reference's 10 s/30 s cloud-status polling (kubelet.go:292-303, :713-731)
+// with sub-millisecond push.
+class EventLoop {
+ public:
+  EventLoop();
+  ~EventLoop();
+
+  // Register a process; both fds are owned by the loop afterwards.
+  void AddProcess(int64_t pid, int pidfd, int ready_fd, uint64_t token);
+  // Stop tracking (e.g. pod force-removed); closes owned fds.
+  void RemoveProcess(int64_t pid);
+  std::vector<Event> Poll(int timeout_ms);
+  void Wake();
+  size_t TrackedCount() const;
+
+ private:
+  struct Entry {
+    int64_t pid;
+    uint64_t token;
+    int pidfd;
+    int ready_fd;
+  };
+  int epfd_;
+  int wakefd_;
+  std::map<int64_t, Entry> procs_;      // pid -> entry
+  std::map<int, int64_t> fd_to_pid_;    // registered fd -> pid
+  mutable std::mutex mu_;
+};
+
+}  // namespace amdvk

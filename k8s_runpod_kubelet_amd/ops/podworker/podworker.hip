@@ -1,0 +1,176 @@
+// podworker — the synthetic GPU pod workload used by tests and bench.py.
+//
+// Stands in for the container image the reference's integration test deploys
+// (reference pkg/virtual_kubelet/runpod_test.go:99 runs a CUDA image with
+// `nvidia-smi -L`): it initializes a HIP context on its *bound* GPUs (the
+// binder scopes visibility via ROCR_VISIBLE_DEVICES), verifies the visible
+// device count matches the pod's amd.com/gpu request, runs a tiny gfx950
+// kernel to prove real GPU access, optionally binds TCP listen ports (for the
+// port-readiness gate), then signals readiness by writing "READY\n" to the
+// AMDVK_READY_FD pipe inherited from the native launcher, and finally holds
+// until SIGTERM (exit 0) or runs for a fixed duration.
+//
+// Build: hipcc --offload-arch=gfx950 -O2 podworker.hip -o podworker
+
+#include <hip/hip_runtime.h>
+
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <signal.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <sys/socket.h>
+#include <time.h>
+#include <unistd.h>
+
+#include <string>
+#include <vector>
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t err_ = (expr);                                               \
+    if (err_ != hipSuccess) {                                               \
+      fprintf(stderr, "podworker: %s failed: %s\n", #expr,                  \
+              hipGetErrorString(err_));                                     \
+      return 11;                                                            \
+    }                                                                       \
+  } while (0)
+
+__global__ void touch_kernel(unsigned int* out, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = static_cast<unsigned int>(i) * 2u + 1u;
+}
+
+static volatile sig_atomic_t g_terminate = 0;
+static void on_term(int) { g_terminate = 1; }
+
+// Touch every visible device: context init + kernel + verify.
+static int touch_gpus(int expect_gpus) {
+  int count = 0;
+  HIP_CHECK(hipGetDeviceCount(&count));
+  if (expect_gpus >= 0 && count != expect_gpus) {
+    fprintf(stderr, "podworker: visible GPU count %d != expected %d\n", count,
+            expect_gpus);
+    return 12;
+  }
+  const int n = 1 << 20;
+  for (int dev = 0; dev < count; ++dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    unsigned int* d = nullptr;
+    HIP_CHECK(hipMalloc(&d, n * sizeof(unsigned int)));
+    dim3 block(256);
+    dim3 grid((n + 255) / 256);
+    hipLaunchKernelGGL(touch_kernel, grid, block, 0, 0, d, n);
+    HIP_CHECK(hipGetLastError());
+    std::vector<unsigned int> host(n);
+    HIP_CHECK(hipMemcpy(host.data(), d, n * sizeof(unsigned int),
+                        hipMemcpyDeviceToHost));
+    HIP_CHECK(hipFree(d));
+    for (int i = 0; i < n; i += 65537) {
+      if (host[i] != static_cast<unsigned int>(i) * 2u + 1u) {
+        fprintf(stderr, "podworker: kernel verify failed at %d on dev %d\n", i,
+                dev);
+        return 13;
+      }
+    }
+    hipDeviceProp_t prop;
+    HIP_CHECK(hipGetDeviceProperties(&prop, dev));
+    printf("podworker: dev %d %s gcnArch=%s vram=%zuMB ok\n", dev, prop.name,
+           prop.gcnArchName, prop.totalGlobalMem >> 20);
+  }
+  return 0;
+}
+
+static int listen_on(int port) {
+  int fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd < 0) return -1;
+  int one = 1;
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  sockaddr_in addr{};
+  addr.sin_family = AF_INET;
+  addr.sin_addr.s_addr = htonl(INADDR_ANY);
+  addr.sin_port = htons(static_cast<uint16_t>(port));
+  if (bind(fd, reinterpret_cast<sockaddr*>(&addr), sizeof(addr)) != 0 ||
+      listen(fd, 8) != 0) {
+    close(fd);
+    return -1;
+  }
+  return fd;
+}
+
+int main(int argc, char** argv) {
+  int expect_gpus = -1;     // -1: skip GPU entirely (CPU pod)
+  double run_for = -1.0;    // seconds; -1 with hold=true means until SIGTERM
+  bool hold = false;
+  int exit_code = 0;
+  double startup_delay = 0.0;
+  std::vector<int> ports;
+
+  for (int i = 1; i < argc; ++i) {
+    std::string a = argv[i];
+    auto next = [&]() -> const char* { return (i + 1 < argc) ? argv[++i] : ""; };
+    if (a == "--expect-gpus") expect_gpus = atoi(next());
+    else if (a == "--run-for") run_for = atof(next());
+    else if (a == "--hold") hold = true;
+    else if (a == "--exit-code") exit_code = atoi(next());
+    else if (a == "--startup-delay") startup_delay = atof(next());
+    else if (a == "--listen-port") ports.push_back(atoi(next()));
+    else if (a == "--fail") {
+      fprintf(stderr, "podworker: simulated failure\n");
+      return 1;
+    } else {
+      fprintf(stderr, "podworker: unknown arg %s\n", a.c_str());
+      return 2;
+    }
+  }
+
+  struct sigaction sa{};
+  sa.sa_handler = on_term;
+  sigaction(SIGTERM, &sa, nullptr);
+  sigaction(SIGINT, &sa, nullptr);
+
+  if (startup_delay > 0) usleep(static_cast<useconds_t>(startup_delay * 1e6));
+
+  if (expect_gpus >= 0) {
+    int rc = touch_gpus(expect_gpus);
+    if (rc != 0) return rc;
+  }
+
+  std::vector<int> listen_fds;
+  for (int p : ports) {
+    int fd = listen_on(p);
+    if (fd < 0) {
+      fprintf(stderr, "podworker: listen on %d failed\n", p);
+      return 3;
+    }
+    listen_fds.push_back(fd);
+  }
+
+  // Readiness: the launcher's pipe (event-driven, replaces the reference's
+  // 10 s status poll as the readiness signal path).
+  const char* ready_env = getenv("AMDVK_READY_FD");
+  if (ready_env) {
+    int fd = atoi(ready_env);
+    const char msg[] = "READY\n";
+    ssize_t w = write(fd, msg, sizeof(msg) - 1);
+    (void)w;
+    close(fd);
+  }
+  printf("podworker: ready (gpus=%d ports=%zu)\n", expect_gpus, ports.size());
+  fflush(stdout);
+
+  if (hold && run_for < 0) {
+    while (!g_terminate) pause();
+  } else if (run_for > 0) {
+    double remaining = run_for;
+    while (remaining > 0 && !g_terminate) {
+      double chunk = remaining > 0.1 ? 0.1 : remaining;
+      usleep(static_cast<useconds_t>(chunk * 1e6));
+      remaining -= chunk;
+    }
+  }
+  for (int fd : listen_fds) close(fd);
+  printf("podworker: exiting code=%d\n", exit_code);
+  return exit_code;
+}
