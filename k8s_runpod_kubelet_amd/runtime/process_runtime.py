@@ -1,0 +1,519 @@
+"""ProcessRuntime — local host-process pod runtime with event-driven status.
+
+The MI355X-native data path that replaces the reference's RunPod cloud
+(reference pkg/virtual_kubelet/runpod_client.go): a pod's containers run as
+host processes launched by the native C++ launcher (ops/csrc/launcher.cpp)
+with:
+
+- GPU scoping via ROCR_VISIBLE_DEVICES/HIP_VISIBLE_DEVICES from the binder
+  (the server-side GPU attach analogue),
+- a per-pod cgroup v2 slice (cpu.max / memory.max) — best-effort when
+  unprivileged,
+- stdout/stderr captured to per-container log files (GetContainerLogs is real
+  here, unlike the reference's "not supported" stub, kubelet.go:2027-2066),
+- lifecycle *pushed* over pidfd+epoll: readiness from the AMDVK_READY_FD
+  pipe, exit from the pidfd — so the provider learns of state changes in
+  microseconds instead of on the next 10 s poll tick (kubelet.go:719).
+
+"Image" handling: there is no container-image store on the node, so the
+runtime executes the container's command/args directly (host binaries); the
+image string is recorded as metadata. Synthetic workloads (tests, bench) use
+the in-tree HIP `podworker` binary. A `command`-less container falls back to
+podworker in hold mode sized to the pod's GPU request — mirroring how the
+reference test pod runs a GPU-probe image (runpod_test.go:99).
+
+Instance state is journaled under <state_dir>/instances/<id>.json so a
+restarted kubelet re-adopts live pods (reference LoadRunning analogue,
+kubelet.go:1380-1535) without double-binding GPUs.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import os
+import secrets
+import shutil
+import threading
+import time
+from pathlib import Path
+from typing import Callable, Dict, List, Optional
+
+from ..gpu.binder import Binder, BindRequest, PlacementError, device_env
+from ..gpu.ledger import Ledger
+from .base import Runtime
+from .types import (
+    ContainerRuntimeInfo,
+    DeployParams,
+    DetailedStatus,
+    Instance,
+    PodStatus,
+)
+
+log = logging.getLogger("runtime.process")
+
+TERM_GRACE_S = 10.0
+
+
+class ProcessRuntime(Runtime):
+    def __init__(
+        self,
+        ledger: Ledger,
+        state_dir: str,
+        cgroup_root: str = "/sys/fs/cgroup",
+        cgroup_parent: str = "amdvk.slice",
+        podworker: Optional[str] = None,
+        enable_cgroups: bool = True,
+    ):
+        from ..ops import load_native
+
+        self._native = load_native()
+        self.ledger = ledger
+        self.binder = Binder(ledger)
+        self.state_dir = Path(state_dir)
+        self.instances_dir = self.state_dir / "instances"
+        self.logs_dir = self.state_dir / "logs"
+        self.instances_dir.mkdir(parents=True, exist_ok=True)
+        self.logs_dir.mkdir(parents=True, exist_ok=True)
+        self.cgroup_base = os.path.join(cgroup_root, cgroup_parent)
+        self.enable_cgroups = enable_cgroups
+        self._podworker = podworker
+
+        self._lock = threading.RLock()
+        self._instances: Dict[str, Instance] = {}
+        self._pid_to_instance: Dict[int, str] = {}
+        self._subscribers: List[Callable[[str], None]] = []
+        self._kill_timers: Dict[str, threading.Timer] = {}
+
+        self._loop = self._native.EventLoop()
+        self._stop = threading.Event()
+        self._watcher = threading.Thread(
+            target=self._watch_events, name="runtime-events", daemon=True
+        )
+        self._watcher.start()
+
+    # ------------- deploy -------------
+
+    def deploy(self, params: DeployParams) -> DetailedStatus:
+        instance_id = "amdvk-" + secrets.token_hex(6)
+        gpu_indices: List[int] = []
+        if params.gpu_count > 0:
+            req = BindRequest(
+                pod_key=params.pod_key,
+                gpu_count=params.gpu_count,
+                total_memory_bytes=params.gpu_memory_bytes,
+                max_cost=params.max_gpu_cost,
+            )
+            gpu_indices = self.binder.bind(req)  # raises PlacementError when full
+
+        inst = Instance(
+            id=instance_id,
+            pod_key=params.pod_key,
+            params=params,
+            gpu_indices=gpu_indices,
+            cost_per_hr=round(0.1 * max(1, params.gpu_count), 4) if params.gpu_count else 0.0,
+        )
+
+        cgroup_dir = ""
+        if self.enable_cgroups:
+            cgroup_dir = os.path.join(self.cgroup_base, instance_id)
+            if not self._native.cgroup_create(
+                cgroup_dir, params.cpu_limit, params.memory_limit
+            ):
+                log.debug("cgroup create failed (unprivileged?); continuing without",
+                          extra={"pod": params.pod_key})
+                cgroup_dir = ""
+        inst.cgroup_dir = cgroup_dir
+
+        try:
+            self._launch_containers(inst)
+        except Exception:
+            if gpu_indices:
+                self.binder.unbind(params.pod_key)
+            if cgroup_dir:
+                self._native.cgroup_remove(cgroup_dir)
+            raise
+
+        with self._lock:
+            self._instances[instance_id] = inst
+        self._persist(inst)
+        log.info(
+            "deployed instance",
+            extra={
+                "instance": instance_id,
+                "pod": params.pod_key,
+                "gpus": gpu_indices,
+                "containers": [c.name for c in inst.containers],
+            },
+        )
+        return self._status_of(inst)
+
+    def _launch_containers(self, inst: Instance) -> None:
+        params = inst.params
+        base_env = dict(os.environ)
+        # Drop our own GPU scoping so the pod's binding is authoritative.
+        base_env.pop("ROCR_VISIBLE_DEVICES", None)
+        base_env.pop("HIP_VISIBLE_DEVICES", None)
+        base_env.update(params.env)
+        base_env.update(device_env(inst.gpu_indices))
+        base_env["AMDVK_INSTANCE_ID"] = inst.id
+        base_env["AMDVK_POD_KEY"] = params.pod_key
+
+        for cspec in params.containers:
+            argv = list(cspec.command) + list(cspec.args)
+            if not argv:
+                argv = [self.podworker_path(), "--hold"]
+                if inst.gpu_indices:
+                    argv += ["--expect-gpus", str(len(inst.gpu_indices))]
+                for port in cspec.tcp_ports:
+                    argv += ["--listen-port", str(port)]
+            elif argv[0] in ("podworker", "amdvk-podworker"):
+                argv[0] = self.podworker_path()
+
+            env = dict(base_env)
+            env.update(cspec.env)
+            envp = [f"{k}={v}" for k, v in env.items()]
+            stdout_path = str(self.logs_dir / f"{inst.id}-{cspec.name}.log")
+
+            pid, pidfd, ready_fd = self._native.launch_process(
+                argv, envp,
+                cspec.working_dir or "",
+                stdout_path, stdout_path,
+                inst.cgroup_dir, True, True,
+            )
+            cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
+            inst.containers.append(cinfo)
+            with self._lock:
+                self._pid_to_instance[pid] = inst.id
+            self._loop.add_process(pid, pidfd, ready_fd, pid)
+        inst.desired_status = PodStatus.RUNNING
+
+    def podworker_path(self) -> str:
+        if self._podworker is None:
+            from ..ops import podworker_binary
+
+            self._podworker = podworker_binary()
+        return self._podworker
+
+    # ------------- events -------------
+
+    def _watch_events(self) -> None:
+        while not self._stop.is_set():
+            try:
+                events = self._loop.poll(500)
+            except Exception:
+                log.exception("event loop poll failed")
+                time.sleep(0.5)
+                continue
+            touched = set()
+            for ev in events:
+                inst_id = self._pid_to_instance.get(ev.pid)
+                if inst_id is None:
+                    continue
+                with self._lock:
+                    inst = self._instances.get(inst_id)
+                if inst is None:
+                    continue
+                cinfo = next((c for c in inst.containers if c.pid == ev.pid), None)
+                if cinfo is None:
+                    continue
+                if ev.type == "ready":
+                    cinfo.ready = True
+                    touched.add(inst_id)
+                elif ev.type == "ready_closed":
+                    # Workload never wrote READY: process start is readiness
+                    # (generic binaries without the pipe protocol). Only
+                    # meaningful if it is still running.
+                    if cinfo.exit_code is None and not cinfo.ready:
+                        cinfo.ready = True
+                        touched.add(inst_id)
+                elif ev.type == "exited":
+                    cinfo.exit_code = ev.exit_code
+                    cinfo.finished_at = time.time()
+                    cinfo.ready = False
+                    if ev.exit_code:
+                        cinfo.message = f"exit code {ev.exit_code}"
+                    self._on_container_exit(inst)
+                    touched.add(inst_id)
+            for inst_id in touched:
+                with self._lock:
+                    inst = self._instances.get(inst_id)
+                if inst is not None:
+                    self._persist(inst)
+                self._notify(inst_id)
+
+    def _on_container_exit(self, inst: Instance) -> None:
+        if any(c.exit_code is None for c in inst.containers):
+            return
+        terminating = inst.desired_status == PodStatus.TERMINATING
+        inst.desired_status = PodStatus.TERMINATED if terminating else PodStatus.EXITED
+        timer = self._kill_timers.pop(inst.id, None)
+        if timer:
+            timer.cancel()
+        # Release GPUs as soon as the workload is gone — HBM headroom returns
+        # to the ledger without waiting for pod deletion.
+        self.binder.unbind(inst.pod_key)
+        if inst.cgroup_dir:
+            self._native.cgroup_remove(inst.cgroup_dir)
+
+    def _notify(self, instance_id: str) -> None:
+        for cb in list(self._subscribers):
+            try:
+                cb(instance_id)
+            except Exception:
+                log.exception("runtime subscriber failed")
+
+    def subscribe(self, callback: Callable[[str], None]) -> None:
+        self._subscribers.append(callback)
+
+    # ------------- status -------------
+
+    def _status_of(self, inst: Instance) -> DetailedStatus:
+        ports: Dict[int, int] = {}
+        if inst.desired_status == PodStatus.RUNNING:
+            want = []
+            for c in inst.params.containers:
+                want.extend(c.tcp_ports)
+            if want:
+                listening = _listening_tcp_ports()
+                ports = {p: p for p in want if p in listening}
+        return DetailedStatus(
+            id=inst.id,
+            desired_status=inst.desired_status,
+            port_mappings=ports,
+            containers=[ContainerRuntimeInfo(**vars(c)) for c in inst.containers],
+            gpu_indices=list(inst.gpu_indices),
+            cost_per_hr=inst.cost_per_hr,
+            last_error=inst.last_error,
+            created_at=inst.created_at,
+        )
+
+    def get_detailed_status(self, instance_id: str) -> DetailedStatus:
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        if inst is None:
+            return DetailedStatus(id=instance_id, desired_status=PodStatus.NOT_FOUND)
+        return self._status_of(inst)
+
+    def list_instances(self, statuses: Optional[List[str]] = None) -> List[DetailedStatus]:
+        with self._lock:
+            insts = list(self._instances.values())
+        out = [self._status_of(i) for i in insts]
+        if statuses:
+            out = [s for s in out if s.desired_status in statuses]
+        return out
+
+    def instance_for_pod(self, pod_key: str) -> Optional[str]:
+        with self._lock:
+            for inst in self._instances.values():
+                if inst.pod_key == pod_key and inst.desired_status not in (
+                    PodStatus.TERMINATED,
+                ):
+                    return inst.id
+        return None
+
+    # ------------- terminate / GC -------------
+
+    def terminate(self, instance_id: str) -> None:
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        if inst is None:
+            return
+        if inst.desired_status in (PodStatus.EXITED, PodStatus.TERMINATED):
+            inst.desired_status = PodStatus.TERMINATED
+            self._persist(inst)
+            return
+        inst.desired_status = PodStatus.TERMINATING
+        self._persist(inst)
+        self._signal_all(inst, 15)  # SIGTERM
+        timer = threading.Timer(TERM_GRACE_S, self._force_kill, args=(instance_id,))
+        timer.daemon = True
+        self._kill_timers[instance_id] = timer
+        timer.start()
+
+    def _force_kill(self, instance_id: str) -> None:
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        if inst is None:
+            return
+        if any(c.exit_code is None for c in inst.containers):
+            log.warning("grace period expired; SIGKILL", extra={"instance": instance_id})
+            self._signal_all(inst, 9)
+
+    def _signal_all(self, inst: Instance, sig: int) -> None:
+        for c in inst.containers:
+            if c.exit_code is None and c.pid > 0:
+                # Whole process group: the container is a session leader.
+                rc = self._native.signal_process(c.pid, sig, True)
+                if rc != 0:
+                    self._native.signal_process(c.pid, sig, False)
+
+    def remove(self, instance_id: str) -> None:
+        with self._lock:
+            inst = self._instances.pop(instance_id, None)
+            if inst is not None:
+                for c in inst.containers:
+                    self._pid_to_instance.pop(c.pid, None)
+        if inst is not None:
+            self.binder.unbind(inst.pod_key)
+            (self.instances_dir / f"{instance_id}.json").unlink(missing_ok=True)
+
+    def healthy(self) -> bool:
+        return self.ledger.any_schedulable() or self.ledger.total_gpus() == 0
+
+    # ------------- logs -------------
+
+    def get_logs(self, instance_id: str, container: str = "", tail: int = -1) -> str:
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        if inst is None:
+            return ""
+        names = [c.name for c in inst.params.containers]
+        if container and container in names:
+            name = container
+        else:
+            name = names[0] if names else ""
+        path = self.logs_dir / f"{instance_id}-{name}.log"
+        if not path.exists():
+            return ""
+        text = path.read_text(errors="replace")
+        if tail > 0:
+            text = "\n".join(text.splitlines()[-tail:]) + "\n"
+        return text
+
+    # ------------- persistence / adoption -------------
+
+    def _persist(self, inst: Instance) -> None:
+        record = {
+            "id": inst.id,
+            "pod_key": inst.pod_key,
+            "gpu_indices": inst.gpu_indices,
+            "desired_status": inst.desired_status,
+            "cgroup_dir": inst.cgroup_dir,
+            "created_at": inst.created_at,
+            "cost_per_hr": inst.cost_per_hr,
+            "gpu_memory_bytes": inst.params.gpu_memory_bytes,
+            "gpu_count": inst.params.gpu_count,
+            "namespace": inst.params.namespace,
+            "name": inst.params.name,
+            "containers": [
+                {
+                    "name": c.name,
+                    "pid": c.pid,
+                    "started_at": c.started_at,
+                    "finished_at": c.finished_at,
+                    "exit_code": c.exit_code,
+                    "ready": c.ready,
+                }
+                for c in inst.containers
+            ],
+            "container_specs": [
+                {
+                    "name": c.name,
+                    "image": c.image,
+                    "command": c.command,
+                    "args": c.args,
+                    "tcp_ports": c.tcp_ports,
+                }
+                for c in inst.params.containers
+            ],
+        }
+        tmp = self.instances_dir / f".{inst.id}.tmp"
+        tmp.write_text(json.dumps(record))
+        tmp.rename(self.instances_dir / f"{inst.id}.json")
+
+    def adopt_persisted(self) -> List[str]:
+        """Rebuild instance state after a kubelet restart: re-open pidfds for
+        still-live processes, mark vanished ones EXITED, re-reserve GPUs.
+        Returns adopted instance ids (reference LoadRunning analogue)."""
+        adopted = []
+        for path in sorted(self.instances_dir.glob("*.json")):
+            try:
+                rec = json.loads(path.read_text())
+            except (OSError, json.JSONDecodeError):
+                continue
+            from .types import ContainerSpec
+
+            params = DeployParams(
+                pod_key=rec["pod_key"],
+                name=rec.get("name", rec["pod_key"]),
+                namespace=rec.get("namespace", "default"),
+                gpu_count=rec.get("gpu_count", 0),
+                gpu_memory_bytes=rec.get("gpu_memory_bytes", 0),
+                containers=[
+                    ContainerSpec(
+                        name=c["name"], image=c.get("image", ""),
+                        command=c.get("command", []), args=c.get("args", []),
+                        tcp_ports=c.get("tcp_ports", []),
+                    )
+                    for c in rec.get("container_specs", [])
+                ],
+            )
+            inst = Instance(
+                id=rec["id"], pod_key=rec["pod_key"], params=params,
+                gpu_indices=rec.get("gpu_indices", []),
+                desired_status=rec.get("desired_status", PodStatus.RUNNING),
+                cgroup_dir=rec.get("cgroup_dir", ""),
+                created_at=rec.get("created_at", time.time()),
+                cost_per_hr=rec.get("cost_per_hr", 0.0),
+            )
+            all_alive = True
+            for c in rec.get("containers", []):
+                cinfo = ContainerRuntimeInfo(
+                    name=c["name"], pid=c["pid"], started_at=c.get("started_at", 0.0),
+                    finished_at=c.get("finished_at", 0.0),
+                    exit_code=c.get("exit_code"), ready=c.get("ready", False),
+                )
+                if cinfo.exit_code is None:
+                    pidfd = self._native.open_pidfd(cinfo.pid)
+                    if pidfd >= 0:
+                        self._loop.add_process(cinfo.pid, pidfd, -1, cinfo.pid)
+                        with self._lock:
+                            self._pid_to_instance[cinfo.pid] = inst.id
+                    else:
+                        # Process died while we were away; exact code unknown.
+                        cinfo.exit_code = -1
+                        cinfo.finished_at = time.time()
+                        cinfo.message = "process vanished during kubelet restart"
+                        all_alive = False
+                inst.containers.append(cinfo)
+            if inst.containers and all(c.exit_code is not None for c in inst.containers):
+                if inst.desired_status not in (PodStatus.TERMINATED,):
+                    inst.desired_status = (
+                        PodStatus.EXITED if all_alive else PodStatus.EXITED
+                    )
+            elif inst.desired_status == PodStatus.RUNNING and inst.gpu_indices:
+                self.ledger.adopt(
+                    inst.pod_key, inst.gpu_indices,
+                    BindRequest(inst.pod_key, len(inst.gpu_indices),
+                                params.gpu_memory_bytes).bytes_per_gpu,
+                )
+            with self._lock:
+                self._instances[inst.id] = inst
+            adopted.append(inst.id)
+        if adopted:
+            log.info("adopted persisted instances", extra={"count": len(adopted)})
+        return adopted
+
+    def close(self) -> None:
+        self._stop.set()
+        self._loop.wake()
+        self._watcher.join(timeout=2.0)
+        for timer in self._kill_timers.values():
+            timer.cancel()
+
+
+def _listening_tcp_ports() -> set:
+    """LISTEN-state local TCP ports from /proc/net/tcp{,6} (state 0A)."""
+    ports = set()
+    for path in ("/proc/net/tcp", "/proc/net/tcp6"):
+        try:
+            with open(path, "r", encoding="ascii") as fh:
+                next(fh, None)
+                for line in fh:
+                    parts = line.split()
+                    if len(parts) > 3 and parts[3] == "0A":
+                        ports.add(int(parts[1].rsplit(":", 1)[1], 16))
+        except OSError:
+            continue
+    return ports

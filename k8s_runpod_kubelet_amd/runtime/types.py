@@ -1,0 +1,129 @@
+"""Backend wire types — local equivalents of the reference's client types
+(reference pkg/virtual_kubelet/runpod_client.go:55-140: PodStatus enum,
+RunPodInstance, InstanceInfo, DetailedStatus/RuntimeInfo/MachineInfo).
+
+The status vocabulary is kept verbatim so the provider's status translation
+(provider/status.py) matches the reference's semantics
+(kubelet.go:1848-2024) state for state.
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+
+class PodStatus:
+    RUNNING = "RUNNING"
+    STARTING = "STARTING"
+    TERMINATING = "TERMINATING"
+    TERMINATED = "TERMINATED"
+    NOT_FOUND = "NOT_FOUND"
+    EXITED = "EXITED"
+
+
+@dataclass
+class ContainerRuntimeInfo:
+    """Per-container runtime record (reference RuntimeInfo is single-container;
+    we track all containers — SURVEY §7.4 parity-plus)."""
+    name: str
+    pid: int = -1
+    started_at: float = 0.0
+    finished_at: float = 0.0
+    exit_code: Optional[int] = None
+    ready: bool = False
+    message: str = ""
+
+
+@dataclass
+class DetailedStatus:
+    """Reference DetailedStatus analogue (runpod_client.go:105-140)."""
+    id: str
+    desired_status: str = PodStatus.NOT_FOUND
+    # "exposed" TCP ports actually listening — the portMappings analogue
+    # (machine.portMappings in the reference, kubelet.go:566-605 gate).
+    port_mappings: Dict[int, int] = field(default_factory=dict)
+    containers: List[ContainerRuntimeInfo] = field(default_factory=list)
+    gpu_indices: List[int] = field(default_factory=list)
+    cost_per_hr: float = 0.0
+    last_error: str = ""
+    created_at: float = 0.0
+
+    @property
+    def exit_code(self) -> Optional[int]:
+        """Aggregate exit code: first nonzero, else 0 once all finished."""
+        codes = [c.exit_code for c in self.containers]
+        if any(c is None for c in codes) or not codes:
+            return None
+        for code in codes:
+            if code:
+                return code
+        return 0
+
+    @property
+    def completion_message(self) -> str:
+        for c in self.containers:
+            if c.exit_code:
+                return c.message or f"container {c.name} exited with code {c.exit_code}"
+        return "completed"
+
+
+def is_successful_completion(status: DetailedStatus) -> bool:
+    """Reference IsSuccessfulCompletion (runpod_client.go:820-843): no runtime
+    info => not successful; exit code 0 => success; else message sniffing."""
+    code = status.exit_code
+    if code is None:
+        return False
+    if code == 0:
+        return True
+    msg = status.completion_message.lower()
+    return "success" in msg or "completed" in msg
+
+
+@dataclass
+class ContainerSpec:
+    name: str
+    image: str = ""
+    command: List[str] = field(default_factory=list)
+    args: List[str] = field(default_factory=list)
+    env: Dict[str, str] = field(default_factory=dict)
+    working_dir: str = ""
+    tcp_ports: List[int] = field(default_factory=list)
+
+
+@dataclass
+class DeployParams:
+    """Local-deploy parameter set — the assembled output of spec translation
+    (reference PrepareRunPodParameters, runpod_client.go:1248-1377)."""
+    pod_key: str  # "<namespace>-<name>" (reference map key convention)
+    name: str
+    namespace: str = "default"
+    containers: List[ContainerSpec] = field(default_factory=list)
+    env: Dict[str, str] = field(default_factory=dict)  # pod-level (merged into all)
+    gpu_count: int = 0
+    gpu_memory_bytes: int = 0  # total across the GPU set
+    max_gpu_cost: float = 0.5
+    requested_ports: List[str] = field(default_factory=list)  # "8080/tcp" style
+    cloud_type: str = "SECURE"
+    datacenter_ids: List[str] = field(default_factory=list)
+    template_id: str = ""
+    registry_auth_id: str = ""
+    cpu_limit: str = ""      # cgroup cpu.max, e.g. "200000 100000"
+    memory_limit: str = ""   # cgroup memory.max bytes or "max"
+    labels: Dict[str, str] = field(default_factory=dict)
+
+
+@dataclass
+class Instance:
+    """A deployed local instance (reference RunPodInstance analogue)."""
+    id: str
+    pod_key: str
+    params: DeployParams
+    gpu_indices: List[int] = field(default_factory=list)
+    desired_status: str = PodStatus.STARTING
+    containers: List[ContainerRuntimeInfo] = field(default_factory=list)
+    cgroup_dir: str = ""
+    created_at: float = field(default_factory=time.time)
+    cost_per_hr: float = 0.0
+    last_error: str = ""
