@@ -1,0 +1,103 @@
+"""Kubelet API server (:10250).
+
+Counterpart of the reference's createAPIServer + api.AttachPodRoutes
+(cmd/virtual_kubelet/main.go:217-248). The reference stubs RunInContainer and
+GetContainerLogs with "not supported by RunPod" (kubelet.go:2027-2066);
+containers here are local processes, so ``/containerLogs`` serves the real
+log files (parity-plus per SURVEY §7 non-goals note) and ``/pods`` serves the
+provider's tracked pods. Exec is non-interactive only (one-shot command in
+the pod's environment) — interactive TTY streaming is not implemented.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import threading
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from typing import Optional
+from urllib.parse import parse_qs, unquote, urlparse
+
+log = logging.getLogger("kube.apiserver")
+
+
+class KubeletApiServer:
+    def __init__(self, provider, internal_ip: str = "127.0.0.1", port: int = 10250):
+        self.provider = provider
+        self.internal_ip = internal_ip
+        self.port = port
+        self._server: Optional[ThreadingHTTPServer] = None
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self) -> None:
+        outer = self
+
+        class Handler(BaseHTTPRequestHandler):
+            def log_message(self, fmt, *args):
+                log.debug(fmt % args)
+
+            def _respond(self, code: int, body: bytes, ctype: str = "text/plain"):
+                self.send_response(code)
+                self.send_header("Content-Type", ctype)
+                self.send_header("Content-Length", str(len(body)))
+                self.end_headers()
+                self.wfile.write(body)
+
+            def do_GET(self):
+                parsed = urlparse(self.path)
+                parts = [unquote(p) for p in parsed.path.strip("/").split("/") if p]
+                query = parse_qs(parsed.query)
+                if parts and parts[0] == "containerLogs" and len(parts) >= 3:
+                    namespace, pod = parts[1], parts[2]
+                    container = parts[3] if len(parts) > 3 else ""
+                    tail = int(query.get("tailLines", ["-1"])[0])
+                    text = outer.provider.get_container_logs(
+                        namespace, pod, container, tail
+                    )
+                    self._respond(200, text.encode())
+                elif parts and parts[0] == "pods":
+                    pods = outer.provider.get_pods()
+                    body = json.dumps(
+                        {"kind": "PodList", "apiVersion": "v1", "items": pods}
+                    ).encode()
+                    self._respond(200, body, "application/json")
+                elif parts and parts[0] == "stats" and len(parts) > 1 and parts[1] == "summary":
+                    self._respond(200, json.dumps({"node": {}, "pods": []}).encode(),
+                                  "application/json")
+                elif parts and parts[0] == "healthz":
+                    self._respond(200, b"ok")
+                else:
+                    self._respond(404, b"not found")
+
+            def do_POST(self):
+                parsed = urlparse(self.path)
+                parts = [unquote(p) for p in parsed.path.strip("/").split("/") if p]
+                query = parse_qs(parsed.query)
+                if parts and parts[0] in ("exec", "run") and len(parts) >= 3:
+                    commands = query.get("command", [])
+                    if not commands:
+                        self._respond(400, b"missing command")
+                        return
+                    self._respond(
+                        501,
+                        b"interactive exec streaming is not supported; "
+                        b"use containerLogs for output",
+                    )
+                else:
+                    self._respond(404, b"not found")
+
+        self._server = ThreadingHTTPServer((self.internal_ip, self.port), Handler)
+        self.port = self._server.server_address[1]
+        self._thread = threading.Thread(
+            target=self._server.serve_forever, name="kubelet-api", daemon=True
+        )
+        self._thread.start()
+        log.info("kubelet API listening",
+                 extra={"addr": f"{self.internal_ip}:{self.port}"})
+
+    def stop(self) -> None:
+        if self._server is not None:
+            self._server.shutdown()
+            self._server.server_close()
+        if self._thread is not None:
+            self._thread.join(timeout=5.0)

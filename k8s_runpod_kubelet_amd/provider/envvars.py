@@ -1,0 +1,149 @@
+"""Env-var extraction from pod specs.
+
+Counterpart of the reference's ExtractEnvVars pipeline
+(reference pkg/virtual_kubelet/runpod_client.go:866-1054):
+- literal container env,
+- ``valueFrom.secretKeyRef`` single keys,
+- ``envFrom.secretRef`` whole-secret imports,
+- secret *volumes* flattened into env vars (items→path/key names,
+  runpod_client.go:949-979),
+- newline escaping ``\\n`` → ``\\\\n`` (runpod_client.go:995, :1016),
+- Kubernetes auto-injected service vars filtered out by substring patterns
+  (runpod_client.go:886-904; stated rationale at :1022 is attack-surface
+  reduction).
+
+Parity-plus: the reference reads only ``Containers[0]`` (:1028-1030); this
+implementation extracts pod-level env from *all* containers (first container
+wins on conflicts, preserving reference behavior for single-container pods),
+and also returns per-container maps for the process runtime.
+"""
+
+from __future__ import annotations
+
+import base64
+import logging
+from typing import Any, Dict, List, Tuple
+
+from ..kube.client import K8sClient, is_not_found
+
+log = logging.getLogger("provider.envvars")
+
+# Substring patterns of auto-injected k8s env vars (runpod_client.go:886-904).
+_AUTO_PATTERNS = ("KUBERNETES_", "_PORT_", "_TCP_", "_SERVICE_PORT_", "_SERVICE_HOST")
+
+
+def is_k8s_auto_injected(name: str) -> bool:
+    return any(p in name for p in _AUTO_PATTERNS)
+
+
+def _escape(value: str) -> str:
+    return value.replace("\n", "\\n")
+
+
+def _decode_secret_data(secret: Dict[str, Any]) -> Dict[str, str]:
+    out: Dict[str, str] = {}
+    for key, value in (secret.get("data") or {}).items():
+        try:
+            out[key] = base64.b64decode(value).decode("utf-8", errors="replace")
+        except Exception:
+            out[key] = str(value)
+    for key, value in (secret.get("stringData") or {}).items():
+        out[key] = str(value)
+    return out
+
+
+class SecretCollector:
+    """Batches secret fetches (reference secretCollector, :906-947)."""
+
+    def __init__(self, client: K8sClient, namespace: str):
+        self.client = client
+        self.namespace = namespace
+        self._cache: Dict[str, Dict[str, str]] = {}
+
+    def get(self, name: str) -> Dict[str, str]:
+        if name not in self._cache:
+            try:
+                secret = self.client.get_secret(self.namespace, name)
+                self._cache[name] = _decode_secret_data(secret)
+            except Exception as exc:
+                if is_not_found(exc):
+                    log.warning("secret not found", extra={"secret": name, "ns": self.namespace})
+                    self._cache[name] = {}
+                else:
+                    raise
+        return self._cache[name]
+
+
+def _container_env(container: Dict[str, Any], collector: SecretCollector) -> Dict[str, str]:
+    env: Dict[str, str] = {}
+    # envFrom secretRef: whole-secret import (runpod_client.go:981-1000)
+    for ef in container.get("envFrom", []) or []:
+        ref = ef.get("secretRef")
+        if ref and ref.get("name"):
+            for key, value in collector.get(ref["name"]).items():
+                env[key] = _escape(value)
+    # explicit env entries (literal + secretKeyRef), later wins
+    for item in container.get("env", []) or []:
+        name = item.get("name", "")
+        if not name:
+            continue
+        if "value" in item:
+            env[name] = _escape(str(item["value"]))
+        else:
+            ref = (item.get("valueFrom") or {}).get("secretKeyRef")
+            if ref and ref.get("name"):
+                data = collector.get(ref["name"])
+                key = ref.get("key", "")
+                if key in data:
+                    env[name] = _escape(data[key])
+    return {k: v for k, v in env.items() if not is_k8s_auto_injected(k)}
+
+
+def _volume_secret_env(pod: Dict[str, Any], collector: SecretCollector) -> Dict[str, str]:
+    """Secret volumes flattened to env vars (runpod_client.go:949-979): each
+    item's path (or key) becomes the var name."""
+    env: Dict[str, str] = {}
+    for vol in pod.get("spec", {}).get("volumes", []) or []:
+        sec = vol.get("secret")
+        if not sec or not sec.get("secretName"):
+            continue
+        data = collector.get(sec["secretName"])
+        items = sec.get("items")
+        if items:
+            for item in items:
+                key = item.get("key", "")
+                name = (item.get("path") or key).replace("/", "_").replace(".", "_")
+                if key in data and name:
+                    env[name] = _escape(data[key])
+        else:
+            for key, value in data.items():
+                name = key.replace("/", "_").replace(".", "_")
+                env[name] = _escape(value)
+    return {k: v for k, v in env.items() if not is_k8s_auto_injected(k)}
+
+
+def extract_env_vars(
+    pod: Dict[str, Any], client: K8sClient
+) -> Tuple[Dict[str, str], List[Dict[str, str]]]:
+    """Returns (pod_level_env, per_container_env_list).
+
+    pod_level_env mirrors the reference's flat map (first container wins);
+    per-container maps preserve each container's own view for the runtime.
+    """
+    namespace = pod.get("metadata", {}).get("namespace", "default")
+    collector = SecretCollector(client, namespace)
+    containers = pod.get("spec", {}).get("containers", []) or []
+
+    per_container: List[Dict[str, str]] = [
+        _container_env(c, collector) for c in containers
+    ]
+    vol_env = _volume_secret_env(pod, collector)
+
+    pod_level: Dict[str, str] = dict(vol_env)
+    for cenv in reversed(per_container):  # first container wins
+        pod_level.update(cenv)
+    for i in range(len(per_container)):
+        merged = dict(vol_env)
+        merged.update(per_container[i])
+        per_container[i] = merged
+    return pod_level, per_container

@@ -1,0 +1,243 @@
+"""Spec translation: Kubernetes pod → local deploy parameters.
+
+Counterpart of the reference's translation layer
+(pkg/virtual_kubelet/runpod_client.go:1023-1393):
+
+- owner-Job resolution + annotation fallback (getOwnerJob :1056-1099,
+  getAnnotationWithFallback :1101-1112),
+- cloud-type validation (validateCloudType :1114-1134 — SECURE default,
+  SECURE/COMMUNITY accepted; locally SECURE means "cgroup-isolated", the only
+  mode implemented, COMMUNITY is accepted and treated the same),
+- datacenter-ID compliance against the node-level allow-list
+  (validateDatacenterIDs :1136-1178 — kept verbatim: the "datacenters" of a
+  single node deployment are its configured partition labels),
+- GPU memory annotation parse with the 16 GB default (extractGPUMemory
+  :1180-1191),
+- port extraction (ports.py), env extraction (envvars.py),
+- parameter assembly (PrepareRunPodParameters :1248-1377) — but targeting the
+  local binder: gpu_count comes from the ``amd.com/gpu`` resource request
+  (the reference picks a cloud GPU *type*; here the type is always MI355X and
+  the choice is *which* GPUs, made by the binder at deploy time).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Any, Dict, List, Optional
+
+from ..config import Config
+from ..kube.client import K8sClient, is_not_found
+from ..kube.objects import (
+    namespace_of,
+    owner_references,
+    pod_key,
+    resource_parse_bytes,
+    resource_parse_cpu,
+)
+from ..runtime.types import ContainerSpec, DeployParams
+from . import annotations as ann
+from .envvars import extract_env_vars
+from .ports import get_requested_ports, tcp_ports_of_container
+
+log = logging.getLogger("provider.translate")
+
+GIB = 1024**3
+
+
+class ValidationError(ValueError):
+    pass
+
+
+def get_owner_job(pod: Dict[str, Any], client: K8sClient) -> Optional[Dict[str, Any]]:
+    """Walk ownerReferences for Kind=Job, fetch it, verify UID
+    (runpod_client.go:1056-1099)."""
+    for ref in owner_references(pod):
+        if ref.get("kind") != "Job":
+            continue
+        try:
+            job = client.get_job(namespace_of(pod), ref.get("name", ""))
+        except Exception as exc:
+            if is_not_found(exc):
+                continue
+            raise
+        if job.get("metadata", {}).get("uid") == ref.get("uid"):
+            return job
+    return None
+
+
+def annotation_with_fallback(
+    pod: Dict[str, Any], job: Optional[Dict[str, Any]], key: str, default: str = ""
+) -> str:
+    """pod annotation → owner-Job annotation → default
+    (runpod_client.go:1101-1112)."""
+    value = pod.get("metadata", {}).get("annotations", {}).get(key, "")
+    if value:
+        return value
+    if job is not None:
+        value = job.get("metadata", {}).get("annotations", {}).get(key, "")
+        if value:
+            return value
+    return default
+
+
+def validate_cloud_type(value: str) -> str:
+    """Default SECURE; normalize upper; only SECURE/COMMUNITY accepted
+    (runpod_client.go:1114-1134)."""
+    if not value:
+        return "SECURE"
+    norm = value.upper()
+    if norm not in ("SECURE", "COMMUNITY"):
+        raise ValidationError(f"invalid cloud type {value!r}: must be SECURE or COMMUNITY")
+    return norm
+
+
+def validate_datacenter_ids(pod_value: str, node_allow_list: List[str]) -> List[str]:
+    """Node-level allow-list semantics (runpod_client.go:1136-1178): pod list
+    filtered to the subset allowed by the node config; empty intersection is a
+    hard error; no node config ⇒ pod value passes through."""
+    pod_ids = [s.strip() for s in pod_value.split(",") if s.strip()] if pod_value else []
+    if not node_allow_list:
+        return pod_ids
+    if not pod_ids:
+        return list(node_allow_list)
+    allowed = [d for d in pod_ids if d in node_allow_list]
+    if not allowed:
+        raise ValidationError(
+            f"pod datacenter ids {pod_ids} not allowed by node config {node_allow_list}"
+        )
+    return allowed
+
+
+def extract_gpu_memory_gb(pod: Dict[str, Any], job: Optional[Dict[str, Any]],
+                          default_gb: int) -> int:
+    """Annotation string → int GiB, default 16 (runpod_client.go:1180-1191).
+    Accepts both runpod.io/required-gpu-memory and runpod.io/gpu-memory, and
+    tolerates quantity suffixes ("256GiB")."""
+    raw = annotation_with_fallback(pod, job, ann.GPU_MEMORY) or annotation_with_fallback(
+        pod, job, ann.GPU_MEMORY_ALT
+    )
+    if not raw:
+        return default_gb
+    text = raw.strip()
+    for suffix in ("GiB", "Gi", "GB", "G"):
+        if text.endswith(suffix):
+            text = text[: -len(suffix)]
+            break
+    try:
+        return max(0, int(float(text)))
+    except ValueError:
+        log.warning("unparseable gpu-memory annotation", extra={"value": raw})
+        return default_gb
+
+
+def gpu_count_of(pod: Dict[str, Any]) -> int:
+    """amd.com/gpu from resource limits/requests across containers."""
+    total = 0
+    for container in pod.get("spec", {}).get("containers", []) or []:
+        res = container.get("resources", {}) or {}
+        for section in ("limits", "requests"):
+            value = (res.get(section) or {}).get(ann.GPU_RESOURCE)
+            if value is not None:
+                total += int(value)
+                break
+    return total
+
+
+def _cgroup_limits(pod: Dict[str, Any]) -> (str, str):
+    """Aggregate container CPU/memory limits into cgroup v2 strings."""
+    cpu_cores = 0.0
+    mem_bytes = 0
+    for container in pod.get("spec", {}).get("containers", []) or []:
+        limits = (container.get("resources", {}) or {}).get("limits") or {}
+        if "cpu" in limits:
+            cpu_cores += resource_parse_cpu(limits["cpu"])
+        if "memory" in limits:
+            mem_bytes += resource_parse_bytes(limits["memory"])
+    cpu_max = f"{int(cpu_cores * 100000)} 100000" if cpu_cores > 0 else ""
+    memory_max = str(mem_bytes) if mem_bytes > 0 else ""
+    return cpu_max, memory_max
+
+
+def prepare_deploy_params(
+    pod: Dict[str, Any],
+    client: K8sClient,
+    config: Config,
+    gpu_offers=None,  # selector.GpuOfferCatalog | None — fail-fast feasibility
+) -> DeployParams:
+    """PrepareRunPodParameters analogue (runpod_client.go:1248-1377)."""
+    metadata = pod.get("metadata", {})
+    job = get_owner_job(pod, client)
+
+    cloud_type = validate_cloud_type(
+        annotation_with_fallback(pod, job, ann.CLOUD_TYPE)
+    )
+    datacenters = validate_datacenter_ids(
+        annotation_with_fallback(pod, job, ann.DATACENTER_IDS), config.datacenter_ids
+    )
+    template_id = annotation_with_fallback(pod, job, ann.TEMPLATE_ID)
+    registry_auth = annotation_with_fallback(pod, job, ann.REGISTRY_AUTH_ID)
+
+    gpu_count = gpu_count_of(pod)
+    gpu_memory_gb = extract_gpu_memory_gb(pod, job, config.gpu_memory_default_gb)
+    # A pod that names a GPU memory requirement but no resource count gets one
+    # GPU (the reference deploys gpuCount 1 implicitly via the type selector).
+    if gpu_count == 0 and (
+        ann.GPU_MEMORY in metadata.get("annotations", {})
+        or ann.GPU_MEMORY_ALT in metadata.get("annotations", {})
+    ):
+        gpu_count = 1
+
+    # Fail-fast feasibility check — the GetGPUTypes analogue
+    # (runpod_client.go:1281: GetGPUTypes(minRAM, maxPrice, cloudType) errors
+    # when no type matches). max_gpu_price is wired (dead in the reference).
+    if gpu_count > 0 and gpu_offers is not None:
+        offers = gpu_offers.offers(
+            min_memory_bytes=(gpu_memory_gb * GIB + gpu_count - 1) // max(gpu_count, 1),
+            max_cost=config.max_gpu_price,
+        )
+        if len(offers) < gpu_count:
+            raise ValidationError(
+                f"no GPU set available: need {gpu_count} × "
+                f"{gpu_memory_gb // max(gpu_count, 1)} GiB headroom at cost <= "
+                f"{config.max_gpu_price}; {len(offers)} GPUs eligible"
+            )
+
+    pod_env, per_container_env = extract_env_vars(pod, client)
+    requested_ports = get_requested_ports(pod)
+
+    containers: List[ContainerSpec] = []
+    for i, c in enumerate(pod.get("spec", {}).get("containers", []) or []):
+        containers.append(
+            ContainerSpec(
+                name=c.get("name", f"c{i}"),
+                image=c.get("image", ""),
+                command=list(c.get("command", []) or []),
+                args=list(c.get("args", []) or []),
+                env=per_container_env[i] if i < len(per_container_env) else {},
+                working_dir=c.get("workingDir", ""),
+                tcp_ports=tcp_ports_of_container(c),
+            )
+        )
+    if not containers:
+        raise ValidationError("pod has no containers")
+
+    cpu_max, memory_max = _cgroup_limits(pod)
+
+    return DeployParams(
+        pod_key=pod_key(pod),
+        name=metadata.get("name", ""),
+        namespace=metadata.get("namespace", "default"),
+        containers=containers,
+        env=pod_env,
+        gpu_count=gpu_count,
+        gpu_memory_bytes=gpu_memory_gb * GIB,
+        max_gpu_cost=config.max_gpu_price,
+        requested_ports=requested_ports,
+        cloud_type=cloud_type,
+        datacenter_ids=datacenters,
+        template_id=template_id,
+        registry_auth_id=registry_auth,
+        cpu_limit=cpu_max,
+        memory_limit=memory_max,
+        labels=dict(metadata.get("labels", {}) or {}),
+    )
