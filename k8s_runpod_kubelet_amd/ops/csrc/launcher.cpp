@@ -168,6 +168,10 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
 
   res.pid = pid;
   res.pidfd = PidfdOpen(pid);
+  if (ready_pipe[0] >= 0) {
+    int fl = fcntl(ready_pipe[0], F_GETFL);
+    fcntl(ready_pipe[0], F_SETFL, fl | O_NONBLOCK);
+  }
   res.ready_fd = ready_pipe[0];
   return res;
 }
@@ -261,8 +265,23 @@ std::vector<Event> EventLoop::Poll(int timeout_ms) {
   if (n <= 0) return out;
 
   std::lock_guard<std::mutex> lock(mu_);
+  // Two passes: readiness pipes first, then pidfd exits. When a fast process
+  // writes READY and exits within one epoll batch, the exit handler closes
+  // the ready fd — draining ready events first keeps them from being lost.
+  std::vector<int> order;
+  order.reserve(static_cast<size_t>(n));
   for (int i = 0; i < n; ++i) {
     int fd = evs[i].data.fd;
+    auto pit = fd_to_pid_.find(fd);
+    bool is_pidfd = false;
+    if (pit != fd_to_pid_.end()) {
+      auto it = procs_.find(pit->second);
+      if (it != procs_.end() && fd == it->second.pidfd) is_pidfd = true;
+    }
+    if (is_pidfd) order.push_back(i); else order.insert(order.begin(), i);
+  }
+  for (int idx : order) {
+    int fd = evs[idx].data.fd;
     if (fd == wakefd_) {
       uint64_t junk;
       while (read(wakefd_, &junk, sizeof(junk)) > 0) {}
@@ -275,6 +294,19 @@ std::vector<Event> EventLoop::Poll(int timeout_ms) {
     Entry& e = it->second;
 
     if (fd == e.pidfd) {
+      // Final drain of the readiness pipe before tearing the entry down.
+      if (e.ready_fd >= 0) {
+        char rbuf[256];
+        ssize_t rr = read(e.ready_fd, rbuf, sizeof(rbuf));
+        if (rr > 0) {
+          Event rev;
+          rev.type = Event::kReady;
+          rev.pid = e.pid;
+          rev.token = e.token;
+          rev.data.assign(rbuf, static_cast<size_t>(rr));
+          out.push_back(rev);
+        }
+      }
       // Process exited — reap it.
       int status = 0;
       pid_t r = waitpid(static_cast<pid_t>(e.pid), &status, WNOHANG);

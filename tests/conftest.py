@@ -1,0 +1,97 @@
+import os
+import sys
+import tempfile
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: tests that need a real MI355X GPU (run via gpurun)"
+    )
+
+
+@pytest.fixture
+def tmp_state_dir():
+    with tempfile.TemporaryDirectory(prefix="amdvk-test-") as d:
+        yield d
+
+
+@pytest.fixture
+def fake_kube():
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+
+    return FakeKube()
+
+
+@pytest.fixture
+def synthetic_ledger():
+    from k8s_runpod_kubelet_amd.gpu.inventory import Inventory
+    from k8s_runpod_kubelet_amd.gpu.ledger import Ledger
+
+    inv = Inventory(synthetic_count=8)
+    inv.discover()
+    ledger = Ledger(inv)
+    ledger.sync_inventory()
+    return ledger
+
+
+@pytest.fixture
+def process_runtime(synthetic_ledger, tmp_state_dir):
+    from k8s_runpod_kubelet_amd.runtime.process_runtime import ProcessRuntime
+
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    yield rt
+    rt.close()
+
+
+def make_pod(name="p1", namespace="default", node="virtual-runpod", gpus=0,
+             annotations=None, command=None, args=None, ports=None,
+             containers=None, labels=None, owner=None):
+    """Test pod factory (dict in wire shape)."""
+    if containers is None:
+        c = {"name": "main", "image": "amdvk/test:latest"}
+        if command is not None:
+            c["command"] = command
+        if args is not None:
+            c["args"] = args
+        if gpus:
+            c["resources"] = {"limits": {"amd.com/gpu": str(gpus)}}
+        if ports:
+            c["ports"] = [{"containerPort": p, "protocol": "TCP"} for p in ports]
+        containers = [c]
+    pod = {
+        "apiVersion": "v1",
+        "kind": "Pod",
+        "metadata": {"name": name, "namespace": namespace,
+                     "annotations": dict(annotations or {}),
+                     "labels": dict(labels or {})},
+        "spec": {"nodeName": node, "containers": containers},
+    }
+    if owner:
+        pod["metadata"]["ownerReferences"] = [owner]
+    return pod
+
+
+@pytest.fixture
+def pod_factory():
+    return make_pod
+
+
+def wait_until(fn, timeout_s=5.0, interval_s=0.01):
+    import time
+
+    deadline = time.time() + timeout_s
+    while time.time() < deadline:
+        result = fn()
+        if result:
+            return result
+        time.sleep(interval_s)
+    return fn()
+
+
+@pytest.fixture
+def waiter():
+    return wait_until

@@ -1,0 +1,180 @@
+"""ProcessRuntime lifecycle tests (CPU: podworker runs in no-GPU mode)."""
+
+import time
+
+import pytest
+
+from k8s_runpod_kubelet_amd.runtime.process_runtime import ProcessRuntime
+from k8s_runpod_kubelet_amd.runtime.types import (
+    ContainerSpec,
+    DeployParams,
+    PodStatus,
+    is_successful_completion,
+)
+
+
+def params(pod_key="default-p1", args=None, containers=None, **kw):
+    if containers is None:
+        containers = [ContainerSpec(name="main", command=["podworker"],
+                                    args=args or ["--hold"])]
+    return DeployParams(pod_key=pod_key, name=pod_key.split("-", 1)[1],
+                        containers=containers, **kw)
+
+
+def wait_status(rt, iid, status, timeout=5.0):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        s = rt.get_detailed_status(iid)
+        if s.desired_status == status:
+            return s
+        time.sleep(0.01)
+    return rt.get_detailed_status(iid)
+
+
+def wait_ready(rt, iid, timeout=5.0):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        s = rt.get_detailed_status(iid)
+        if s.containers and all(c.ready for c in s.containers):
+            return s
+        time.sleep(0.005)
+    return rt.get_detailed_status(iid)
+
+
+def test_deploy_ready_exit(process_runtime):
+    rt = process_runtime
+    st = rt.deploy(params(args=["--run-for", "0.2"]))
+    assert st.desired_status == PodStatus.RUNNING
+    s = wait_ready(rt, st.id)
+    assert s.containers[0].ready
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.desired_status == PodStatus.EXITED
+    assert s.exit_code == 0
+    assert is_successful_completion(s)
+
+
+def test_failure_exit_code(process_runtime):
+    rt = process_runtime
+    st = rt.deploy(params(args=["--exit-code", "3", "--run-for", "0.05"]))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.exit_code == 3
+    assert not is_successful_completion(s)
+
+
+def test_terminate_sigterm(process_runtime):
+    rt = process_runtime
+    st = rt.deploy(params(args=["--hold"]))
+    wait_ready(rt, st.id)
+    rt.terminate(st.id)
+    s = wait_status(rt, st.id, PodStatus.TERMINATED)
+    assert s.desired_status == PodStatus.TERMINATED
+
+
+def test_logs(process_runtime):
+    rt = process_runtime
+    st = rt.deploy(params(args=["--run-for", "0.05"]))
+    wait_status(rt, st.id, PodStatus.EXITED)
+    logs = rt.get_logs(st.id)
+    assert "podworker: ready" in logs
+    assert rt.get_logs("nonexistent") == ""
+
+
+def test_gpu_reservation_lifecycle(synthetic_ledger, tmp_state_dir):
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    try:
+        st = rt.deploy(params(gpu_count=2, gpu_memory_bytes=64 * 1024**3,
+                              args=["--run-for", "0.2"]))
+        assert len(st.gpu_indices) == 2
+        res = synthetic_ledger.get_reservation("default-p1")
+        assert res is not None and res.bytes_per_gpu == 32 * 1024**3
+        wait_status(rt, st.id, PodStatus.EXITED)
+        # HBM returns to the ledger on exit, before pod deletion
+        assert synthetic_ledger.get_reservation("default-p1") is None
+    finally:
+        rt.close()
+
+
+def test_subscriber_push(process_runtime):
+    events = []
+    process_runtime.subscribe(lambda iid: events.append(iid))
+    st = process_runtime.deploy(params(args=["--run-for", "0.05"]))
+    wait_status(process_runtime, st.id, PodStatus.EXITED)
+    time.sleep(0.05)
+    assert events.count(st.id) >= 2  # ready + exit
+
+
+def test_not_found(process_runtime):
+    assert process_runtime.get_detailed_status("nope").desired_status == PodStatus.NOT_FOUND
+    assert process_runtime.get_status("nope") == PodStatus.NOT_FOUND
+
+
+def test_multi_container_pod(process_runtime):
+    rt = process_runtime
+    p = params(containers=[
+        ContainerSpec(name="a", command=["podworker"], args=["--run-for", "0.1"]),
+        ContainerSpec(name="b", command=["podworker"], args=["--run-for", "0.3"]),
+    ])
+    st = rt.deploy(p)
+    assert len(st.containers) == 2
+    # EXITED only after *all* containers exit
+    s = wait_status(rt, st.id, PodStatus.EXITED, timeout=5)
+    assert s.desired_status == PodStatus.EXITED
+    assert {c.name for c in s.containers} == {"a", "b"}
+    assert rt.get_logs(st.id, "b")
+
+
+def test_persistence_and_adoption(synthetic_ledger, tmp_state_dir):
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    st = rt.deploy(params(gpu_count=1, args=["--hold"]))
+    wait_ready(rt, st.id)
+    # Simulate kubelet restart: stop watcher, drop state, re-adopt from disk.
+    rt._stop.set()
+    rt._loop.wake()
+    rt._watcher.join(timeout=2)
+
+    from k8s_runpod_kubelet_amd.gpu.inventory import Inventory
+    from k8s_runpod_kubelet_amd.gpu.ledger import Ledger
+
+    inv2 = Inventory(synthetic_count=8)
+    inv2.discover()
+    ledger2 = Ledger(inv2)
+    ledger2.sync_inventory()
+    rt2 = ProcessRuntime(ledger2, tmp_state_dir, enable_cgroups=False)
+    try:
+        adopted = rt2.adopt_persisted()
+        assert st.id in adopted
+        s = rt2.get_detailed_status(st.id)
+        assert s.desired_status == PodStatus.RUNNING
+        assert ledger2.get_reservation("default-p1") is not None  # no double-bind
+        rt2.terminate(st.id)
+        s = wait_status(rt2, st.id, PodStatus.TERMINATED)
+        assert s.desired_status == PodStatus.TERMINATED
+    finally:
+        rt2.close()
+        rt.close()
+
+
+def test_adoption_of_vanished_process(synthetic_ledger, tmp_state_dir):
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    st = rt.deploy(params(args=["--run-for", "30"]))
+    wait_ready(rt, st.id)
+    # Kill the process outside the runtime AND stop the watcher first so the
+    # exit is never observed (simulates dying while the kubelet is down).
+    rt._stop.set(); rt._loop.wake(); rt._watcher.join(timeout=2)
+    import os, signal
+    pid = rt.get_detailed_status(st.id).containers[0].pid
+    os.kill(pid, signal.SIGKILL)
+    os.waitpid(pid, 0)  # reap: in a real restart the process is fully gone
+
+    rt2 = ProcessRuntime(synthetic_ledger.__class__(rt.ledger.inventory), tmp_state_dir,
+                         enable_cgroups=False)
+    # note: fresh ledger instance over same inventory
+    rt2.ledger.sync_inventory()
+    try:
+        rt2.adopt_persisted()
+        s = rt2.get_detailed_status(st.id)
+        assert s.desired_status == PodStatus.EXITED
+        assert s.containers[0].exit_code == -1  # vanished marker
+    finally:
+        rt2.close()
+        rt.close()

@@ -1,0 +1,279 @@
+"""Spec-translation tests — the param-level suite the reference gates behind
+real API keys (reference annotations_test.go), here fully hermetic."""
+
+import base64
+
+import pytest
+
+from k8s_runpod_kubelet_amd.config import Config
+from k8s_runpod_kubelet_amd.provider import annotations as ann
+from k8s_runpod_kubelet_amd.provider.envvars import extract_env_vars, is_k8s_auto_injected
+from k8s_runpod_kubelet_amd.provider.ports import (
+    check_ports_exposed,
+    extract_ports_from_pod,
+    get_requested_ports,
+)
+from k8s_runpod_kubelet_amd.provider.translate import (
+    ValidationError,
+    annotation_with_fallback,
+    extract_gpu_memory_gb,
+    gpu_count_of,
+    prepare_deploy_params,
+    validate_cloud_type,
+    validate_datacenter_ids,
+)
+from tests.conftest import make_pod
+
+
+def job_with(annotations, name="job1", uid="u-1"):
+    return {"metadata": {"name": name, "namespace": "default", "uid": uid,
+                         "annotations": annotations}}
+
+
+def owner_ref(name="job1", uid="u-1"):
+    return {"kind": "Job", "name": name, "uid": uid, "apiVersion": "batch/v1"}
+
+
+# --- annotation fallback (reference annotations_test.go:27-239 scenarios) ---
+
+def test_pod_annotation_overrides_job(fake_kube):
+    fake_kube.put_job("default", job_with({ann.CLOUD_TYPE: "COMMUNITY"}))
+    pod = make_pod(annotations={ann.CLOUD_TYPE: "SECURE"}, owner=owner_ref())
+    from k8s_runpod_kubelet_amd.provider.translate import get_owner_job
+
+    job = get_owner_job(pod, fake_kube)
+    assert job is not None
+    assert annotation_with_fallback(pod, job, ann.CLOUD_TYPE) == "SECURE"
+
+
+def test_job_annotation_fallback(fake_kube):
+    fake_kube.put_job("default", job_with({
+        ann.REGISTRY_AUTH_ID: "auth-123",
+        ann.DATACENTER_IDS: "dc1,dc2",
+        ann.GPU_MEMORY: "48",
+        ann.PORTS: "9999/tcp",
+    }))
+    pod = make_pod(owner=owner_ref())
+    params = prepare_deploy_params(pod, fake_kube, Config())
+    assert params.registry_auth_id == "auth-123"
+    assert params.datacenter_ids == ["dc1", "dc2"]
+    assert params.gpu_memory_bytes == 48 * 1024**3
+    assert params.requested_ports == ["9999/tcp"]
+
+
+def test_owner_job_uid_mismatch_ignored(fake_kube):
+    fake_kube.put_job("default", job_with({ann.TEMPLATE_ID: "tpl"}, uid="other-uid"))
+    pod = make_pod(owner=owner_ref(uid="u-1"))
+    from k8s_runpod_kubelet_amd.provider.translate import get_owner_job
+
+    assert get_owner_job(pod, fake_kube) is None
+
+
+def test_template_and_auth_coexist(fake_kube):
+    pod = make_pod(annotations={ann.TEMPLATE_ID: "tpl-1",
+                                ann.REGISTRY_AUTH_ID: "auth-1"})
+    params = prepare_deploy_params(pod, fake_kube, Config())
+    assert params.template_id == "tpl-1"
+    assert params.registry_auth_id == "auth-1"
+
+
+# --- cloud type (runpod_client.go:1114-1134) ---
+
+def test_cloud_type():
+    assert validate_cloud_type("") == "SECURE"
+    assert validate_cloud_type("secure") == "SECURE"
+    assert validate_cloud_type("Community") == "COMMUNITY"
+    with pytest.raises(ValidationError):
+        validate_cloud_type("SPOT")
+
+
+# --- datacenter allow-list (runpod_client.go:1136-1178) ---
+
+def test_datacenter_validation():
+    assert validate_datacenter_ids("", []) == []
+    assert validate_datacenter_ids("a,b", []) == ["a", "b"]
+    assert validate_datacenter_ids("", ["x"]) == ["x"]
+    assert validate_datacenter_ids("a,x", ["x", "y"]) == ["x"]
+    with pytest.raises(ValidationError):
+        validate_datacenter_ids("a,b", ["x"])
+
+
+# --- gpu memory (runpod_client.go:1180-1191) ---
+
+def test_gpu_memory_parse():
+    assert extract_gpu_memory_gb(make_pod(), None, 16) == 16  # default
+    pod = make_pod(annotations={ann.GPU_MEMORY: "64"})
+    assert extract_gpu_memory_gb(pod, None, 16) == 64
+    pod = make_pod(annotations={ann.GPU_MEMORY_ALT: "256GiB"})
+    assert extract_gpu_memory_gb(pod, None, 16) == 256
+    pod = make_pod(annotations={ann.GPU_MEMORY: "garbage"})
+    assert extract_gpu_memory_gb(pod, None, 16) == 16
+
+
+def test_gpu_count_from_resources():
+    assert gpu_count_of(make_pod(gpus=2)) == 2
+    assert gpu_count_of(make_pod()) == 0
+    pod = make_pod(containers=[
+        {"name": "a", "resources": {"requests": {"amd.com/gpu": "1"}}},
+        {"name": "b", "resources": {"limits": {"amd.com/gpu": "3"}}},
+    ])
+    assert gpu_count_of(pod) == 4
+
+
+def test_memory_annotation_implies_one_gpu(fake_kube):
+    pod = make_pod(annotations={ann.GPU_MEMORY: "32"})
+    params = prepare_deploy_params(pod, fake_kube, Config())
+    assert params.gpu_count == 1
+
+
+# --- ports (runpod_client.go:1193-1246, kubelet.go:566-605) ---
+
+def test_port_extraction_http_autodetect():
+    pod = make_pod(containers=[
+        {"name": "a", "ports": [
+            {"containerPort": 8080, "protocol": "TCP"},
+            {"containerPort": 9222},
+            {"containerPort": 53, "protocol": "UDP"},  # skipped
+        ]},
+        {"name": "b", "ports": [{"containerPort": 443}]},
+    ])
+    assert extract_ports_from_pod(pod) == ["8080/http", "9222/tcp", "443/http"]
+
+
+def test_requested_ports_annotation_override():
+    pod = make_pod(ports=[1234], annotations={ann.PORTS: "8080/http, 22/tcp"})
+    assert get_requested_ports(pod) == ["8080/http", "22/tcp"]
+    pod = make_pod(ports=[1234])
+    assert get_requested_ports(pod) == ["1234/tcp"]
+
+
+def test_check_ports_exposed():
+    assert check_ports_exposed([], {})  # no ports => ready
+    assert check_ports_exposed(["80/http"], {})  # http assumed ready
+    assert not check_ports_exposed(["22/tcp"], {})
+    assert check_ports_exposed(["22/tcp"], {22: 22})
+    assert check_ports_exposed(["22/tcp", "80/http"], {22: 10022})
+
+
+# --- env extraction (runpod_client.go:866-1054) ---
+
+def b64(s):
+    return base64.b64encode(s.encode()).decode()
+
+
+def test_env_literal_and_filtering(fake_kube):
+    pod = make_pod(containers=[{
+        "name": "a",
+        "env": [
+            {"name": "FOO", "value": "bar"},
+            {"name": "MULTI", "value": "a\nb"},
+            {"name": "KUBERNETES_SERVICE_HOST", "value": "x"},  # filtered
+            {"name": "MYAPP_PORT_8080_TCP_ADDR", "value": "x"},  # filtered
+        ],
+    }])
+    env, per = extract_env_vars(pod, fake_kube)
+    assert env["FOO"] == "bar"
+    assert env["MULTI"] == "a\\nb"  # newline escaping
+    assert "KUBERNETES_SERVICE_HOST" not in env
+    assert "MYAPP_PORT_8080_TCP_ADDR" not in env
+
+
+def test_env_secret_key_ref(fake_kube):
+    fake_kube.put_secret("default", {"metadata": {"name": "s1"},
+                                     "data": {"tok": b64("sekrit")}})
+    pod = make_pod(containers=[{
+        "name": "a",
+        "env": [{"name": "TOKEN",
+                 "valueFrom": {"secretKeyRef": {"name": "s1", "key": "tok"}}}],
+    }])
+    env, _ = extract_env_vars(pod, fake_kube)
+    assert env["TOKEN"] == "sekrit"
+
+
+def test_env_from_whole_secret(fake_kube):
+    fake_kube.put_secret("default", {"metadata": {"name": "s2"},
+                                     "data": {"A": b64("1"), "B": b64("2")}})
+    pod = make_pod(containers=[{"name": "a",
+                                "envFrom": [{"secretRef": {"name": "s2"}}]}])
+    env, _ = extract_env_vars(pod, fake_kube)
+    assert env["A"] == "1" and env["B"] == "2"
+
+
+def test_secret_volume_flattened(fake_kube):
+    fake_kube.put_secret("default", {"metadata": {"name": "vs"},
+                                     "data": {"key1": b64("v1"), "key2": b64("v2")}})
+    pod = make_pod()
+    pod["spec"]["volumes"] = [{"name": "v", "secret": {
+        "secretName": "vs", "items": [{"key": "key1", "path": "renamed.txt"}]}}]
+    env, _ = extract_env_vars(pod, fake_kube)
+    assert env["renamed_txt"] == "v1"
+    assert "key2" not in env
+    # no items => all keys
+    pod["spec"]["volumes"] = [{"name": "v", "secret": {"secretName": "vs"}}]
+    env, _ = extract_env_vars(pod, fake_kube)
+    assert env["key1"] == "v1" and env["key2"] == "v2"
+
+
+def test_env_all_containers_parity_plus(fake_kube):
+    pod = make_pod(containers=[
+        {"name": "a", "env": [{"name": "X", "value": "from-a"}]},
+        {"name": "b", "env": [{"name": "X", "value": "from-b"},
+                              {"name": "Y", "value": "y"}]},
+    ])
+    env, per = extract_env_vars(pod, fake_kube)
+    assert env["X"] == "from-a"  # first container wins (reference: only c0)
+    assert env["Y"] == "y"
+    assert per[0] == {"X": "from-a"}
+    assert per[1]["X"] == "from-b"
+
+
+def test_missing_secret_tolerated(fake_kube):
+    pod = make_pod(containers=[{"name": "a",
+                                "envFrom": [{"secretRef": {"name": "nope"}}]}])
+    env, _ = extract_env_vars(pod, fake_kube)
+    assert env == {}
+
+
+def test_auto_injected_patterns():
+    assert is_k8s_auto_injected("KUBERNETES_PORT")
+    assert is_k8s_auto_injected("FOO_SERVICE_HOST")
+    assert not is_k8s_auto_injected("MY_TOKEN")
+
+
+# --- full assembly (runpod_client.go:1248-1377) ---
+
+def test_prepare_full(fake_kube):
+    cfg = Config(datacenter_ids=["dc1"])
+    pod = make_pod(gpus=2, ports=[8080],
+                   annotations={ann.GPU_MEMORY: "128", ann.CLOUD_TYPE: "secure"})
+    pod["spec"]["containers"][0]["resources"]["limits"].update(
+        {"cpu": "2", "memory": "4Gi"})
+    params = prepare_deploy_params(pod, fake_kube, cfg)
+    assert params.pod_key == "default-p1"
+    assert params.gpu_count == 2
+    assert params.gpu_memory_bytes == 128 * 1024**3
+    assert params.cloud_type == "SECURE"
+    assert params.datacenter_ids == ["dc1"]
+    assert params.requested_ports == ["8080/http"]
+    assert params.cpu_limit == "200000 100000"
+    assert params.memory_limit == str(4 * 1024**3)
+    assert params.max_gpu_cost == 0.5
+
+
+def test_prepare_no_containers_rejected(fake_kube):
+    pod = make_pod()
+    pod["spec"]["containers"] = []
+    with pytest.raises(ValidationError):
+        prepare_deploy_params(pod, fake_kube, Config())
+
+
+def test_prepare_fail_fast_when_no_offers(fake_kube, synthetic_ledger):
+    from k8s_runpod_kubelet_amd.provider.selector import GpuOfferCatalog
+
+    catalog = GpuOfferCatalog(synthetic_ledger)
+    pod = make_pod(gpus=1, annotations={ann.GPU_MEMORY: "10000"})  # > 288 GB
+    with pytest.raises(ValidationError, match="no GPU set available"):
+        prepare_deploy_params(pod, fake_kube, Config(), catalog)
+    ok = make_pod(gpus=8, annotations={ann.GPU_MEMORY: "2000"})  # 250/GPU
+    params = prepare_deploy_params(ok, fake_kube, Config(), catalog)
+    assert params.gpu_count == 8
