@@ -64,8 +64,11 @@ class DetailedStatus:
     @property
     def completion_message(self) -> str:
         for c in self.containers:
+            if c.message:
+                return c.message
+        for c in self.containers:
             if c.exit_code:
-                return c.message or f"container {c.name} exited with code {c.exit_code}"
+                return f"container {c.name} exited with code {c.exit_code}"
         return "completed"
 
 
