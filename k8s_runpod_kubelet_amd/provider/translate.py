@@ -203,7 +203,12 @@ def prepare_deploy_params(
             )
 
     pod_env, per_container_env = extract_env_vars(pod, client)
-    requested_ports = get_requested_ports(pod)
+    # ports override with owner-Job fallback (runpod_client.go:1310-1330)
+    ports_override = annotation_with_fallback(pod, job, ann.PORTS)
+    if ports_override:
+        requested_ports = [p.strip() for p in ports_override.split(",") if p.strip()]
+    else:
+        requested_ports = get_requested_ports(pod)
 
     containers: List[ContainerSpec] = []
     for i, c in enumerate(pod.get("spec", {}).get("containers", []) or []):
