@@ -65,7 +65,9 @@ class Binder:
         need = req.bytes_per_gpu
         eligible = [
             s for s in states
-            if s.headroom_bytes >= need and s.cost() <= req.max_cost
+            if not s.occupied  # exclusive GPU claims (device-plugin semantics)
+            and s.headroom_bytes >= need
+            and s.cost() <= req.max_cost
         ]
         if len(eligible) < req.gpu_count:
             raise PlacementError(

@@ -38,6 +38,12 @@ class GpuState:
         return self.gpu.healthy and not self.cordoned
 
     @property
+    def occupied(self) -> bool:
+        """``amd.com/gpu`` is an exclusive countable resource (Kubernetes
+        device-plugin semantics): one pod's GPU claim owns the whole GPU."""
+        return bool(self.pod_keys)
+
+    @property
     def headroom_bytes(self) -> int:
         """Free HBM available for new reservations: total minus the larger of
         (ledger reservations, live measured use)."""
@@ -81,6 +87,8 @@ class Ledger:
                     raise ValueError(f"unknown GPU index {idx}")
                 if not state.schedulable:
                     raise ValueError(f"GPU {idx} is not schedulable")
+                if state.occupied:
+                    raise ValueError(f"GPU {idx} is already bound to {state.pod_keys}")
             for idx in gpu_indices:
                 state = self.states[idx]
                 state.reserved_bytes += bytes_per_gpu

@@ -21,3 +21,4 @@ class InstanceInfo:
     cost_per_hr: float = 0.0
     last_error: str = ""
     ready_time: Optional[float] = None  # first time the pod went Ready (metrics)
+    deploying: bool = False  # claim flag: a deploy for this pod is in flight

@@ -111,6 +111,16 @@ class Provider:
         if self._started:
             return
         self._started = True
+        # Rebuild runtime state from the instance journal BEFORE any
+        # controller can call create_pod, so existing pods adopt instead of
+        # redeploying (reference orders this differently — main.go:410 starts
+        # controllers before LoadRunning at :426 — and relies on timing).
+        adopt = getattr(self.runtime, "adopt_persisted", None)
+        if callable(adopt):
+            try:
+                adopt()
+            except Exception:
+                log.exception("instance adoption failed")
         self.check_backend_health()
         self.cleanup_stuck_terminating_pods()
         self.registrar.register()  # optional, non-fatal (unlike kubelet.go:369)
@@ -169,14 +179,32 @@ class Provider:
         from .ports import get_requested_ports
 
         requested = get_requested_ports(pod)
+        info = InstanceInfo(
+            status=PodStatus.STARTING,
+            requested_ports=requested,
+            creation_time=self._creation_ts(pod),
+        )
         with self._pods_lock:
             self._pods[key] = pod
-            self._pod_status[key] = InstanceInfo(
-                status=PodStatus.STARTING,
-                requested_ports=requested,
-                creation_time=self._creation_ts(pod),
-            )
+            self._pod_status[key] = info
         metrics.pods_created.inc()
+        # Adoption path: the pod already names a live instance (kubelet
+        # restarted between deploy and now) — do not redeploy.
+        existing_id = obj_annotations(pod).get(ann.POD_ID, "")
+        if existing_id:
+            detailed = self.runtime.get_detailed_status(existing_id)
+            if detailed.desired_status != PodStatus.NOT_FOUND:
+                with self._pods_lock:
+                    info.instance_id = existing_id
+                    info.status = detailed.desired_status
+                    info.gpu_indices = list(detailed.gpu_indices)
+                    info.cost_per_hr = detailed.cost_per_hr
+                log.info("adopted existing instance in CreatePod",
+                         extra={"pod": key, "instance": existing_id})
+                self._sync_pod_status(key)
+                return
+            self.handle_missing_instance(pod, info)
+            return
         try:
             self.deploy_pod(pod)
         except Exception as exc:
@@ -260,6 +288,25 @@ class Provider:
     # ------------------------------------------------------------------
 
     def deploy_pod(self, pod: Dict[str, Any]) -> None:
+        key = pod_key_of(pod)
+        # Claim the deploy: the pending-retry ticker and the controller can
+        # race here (the reference's single worker serializes this at the cost
+        # of throughput; with N workers the claim flag does it instead).
+        with self._pods_lock:
+            claim = self._pod_status.get(key)
+            if claim is not None:
+                if claim.instance_id or claim.deploying:
+                    return
+                claim.deploying = True
+        try:
+            self._deploy_pod_locked(pod)
+        finally:
+            with self._pods_lock:
+                claim = self._pod_status.get(key)
+                if claim is not None:
+                    claim.deploying = False
+
+    def _deploy_pod_locked(self, pod: Dict[str, Any]) -> None:
         key = pod_key_of(pod)
         # Inject node-level datacenter annotation if missing
         # (kubelet.go:437-455).

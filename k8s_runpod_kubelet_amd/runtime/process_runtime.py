@@ -206,6 +206,21 @@ class ProcessRuntime(Runtime):
                 time.sleep(0.5)
                 continue
             touched = set()
+            try:
+                self._handle_events(events, touched)
+            except Exception:
+                log.exception("event handling failed")
+            for inst_id in touched:
+                with self._lock:
+                    inst = self._instances.get(inst_id)
+                if inst is not None:
+                    try:
+                        self._persist(inst)
+                    except Exception:
+                        log.exception("instance persist failed")
+                self._notify(inst_id)
+
+    def _handle_events(self, events, touched) -> None:
             for ev in events:
                 inst_id = self._pid_to_instance.get(ev.pid)
                 if inst_id is None:
@@ -235,12 +250,6 @@ class ProcessRuntime(Runtime):
                         cinfo.message = f"exit code {ev.exit_code}"
                     self._on_container_exit(inst)
                     touched.add(inst_id)
-            for inst_id in touched:
-                with self._lock:
-                    inst = self._instances.get(inst_id)
-                if inst is not None:
-                    self._persist(inst)
-                self._notify(inst_id)
 
     def _on_container_exit(self, inst: Instance) -> None:
         if any(c.exit_code is None for c in inst.containers):
@@ -418,9 +427,14 @@ class ProcessRuntime(Runtime):
                 for c in inst.params.containers
             ],
         }
-        tmp = self.instances_dir / f".{inst.id}.tmp"
+        # Unique temp name: the event thread and API threads may persist the
+        # same instance concurrently; rename is atomic, last writer wins.
+        tmp = self.instances_dir / f".{inst.id}.{threading.get_ident()}.tmp"
         tmp.write_text(json.dumps(record))
-        tmp.rename(self.instances_dir / f"{inst.id}.json")
+        try:
+            tmp.rename(self.instances_dir / f"{inst.id}.json")
+        except FileNotFoundError:
+            pass  # state dir torn down during shutdown
 
     def adopt_persisted(self) -> List[str]:
         """Rebuild instance state after a kubelet restart: re-open pidfds for

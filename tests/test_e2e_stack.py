@@ -1,0 +1,235 @@
+"""Full-stack end-to-end tests: fake apiserver → informer → PodController →
+Provider → ProcessRuntime (real processes, CPU podworker) — the hermetic
+version of the reference's TestRunPodIntegration 7-step lifecycle
+(reference runpod_test.go:182-390), plus the BASELINE burst config."""
+
+import time
+
+import pytest
+
+from k8s_runpod_kubelet_amd.app import Stack, build_stack
+from k8s_runpod_kubelet_amd.config import Config
+from k8s_runpod_kubelet_amd.kube.client import NotFoundError
+from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+from k8s_runpod_kubelet_amd.provider import annotations as ann
+from tests.conftest import make_pod, wait_until
+
+
+@pytest.fixture
+def stack(tmp_state_dir):
+    cfg = Config(
+        state_dir=tmp_state_dir,
+        gpu_count_override=8,
+        pending_retry_interval_s=0.2,
+        reconcile_interval_s=30,
+        notify_interval_s=0,
+        pod_controller_workers=8,
+    )
+    kube = FakeKube()
+    s = build_stack(cfg, client=kube)
+    # tests run unprivileged paths: no cgroups
+    s.runtime.enable_cgroups = False
+    s.start(serve_http=False)
+    yield s, kube
+    s.stop()
+
+
+def ready(kube, name, ns="default"):
+    try:
+        pod = kube.get_pod(ns, name)
+    except NotFoundError:
+        return None
+    conds = {c["type"]: c["status"]
+             for c in pod.get("status", {}).get("conditions", [])}
+    return pod if conds.get("Ready") == "True" else None
+
+
+def gone(kube, name, ns="default"):
+    try:
+        kube.get_pod(ns, name)
+        return False
+    except NotFoundError:
+        return True
+
+
+def test_pod_lifecycle_end_to_end(stack):
+    s, kube = stack
+    pod = make_pod("e2e", gpus=1, command=["podworker"], args=["--hold"])
+    t0 = time.time()
+    kube.create_pod("default", pod)
+    got = wait_until(lambda: ready(kube, "e2e"), timeout_s=10)
+    latency = time.time() - t0
+    assert got is not None, "pod never went Ready"
+    assert latency < 5.0
+    assert got["status"]["phase"] == "Running"
+    anns = got["metadata"]["annotations"]
+    assert anns[ann.POD_ID].startswith("amdvk-")
+    assert anns[ann.GPU_IDS] != ""
+    # real logs through the provider (reference stubs this)
+    logs = wait_until(
+        lambda: s.provider.get_container_logs("default", "e2e") or None,
+        timeout_s=5)
+    assert "podworker: ready" in logs
+
+    kube.delete_pod("default", "e2e")
+    assert wait_until(lambda: gone(kube, "e2e"), timeout_s=15)
+    # GPU returned to the ledger
+    assert wait_until(
+        lambda: s.ledger.get_reservation("default-e2e") is None, timeout_s=5)
+
+
+def test_run_to_completion_succeeds(stack):
+    s, kube = stack
+    pod = make_pod("job1", command=["podworker"], args=["--run-for", "0.2"])
+    kube.create_pod("default", pod)
+    got = wait_until(
+        lambda: (kube.get_pod("default", "job1")
+                 if kube.get_pod("default", "job1").get("status", {}).get("phase")
+                 == "Succeeded" else None),
+        timeout_s=10)
+    assert got is not None
+    term = got["status"]["containerStatuses"][0]["state"]["terminated"]
+    assert term["exitCode"] == 0
+
+
+def test_failing_pod_reports_failed(stack):
+    s, kube = stack
+    pod = make_pod("bad", command=["podworker"],
+                   args=["--exit-code", "3", "--run-for", "0.05"])
+    kube.create_pod("default", pod)
+    got = wait_until(
+        lambda: (kube.get_pod("default", "bad")
+                 if kube.get_pod("default", "bad").get("status", {}).get("phase")
+                 == "Failed" else None),
+        timeout_s=10)
+    assert got is not None
+    assert got["status"]["containerStatuses"][0]["state"]["terminated"]["exitCode"] == 3
+
+
+def test_port_gated_readiness(stack):
+    s, kube = stack
+    pod = make_pod("srv", ports=[39131], command=["podworker"],
+                   args=["--listen-port", "39131", "--hold"])
+    kube.create_pod("default", pod)
+    got = wait_until(lambda: ready(kube, "srv"), timeout_s=10)
+    assert got is not None  # listening socket detected via /proc/net/tcp
+    kube.delete_pod("default", "srv")
+    wait_until(lambda: gone(kube, "srv"), timeout_s=15)
+
+
+def test_eight_concurrent_gpu_pods(stack):
+    # BASELINE config 3: 8 concurrent 1-GPU pods saturating the node.
+    s, kube = stack
+    names = [f"sat{i}" for i in range(8)]
+    for n in names:
+        kube.create_pod("default", make_pod(n, gpus=1, command=["podworker"],
+                                            args=["--hold"]))
+    for n in names:
+        assert wait_until(lambda n=n: ready(kube, n), timeout_s=15), n
+    # all 8 GPUs reserved, each exactly once
+    used = set()
+    for n in names:
+        ids = kube.get_pod("default", n)["metadata"]["annotations"][ann.GPU_IDS]
+        used.update(int(x) for x in ids.split(","))
+    assert used == set(range(8))
+    # ninth pod must stay Pending (no free GPU)
+    kube.create_pod("default", make_pod("ninth", gpus=1, command=["podworker"],
+                                        args=["--hold"]))
+    time.sleep(0.6)
+    ninth = kube.get_pod("default", "ninth")
+    assert ninth.get("status", {}).get("phase", "Pending") in ("Pending", "")
+    # free one GPU -> ninth gets placed by the retry loop
+    kube.delete_pod("default", names[0])
+    assert wait_until(lambda: ready(kube, "ninth"), timeout_s=15)
+    for n in names[1:] + ["ninth"]:
+        kube.delete_pod("default", n)
+        wait_until(lambda n=n: gone(kube, n), timeout_s=15)
+
+
+def test_multi_gpu_pod_with_memory_annotation(stack):
+    # BASELINE config 4: 4x amd.com/gpu with >=256GiB memory annotation.
+    s, kube = stack
+    pod = make_pod("big", gpus=4, command=["podworker"], args=["--hold"],
+                   annotations={ann.GPU_MEMORY_ALT: "256GiB"})
+    kube.create_pod("default", pod)
+    got = wait_until(lambda: ready(kube, "big"), timeout_s=10)
+    assert got is not None
+    ids = [int(x) for x in
+           got["metadata"]["annotations"][ann.GPU_IDS].split(",")]
+    assert len(ids) == 4
+    res = s.ledger.get_reservation("default-big")
+    assert res.bytes_per_gpu == 64 * 1024**3
+    kube.delete_pod("default", "big")
+    wait_until(lambda: gone(kube, "big"), timeout_s=15)
+
+
+def test_burst_32_pods_fifo_drain(stack):
+    # BASELINE config 5: 32 queued run-to-completion GPU pods FIFO-drained
+    # across 8 GPUs (backpressure + reconcile throughput).
+    s, kube = stack
+    names = [f"burst{i:02d}" for i in range(32)]
+    t0 = time.time()
+    for n in names:
+        kube.create_pod("default", make_pod(
+            n, gpus=1, command=["podworker"], args=["--run-for", "0.05"]))
+
+    def all_done():
+        done = 0
+        for n in names:
+            try:
+                pod = kube.get_pod("default", n)
+            except NotFoundError:
+                continue
+            if pod.get("status", {}).get("phase") == "Succeeded":
+                done += 1
+        return done == 32 or None
+
+    assert wait_until(all_done, timeout_s=60), "burst did not drain"
+    elapsed = time.time() - t0
+    # 32 pods over 8 GPUs with ~50ms workloads must drain fast
+    assert elapsed < 40
+    # no reservations left
+    assert all(s.ledger.get_reservation(f"default-{n}") is None for n in names)
+
+
+def test_node_registered_and_ready(stack):
+    s, kube = stack
+    node = kube.get_node("virtual-runpod")
+    assert node["status"]["capacity"][ann.GPU_RESOURCE] == "8"
+    ready_cond = [c for c in node["status"]["conditions"] if c["type"] == "Ready"][0]
+    assert ready_cond["status"] == "True"
+    lease = kube.get_lease("kube-node-lease", "virtual-runpod")
+    assert lease["spec"]["holderIdentity"] == "virtual-runpod"
+
+
+def test_restart_adoption_end_to_end(tmp_state_dir):
+    # kubelet restarts; running pod is adopted, not redeployed
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=8,
+                 notify_interval_s=0, pending_retry_interval_s=0.2)
+    kube = FakeKube()
+    s1 = build_stack(cfg, client=kube)
+    s1.runtime.enable_cgroups = False
+    s1.start(serve_http=False)
+    kube.create_pod("default", make_pod("persist", gpus=1,
+                                        command=["podworker"], args=["--hold"]))
+    got = wait_until(lambda: ready(kube, "persist"), timeout_s=10)
+    assert got
+    iid = got["metadata"]["annotations"][ann.POD_ID]
+    pid = s1.runtime.get_detailed_status(iid).containers[0].pid
+    # stop the stack without terminating pods (kubelet crash)
+    s1.pod_controller.stop(); s1.node_controller.stop(); s1.provider.stop()
+    s1.runtime._stop.set(); s1.runtime._loop.wake()
+
+    s2 = build_stack(cfg, client=kube)
+    s2.runtime.enable_cgroups = False
+    s2.start(serve_http=False)
+    try:
+        info = s2.provider.instance_info("default", "persist")
+        assert info is not None and info.instance_id == iid
+        assert s2.runtime.get_detailed_status(iid).containers[0].pid == pid
+        # exactly one reservation (no double-bind)
+        assert s2.ledger.get_reservation("default-persist") is not None
+        kube.delete_pod("default", "persist")
+        assert wait_until(lambda: gone(kube, "persist"), timeout_s=15)
+    finally:
+        s2.stop()

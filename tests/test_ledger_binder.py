@@ -39,10 +39,23 @@ def test_double_reserve_rejected():
         ledger.reserve("p1", [1], GIB)
 
 
-def test_best_fit_prefers_smallest_sufficient_headroom():
+def test_exclusive_gpu_claims():
+    # amd.com/gpu is exclusive: a bound GPU is ineligible for other pods.
     _, ledger = make_ledger(count=3)
-    ledger.reserve("fill0", [0], 200 * GIB)  # headroom 88
-    ledger.reserve("fill1", [1], 100 * GIB)  # headroom 188
+    binder = Binder(ledger)
+    first = binder.bind(BindRequest("p1", 1, GIB))
+    second = binder.bind(BindRequest("p2", 1, GIB))
+    assert set(first).isdisjoint(second)
+    with pytest.raises(ValueError, match="already bound"):
+        ledger.reserve("p3", first, GIB)
+
+
+def test_best_fit_prefers_smallest_sufficient_headroom():
+    inv, ledger = make_ledger(count=3)
+    # Live VRAM use (other tenants/system) differentiates headroom.
+    inv.gpus[0].vram_used_bytes = 200 * GIB  # headroom 88
+    inv.gpus[1].vram_used_bytes = 100 * GIB  # headroom 188
+    ledger.sync_inventory()
     binder = Binder(ledger)
     # 50 GiB fits on GPU 0's 88 GiB → best-fit picks 0
     chosen = binder.select(BindRequest("p", 1, 50 * GIB, max_cost=1.0))
@@ -53,15 +66,15 @@ def test_best_fit_prefers_smallest_sufficient_headroom():
 
 
 def test_cost_gating_respects_max_gpu_price():
-    _, ledger = make_ledger(count=2)
-    # Make GPU 0 heavily reserved → cost = 0.5 * (250/288) ≈ 0.43
-    ledger.reserve("fill", [0], 250 * GIB)
+    inv, ledger = make_ledger(count=2)
+    inv.gpus[0].busy_percent = 60  # cost 0.3
+    ledger.sync_inventory()
     binder = Binder(ledger)
     chosen = binder.select(BindRequest("p", 1, GIB, max_cost=0.2))
     assert chosen == [1]  # GPU 0 too "expensive"
-    # Both eligible at default 0.5; best fit picks the fuller GPU 0
+    # Both eligible at default 0.5; equal headroom → lower cost wins
     chosen = binder.select(BindRequest("p", 1, GIB, max_cost=0.5))
-    assert chosen == [0]
+    assert chosen == [1]
 
 
 def test_multi_gpu_set_is_xgmi_connected():
@@ -88,12 +101,10 @@ def test_placement_error_when_full():
     _, ledger = make_ledger(count=2)
     binder = Binder(ledger)
     binder.bind(BindRequest("p1", 2, 0))
-    # GPUs are reserved but memory-free; a 2-GPU all-memory pod still fits
-    # by headroom, so fill memory instead:
-    binder.unbind("p1")
-    binder.bind(BindRequest("p2", 2, 2 * 280 * GIB))
     with pytest.raises(PlacementError):
-        binder.bind(BindRequest("p3", 1, 100 * GIB))
+        binder.bind(BindRequest("p2", 1, GIB))
+    binder.unbind("p1")
+    assert len(binder.bind(BindRequest("p3", 2, 2 * 280 * GIB))) == 2
 
 
 def test_unhealthy_gpu_not_schedulable():
