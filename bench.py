@@ -1,0 +1,250 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GPU-pod scheduling throughput + pod-Ready latency.
+
+Measures the BASELINE.json metric ("p50 pod-Ready latency + GPU-pods
+scheduled/sec at 1/2/4/8 concurrent") on one node: an in-process apiserver
+(hermetic, no cluster available in this environment — the reference's own
+integration test needs a live cluster and cloud account and publishes no
+numbers), the real informer → PodController → Provider pipeline,
+and the real ProcessRuntime launching one HIP `podworker` per pod bound to
+its GPU via ROCR_VISIBLE_DEVICES.
+
+One *step* = a wave of N pods (one per GPU, each requesting 1×amd.com/gpu):
+create all → wait all Ready (podworker initializes HIP on its bound GPU,
+runs a gfx950 kernel, then signals readiness) → delete all → wait all gone.
+Throughput counts the FULL lifecycle; latency is per-pod create→Ready.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  torchrun --nnodes=1 --nproc-per-node N ... bench.py --gpus N ...
+Under torchrun, rank 0 runs the kubelet stack managing GPUs 0..N-1 (one
+kubelet per node is the product architecture); other ranks only participate
+in barriers. Weak scaling: each extra GPU adds one pod per wave.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import tempfile
+import time
+
+REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO_ROOT)
+
+
+def _dist_env():
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    return rank, world
+
+
+def _barrier(dist, device_sync: bool):
+    if device_sync:
+        import torch
+
+        torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    if device_sync:
+        import torch
+
+        torch.cuda.synchronize()
+
+
+def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
+              hold_pods: bool = True):
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.client import NotFoundError
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.kube.objects import parse_rfc3339
+
+    state_dir = tempfile.mkdtemp(prefix="amdvk-bench-")
+    cfg = Config(
+        state_dir=state_dir,
+        gpu_count_override=(-1 if use_gpu else max(n_gpus, 8)),
+        pending_retry_interval_s=0.2,
+        reconcile_interval_s=30.0,
+        notify_interval_s=0.0,  # push path only; no polling needed
+        pod_controller_workers=max(8, n_gpus),
+    )
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    if stack.ledger is not None and stack.ledger.total_gpus() < n_gpus:
+        raise SystemExit(
+            f"bench needs {n_gpus} GPUs; inventory has {stack.ledger.total_gpus()}"
+        )
+    stack.start(serve_http=False)
+
+    args = ["--hold"] if hold_pods else ["--run-for", "0.05"]
+    if use_gpu:
+        args = ["--expect-gpus", "1"] + args
+
+    def make_pod(name):
+        return {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": name, "namespace": "default"},
+            "spec": {
+                "nodeName": cfg.node_name,
+                "containers": [{
+                    "name": "main",
+                    "image": "amdvk/podworker:bench",
+                    "command": ["podworker"],
+                    "args": args,
+                    "resources": {"limits": {"amd.com/gpu": "1"}},
+                }],
+            },
+        }
+
+    def pod_ready(name):
+        try:
+            pod = kube.get_pod("default", name)
+        except NotFoundError:
+            return None
+        conds = {c["type"]: c["status"]
+                 for c in pod.get("status", {}).get("conditions", [])}
+        return pod if conds.get("Ready") == "True" else None
+
+    def pod_gone(name):
+        try:
+            kube.get_pod("default", name)
+            return False
+        except NotFoundError:
+            return True
+
+    def wave(tag, timeout_s=120.0):
+        names = [f"bench-{tag}-{i}" for i in range(n_gpus)]
+        created_at = {}
+        ready_at = {}
+        for name in names:
+            created_at[name] = time.monotonic()
+            kube.create_pod("default", make_pod(name))
+        deadline = time.monotonic() + timeout_s
+        pending = set(names)
+        while pending and time.monotonic() < deadline:
+            for name in list(pending):
+                if pod_ready(name) is not None:
+                    ready_at[name] = time.monotonic()
+                    pending.remove(name)
+            if pending:
+                time.sleep(0.002)
+        if pending:
+            raise SystemExit(f"pods never became Ready: {sorted(pending)}")
+        for name in names:
+            kube.delete_pod("default", name)
+        pending = set(names)
+        while pending and time.monotonic() < deadline:
+            for name in list(pending):
+                if pod_gone(name):
+                    pending.remove(name)
+            if pending:
+                time.sleep(0.002)
+        if pending:
+            raise SystemExit(f"pods never finalized: {sorted(pending)}")
+        return [(ready_at[n] - created_at[n]) for n in names]
+
+    try:
+        for w in range(warmup):
+            wave(f"w{w}")
+        latencies = []
+        t0 = time.monotonic()
+        for k in range(steps):
+            latencies.extend(wave(f"s{k}"))
+        elapsed = time.monotonic() - t0
+    finally:
+        stack.stop()
+
+    lat_ms = sorted(x * 1000 for x in latencies)
+    return {
+        "elapsed_s": elapsed,
+        "pods": n_gpus * steps,
+        "pods_per_sec": (n_gpus * steps) / elapsed,
+        "p50_ready_ms": statistics.median(lat_ms),
+        "p99_ready_ms": lat_ms[min(len(lat_ms) - 1, int(len(lat_ms) * 0.99))],
+        "max_ready_ms": lat_ms[-1],
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--mode", choices=["auto", "gpu", "cpu"], default="auto")
+    args = ap.parse_args()
+
+    rank, world = _dist_env()
+    dist = None
+    device_sync = False
+    use_gpu = False
+    if args.mode == "auto":
+        try:
+            import torch
+
+            use_gpu = torch.cuda.is_available()
+        except ImportError:
+            use_gpu = False
+    else:
+        use_gpu = args.mode == "gpu"
+
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        # Control plane has no collectives; gloo barriers coordinate ranks
+        # without claiming HBM that the benchmarked pods need.
+        tdist.init_process_group(backend="gloo")
+        dist = tdist
+        device_sync = use_gpu
+        if use_gpu:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+
+    _barrier(dist, device_sync)
+    result = None
+    if rank == 0:
+        result = run_bench(args.gpus, args.steps, args.warmup, use_gpu)
+    _barrier(dist, device_sync)
+
+    if rank == 0 and result is not None:
+        out = {
+            "metric": "gpu_pods_scheduled_per_sec",
+            "value": round(result["pods_per_sec"], 3),
+            "unit": "pods/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(result["elapsed_s"] / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "n/a",  # control-plane benchmark: no tensor math
+            "data": "synthetic",
+            "p50_pod_ready_ms": round(result["p50_ready_ms"], 3),
+            "p99_pod_ready_ms": round(result["p99_ready_ms"], 3),
+            "config": {
+                "model": "gpu-pod-lifecycle",
+                "global_batch": args.gpus,  # pods per wave
+                "seq_len": 0,
+                "parallelism": f"1 kubelet x {args.gpus} GPUs, "
+                               f"{max(8, args.gpus)} sync workers",
+                "workload": ("podworker HIP init+kernel per pod (gfx950)"
+                             if use_gpu else "podworker CPU mode"),
+                "lifecycle": "create->bind->launch->Ready->delete->finalized",
+                "pods_total": result["pods"],
+                "p50_pod_ready_ms": round(result["p50_ready_ms"], 3),
+                "p99_pod_ready_ms": round(result["p99_ready_ms"], 3),
+                "reference_poll_floor_ms": 10000.0,  # kubelet.go:719 10 s tick
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
