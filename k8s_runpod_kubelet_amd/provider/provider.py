@@ -393,6 +393,27 @@ class Provider:
                 self._sync_pod_status(key)
             except Exception:
                 log.exception("event-driven sync failed", extra={"pod": key})
+        # A terminal instance freed its GPUs: place pending pods NOW instead
+        # of on the next retry tick (this is what keeps back-to-back waves —
+        # BASELINE config 5's FIFO burst — GPU-bound rather than tick-bound).
+        try:
+            status = self.runtime.get_status(instance_id)
+        except Exception:
+            status = PodStatus.NOT_FOUND
+        if status in (PodStatus.EXITED, PodStatus.TERMINATED, PodStatus.NOT_FOUND):
+            self._nudge_pending()
+
+    def _nudge_pending(self) -> None:
+        with self._pods_lock:
+            has_pending = any(
+                not info.instance_id and not info.deploying
+                for info in self._pod_status.values()
+            )
+        if has_pending:
+            try:
+                self.process_pending_pods()
+            except Exception:
+                log.exception("event-driven pending processing failed")
 
     def update_all_pod_statuses(self) -> None:
         with self._pods_lock:

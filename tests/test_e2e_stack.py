@@ -233,3 +233,31 @@ def test_restart_adoption_end_to_end(tmp_state_dir):
         assert wait_until(lambda: gone(kube, "persist"), timeout_s=15)
     finally:
         s2.stop()
+
+
+def test_event_driven_pending_placement(tmp_state_dir):
+    """When a GPU frees, pending pods place on the exit *event*, not on the
+    retry tick (interval here is 999 s — a tick-based design would hang)."""
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=2,
+                 notify_interval_s=0, pending_retry_interval_s=999)
+    kube = FakeKube()
+    s = build_stack(cfg, client=kube)
+    s.runtime.enable_cgroups = False
+    s.start(serve_http=False)
+    try:
+        for n in ("h0", "h1"):
+            kube.create_pod("default", make_pod(n, gpus=1, command=["podworker"],
+                                                args=["--hold"]))
+        for n in ("h0", "h1"):
+            assert wait_until(lambda n=n: ready(kube, n), timeout_s=10), n
+        kube.create_pod("default", make_pod("queued", gpus=1,
+                                            command=["podworker"],
+                                            args=["--hold"]))
+        time.sleep(0.3)
+        assert ready(kube, "queued") is None  # both GPUs taken
+        t0 = time.time()
+        kube.delete_pod("default", "h0")
+        assert wait_until(lambda: ready(kube, "queued"), timeout_s=10)
+        assert time.time() - t0 < 5  # placed via exit event, no 999 s tick
+    finally:
+        s.stop()
