@@ -78,11 +78,13 @@ class KubeletApiServer:
                     if not commands:
                         self._respond(400, b"missing command")
                         return
-                    self._respond(
-                        501,
-                        b"interactive exec streaming is not supported; "
-                        b"use containerLogs for output",
+                    namespace, pod = parts[1], parts[2]
+                    code, output = outer.provider.run_in_container(
+                        namespace, pod, commands
                     )
+                    body = json.dumps({"exitCode": code, "output": output}).encode()
+                    self._respond(200 if code == 0 else 500, body,
+                                  "application/json")
                 else:
                     self._respond(404, b"not found")
 

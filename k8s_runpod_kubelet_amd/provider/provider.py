@@ -900,6 +900,22 @@ class Provider:
             return ""
         return self.runtime.get_logs(info.instance_id, container, tail)
 
+    def run_in_container(self, namespace: str, name: str, command: List[str],
+                         timeout_s: float = 30.0) -> tuple:
+        """One-shot exec in the pod's environment (GPU binding included).
+
+        The reference returns "not supported by RunPod" (kubelet.go:2027-2047);
+        local pods make a non-interactive exec cheap. Returns
+        (exit_code, combined_output)."""
+        with self._pods_lock:
+            info = self._pod_status.get(f"{namespace}-{name}")
+        if info is None:
+            return 127, f"pod {namespace}/{name} not tracked by this node"
+        exec_fn = getattr(self.runtime, "exec_in_instance", None)
+        if exec_fn is None:
+            return 501, "runtime does not support exec"
+        return exec_fn(info.instance_id, command, timeout_s)
+
     def instance_info(self, namespace: str, name: str) -> Optional[InstanceInfo]:
         with self._pods_lock:
             return self._pod_status.get(f"{namespace}-{name}")

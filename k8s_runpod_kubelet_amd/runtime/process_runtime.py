@@ -390,6 +390,42 @@ class ProcessRuntime(Runtime):
             text = "\n".join(text.splitlines()[-tail:]) + "\n"
         return text
 
+    def exec_in_instance(self, instance_id: str, command: List[str],
+                         timeout_s: float = 30.0) -> tuple:
+        """Non-interactive exec with the instance's environment (same GPU
+        binding, same cgroup). Returns (exit_code, combined_output)."""
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        if inst is None:
+            return 127, f"instance {instance_id} not found"
+        env = dict(os.environ)
+        env.pop("ROCR_VISIBLE_DEVICES", None)
+        env.pop("HIP_VISIBLE_DEVICES", None)
+        env.update(inst.params.env)
+        env.update(device_env(inst.gpu_indices))
+        out_path = self.logs_dir / f".exec-{inst.id}-{secrets.token_hex(4)}.log"
+        try:
+            pid, pidfd, _ = self._native.launch_process(
+                list(command), [f"{k}={v}" for k, v in env.items()],
+                "", str(out_path), str(out_path), inst.cgroup_dir, True, False,
+            )
+        except RuntimeError as exc:
+            return 127, str(exc)
+        loop = self._native.EventLoop()
+        loop.add_process(pid, pidfd, -1, 0)
+        deadline = time.time() + timeout_s
+        exit_code = None
+        while time.time() < deadline and exit_code is None:
+            for ev in loop.poll(100):
+                if ev.type == "exited":
+                    exit_code = ev.exit_code
+        if exit_code is None:
+            self._native.signal_process(pid, 9, True)
+            exit_code = 124
+        output = out_path.read_text(errors="replace") if out_path.exists() else ""
+        out_path.unlink(missing_ok=True)
+        return exit_code, output
+
     # ------------- persistence / adoption -------------
 
     def _persist(self, inst: Instance) -> None:

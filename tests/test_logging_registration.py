@@ -1,0 +1,84 @@
+import json
+import logging
+
+from k8s_runpod_kubelet_amd.logging_setup import (
+    JSONFormatter,
+    KVFormatter,
+    MultiHandler,
+    initialize_logger,
+)
+from k8s_runpod_kubelet_amd.provider.registration import Registrar, build_payload
+
+
+def test_multihandler_fans_out():
+    records_a, records_b = [], []
+
+    class Sink(logging.Handler):
+        def __init__(self, store, level=logging.INFO):
+            super().__init__(level)
+            self.store = store
+
+        def emit(self, record):
+            self.store.append(record.getMessage())
+
+    mh = MultiHandler([Sink(records_a), Sink(records_b, level=logging.ERROR)])
+    logger = logging.Logger("t")
+    logger.addHandler(mh)
+    logger.info("hello")
+    logger.error("bad")
+    assert records_a == ["hello", "bad"]
+    assert records_b == ["bad"]  # per-handler level respected
+
+
+def test_log_level_applied(tmp_path):
+    # reference quirk: --log-level parsed but never applied (main.go:69);
+    # here it must take effect.
+    root = initialize_logger("error")
+    assert root.level == logging.ERROR
+    root = initialize_logger("debug")
+    assert root.level == logging.DEBUG
+    initialize_logger("info")
+
+
+def test_kv_and_json_formatters():
+    record = logging.LogRecord("x", logging.INFO, "f.py", 1, "msg here", (), None)
+    record.pod = "ns/p"
+    text = KVFormatter().format(record)
+    assert 'msg="msg here"' in text and "pod='ns/p'" in text
+    payload = json.loads(JSONFormatter().format(record))
+    assert payload["msg"] == "msg here"
+    assert payload["pod"] == "ns/p"
+
+
+def test_json_file_sink(tmp_path):
+    path = tmp_path / "log.json"
+    initialize_logger("info", str(path))
+    logging.getLogger("t2").info("to file", extra={"k": 1})
+    initialize_logger("info")  # reset handlers / flush
+    lines = [json.loads(l) for l in path.read_text().splitlines()]
+    assert any(l["msg"] == "to file" and l["k"] == 1 for l in lines)
+
+
+def test_registration_payload_shape():
+    p = build_payload("node1", "kube-system", 8)
+    # reference RegistrationPayload fields (kubelet.go:88-127)
+    assert set(p) == {"clusterName", "namespace", "nodeName", "version",
+                      "capabilities", "metadata"}
+    assert p["metadata"]["gpuCount"] == 8
+
+
+def test_registrar_disabled_by_default():
+    r = Registrar("", "n", "ns", 8)
+    assert r.register() is True  # no endpoint: registration is a no-op
+    r.start_heartbeat()
+    assert r._ticker is None
+    r.stop_heartbeat()
+
+
+def test_registrar_failure_not_fatal(monkeypatch):
+    # endpoint set but unreachable: register() returns False, no raise
+    # (unlike the reference, which fails provider construction,
+    # kubelet.go:369-371)
+    r = Registrar("http://127.0.0.1:1", "n", "ns", 8, heartbeat_interval_s=0)
+    assert r.register() is False
+    r.stop_heartbeat()

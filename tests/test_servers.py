@@ -1,0 +1,101 @@
+"""Health server + kubelet API server tests (reference health.go:11-74 and
+main.go:217-248 surfaces)."""
+
+import json
+import urllib.request
+
+import pytest
+
+from k8s_runpod_kubelet_amd.config import Config
+from k8s_runpod_kubelet_amd.kube.apiserver import KubeletApiServer
+from k8s_runpod_kubelet_amd.provider.provider import Provider
+from k8s_runpod_kubelet_amd.runtime.fake import FakeRuntime
+from k8s_runpod_kubelet_amd.server.health import HealthServer
+from tests.conftest import make_pod
+
+
+def get(url, expect_error=False):
+    try:
+        with urllib.request.urlopen(url, timeout=5) as resp:
+            return resp.status, resp.read().decode()
+    except urllib.error.HTTPError as exc:
+        return exc.code, exc.read().decode()
+
+
+@pytest.fixture
+def provider(fake_kube):
+    cfg = Config(notify_interval_s=0)
+    rt = FakeRuntime(gpu_count=8)
+    prov = Provider(fake_kube, cfg, rt)
+    yield prov, rt, fake_kube
+    prov.stop()
+
+
+def test_health_endpoints(provider):
+    prov, rt, _ = provider
+    hs = HealthServer("127.0.0.1:0", prov.ping)
+    hs.start()
+    base = f"http://127.0.0.1:{hs.port}"
+    try:
+        assert get(f"{base}/healthz") == (200, "ok")
+        code, _ = get(f"{base}/readyz")
+        assert code == 200
+        rt.set_healthy(False)
+        code, body = get(f"{base}/readyz")
+        assert code == 503 and "not ready" in body
+        hs.set_alive(False)
+        assert get(f"{base}/healthz")[0] == 503
+        hs.set_alive(True)
+        code, body = get(f"{base}/metrics")
+        assert code == 200 and "amdvk_pod_ready_seconds" in body
+        code, body = get(f"{base}/debug/threads")
+        assert code == 200 and "MainThread" in body
+        assert get(f"{base}/nope")[0] == 404
+    finally:
+        hs.stop()
+
+
+def test_kubelet_api(provider, process_runtime):
+    prov, _, kube = provider
+    # swap in the process runtime so logs/exec are real
+    prov.runtime = process_runtime
+    pod = make_pod("api1", command=["podworker"], args=["--hold"])
+    kube.create_pod("default", pod)
+    prov.create_pod(kube.get_pod("default", "api1"))
+
+    srv = KubeletApiServer(prov, "127.0.0.1", 0)
+    srv.start()
+    base = f"http://127.0.0.1:{srv.port}"
+    try:
+        import time
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            code, body = get(f"{base}/containerLogs/default/api1/main")
+            if "podworker: ready" in body:
+                break
+            time.sleep(0.05)
+        assert "podworker: ready" in body
+
+        code, body = get(f"{base}/pods")
+        assert code == 200
+        pods = json.loads(body)
+        assert pods["kind"] == "PodList"
+        assert any(p["metadata"]["name"] == "api1" for p in pods["items"])
+
+        # one-shot exec in the pod's env (reference: "not supported")
+        req = urllib.request.Request(
+            f"{base}/exec/default/api1/main?command=/bin/sh&command=-c&command=echo%20hi",
+            method="POST")
+        with urllib.request.urlopen(req, timeout=10) as resp:
+            out = json.loads(resp.read())
+        assert out["exitCode"] == 0
+        assert "hi" in out["output"]
+
+        assert get(f"{base}/healthz") == (200, "ok")
+        code, body = get(f"{base}/stats/summary")
+        assert code == 200
+    finally:
+        srv.stop()
+        pod_obj = prov.get_pod("default", "api1")
+        if pod_obj:
+            prov.delete_pod(pod_obj)
