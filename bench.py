@@ -56,7 +56,7 @@ def _barrier(dist, device_sync: bool):
 
 
 def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
-              hold_pods: bool = True):
+              hold_pods: bool = True, burst_size: int = 0):
     from k8s_runpod_kubelet_amd.app import build_stack
     from k8s_runpod_kubelet_amd.config import Config
     from k8s_runpod_kubelet_amd.kube.client import NotFoundError
@@ -147,7 +147,62 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
             raise SystemExit(f"pods never finalized: {sorted(pending)}")
         return [(ready_at[n] - created_at[n]) for n in names]
 
+    def burst(tag, count, timeout_s=600.0):
+        """BASELINE config 5: queue `count` run-to-completion pods at once,
+        FIFO-drain across the node's GPUs (placement backpressure)."""
+        names = [f"burst-{tag}-{i:03d}" for i in range(count)]
+        t0 = time.monotonic()
+        for name in names:
+            pod = make_pod(name)
+            pod["spec"]["containers"][0]["args"] = (
+                (["--expect-gpus", "1"] if use_gpu else []) + ["--run-for", "0.05"]
+            )
+            kube.create_pod("default", pod)
+        deadline = time.monotonic() + timeout_s
+        pending = set(names)
+        while pending and time.monotonic() < deadline:
+            for name in list(pending):
+                try:
+                    pod = kube.get_pod("default", name)
+                except NotFoundError:
+                    pending.remove(name)
+                    continue
+                if pod.get("status", {}).get("phase") in ("Succeeded", "Failed"):
+                    pending.remove(name)
+            if pending:
+                time.sleep(0.005)
+        if pending:
+            raise SystemExit(f"burst never drained: {len(pending)} left")
+        drain = time.monotonic() - t0
+        for name in names:
+            try:
+                kube.delete_pod("default", name)
+            except NotFoundError:
+                pass
+        t_gone = time.monotonic() + timeout_s
+        while time.monotonic() < t_gone:
+            if all(pod_gone(n) for n in names):
+                break
+            time.sleep(0.005)
+        return drain
+
     try:
+        if burst_size > 0:
+            for w in range(warmup):
+                burst(f"w{w}", max(n_gpus, burst_size // 4))
+            t0 = time.monotonic()
+            drains = [burst(f"s{k}", burst_size) for k in range(steps)]
+            elapsed = time.monotonic() - t0
+            total_pods = burst_size * steps
+            return {
+                "elapsed_s": elapsed,
+                "pods": total_pods,
+                "pods_per_sec": total_pods / sum(drains),
+                "p50_ready_ms": statistics.median(drains) * 1000 / burst_size,
+                "p99_ready_ms": max(drains) * 1000 / burst_size,
+                "max_ready_ms": max(drains) * 1000,
+                "drain_s_per_burst": statistics.median(drains),
+            }
         for w in range(warmup):
             wave(f"w{w}")
         latencies = []
@@ -175,6 +230,8 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--mode", choices=["auto", "gpu", "cpu"], default="auto")
+    ap.add_argument("--burst", type=int, default=0,
+                    help="BASELINE config 5: pods per burst (0 = wave mode)")
     args = ap.parse_args()
 
     rank, world = _dist_env()
@@ -207,7 +264,8 @@ def main():
     _barrier(dist, device_sync)
     result = None
     if rank == 0:
-        result = run_bench(args.gpus, args.steps, args.warmup, use_gpu)
+        result = run_bench(args.gpus, args.steps, args.warmup, use_gpu,
+                           burst_size=args.burst)
     _barrier(dist, device_sync)
 
     if rank == 0 and result is not None:
@@ -234,7 +292,9 @@ def main():
                                f"{max(8, args.gpus)} sync workers",
                 "workload": ("podworker HIP init+kernel per pod (gfx950)"
                              if use_gpu else "podworker CPU mode"),
-                "lifecycle": "create->bind->launch->Ready->delete->finalized",
+                "lifecycle": ("32-pod FIFO burst drain" if args.burst
+                              else "create->bind->launch->Ready->delete->finalized"),
+                "burst_size": args.burst,
                 "pods_total": result["pods"],
                 "p50_pod_ready_ms": round(result["p50_ready_ms"], 3),
                 "p99_pod_ready_ms": round(result["p99_ready_ms"], 3),
