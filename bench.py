@@ -214,6 +214,10 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
         stack.stop()
 
     lat_ms = sorted(x * 1000 for x in latencies)
+    from k8s_runpod_kubelet_amd.server import metrics as m
+
+    deploy_sum = m.deploy_seconds._sum.get()
+    deploy_count = sum(b.get() for b in m.deploy_seconds._buckets) or 1
     return {
         "elapsed_s": elapsed,
         "pods": n_gpus * steps,
@@ -221,6 +225,9 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
         "p50_ready_ms": statistics.median(lat_ms),
         "p99_ready_ms": lat_ms[min(len(lat_ms) - 1, int(len(lat_ms) * 0.99))],
         "max_ready_ms": lat_ms[-1],
+        # control-plane share of Ready latency: translate+bind+launch
+        # (the rest is the workload's own HIP context init)
+        "mean_deploy_ms": deploy_sum / max(deploy_count, 1) * 1000,
     }
 
 
@@ -296,6 +303,7 @@ def main():
                               else "create->bind->launch->Ready->delete->finalized"),
                 "burst_size": args.burst,
                 "pods_total": result["pods"],
+                "mean_deploy_ms": round(result.get("mean_deploy_ms", 0), 3),
                 "p50_pod_ready_ms": round(result["p50_ready_ms"], 3),
                 "p99_pod_ready_ms": round(result["p99_ready_ms"], 3),
                 "reference_poll_floor_ms": 10000.0,  # kubelet.go:719 10 s tick

@@ -207,9 +207,13 @@ class Provider:
             return
         try:
             self.deploy_pod(pod)
+        except (PlacementError, ValidationError) as exc:
+            # Normal backpressure: no GPU free right now — the pod queues and
+            # places on the next free event (reference kubelet.go:412-415
+            # semantics: CreatePod returns nil, retry loop re-deploys).
+            log.info("pod queued: no placement yet",
+                     extra={"pod": key, "reason": str(exc)})
         except Exception as exc:
-            # Reference behavior: CreatePod still returns nil on deploy
-            # failure (kubelet.go:412-415); the pending loop retries.
             log.warning("deploy failed; pod stays pending",
                         extra={"pod": key, "err": str(exc)})
 
