@@ -53,7 +53,7 @@ def test_ready_pipe_protocol(native, tmp_path):
 
 
 def test_exec_failure_reported(native, tmp_path):
-    with pytest.raises(RuntimeError, match="execvpe"):
+    with pytest.raises(RuntimeError, match="spawn"):
         native.launch_process(
             ["/no/such/binary"], base_env(), "", "", "", "", True, True)
 
