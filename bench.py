@@ -216,9 +216,10 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
     lat_ms = sorted(x * 1000 for x in latencies)
     from k8s_runpod_kubelet_amd.server import metrics as m
 
-    deploy_sum = m.deploy_seconds._sum.get()
-    deploy_count = sum(b.get() for b in m.deploy_seconds._buckets) or 1
     return {
+        "mean_translate_ms": m.hist_mean_ms(m.translate_seconds),
+        "mean_bind_ms": m.hist_mean_ms(m.bind_seconds),
+        "mean_launch_ms": m.hist_mean_ms(m.launch_seconds),
         "elapsed_s": elapsed,
         "pods": n_gpus * steps,
         "pods_per_sec": (n_gpus * steps) / elapsed,
@@ -227,7 +228,7 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
         "max_ready_ms": lat_ms[-1],
         # control-plane share of Ready latency: translate+bind+launch
         # (the rest is the workload's own HIP context init)
-        "mean_deploy_ms": deploy_sum / max(deploy_count, 1) * 1000,
+        "mean_deploy_ms": m.hist_mean_ms(m.deploy_seconds),
     }
 
 
@@ -304,6 +305,9 @@ def main():
                 "burst_size": args.burst,
                 "pods_total": result["pods"],
                 "mean_deploy_ms": round(result.get("mean_deploy_ms", 0), 3),
+                "mean_translate_ms": round(result.get("mean_translate_ms", 0), 3),
+                "mean_bind_ms": round(result.get("mean_bind_ms", 0), 3),
+                "mean_launch_ms": round(result.get("mean_launch_ms", 0), 3),
                 "p50_pod_ready_ms": round(result["p50_ready_ms"], 3),
                 "p99_pod_ready_ms": round(result["p99_ready_ms"], 3),
                 "reference_poll_floor_ms": 10000.0,  # kubelet.go:719 10 s tick

@@ -331,6 +331,7 @@ class Provider:
 
         t0 = time.monotonic()
         params = prepare_deploy_params(pod, self.client, self.config, self.catalog)
+        metrics.translate_seconds.observe(time.monotonic() - t0)
         env_count = len(params.env)
         log.info(
             "deploying pod",

@@ -95,8 +95,11 @@ class ProcessRuntime(Runtime):
     # ------------- deploy -------------
 
     def deploy(self, params: DeployParams) -> DetailedStatus:
+        from ..server import metrics
+
         instance_id = "amdvk-" + secrets.token_hex(6)
         gpu_indices: List[int] = []
+        t_bind = time.monotonic()
         if params.gpu_count > 0:
             req = BindRequest(
                 pod_key=params.pod_key,
@@ -105,6 +108,8 @@ class ProcessRuntime(Runtime):
                 max_cost=params.max_gpu_cost,
             )
             gpu_indices = self.binder.bind(req)  # raises PlacementError when full
+        metrics.bind_seconds.observe(time.monotonic() - t_bind)
+        t_launch = time.monotonic()
 
         inst = Instance(
             id=instance_id,
@@ -134,6 +139,7 @@ class ProcessRuntime(Runtime):
                 self._native.cgroup_remove(cgroup_dir)
             raise
 
+        metrics.launch_seconds.observe(time.monotonic() - t_launch)
         with self._lock:
             self._instances[instance_id] = inst
         self._persist(inst)

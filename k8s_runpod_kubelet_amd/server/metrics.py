@@ -30,12 +30,30 @@ pod_ready_seconds = Histogram(
     buckets=(0.005, 0.01, 0.025, 0.05, 0.1, 0.25, 0.5, 1, 2.5, 5, 10, 30, 60, 300),
     registry=registry,
 )
+_PHASE_BUCKETS = (0.0005, 0.001, 0.0025, 0.005, 0.01, 0.025, 0.05, 0.1,
+                  0.25, 0.5, 1, 2.5, 5)
 deploy_seconds = Histogram(
     "amdvk_deploy_seconds",
     "Spec translation + bind + launch latency",
-    buckets=(0.001, 0.0025, 0.005, 0.01, 0.025, 0.05, 0.1, 0.25, 0.5, 1, 2.5, 5),
+    buckets=_PHASE_BUCKETS,
     registry=registry,
 )
+# Per-phase latency histograms (the reference has no tracing at
+# all; the north-star metric is latency, so every deploy phase is measured).
+translate_seconds = Histogram(
+    "amdvk_translate_seconds", "Spec translation (annotations/env/ports)",
+    buckets=_PHASE_BUCKETS, registry=registry)
+bind_seconds = Histogram(
+    "amdvk_bind_seconds", "GPU set selection + ledger reservation",
+    buckets=_PHASE_BUCKETS, registry=registry)
+launch_seconds = Histogram(
+    "amdvk_launch_seconds", "cgroup setup + posix_spawn of containers",
+    buckets=_PHASE_BUCKETS, registry=registry)
+
+
+def hist_mean_ms(h) -> float:
+    count = sum(b.get() for b in h._buckets)
+    return (h._sum.get() / count * 1000.0) if count else 0.0
 
 gpu_hbm_free = Gauge("amdvk_gpu_hbm_free_bytes", "Free HBM headroom per GPU",
                      ["gpu"], registry=registry)
