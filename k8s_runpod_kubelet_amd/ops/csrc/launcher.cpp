@@ -25,6 +25,7 @@
 #include <unistd.h>
 
 #include <cerrno>
+#include <chrono>
 #include <cstdio>
 #include <fstream>
 #include <stdexcept>
@@ -128,6 +129,7 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
   posix_spawnattr_setflags(&attr, flags);
 
   pid_t pid = -1;
+  auto t0 = std::chrono::steady_clock::now();
   int rc = posix_spawnp(&pid, argv[0], &fa, &attr, argv.data(), envp.data());
   posix_spawn_file_actions_destroy(&fa);
   posix_spawnattr_destroy(&attr);
@@ -145,6 +147,9 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
     // posix_spawn has no pre-exec hook for it.)
     WriteFileString(spec.cgroup_dir + "/cgroup.procs", std::to_string(pid));
   }
+  res.spawn_ns = std::chrono::duration_cast<std::chrono::nanoseconds>(
+                     std::chrono::steady_clock::now() - t0)
+                     .count();
 
   res.pid = pid;
   res.pidfd = PidfdOpen(pid);

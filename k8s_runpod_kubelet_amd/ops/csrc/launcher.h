@@ -23,6 +23,10 @@ struct LaunchResult {
   int64_t pid = -1;
   int pidfd = -1;
   int ready_fd = -1;  // parent read end of the readiness pipe (-1 if disabled)
+  // Spawn cost measured around posix_spawnp + cgroup migration on the native
+  // side: Python-side wall timing overstates it by ~100 ms whenever the
+  // launching thread loses the GIL to the already-running child.
+  int64_t spawn_ns = 0;
   std::string error;
 };
 

@@ -29,7 +29,7 @@ def drain(loop, want, timeout_s=5.0):
 
 def test_launch_exit_code(native, tmp_path):
     loop = native.EventLoop()
-    pid, pidfd, ready_fd = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
         ["/bin/sh", "-c", "exit 7"], base_env(),
         "", str(tmp_path / "out.log"), "", "", True, True)
     loop.add_process(pid, pidfd, ready_fd, 1)
@@ -41,7 +41,7 @@ def test_launch_exit_code(native, tmp_path):
 
 def test_ready_pipe_protocol(native, tmp_path):
     loop = native.EventLoop()
-    pid, pidfd, ready_fd = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
         ["/bin/bash", "-c", 'echo READY >&$AMDVK_READY_FD; sleep 0.1'],
         base_env(), "", str(tmp_path / "out.log"), "", "", True, True)
     loop.add_process(pid, pidfd, ready_fd, 2)
@@ -60,7 +60,7 @@ def test_exec_failure_reported(native, tmp_path):
 
 def test_signal_process(native, tmp_path):
     loop = native.EventLoop()
-    pid, pidfd, ready_fd = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
         ["/bin/sleep", "30"], base_env(), "", str(tmp_path / "o.log"), "", "",
         True, True)
     loop.add_process(pid, pidfd, ready_fd, 3)
@@ -73,7 +73,7 @@ def test_signal_process(native, tmp_path):
 def test_stdout_redirect(native, tmp_path):
     log = tmp_path / "redir.log"
     loop = native.EventLoop()
-    pid, pidfd, ready_fd = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
         ["/bin/sh", "-c", "echo hello-out; echo hello-err >&2"],
         base_env(), "", str(log), "", "", True, True)
     loop.add_process(pid, pidfd, ready_fd, 4)
@@ -83,7 +83,7 @@ def test_stdout_redirect(native, tmp_path):
 
 
 def test_open_pidfd_adoption(native):
-    pid, pidfd, ready_fd = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
         ["/bin/sleep", "0.2"], base_env(), "", "", "", "", True, False)
     os.close(pidfd)
     # Re-open (adoption path) and watch the exit through a fresh loop.
