@@ -25,6 +25,7 @@ on container start, not on placement.
 from __future__ import annotations
 
 import logging
+import time
 from dataclasses import dataclass
 from typing import List, Optional, Sequence
 
@@ -33,6 +34,10 @@ from .ledger import GpuState, Ledger
 log = logging.getLogger("gpu.binder")
 
 GIB = 1024**3
+
+# How long a freed GPU is considered "settling" (see select()); the measured
+# init penalty decays within ~1 s of the previous pod's exit.
+SETTLE_S = 1.0
 
 
 @dataclass
@@ -74,8 +79,16 @@ class Binder:
                 f"need {req.gpu_count} GPUs with {need / GIB:.0f} GiB headroom "
                 f"and cost<={req.max_cost}; only {len(eligible)}/{len(states)} eligible"
             )
-        # Best-fit: smallest sufficient headroom first, then lowest cost.
-        eligible.sort(key=lambda s: (s.headroom_bytes, s.cost(), s.gpu.index))
+        # Settled GPUs first: a GPU freed < SETTLE_S ago still runs the
+        # previous pod's asynchronous KFD context teardown, which serializes
+        # against the next pod's HIP init (~300 ms vs ~160 ms measured on
+        # MI355X, profiles/pw_timing.txt). Within each group, best-fit:
+        # smallest sufficient headroom first, then lowest cost.
+        now = time.monotonic()
+        eligible.sort(key=lambda s: (
+            1 if (now - s.last_freed_at) < SETTLE_S else 0,
+            s.headroom_bytes, s.cost(), s.gpu.index,
+        ))
         if req.gpu_count == 1:
             return [eligible[0].gpu.index]
         return self._select_set(eligible, req.gpu_count)
@@ -142,22 +155,41 @@ class Binder:
         self.ledger.release(pod_key)
 
 
-def device_env(gpu_indices: List[int]) -> dict:
+def device_env(gpu_indices: List[int], inventory=None) -> dict:
     """Environment that scopes a pod process to its bound GPUs.
 
     ROCR_VISIBLE_DEVICES is enforced by the ROCr runtime (the ROCm-native
     equivalent of the reference backend's server-side GPU attach);
     HIP_VISIBLE_DEVICES is set too for HIP-level tools. Indices are renumbered
     from the pod's perspective (device 0..N-1 inside the pod).
+
+    For multi-GPU pods the xGMI adjacency of the bound set is exported as
+    ``AMDVK_XGMI_PEERS`` in pod-local indices (``a:b@w,c@w;...`` — peer list
+    per device with link weights), so an in-pod RCCL workload can lay its
+    collectives out for the actual point-to-point topology (the
+    kubelet runs no collectives itself; it hands the workload the map).
     """
     if not gpu_indices:
         return {"ROCR_VISIBLE_DEVICES": "", "HIP_VISIBLE_DEVICES": ""}
     joined = ",".join(str(i) for i in gpu_indices)
-    return {
+    env = {
         "ROCR_VISIBLE_DEVICES": joined,
         "HIP_VISIBLE_DEVICES": joined,
         "AMDVK_GPU_IDS": joined,
     }
+    if inventory is not None and len(gpu_indices) > 1:
+        local = {g: i for i, g in enumerate(gpu_indices)}
+        parts = []
+        for g in gpu_indices:
+            gpu = inventory.get(g)
+            peers = []
+            if gpu is not None:
+                for peer, w in sorted(gpu.xgmi_peers.items()):
+                    if peer in local:
+                        peers.append(f"{local[peer]}@{w}")
+            parts.append(f"{local[g]}:{','.join(peers)}")
+        env["AMDVK_XGMI_PEERS"] = ";".join(parts)
+    return env
 
 
 def render_nodes(gpu_indices: List[int], inventory) -> List[str]:

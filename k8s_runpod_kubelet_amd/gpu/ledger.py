@@ -13,6 +13,7 @@ kubelet.go:1380).
 from __future__ import annotations
 
 import threading
+import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
@@ -32,6 +33,13 @@ class GpuState:
     reserved_bytes: int = 0
     pod_keys: List[str] = field(default_factory=list)
     cordoned: bool = False  # operator/admin gate, separate from health
+    # monotonic timestamp of the last release on this GPU. A freshly freed
+    # GPU is "settling": the exited pod's KFD context teardown (VRAM unmap,
+    # queue destruction) runs asynchronously in the kernel for a few hundred
+    # ms and serializes against the next pod's HIP init — measured on MI355X
+    # as ~300 ms init back-to-back vs ~160 ms after a 1 s gap
+    # (profiles/pw_timing.txt). The binder prefers settled GPUs.
+    last_freed_at: float = 0.0
 
     @property
     def schedulable(self) -> bool:
@@ -100,12 +108,14 @@ class Ledger:
             res = self.reservations.pop(pod_key, None)
             if res is None:
                 return None
+            now = time.monotonic()
             for idx in res.gpu_indices:
                 state = self.states.get(idx)
                 if state is not None:
                     state.reserved_bytes = max(0, state.reserved_bytes - res.bytes_per_gpu)
                     if pod_key in state.pod_keys:
                         state.pod_keys.remove(pod_key)
+                    state.last_freed_at = now
             return res
 
     def adopt(self, pod_key: str, gpu_indices: List[int], bytes_per_gpu: int) -> None:

@@ -159,7 +159,7 @@ class ProcessRuntime(Runtime):
         base_env.pop("ROCR_VISIBLE_DEVICES", None)
         base_env.pop("HIP_VISIBLE_DEVICES", None)
         base_env.update(params.env)
-        base_env.update(device_env(inst.gpu_indices))
+        base_env.update(device_env(inst.gpu_indices, self.binder.ledger.inventory))
         base_env["AMDVK_INSTANCE_ID"] = inst.id
         base_env["AMDVK_POD_KEY"] = params.pod_key
 
@@ -456,7 +456,7 @@ class ProcessRuntime(Runtime):
         env.pop("ROCR_VISIBLE_DEVICES", None)
         env.pop("HIP_VISIBLE_DEVICES", None)
         env.update(inst.params.env)
-        env.update(device_env(inst.gpu_indices))
+        env.update(device_env(inst.gpu_indices, self.binder.ledger.inventory))
         out_path = self.logs_dir / f".exec-{inst.id}-{secrets.token_hex(4)}.log"
         try:
             pid, pidfd, _, _ = self._native.launch_process(

@@ -131,3 +131,28 @@ def test_device_env():
     assert env["ROCR_VISIBLE_DEVICES"] == "2,5"
     assert env["HIP_VISIBLE_DEVICES"] == "2,5"
     assert device_env([])["ROCR_VISIBLE_DEVICES"] == ""
+
+
+def test_device_env_exports_xgmi_topology():
+    inv, _ = make_ledger(count=4)
+    env = device_env([1, 3], inv)
+    # Pod-local adjacency: devices 0 (=GPU 1) and 1 (=GPU 3) are peers.
+    assert env["AMDVK_XGMI_PEERS"] == "0:1@1;1:0@1"
+    # Single-GPU pods get no topology map.
+    assert "AMDVK_XGMI_PEERS" not in device_env([2], inv)
+
+
+def test_binder_prefers_settled_gpu():
+    """A GPU freed < 1 s ago still runs the previous pod's KFD teardown
+    (measured ~140 ms HIP-init penalty, profiles/pw_timing.txt): with a
+    settled alternative available, the binder must pick the alternative."""
+    _, ledger = make_ledger(count=2)
+    binder = Binder(ledger)
+    assert binder.bind(BindRequest("p1", 1, GIB)) == [0]
+    binder.unbind("p1")  # GPU 0 now settling
+    assert binder.select(BindRequest("p2", 1, GIB)) == [1]
+    # When every eligible GPU is settling, placement still proceeds.
+    assert binder.bind(BindRequest("p3", 1, GIB)) == [1]
+    binder.unbind("p3")
+    chosen = binder.select(BindRequest("p4", 2, 2 * GIB))
+    assert sorted(chosen) == [0, 1]
