@@ -424,6 +424,10 @@ class ProcessRuntime(Runtime):
     def healthy(self) -> bool:
         return self.ledger.any_schedulable() or self.ledger.total_gpus() == 0
 
+    def tracked_process_count(self) -> int:
+        """Processes the native event loop still watches (leak check)."""
+        return int(self._loop.tracked_count())
+
     # ------------- logs -------------
 
     def get_logs(self, instance_id: str, container: str = "", tail: int = -1) -> str:

@@ -167,3 +167,184 @@ def test_graft_smoke_entry():
     import __graft_entry__
 
     __graft_entry__.smoke()
+
+
+def _wait_phase(kube, ns, name, phases, timeout_s=90):
+    from k8s_runpod_kubelet_amd.kube.client import NotFoundError
+
+    deadline = time.time() + timeout_s
+    last = None
+    while time.time() < deadline:
+        try:
+            pod = kube.get_pod(ns, name)
+        except NotFoundError:
+            return None
+        last = pod
+        if pod.get("status", {}).get("phase") in phases:
+            return pod
+        time.sleep(0.02)
+    return last
+
+
+def test_gpu_pod_failure_reflected(tmp_path):
+    """A GPU pod whose workload exits nonzero must surface phase=Failed with
+    the real exit code in the terminated container state (status-translation
+    fidelity, reference kubelet.go:1848-2024 semantics)."""
+    _require_gpu()
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from tests.conftest import make_pod
+
+    cfg = Config(state_dir=str(tmp_path), notify_interval_s=0,
+                 pending_retry_interval_s=0.2)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.start(serve_http=False)
+    try:
+        kube.create_pod("default", make_pod(
+            "gpufail", gpus=1, command=["podworker"],
+            args=["--expect-gpus", "1", "--run-for", "0.05",
+                  "--exit-code", "3"]))
+        pod = _wait_phase(kube, "default", "gpufail",
+                          ("Failed", "Succeeded"))
+        assert pod is not None
+        assert pod["status"]["phase"] == "Failed"
+        cs = pod["status"]["containerStatuses"][0]
+        assert cs["state"]["terminated"]["exitCode"] == 3
+    finally:
+        stack.stop()
+
+
+def test_gpu_port_readiness_and_exec(tmp_path):
+    """Port-gated readiness against a real listening socket, then a one-shot
+    exec in the pod's GPU environment (parity-plus vs the reference's
+    'not supported' stubs)."""
+    _require_gpu()
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from tests.conftest import make_pod, wait_until
+
+    cfg = Config(state_dir=str(tmp_path), notify_interval_s=0,
+                 pending_retry_interval_s=0.2)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.start(serve_http=False)
+    try:
+        pod = make_pod("gpuport", gpus=1, command=["podworker"],
+                       args=["--expect-gpus", "1",
+                             "--listen-port", "18081", "--hold"],
+                       ports=[18081])
+        kube.create_pod("default", pod)
+
+        def ready():
+            p = _wait_phase(kube, "default", "gpuport", ("Running",),
+                            timeout_s=0.1)
+            if not p:
+                return None
+            conds = {c["type"]: c["status"]
+                     for c in p.get("status", {}).get("conditions", [])}
+            return p if conds.get("Ready") == "True" else None
+
+        assert wait_until(ready, timeout_s=90) is not None
+        info = stack.provider.instance_info("default", "gpuport")
+        detailed = stack.runtime.get_detailed_status(info.instance_id)
+        assert detailed.port_mappings.get(18081) == 18081
+
+        code, out = stack.provider.run_in_container(
+            "default", "gpuport", ["/usr/bin/env"])
+        assert code == 0, out
+        assert "ROCR_VISIBLE_DEVICES=" in out
+        assert "AMDVK_GPU_IDS=" in out
+    finally:
+        stack.stop()
+
+
+def test_gpu_metrics_reflect_real_hbm(real_inventory):
+    """Prometheus gauges carry the probe's real HBM numbers (SURVEY §5.5:
+    reference exports no metrics at all)."""
+    from k8s_runpod_kubelet_amd.gpu.ledger import Ledger
+    from k8s_runpod_kubelet_amd.server import metrics as m
+
+    ledger = Ledger(real_inventory)
+    ledger.sync_inventory()
+    m.observe_gpus(ledger.snapshot())
+    text = m.render().decode()
+    assert "amdvk_gpu_hbm_total_bytes" in text
+    line = [l for l in text.splitlines()
+            if l.startswith('amdvk_gpu_hbm_total_bytes{gpu="0"}')][0]
+    assert float(line.split()[-1]) > 200 * GIB  # MI355X: 288 GB HBM3E
+
+
+def test_gpu_churn_soak(real_inventory, tmp_path):
+    """25 back-to-back pod lifecycles on one GPU (mixed success/failure):
+    ledger drains to zero, the event loop tracks nothing, no instance leaks —
+    the fd/reservation-leak check for the event-driven runtime path."""
+    from k8s_runpod_kubelet_amd.gpu.ledger import Ledger
+    from k8s_runpod_kubelet_amd.runtime.process_runtime import ProcessRuntime
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+
+    ledger = Ledger(real_inventory)
+    ledger.sync_inventory()
+    rt = ProcessRuntime(ledger, str(tmp_path), enable_cgroups=True)
+    try:
+        for i in range(25):
+            fail = (i % 5 == 4)
+            args = ["--expect-gpus", "1", "--run-for", "0.02"]
+            if fail:
+                args += ["--exit-code", "7"]
+            st = rt.deploy(DeployParams(
+                pod_key=f"default-churn{i}", name=f"churn{i}", gpu_count=1,
+                containers=[ContainerSpec(name="main", command=["podworker"],
+                                          args=args)],
+            ))
+            deadline = time.time() + 60
+            while time.time() < deadline:
+                s = rt.get_detailed_status(st.id)
+                if s.desired_status == PodStatus.EXITED:
+                    break
+                time.sleep(0.01)
+            assert s.desired_status == PodStatus.EXITED, f"pod {i} stuck"
+            assert s.exit_code == (7 if fail else 0), rt.get_logs(st.id)
+            rt.remove(st.id)
+        assert not ledger.reservations
+        assert rt.tracked_process_count() == 0
+        assert rt.list_instances() == []
+    finally:
+        rt.close()
+
+
+def test_gpu_cgroup_limits_applied(real_inventory, tmp_path):
+    """As root on the GPU box, the pod lands in a cgroup v2 slot with its
+    memory.max applied and its pid migrated."""
+    import os as _os
+
+    if _os.geteuid() != 0:
+        pytest.skip("cgroup test needs root")
+    from k8s_runpod_kubelet_amd.gpu.ledger import Ledger
+    from k8s_runpod_kubelet_amd.runtime.process_runtime import ProcessRuntime
+    from k8s_runpod_kubelet_amd.runtime.types import ContainerSpec, DeployParams
+
+    ledger = Ledger(real_inventory)
+    ledger.sync_inventory()
+    rt = ProcessRuntime(ledger, str(tmp_path), enable_cgroups=True)
+    try:
+        st = rt.deploy(DeployParams(
+            pod_key="default-cg", name="cg", gpu_count=0,
+            memory_limit=str(1 << 30),
+            containers=[ContainerSpec(name="main", command=["podworker"],
+                                      args=["--hold"])],
+        ))
+        inst = rt._instances[st.id]
+        if not inst.cgroup_dir:
+            pytest.skip("cgroupfs not writable on this box")
+        with open(inst.cgroup_dir + "/memory.max") as fh:
+            assert fh.read().strip() == str(1 << 30)
+        with open(inst.cgroup_dir + "/cgroup.procs") as fh:
+            procs = [int(l) for l in fh.read().split()]
+        assert inst.containers[0].pid in procs
+        rt.terminate(st.id)
+    finally:
+        rt.close()
