@@ -220,6 +220,7 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
         "mean_translate_ms": m.hist_mean_ms(m.translate_seconds),
         "mean_bind_ms": m.hist_mean_ms(m.bind_seconds),
         "mean_launch_ms": m.hist_mean_ms(m.launch_seconds),
+        "mean_cgroup_ms": m.hist_mean_ms(m.cgroup_migrate_seconds),
         "elapsed_s": elapsed,
         "pods": n_gpus * steps,
         "pods_per_sec": (n_gpus * steps) / elapsed,
@@ -308,6 +309,7 @@ def main():
                 "mean_translate_ms": round(result.get("mean_translate_ms", 0), 3),
                 "mean_bind_ms": round(result.get("mean_bind_ms", 0), 3),
                 "mean_launch_ms": round(result.get("mean_launch_ms", 0), 3),
+                "mean_cgroup_ms": round(result.get("mean_cgroup_ms", 0), 3),
                 "p50_pod_ready_ms": round(result["p50_ready_ms"], 3),
                 "p99_pod_ready_ms": round(result["p99_ready_ms"], 3),
                 "reference_poll_floor_ms": 10000.0,  # kubelet.go:719 10 s tick

@@ -140,16 +140,19 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
     return res;
   }
 
+  auto t1 = std::chrono::steady_clock::now();
+  res.spawn_ns =
+      std::chrono::duration_cast<std::chrono::nanoseconds>(t1 - t0).count();
   if (!spec.cgroup_dir.empty()) {
     // cgroup v2 migration from the parent: writing the pid moves the whole
     // process (all threads); done immediately after spawn, before the
     // workload can matter. (The fork path did this pre-exec in the child;
     // posix_spawn has no pre-exec hook for it.)
     WriteFileString(spec.cgroup_dir + "/cgroup.procs", std::to_string(pid));
+    res.cgroup_ns = std::chrono::duration_cast<std::chrono::nanoseconds>(
+                        std::chrono::steady_clock::now() - t1)
+                        .count();
   }
-  res.spawn_ns = std::chrono::duration_cast<std::chrono::nanoseconds>(
-                     std::chrono::steady_clock::now() - t0)
-                     .count();
 
   res.pid = pid;
   res.pidfd = PidfdOpen(pid);

@@ -23,10 +23,13 @@ struct LaunchResult {
   int64_t pid = -1;
   int pidfd = -1;
   int ready_fd = -1;  // parent read end of the readiness pipe (-1 if disabled)
-  // Spawn cost measured around posix_spawnp + cgroup migration on the native
-  // side: Python-side wall timing overstates it by ~100 ms whenever the
-  // launching thread loses the GIL to the already-running child.
-  int64_t spawn_ns = 0;
+  // Costs measured on the native side (Python-side wall timing overstates
+  // them by ~100 ms whenever the launching thread loses the GIL to the
+  // already-running child). Split so the cgroup.procs migration — which
+  // serializes on the kernel's cgroup_mutex against concurrent teardown —
+  // is attributable separately from posix_spawnp itself.
+  int64_t spawn_ns = 0;   // posix_spawnp only
+  int64_t cgroup_ns = 0;  // cgroup.procs migration write
   std::string error;
 };
 

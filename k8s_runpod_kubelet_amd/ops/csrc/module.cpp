@@ -72,7 +72,7 @@ PYBIND11_MODULE(_native, m) {
         }
         if (!res.error.empty()) throw std::runtime_error(res.error);
         return py::make_tuple(res.pid, res.pidfd, res.ready_fd,
-                              res.spawn_ns * 1e-9);
+                              res.spawn_ns * 1e-9, res.cgroup_ns * 1e-9);
       },
       py::arg("argv"), py::arg("env"), py::arg("cwd") = "",
       py::arg("stdout_path") = "", py::arg("stderr_path") = "",

@@ -151,7 +151,11 @@ class ProcessRuntime(Runtime):
 
     def _launch_containers(self, inst: Instance) -> float:
         """Spawn every container of the pod; returns the summed native
-        spawn time in seconds (posix_spawnp + cgroup migration)."""
+        posix_spawnp time in seconds. The cgroup.procs migration is timed
+        separately (cgroup_migrate_seconds) — it serializes on the kernel's
+        cgroup_mutex and dominates launch cost under churn."""
+        from ..server import metrics
+
         spawn_total_s = 0.0
         params = inst.params
         base_env = dict(os.environ)
@@ -179,13 +183,15 @@ class ProcessRuntime(Runtime):
             envp = [f"{k}={v}" for k, v in env.items()]
             stdout_path = str(self.logs_dir / f"{inst.id}-{cspec.name}.log")
 
-            pid, pidfd, ready_fd, spawn_s = self._native.launch_process(
+            pid, pidfd, ready_fd, spawn_s, cgroup_s = self._native.launch_process(
                 argv, envp,
                 cspec.working_dir or "",
                 stdout_path, stdout_path,
                 inst.cgroup_dir, True, True,
             )
             spawn_total_s += spawn_s
+            if inst.cgroup_dir:
+                metrics.cgroup_migrate_seconds.observe(cgroup_s)
             cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
             inst.containers.append(cinfo)
             with self._lock:
@@ -463,7 +469,7 @@ class ProcessRuntime(Runtime):
         env.update(device_env(inst.gpu_indices, self.binder.ledger.inventory))
         out_path = self.logs_dir / f".exec-{inst.id}-{secrets.token_hex(4)}.log"
         try:
-            pid, pidfd, _, _ = self._native.launch_process(
+            pid, pidfd, _, _, _ = self._native.launch_process(
                 list(command), [f"{k}={v}" for k, v in env.items()],
                 "", str(out_path), str(out_path), inst.cgroup_dir, True, False,
             )

@@ -47,7 +47,11 @@ bind_seconds = Histogram(
     "amdvk_bind_seconds", "GPU set selection + ledger reservation",
     buckets=_PHASE_BUCKETS, registry=registry)
 launch_seconds = Histogram(
-    "amdvk_launch_seconds", "cgroup setup + posix_spawn of containers",
+    "amdvk_launch_seconds", "posix_spawn of containers (native clock)",
+    buckets=_PHASE_BUCKETS, registry=registry)
+cgroup_migrate_seconds = Histogram(
+    "amdvk_cgroup_migrate_seconds",
+    "cgroup.procs migration write (native clock; serializes on cgroup_mutex)",
     buckets=_PHASE_BUCKETS, registry=registry)
 
 

@@ -29,7 +29,7 @@ def drain(loop, want, timeout_s=5.0):
 
 def test_launch_exit_code(native, tmp_path):
     loop = native.EventLoop()
-    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s, _cg_s = native.launch_process(
         ["/bin/sh", "-c", "exit 7"], base_env(),
         "", str(tmp_path / "out.log"), "", "", True, True)
     loop.add_process(pid, pidfd, ready_fd, 1)
@@ -41,7 +41,7 @@ def test_launch_exit_code(native, tmp_path):
 
 def test_ready_pipe_protocol(native, tmp_path):
     loop = native.EventLoop()
-    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s, _cg_s = native.launch_process(
         ["/bin/bash", "-c", 'echo READY >&$AMDVK_READY_FD; sleep 0.1'],
         base_env(), "", str(tmp_path / "out.log"), "", "", True, True)
     loop.add_process(pid, pidfd, ready_fd, 2)
@@ -60,7 +60,7 @@ def test_exec_failure_reported(native, tmp_path):
 
 def test_signal_process(native, tmp_path):
     loop = native.EventLoop()
-    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s, _cg_s = native.launch_process(
         ["/bin/sleep", "30"], base_env(), "", str(tmp_path / "o.log"), "", "",
         True, True)
     loop.add_process(pid, pidfd, ready_fd, 3)
@@ -73,7 +73,7 @@ def test_signal_process(native, tmp_path):
 def test_stdout_redirect(native, tmp_path):
     log = tmp_path / "redir.log"
     loop = native.EventLoop()
-    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s, _cg_s = native.launch_process(
         ["/bin/sh", "-c", "echo hello-out; echo hello-err >&2"],
         base_env(), "", str(log), "", "", True, True)
     loop.add_process(pid, pidfd, ready_fd, 4)
@@ -83,7 +83,7 @@ def test_stdout_redirect(native, tmp_path):
 
 
 def test_open_pidfd_adoption(native):
-    pid, pidfd, ready_fd, _spawn_s = native.launch_process(
+    pid, pidfd, ready_fd, _spawn_s, _cg_s = native.launch_process(
         ["/bin/sleep", "0.2"], base_env(), "", "", "", "", True, False)
     os.close(pidfd)
     # Re-open (adoption path) and watch the exit through a fresh loop.
@@ -103,3 +103,34 @@ def test_cgroup_helpers_best_effort(native, tmp_path):
     assert os.path.isdir(path)
     assert native.cgroup_proc_count(path) == -1  # no cgroup.procs file
     assert native.cgroup_remove(path)
+
+
+@pytest.mark.parametrize("sanitizer", ["thread", "address,undefined"])
+def test_native_sanitizer_stress(sanitizer, tmp_path):
+    """Build the launcher + event loop with TSan/ASan+UBSan and hammer them
+    from 8 threads (spawn + mid-flight removal + concurrent polling). The
+    reference ships no race detection at all and has known data races
+    (SURVEY §5.2); the native hot path here must be clean under both."""
+    import shutil
+    import subprocess
+
+    if shutil.which("g++") is None:
+        pytest.skip("no g++ on this box")
+    csrc = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "k8s_runpod_kubelet_amd", "ops", "csrc")
+    binary = tmp_path / f"stress_{sanitizer.split(',')[0]}"
+    build = subprocess.run(
+        ["g++", "-O1", "-g", f"-fsanitize={sanitizer}", "-std=c++17",
+         os.path.join(csrc, "launcher.cpp"),
+         os.path.join(csrc, "stress_launcher.cpp"),
+         "-o", str(binary), "-lpthread"],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert build.returncode == 0, build.stderr[-2000:]
+    run = subprocess.run(
+        [str(binary), "8", "25"], capture_output=True, text=True, timeout=120,
+        env={"PATH": "/usr/bin:/bin", "TSAN_OPTIONS": "halt_on_error=1",
+             "ASAN_OPTIONS": "detect_leaks=0"},
+    )
+    assert run.returncode == 0, (run.stdout + run.stderr)[-4000:]
+    assert run.stdout.startswith("ok:"), run.stdout
