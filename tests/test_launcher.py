@@ -134,3 +134,53 @@ def test_native_sanitizer_stress(sanitizer, tmp_path):
     )
     assert run.returncode == 0, (run.stdout + run.stderr)[-4000:]
     assert run.stdout.startswith("ok:"), run.stdout
+
+
+def test_clone3_born_in_cgroup(native, tmp_path):
+    """clone3(CLONE_INTO_CGROUP) fast path: the child's first instruction
+    already runs inside the pod cgroup — no cgroup.procs migration (measured
+    ~130 ms under churn on MI355X, profiles/). Verified by reading the
+    child's /proc/<pid>/cgroup."""
+    import secrets
+
+    cgroot = None
+    for cand in ("/sys/fs/cgroup", "/sys/fs/cgroup/unified"):
+        if os.path.exists(os.path.join(cand, "cgroup.procs")) and \
+                os.access(cand, os.W_OK):
+            cgroot = cand
+            break
+    if cgroot is None:
+        pytest.skip("no writable cgroup2 hierarchy on this box")
+    cgdir = os.path.join(cgroot, f"amdvk-test-{secrets.token_hex(4)}")
+    try:
+        os.mkdir(cgdir)
+    except OSError:
+        pytest.skip("cgroup2 root not writable")
+    try:
+        out = tmp_path / "out.log"
+        pid, pidfd, ready_fd, spawn_s, cgroup_s = native.launch_process(
+            ["/bin/bash", "-c",
+             "echo READY >&${AMDVK_READY_FD}; sleep 5"],
+            base_env(), "", str(out), str(out), cgdir, True, True,
+        )
+        try:
+            with open(f"/proc/{pid}/cgroup") as fh:
+                line = fh.read().strip()
+            # cgroup2 entry: "0::<path>"; the path must be our pod cgroup.
+            assert line.splitlines()[-1].startswith("0::"), line
+            assert line.rsplit("::", 1)[-1].endswith(os.path.basename(cgdir)), line
+            assert cgroup_s == 0.0  # born attached: nothing was migrated
+            loop = native.EventLoop()
+            loop.add_process(pid, pidfd, ready_fd, pid)
+            events = drain(loop, "ready")
+            assert any(e.type == "ready" for e in events)
+        finally:
+            native.signal_process(pid, 9, True)
+            time.sleep(0.1)
+    finally:
+        for _ in range(50):
+            try:
+                os.rmdir(cgdir)
+                break
+            except OSError:
+                time.sleep(0.05)
