@@ -84,14 +84,19 @@ class Ticker:
 
 
 def parse_duration_s(value: object, default: float) -> float:
+    """Parse "30s"/"5m"/"1h"/"500ms"/bare numbers; malformed input returns
+    ``default`` instead of raising (a flag typo must not crash the kubelet)."""
     if value is None:
         return default
     if isinstance(value, (int, float)):
         return float(value)
     text = str(value).strip()
-    if text.endswith("ms"):
-        return float(text[:-2]) / 1000.0
-    for suffix, mult in (("s", 1.0), ("m", 60.0), ("h", 3600.0)):
-        if text.endswith(suffix):
-            return float(text[: -len(suffix)]) * mult
-    return float(text)
+    try:
+        if text.endswith("ms"):
+            return float(text[:-2]) / 1000.0
+        for suffix, mult in (("s", 1.0), ("m", 60.0), ("h", 3600.0)):
+            if text.endswith(suffix):
+                return float(text[: -len(suffix)]) * mult
+        return float(text)
+    except ValueError:
+        return default

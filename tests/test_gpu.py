@@ -358,3 +358,129 @@ def test_gpu_cgroup_limits_applied(real_inventory, tmp_path):
         rt.terminate(st.id)
     finally:
         rt.close()
+
+
+def test_gpu_crash_detection_and_replacement(tmp_path):
+    """Fault injection on real hardware: SIGKILL a running GPU pod's process
+    out-of-band. The pidfd event must mark the pod Failed within ms and the
+    freed GPU must immediately place a queued pod (event-driven pending
+    placement — retry tick is set to 999 s so only the event path can win)."""
+    _require_gpu()
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.provider import annotations as ann
+    from tests.conftest import make_pod, wait_until
+
+    cfg = Config(state_dir=str(tmp_path), notify_interval_s=0,
+                 pending_retry_interval_s=999)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    if stack.ledger.total_gpus() > 1:
+        pytest.skip("test assumes a 1-GPU box (queued pod must wait)")
+    stack.start(serve_http=False)
+    try:
+        kube.create_pod("default", make_pod(
+            "victim", gpus=1, command=["podworker"],
+            args=["--expect-gpus", "1", "--hold"]))
+
+        def victim_ready():
+            p = kube.get_pod("default", "victim")
+            conds = {c["type"]: c["status"]
+                     for c in p.get("status", {}).get("conditions", [])}
+            return p if conds.get("Ready") == "True" else None
+
+        got = wait_until(victim_ready, timeout_s=60)
+        assert got is not None
+        # Queue a second pod: the only GPU is taken.
+        kube.create_pod("default", make_pod(
+            "heir", gpus=1, command=["podworker"],
+            args=["--expect-gpus", "1", "--hold"]))
+        time.sleep(0.3)
+
+        iid = got["metadata"]["annotations"][ann.POD_ID]
+        pid = stack.runtime.get_detailed_status(iid).containers[0].pid
+        t0 = time.time()
+        os.kill(pid, 9)  # out-of-band crash
+
+        def victim_failed():
+            p = kube.get_pod("default", "victim")
+            return p if p.get("status", {}).get("phase") == "Failed" else None
+
+        failed = wait_until(victim_failed, timeout_s=30)
+        assert failed is not None
+        cs = failed["status"]["containerStatuses"][0]
+        assert cs["state"]["terminated"]["exitCode"] == 137  # 128+SIGKILL
+
+        def heir_ready():
+            p = kube.get_pod("default", "heir")
+            conds = {c["type"]: c["status"]
+                     for c in p.get("status", {}).get("conditions", [])}
+            return p if conds.get("Ready") == "True" else None
+
+        assert wait_until(heir_ready, timeout_s=60) is not None
+        assert time.time() - t0 < 30  # event path, not the 999 s tick
+    finally:
+        stack.stop()
+
+
+def test_gpu_restart_adoption_live_process(tmp_path):
+    """Kubelet crash with a live GPU pod: the restarted stack must adopt the
+    still-running podworker (same pid, same instance id), rebuild the GPU
+    ledger without double-binding, and keep serving logs."""
+    _require_gpu()
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.provider import annotations as ann
+    from tests.conftest import make_pod, wait_until
+
+    cfg = Config(state_dir=str(tmp_path), notify_interval_s=0,
+                 pending_retry_interval_s=0.2)
+    kube = FakeKube()
+    s1 = build_stack(cfg, client=kube)
+    s1.start(serve_http=False)
+    kube.create_pod("default", make_pod(
+        "survivor", gpus=1, command=["podworker"],
+        args=["--expect-gpus", "1", "--hold"]))
+
+    def ready():
+        p = kube.get_pod("default", "survivor")
+        conds = {c["type"]: c["status"]
+                 for c in p.get("status", {}).get("conditions", [])}
+        return p if conds.get("Ready") == "True" else None
+
+    got = wait_until(ready, timeout_s=60)
+    assert got is not None
+    iid = got["metadata"]["annotations"][ann.POD_ID]
+    pid = s1.runtime.get_detailed_status(iid).containers[0].pid
+    # Simulate kubelet crash: stop controllers/event loop, leave pods alive.
+    s1.pod_controller.stop()
+    s1.node_controller.stop()
+    s1.provider.stop()
+    s1.runtime._stop.set()
+    s1.runtime._loop.wake()
+
+    s2 = build_stack(cfg, client=kube)
+    s2.start(serve_http=False)
+    try:
+        info = s2.provider.instance_info("default", "survivor")
+        assert info is not None and info.instance_id == iid
+        st = s2.runtime.get_detailed_status(iid)
+        assert st.containers[0].pid == pid
+        os.kill(pid, 0)  # the process must actually still be alive
+        assert s2.ledger.get_reservation("default-survivor") is not None
+        assert "ok" in s2.provider.get_container_logs("default", "survivor")
+        kube.delete_pod("default", "survivor")
+
+        def gone():
+            from k8s_runpod_kubelet_amd.kube.client import NotFoundError
+            try:
+                kube.get_pod("default", "survivor")
+                return False
+            except NotFoundError:
+                return True
+
+        assert wait_until(gone, timeout_s=30)
+    finally:
+        s2.stop()

@@ -1,0 +1,178 @@
+"""Property-based tests (hypothesis) for the pure-logic layers.
+
+The reference has no property testing at all (integration-first,
+no hermetic suite); these pin down invariants of the strategic-merge patch
+engine, env escaping, duration parsing and port gating that example-based
+tests under-cover."""
+
+import string
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from k8s_runpod_kubelet_amd.kube.patch import json_merge, strategic_merge
+from k8s_runpod_kubelet_amd.provider.envvars import _escape, is_k8s_auto_injected
+from k8s_runpod_kubelet_amd.provider.ports import check_ports_exposed
+from k8s_runpod_kubelet_amd.utils.backoff import parse_duration_s
+
+# -- strategies --------------------------------------------------------------
+
+scalars = st.one_of(
+    st.integers(-1000, 1000),
+    st.text(string.ascii_letters, max_size=8),
+    st.booleans(),
+)
+json_maps = st.recursive(
+    st.dictionaries(st.text(string.ascii_lowercase, min_size=1, max_size=6),
+                    scalars, max_size=4),
+    lambda children: st.dictionaries(
+        st.text(string.ascii_lowercase, min_size=1, max_size=6),
+        st.one_of(scalars, children), max_size=4),
+    max_leaves=12,
+)
+
+
+# -- strategic merge ---------------------------------------------------------
+
+@settings(max_examples=200)
+@given(original=json_maps, patch=json_maps)
+def test_patch_values_win(original, patch):
+    """Every key present in the patch ends up with the patch's value (or is
+    deleted when the patch value is None); keys absent from the patch keep
+    their original value."""
+    merged = strategic_merge(original, patch)
+    for key, value in patch.items():
+        if value is None:
+            assert key not in merged
+        elif isinstance(value, dict) and isinstance(original.get(key), dict):
+            inner = strategic_merge(original[key], value)
+            assert merged[key] == inner
+        else:
+            assert merged[key] == value
+    for key, value in original.items():
+        if key not in patch:
+            assert merged[key] == value
+
+
+@settings(max_examples=100)
+@given(original=json_maps, patch=json_maps)
+def test_merge_is_idempotent(original, patch):
+    once = strategic_merge(original, patch)
+    twice = strategic_merge(once, patch)
+    assert once == twice
+
+
+@settings(max_examples=100)
+@given(original=json_maps, patch=json_maps)
+def test_merge_does_not_mutate_inputs(original, patch):
+    import copy
+
+    o2, p2 = copy.deepcopy(original), copy.deepcopy(patch)
+    strategic_merge(original, patch)
+    assert original == o2 and patch == p2
+
+
+@settings(max_examples=100)
+@given(original=json_maps, patch=json_maps)
+def test_json_merge_agrees_on_flat_scalars(original, patch):
+    """For scalar-only patches RFC 7386 json-merge and SMP agree."""
+    flat_patch = {k: v for k, v in patch.items() if not isinstance(v, dict)}
+    a = strategic_merge(original, flat_patch)
+    b = json_merge(original, flat_patch)
+    for key in flat_patch:
+        assert a.get(key) == b.get(key)
+
+
+conditions = st.lists(
+    st.fixed_dictionaries({
+        "type": st.sampled_from(["Ready", "PodScheduled", "Initialized"]),
+        "status": st.sampled_from(["True", "False", "Unknown"]),
+    }),
+    max_size=4,
+    unique_by=lambda c: c["type"],
+)
+
+
+@settings(max_examples=200)
+@given(original=conditions, patch=conditions)
+def test_conditions_merge_by_type(original, patch):
+    merged = strategic_merge({"conditions": original},
+                             {"conditions": patch})["conditions"]
+    by_type = {c["type"]: c for c in merged}
+    # no duplicate types after merge
+    assert len(by_type) == len(merged)
+    for cond in patch:  # patched conditions win
+        assert by_type[cond["type"]]["status"] == cond["status"]
+    for cond in original:  # untouched conditions survive
+        if cond["type"] not in {c["type"] for c in patch}:
+            assert by_type[cond["type"]]["status"] == cond["status"]
+
+
+# -- env escaping ------------------------------------------------------------
+
+@settings(max_examples=200)
+@given(st.text(max_size=64))
+def test_env_escape_removes_raw_newlines(value):
+    escaped = _escape(value)
+    assert "\n" not in escaped.replace("\\n", "")
+
+
+@given(st.text(string.ascii_uppercase + "_", min_size=1, max_size=20))
+def test_auto_injected_filter_matches_reference_patterns(name):
+    """isK8sAutoInjectedVar semantics (reference runpod_client.go:886-904):
+    substring patterns, not prefixes."""
+    flagged = is_k8s_auto_injected(name)
+    expected = name.startswith("KUBERNETES_") or any(
+        p in name for p in ("_PORT_", "_TCP_", "_SERVICE_PORT_", "_SERVICE_HOST")
+    ) or name.endswith("_PORT") or name.endswith("_SERVICE_HOST")
+    # our filter must never let a k8s-injected pattern through
+    if expected:
+        assert flagged
+    if not flagged:
+        assert not expected
+
+
+# -- duration parsing --------------------------------------------------------
+
+@given(st.integers(0, 10**6))
+def test_parse_duration_seconds(n):
+    assert parse_duration_s(f"{n}s", -1) == float(n)
+    assert parse_duration_s(str(n), -1) == float(n)
+
+
+@given(st.integers(0, 10**4))
+def test_parse_duration_minutes_hours(n):
+    assert parse_duration_s(f"{n}m", -1) == float(n * 60)
+    assert parse_duration_s(f"{n}h", -1) == float(n * 3600)
+
+
+@given(st.text(max_size=12))
+def test_parse_duration_never_raises(junk):
+    """Malformed durations fall back to the default instead of crashing the
+    CLI (bug found by this property: float('abc') used to propagate)."""
+    out = parse_duration_s(junk, 42.0)
+    assert isinstance(out, float)
+
+
+# -- port gating -------------------------------------------------------------
+
+ports = st.integers(1, 65535)
+
+
+@settings(max_examples=200)
+@given(requested=st.lists(ports, max_size=6, unique=True),
+       listening=st.sets(ports, max_size=8))
+def test_tcp_ports_gate_readiness(requested, listening):
+    """TCP ports must all be exposed; HTTP ports are assumed proxied
+    (reference kubelet.go:566-605 semantics)."""
+    reqs = [f"{p}/tcp" for p in requested]
+    mappings = {p: p for p in listening}
+    ok = check_ports_exposed(reqs, mappings)
+    assert ok == all(p in listening for p in requested)
+
+
+@settings(max_examples=50)
+@given(requested=st.lists(ports, max_size=6, unique=True))
+def test_http_ports_always_ready(requested):
+    reqs = [f"{p}/http" for p in requested]
+    assert check_ports_exposed(reqs, {})
