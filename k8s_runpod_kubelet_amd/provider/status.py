@@ -125,11 +125,6 @@ def translate_status(
         "podIP": node_ip,
         "startTime": start_time,
     }
-    if info.gpu_indices:
-        # surfaced for kubectl debugging; authoritative copy lives in the
-        # amd.com/gpu-ids annotation
-        base["nominatedNodeName"] = ""
-
     if status == PodStatus.RUNNING:
         ready = info.ports_exposed
         if ready:

@@ -484,3 +484,88 @@ def test_gpu_restart_adoption_live_process(tmp_path):
         assert wait_until(gone, timeout_s=30)
     finally:
         s2.stop()
+
+
+def test_gpu_stack_soak(tmp_path):
+    """Sustained full-stack churn on real hardware: 40 pods through the
+    kubelet in mixed modes (hold+delete, run-to-completion success and
+    failure, out-of-band SIGKILL), overlapping via the pending queue.
+    Afterwards: no reservations, no tracked processes, no instance records,
+    and no fd leak in the control plane."""
+    _require_gpu()
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.client import NotFoundError
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.provider import annotations as ann
+    from tests.conftest import make_pod, wait_until
+
+    cfg = Config(state_dir=str(tmp_path), notify_interval_s=0,
+                 pending_retry_interval_s=0.5)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.start(serve_http=False)
+
+    def fd_count():
+        return len(os.listdir("/proc/self/fd"))
+
+    def phase(name):
+        try:
+            return kube.get_pod("default", name).get("status", {}).get("phase")
+        except NotFoundError:
+            return "Gone"
+
+    def ready(name):
+        try:
+            p = kube.get_pod("default", name)
+        except NotFoundError:
+            return None
+        conds = {c["type"]: c["status"]
+                 for c in p.get("status", {}).get("conditions", [])}
+        return p if conds.get("Ready") == "True" else None
+
+    try:
+        fd0 = fd_count()
+        for i in range(40):
+            name = f"soak{i:02d}"
+            mode = i % 4
+            if mode == 0:      # hold, then delete
+                args = ["--expect-gpus", "1", "--hold"]
+            elif mode == 1:    # run to successful completion
+                args = ["--expect-gpus", "1", "--run-for", "0.05"]
+            elif mode == 2:    # run to failure
+                args = ["--expect-gpus", "1", "--run-for", "0.05",
+                        "--exit-code", "9"]
+            else:              # hold, then crash via SIGKILL
+                args = ["--expect-gpus", "1", "--hold"]
+            kube.create_pod("default", make_pod(
+                name, gpus=1, command=["podworker"], args=args))
+
+            if mode in (0, 3):
+                assert wait_until(lambda: ready(name), timeout_s=60), name
+                if mode == 3:
+                    p = kube.get_pod("default", name)
+                    iid = p["metadata"]["annotations"][ann.POD_ID]
+                    pid = stack.runtime.get_detailed_status(iid).containers[0].pid
+                    os.kill(pid, 9)
+                    assert wait_until(
+                        lambda: phase(name) == "Failed", timeout_s=30), name
+                kube.delete_pod("default", name)
+                assert wait_until(lambda: phase(name) == "Gone",
+                                  timeout_s=30), name
+            else:
+                want = "Succeeded" if mode == 1 else "Failed"
+                assert wait_until(lambda: phase(name) == want,
+                                  timeout_s=60), (name, phase(name))
+                kube.delete_pod("default", name)
+                assert wait_until(lambda: phase(name) == "Gone",
+                                  timeout_s=30), name
+
+        assert not stack.ledger.reservations
+        assert stack.runtime.tracked_process_count() == 0
+        assert stack.runtime.list_instances() == []
+        # generous margin: loggers/sockets fluctuate, leaks of 40 pods would
+        # show as 40-120 extra fds
+        assert fd_count() <= fd0 + 15, (fd0, fd_count())
+    finally:
+        stack.stop()
