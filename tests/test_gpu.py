@@ -563,6 +563,9 @@ def test_gpu_stack_soak(tmp_path):
 
         assert not stack.ledger.reservations
         assert stack.runtime.tracked_process_count() == 0
+        # Terminated instance records persist until the periodic GC
+        # (reference cleanupDeletedPods runs on a 5 min ticker); drive it.
+        stack.provider.cleanup_deleted_pods()
         assert stack.runtime.list_instances() == []
         # generous margin: loggers/sockets fluctuate, leaks of 40 pods would
         # show as 40-120 extra fds
