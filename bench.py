@@ -56,7 +56,8 @@ def _barrier(dist, device_sync: bool):
 
 
 def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
-              hold_pods: bool = True, burst_size: int = 0):
+              hold_pods: bool = True, burst_size: int = 0,
+              settle_gap_s: float = 0.0):
     from k8s_runpod_kubelet_amd.app import build_stack
     from k8s_runpod_kubelet_amd.config import Config
     from k8s_runpod_kubelet_amd.kube.client import NotFoundError
@@ -208,6 +209,12 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
         latencies = []
         t0 = time.monotonic()
         for k in range(steps):
+            if settle_gap_s > 0:
+                # Experiment mode (never set by the driver): let the freed
+                # GPU's KFD teardown drain so p50 reflects a non-churn pod
+                # (profiles/pw_timing.txt). Throughput output still includes
+                # the gaps — only the latency numbers are meaningful here.
+                time.sleep(settle_gap_s)
             latencies.extend(wave(f"s{k}"))
         elapsed = time.monotonic() - t0
     finally:
@@ -241,6 +248,9 @@ def main():
     ap.add_argument("--mode", choices=["auto", "gpu", "cpu"], default="auto")
     ap.add_argument("--burst", type=int, default=0,
                     help="BASELINE config 5: pods per burst (0 = wave mode)")
+    ap.add_argument("--settle-gap", type=float, default=0.0,
+                    help="experiment only: sleep S s between waves so pod-Ready "
+                         "reflects a settled GPU (throughput then meaningless)")
     args = ap.parse_args()
 
     rank, world = _dist_env()
@@ -274,7 +284,8 @@ def main():
     result = None
     if rank == 0:
         result = run_bench(args.gpus, args.steps, args.warmup, use_gpu,
-                           burst_size=args.burst)
+                           burst_size=args.burst,
+                           settle_gap_s=args.settle_gap)
     _barrier(dist, device_sync)
 
     if rank == 0 and result is not None:
@@ -304,6 +315,7 @@ def main():
                 "lifecycle": ("32-pod FIFO burst drain" if args.burst
                               else "create->bind->launch->Ready->delete->finalized"),
                 "burst_size": args.burst,
+                "settle_gap_s": args.settle_gap,
                 "pods_total": result["pods"],
                 "mean_deploy_ms": round(result.get("mean_deploy_ms", 0), 3),
                 "mean_translate_ms": round(result.get("mean_translate_ms", 0), 3),
