@@ -381,6 +381,109 @@ bool CgroupCreate(const std::string& path, const std::string& cpu_max,
 
 bool CgroupRemove(const std::string& path) { return rmdir(path.c_str()) == 0; }
 
+// ---------------- cgroup-v2 eBPF device filter ----------------
+//
+// Per-pod GPU device isolation (the reference's backend attaches GPUs
+// server-side; locally ROCR_VISIBLE_DEVICES is cooperative, not enforced).
+// ROCm userspace must open the GPU's /dev/dri/renderD<minor> to acquire a
+// KFD VM, so denying the other GPUs' render nodes at the cgroup level is
+// real isolation. cgroup v2 has no devices controller file — policy is a
+// BPF_PROG_TYPE_CGROUP_DEVICE program attached to the pod cgroup. The
+// program is hand-assembled here (a dozen instructions; no libbpf in the
+// image):
+//
+//   r2 = ctx->major; r3 = ctx->minor
+//   if (r2 != denied_major) return ALLOW        // all non-DRM devices
+//   for m in allowed_minors: if (r3 == m) return ALLOW
+//   return DENY
+//
+// BPF_F_ALLOW_OVERRIDE lets the next pod reusing the cgroup slot replace
+// the program. Returns false (without throwing) when the kernel refuses
+// (unprivileged test runs) — isolation is then best-effort, as with the
+// cgroup limits themselves.
+
+bool CgroupAttachDeviceFilter(const std::string& cgroup_dir, int denied_major,
+                              const std::vector<int>& allowed_minors) {
+  struct Insn {  // struct bpf_insn without needing ASM macros
+    uint8_t code;
+    uint8_t regs;  // dst | (src << 4)
+    int16_t off;
+    int32_t imm;
+  };
+  std::vector<Insn> prog;
+  auto emit = [&](uint8_t code, uint8_t dst, uint8_t src, int16_t off,
+                  int32_t imm) {
+    prog.push_back(Insn{code, static_cast<uint8_t>(dst | (src << 4)), off, imm});
+  };
+  // opcodes (linux/bpf_common.h values, spelled out):
+  constexpr uint8_t kLdxW = 0x61;   // BPF_LDX | BPF_MEM | BPF_W
+  constexpr uint8_t kJneK = 0x55;   // BPF_JMP | BPF_JNE | BPF_K
+  constexpr uint8_t kJeqK = 0x15;   // BPF_JMP | BPF_JEQ | BPF_K
+  constexpr uint8_t kMovK = 0xb7;   // BPF_ALU64 | BPF_MOV | BPF_K
+  constexpr uint8_t kExit = 0x95;   // BPF_JMP | BPF_EXIT
+
+  const int n = static_cast<int>(allowed_minors.size());
+  // layout: [0] ldx r2=major  [1] ldx r3=minor  [2] jne major,+ (n+1 -> ALLOW)
+  // [3..3+n-1] jeq minor,allow_i  [3+n] mov r0,0; exit  [5+n] ALLOW mov r0,1; exit
+  emit(kLdxW, 2, 1, 4, 0);                       // r2 = major
+  emit(kLdxW, 3, 1, 8, 0);                       // r3 = minor
+  emit(kJneK, 2, 0, static_cast<int16_t>(n + 2), denied_major);  // -> ALLOW
+  for (int i = 0; i < n; ++i)
+    emit(kJeqK, 3, 0, static_cast<int16_t>(n - i + 1), allowed_minors[i]);
+  emit(kMovK, 0, 0, 0, 0);                       // DENY: r0 = 0
+  emit(kExit, 0, 0, 0, 0);
+  emit(kMovK, 0, 0, 0, 1);                       // ALLOW: r0 = 1
+  emit(kExit, 0, 0, 0, 0);
+
+  union bpf_attr_shim {
+    struct {  // BPF_PROG_LOAD
+      uint32_t prog_type;
+      uint32_t insn_cnt;
+      uint64_t insns;
+      uint64_t license;
+      uint32_t log_level;
+      uint32_t log_size;
+      uint64_t log_buf;
+      uint32_t kern_version;
+      uint32_t prog_flags;
+    } load;
+    struct {  // BPF_PROG_ATTACH
+      uint32_t target_fd;
+      uint32_t attach_bpf_fd;
+      uint32_t attach_type;
+      uint32_t attach_flags;
+    } attach;
+    char pad[128];
+  };
+  static const char kLicense[] = "GPL";
+
+  bpf_attr_shim attr;
+  memset(&attr, 0, sizeof(attr));
+  attr.load.prog_type = 15;  // BPF_PROG_TYPE_CGROUP_DEVICE
+  attr.load.insn_cnt = static_cast<uint32_t>(prog.size());
+  attr.load.insns = reinterpret_cast<uint64_t>(prog.data());
+  attr.load.license = reinterpret_cast<uint64_t>(kLicense);
+  int progfd = static_cast<int>(
+      syscall(SYS_bpf, 5 /*BPF_PROG_LOAD*/, &attr, sizeof(attr)));
+  if (progfd < 0) return false;
+
+  int cgfd = open(cgroup_dir.c_str(), O_DIRECTORY | O_RDONLY | O_CLOEXEC);
+  if (cgfd < 0) {
+    close(progfd);
+    return false;
+  }
+  memset(&attr, 0, sizeof(attr));
+  attr.attach.target_fd = static_cast<uint32_t>(cgfd);
+  attr.attach.attach_bpf_fd = static_cast<uint32_t>(progfd);
+  attr.attach.attach_type = 6;  // BPF_CGROUP_DEVICE
+  attr.attach.attach_flags = 1; // BPF_F_ALLOW_OVERRIDE (slot reuse replaces)
+  int rc = static_cast<int>(
+      syscall(SYS_bpf, 8 /*BPF_PROG_ATTACH*/, &attr, sizeof(attr)));
+  close(cgfd);
+  close(progfd);  // attachment keeps the program alive
+  return rc == 0;
+}
+
 long CgroupProcCount(const std::string& path) {
   std::ifstream f(path + "/cgroup.procs");
   if (!f.good()) return -1;

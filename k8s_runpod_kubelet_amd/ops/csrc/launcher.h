@@ -48,6 +48,14 @@ bool CgroupCreate(const std::string& path, const std::string& cpu_max,
 bool CgroupRemove(const std::string& path);
 long CgroupProcCount(const std::string& path);
 
+// Attach a cgroup-v2 eBPF device filter to the pod cgroup: devices with
+// major == denied_major are only accessible at the listed minors; all other
+// majors pass. Used to make GPU binding *enforced* (deny other GPUs'
+// /dev/dri/renderD<minor> nodes — ROCm cannot acquire a KFD VM without the
+// render node). Best-effort: false when the kernel/permissions refuse.
+bool CgroupAttachDeviceFilter(const std::string& cgroup_dir, int denied_major,
+                              const std::vector<int>& allowed_minors);
+
 struct Event {
   enum Type { kExited = 0, kReady = 1, kReadyClosed = 2 };
   Type type;

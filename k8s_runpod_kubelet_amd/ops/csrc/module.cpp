@@ -86,6 +86,9 @@ PYBIND11_MODULE(_native, m) {
         py::arg("memory_max") = "");
   m.def("cgroup_remove", &CgroupRemove, py::arg("path"));
   m.def("cgroup_proc_count", &CgroupProcCount, py::arg("path"));
+  m.def("cgroup_attach_device_filter", &CgroupAttachDeviceFilter,
+        py::arg("cgroup_dir"), py::arg("denied_major"),
+        py::arg("allowed_minors"));
 
   py::class_<Event>(m, "Event")
       .def_property_readonly("type",

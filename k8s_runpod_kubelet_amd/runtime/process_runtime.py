@@ -122,6 +122,7 @@ class ProcessRuntime(Runtime):
         )
 
         inst.cgroup_dir = self._acquire_cgroup(params)
+        self._attach_device_filter(inst)
 
         try:
             spawn_s = self._launch_containers(inst)
@@ -148,6 +149,32 @@ class ProcessRuntime(Runtime):
             },
         )
         return self._status_of(inst)
+
+    DRM_MAJOR = 226
+
+    def _attach_device_filter(self, inst: Instance) -> None:
+        """Enforced GPU isolation (SURVEY §7 hard part (a)): a cgroup-v2
+        eBPF device filter denies every DRM render node except the bound
+        GPUs' — ROCm cannot acquire a KFD VM without opening the GPU's
+        /dev/dri/renderD<minor>, so ROCR_VISIBLE_DEVICES stops being merely
+        cooperative. Best-effort like the cgroup limits (no-op when
+        unprivileged)."""
+        if not inst.cgroup_dir:
+            return
+        minors = []
+        inv = self.binder.ledger.inventory
+        for idx in inst.gpu_indices:
+            gpu = inv.get(idx)
+            if gpu is not None and gpu.render_minor >= 0:
+                minors.append(gpu.render_minor)
+        ok = self._native.cgroup_attach_device_filter(
+            inst.cgroup_dir, self.DRM_MAJOR, minors)
+        if ok:
+            log.debug("device filter attached",
+                      extra={"pod": inst.pod_key, "render_minors": minors})
+        else:
+            log.debug("device filter unavailable (unprivileged?)",
+                      extra={"pod": inst.pod_key})
 
     def _launch_containers(self, inst: Instance) -> float:
         """Spawn every container of the pod; returns the summed native

@@ -184,3 +184,55 @@ def test_clone3_born_in_cgroup(native, tmp_path):
                 break
             except OSError:
                 time.sleep(0.05)
+
+
+def test_cgroup_device_filter_enforced(native, tmp_path):
+    """The hand-assembled BPF_PROG_TYPE_CGROUP_DEVICE program actually
+    enforces: inside the filtered cgroup, major 1 (mem devices) is denied
+    except minor 5 (/dev/zero) — so /dev/zero opens and /dev/null fails
+    with EPERM. (In production the denied major is 226/DRM: only the bound
+    GPUs' render nodes are reachable.)"""
+    import secrets
+    import subprocess
+    import sys
+
+    cgroot = None
+    for cand in ("/sys/fs/cgroup", "/sys/fs/cgroup/unified"):
+        if os.path.exists(os.path.join(cand, "cgroup.procs")) and \
+                os.access(cand, os.W_OK):
+            cgroot = cand
+            break
+    if cgroot is None:
+        pytest.skip("no writable cgroup2 hierarchy")
+    cgdir = os.path.join(cgroot, f"amdvk-dev-{secrets.token_hex(4)}")
+    os.mkdir(cgdir)
+    try:
+        if not native.cgroup_attach_device_filter(cgdir, 1, [5]):
+            pytest.skip("BPF device filter not permitted on this box")
+        probe = (
+            "import os\n"
+            "os.close(os.open('/dev/zero', os.O_RDONLY))\n"  # allowed minor
+            "try:\n"
+            "    os.open('/dev/null', os.O_WRONLY)\n"
+            "except PermissionError:\n"
+            "    print('DENIED-OK'); raise SystemExit(0)\n"
+            "print('NOT-DENIED'); raise SystemExit(1)\n"
+        )
+        out = tmp_path / "probe.log"
+        pid, pidfd, _, _, _ = native.launch_process(
+            [sys.executable, "-c", probe], base_env(), "",
+            str(out), str(out), cgdir, True, False,
+        )
+        loop = native.EventLoop()
+        loop.add_process(pid, pidfd, -1, pid)
+        events = drain(loop, "exited", timeout_s=15)
+        exit_ev = [e for e in events if e.type == "exited"]
+        assert exit_ev and exit_ev[0].exit_code == 0, out.read_text()
+        assert "DENIED-OK" in out.read_text()
+    finally:
+        for _ in range(50):
+            try:
+                os.rmdir(cgdir)
+                break
+            except OSError:
+                time.sleep(0.05)
