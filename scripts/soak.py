@@ -1,0 +1,189 @@
+#!/usr/bin/env python3
+"""Continuous-churn robustness soak for the full stack.
+
+Runs the production path — FakeApiServer (HTTP) ← HttpK8sClient ← informer
+← PodController ← Provider ← ProcessRuntime — under randomized pod churn
+for --duration seconds: a rolling population of 1-GPU (or CPU) pods in
+mixed modes (hold-then-delete, run-to-completion success/failure,
+out-of-band SIGKILL), then checks for leaks and prints one JSON summary.
+
+    python scripts/soak.py --duration 240          # GPU box
+    python scripts/soak.py --duration 20 --cpu     # dev container
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import sys
+import tempfile
+import time
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO_ROOT)
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--duration", type=float, default=240.0)
+    ap.add_argument("--max-active", type=int, default=6)
+    ap.add_argument("--cpu", action="store_true",
+                    help="synthetic inventory + CPU podworker")
+    ap.add_argument("--seed", type=int, default=12345)
+    args = ap.parse_args()
+    rng = random.Random(args.seed)
+
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.client import NotFoundError
+    from k8s_runpod_kubelet_amd.kube.fake_apiserver import FakeApiServer
+    from k8s_runpod_kubelet_amd.kube.real import ClusterConfig, HttpK8sClient
+    from k8s_runpod_kubelet_amd.provider import annotations as ann
+    from k8s_runpod_kubelet_amd.server import metrics as m
+
+    srv = FakeApiServer().start()
+    client = HttpK8sClient(ClusterConfig(server=srv.url))
+    cfg = Config(
+        state_dir=tempfile.mkdtemp(prefix="amdvk-soak-"),
+        gpu_count_override=(8 if args.cpu else -1),
+        pending_retry_interval_s=0.5,
+        notify_interval_s=0,
+        pod_controller_workers=8,
+    )
+    stack = build_stack(cfg, client=client)
+    stack.start(serve_http=False)
+
+    gpus = 0 if args.cpu else 1
+    pw_args = (["--expect-gpus", "1"] if gpus else [])
+
+    def make(name, mode):
+        if mode == "hold" or mode == "crash":
+            a = pw_args + ["--hold"]
+        elif mode == "ok":
+            a = pw_args + ["--run-for", f"{rng.uniform(0.05, 0.4):.2f}"]
+        else:  # fail
+            a = pw_args + ["--run-for", "0.05", "--exit-code", "7"]
+        return {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": name, "namespace": "default"},
+            "spec": {
+                "nodeName": cfg.node_name,
+                "containers": [{
+                    "name": "main", "image": "amdvk/podworker:soak",
+                    "command": ["podworker"], "args": a,
+                    **({"resources": {"limits": {"amd.com/gpu": str(gpus)}}}
+                       if gpus else {}),
+                }],
+            },
+        }
+
+    def pod_state(name):
+        try:
+            p = client.get_pod("default", name)
+        except NotFoundError:
+            return "Gone", None
+        conds = {c["type"]: c["status"]
+                 for c in p.get("status", {}).get("conditions", [])}
+        phase = p.get("status", {}).get("phase", "")
+        if conds.get("Ready") == "True":
+            return "Ready", p
+        return phase or "Pending", p
+
+    def fd_count():
+        return len(os.listdir("/proc/self/fd"))
+
+    fd0 = fd_count()
+    active = {}  # name -> dict(mode, created, state)
+    counters = {"created": 0, "succeeded": 0, "failed": 0, "crashed": 0,
+                "deleted_holds": 0, "timeouts": 0}
+    seq = 0
+    deadline = time.time() + args.duration
+    try:
+        while time.time() < deadline or active:
+            now = time.time()
+            # top up population (only while inside the window)
+            while now < deadline and len(active) < args.max_active:
+                mode = rng.choices(["hold", "ok", "fail", "crash"],
+                                   weights=[3, 4, 2, 1])[0]
+                name = f"soak-{seq:05d}"
+                seq += 1
+                client.create_pod("default", make(name, mode))
+                active[name] = {"mode": mode, "created": now,
+                                "dwell": rng.uniform(0.2, 1.5),
+                                "killed": False, "deleted": False}
+                counters["created"] += 1
+            for name, st in list(active.items()):
+                state, pod = pod_state(name)
+                age = now - st["created"]
+                if state == "Gone":
+                    del active[name]
+                    continue
+                if age > 120:
+                    counters["timeouts"] += 1
+                    print(f"TIMEOUT: {name} mode={st['mode']} state={state}",
+                          file=sys.stderr)
+                    try:
+                        client.delete_pod("default", name)
+                    except NotFoundError:
+                        pass
+                    del active[name]
+                    continue
+                if st["mode"] in ("ok", "fail"):
+                    want = "Succeeded" if st["mode"] == "ok" else "Failed"
+                    if state == want and not st["deleted"]:
+                        counters["succeeded" if want == "Succeeded"
+                                 else "failed"] += 1
+                        st["deleted"] = True
+                        client.delete_pod("default", name)
+                elif st["mode"] == "crash":
+                    if state == "Ready" and not st["killed"]:
+                        iid = pod["metadata"]["annotations"].get(ann.POD_ID)
+                        det = stack.runtime.get_detailed_status(iid)
+                        if det.containers:
+                            os.kill(det.containers[0].pid, 9)
+                            st["killed"] = True
+                    elif state == "Failed" and not st["deleted"]:
+                        counters["crashed"] += 1
+                        st["deleted"] = True
+                        client.delete_pod("default", name)
+                else:  # hold
+                    if state == "Ready" and age > st["dwell"] and not st["deleted"]:
+                        counters["deleted_holds"] += 1
+                        st["deleted"] = True
+                        client.delete_pod("default", name)
+            time.sleep(0.02)
+
+        stack.provider.cleanup_deleted_pods()
+        leftovers = stack.runtime.list_instances()
+        summary = {
+            "duration_s": args.duration,
+            "mode": "cpu" if args.cpu else "gpu",
+            **counters,
+            "pods_done": counters["succeeded"] + counters["failed"]
+            + counters["crashed"] + counters["deleted_holds"],
+            "mean_ready_ms": round(m.hist_mean_ms(m.pod_ready_seconds), 1),
+            "leaks": {
+                "reservations": len(stack.ledger.reservations),
+                "tracked_processes": stack.runtime.tracked_process_count(),
+                "instances": len(leftovers),
+                "fds_delta": fd_count() - fd0,
+            },
+        }
+        ok = (counters["timeouts"] == 0
+              and summary["leaks"]["reservations"] == 0
+              and summary["leaks"]["tracked_processes"] == 0
+              and summary["leaks"]["instances"] == 0
+              and summary["leaks"]["fds_delta"] < 20)
+        summary["ok"] = ok
+        print(json.dumps(summary))
+        return 0 if ok else 1
+    finally:
+        stack.stop()
+        client.close()
+        srv.stop()
+
+
+if __name__ == "__main__":
+    sys.exit(main())
