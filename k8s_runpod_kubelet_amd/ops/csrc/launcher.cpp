@@ -130,6 +130,8 @@ struct ChildCtx {
   const char* stderr_path; // may be empty
   int ready_fd;            // write end to keep across exec (-1 = none)
   bool new_session;
+  int64_t uid;             // -1 = inherit
+  int64_t gid;
   const sigset_t* parent_mask;  // restored just before exec
   volatile int* exec_errno;     // shared (CLONE_VM): child reports failure
 };
@@ -160,6 +162,22 @@ int ChildMain(void* p) {
   }
   if (c->ready_fd >= 0) syscall(SYS_fcntl, c->ready_fd, F_SETFD, 0);
   if (c->cwd && c->cwd[0]) syscall(SYS_chdir, c->cwd);
+  // securityContext.runAsGroup/runAsUser: drop credentials last (after the
+  // log-file opens, which may need root). Order matters — gid while still
+  // privileged, uid last. A failed drop must NOT exec a root process.
+  if (c->gid >= 0) {
+    if (syscall(SYS_setgroups, 0, nullptr) != 0 ||
+        syscall(SYS_setgid, static_cast<gid_t>(c->gid)) != 0) {
+      *c->exec_errno = errno ? errno : EPERM;
+      return 126;
+    }
+  }
+  if (c->uid >= 0) {
+    if (syscall(SYS_setuid, static_cast<uid_t>(c->uid)) != 0) {
+      *c->exec_errno = errno ? errno : EPERM;
+      return 126;
+    }
+  }
   syscall(SYS_rt_sigprocmask, SIG_SETMASK, c->parent_mask, nullptr, 8);
   syscall(SYS_execve, c->path, c->argv, c->envp);
   *c->exec_errno = errno;
@@ -168,26 +186,33 @@ int ChildMain(void* p) {
 
 // Returns child pid (>0) on success with *pidfd_out set, 0 when the fast
 // path is unavailable (caller falls back to posix_spawn + migrate), or
-// -errno on a real launch failure.
+// -errno on a real launch failure. cgroup_dir may be empty (plain
+// vfork-style clone3, used when only credential dropping is needed).
 long SpawnIntoCgroup(const char* cgroup_dir, ChildCtx* ctx, int* pidfd_out) {
-  int cgfd = open(cgroup_dir, O_DIRECTORY | O_RDONLY | O_CLOEXEC);
-  if (cgfd < 0) return 0;
+  int cgfd = -1;
+  if (cgroup_dir && cgroup_dir[0]) {
+    cgfd = open(cgroup_dir, O_DIRECTORY | O_RDONLY | O_CLOEXEC);
+    if (cgfd < 0) return 0;
+  }
   constexpr size_t kStackSize = 256 * 1024;
   void* stack = mmap(nullptr, kStackSize, PROT_READ | PROT_WRITE,
                      MAP_PRIVATE | MAP_ANONYMOUS | MAP_STACK, -1, 0);
   if (stack == MAP_FAILED) {
-    close(cgfd);
+    if (cgfd >= 0) close(cgfd);
     return 0;
   }
   int pidfd = -1;
   struct clone_args ca;
   memset(&ca, 0, sizeof(ca));
-  ca.flags = CLONE_VM | CLONE_VFORK | CLONE_INTO_CGROUP | CLONE_PIDFD;
+  ca.flags = CLONE_VM | CLONE_VFORK | CLONE_PIDFD;
+  if (cgfd >= 0) {
+    ca.flags |= CLONE_INTO_CGROUP;
+    ca.cgroup = static_cast<uint64_t>(cgfd);
+  }
   ca.pidfd = reinterpret_cast<uint64_t>(&pidfd);
   ca.exit_signal = SIGCHLD;
   ca.stack = reinterpret_cast<uint64_t>(stack);
   ca.stack_size = kStackSize;
-  ca.cgroup = static_cast<uint64_t>(cgfd);
 
   // Block every signal across the clone window: the child shares our
   // address space until execve, so no handler may run in it. ChildMain
@@ -200,7 +225,7 @@ long SpawnIntoCgroup(const char* cgroup_dir, ChildCtx* ctx, int* pidfd_out) {
   pthread_sigmask(SIG_SETMASK, &saved, nullptr);
 
   munmap(stack, kStackSize);  // child has execed (or exited): mapping is ours
-  close(cgfd);
+  if (cgfd >= 0) close(cgfd);
   if (rv < 0) {
     int err = static_cast<int>(-rv);
     // Kernel without CLONE_INTO_CGROUP support / cgroup v1 fd / no
@@ -259,9 +284,10 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
 
 #if defined(__x86_64__)
   // Fast path: child born inside its cgroup (no ~130 ms cgroup.procs
-  // migration; see comment above SpawnIntoCgroup). Needs an absolute/
-  // relative path in argv[0] — execve does no PATH search.
-  if (!spec.cgroup_dir.empty() &&
+  // migration; see comment above SpawnIntoCgroup) and/or with dropped
+  // credentials (posix_spawn cannot setuid). Needs an absolute/relative
+  // path in argv[0] — execve does no PATH search.
+  if ((!spec.cgroup_dir.empty() || spec.uid >= 0 || spec.gid >= 0) &&
       spec.argv[0].find('/') != std::string::npos) {
     volatile int exec_errno = 0;
     ChildCtx ctx{};
@@ -273,6 +299,8 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
     ctx.stderr_path = spec.stderr_path.c_str();
     ctx.ready_fd = spec.ready_pipe ? ready_pipe[1] : -1;
     ctx.new_session = spec.new_session;
+    ctx.uid = spec.uid;
+    ctx.gid = spec.gid;
     ctx.exec_errno = &exec_errno;
     auto t0 = std::chrono::steady_clock::now();
     int pidfd = -1;
@@ -300,9 +328,24 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
       return res;
     }
     // rv == 0: fast path unavailable here — fall through to
-    // posix_spawn + cgroup.procs migration.
+    // posix_spawn + cgroup.procs migration. Never silently run a pod that
+    // asked for dropped credentials as the kubelet's own user.
+    if (spec.uid >= 0 || spec.gid >= 0) {
+      if (ready_pipe[0] >= 0) close(ready_pipe[0]);
+      if (ready_pipe[1] >= 0) close(ready_pipe[1]);
+      res.error = "runAsUser/runAsGroup requires the clone3 spawn path, "
+                  "which is unavailable here";
+      return res;
+    }
   }
 #endif
+  if (spec.uid >= 0 || spec.gid >= 0) {
+    if (ready_pipe[0] >= 0) close(ready_pipe[0]);
+    if (ready_pipe[1] >= 0) close(ready_pipe[1]);
+    res.error = "runAsUser/runAsGroup requires an absolute path in argv[0] "
+                "(no PATH search on the credential-dropping spawn path)";
+    return res;
+  }
 
   posix_spawn_file_actions_t fa;
   posix_spawn_file_actions_init(&fa);

@@ -17,6 +17,11 @@ struct LaunchSpec {
   std::string cgroup_dir;  // if non-empty, child is placed here before exec
   bool new_session = true;
   bool ready_pipe = true;  // create readiness pipe, exported as AMDVK_READY_FD
+  // Pod securityContext runAsUser/runAsGroup: credentials dropped in the
+  // child just before exec (-1 = inherit). Requires the clone3 fast path
+  // (argv[0] with a '/'): posix_spawn has no setuid file action.
+  int64_t uid = -1;
+  int64_t gid = -1;
 };
 
 struct LaunchResult {

@@ -55,7 +55,7 @@ PYBIND11_MODULE(_native, m) {
       [](const std::vector<std::string>& argv, const std::vector<std::string>& env,
          const std::string& cwd, const std::string& stdout_path,
          const std::string& stderr_path, const std::string& cgroup_dir,
-         bool new_session, bool ready_pipe) {
+         bool new_session, bool ready_pipe, int64_t uid, int64_t gid) {
         LaunchSpec spec;
         spec.argv = argv;
         spec.env = env;
@@ -65,6 +65,8 @@ PYBIND11_MODULE(_native, m) {
         spec.cgroup_dir = cgroup_dir;
         spec.new_session = new_session;
         spec.ready_pipe = ready_pipe;
+        spec.uid = uid;
+        spec.gid = gid;
         LaunchResult res;
         {
           py::gil_scoped_release release;
@@ -77,7 +79,7 @@ PYBIND11_MODULE(_native, m) {
       py::arg("argv"), py::arg("env"), py::arg("cwd") = "",
       py::arg("stdout_path") = "", py::arg("stderr_path") = "",
       py::arg("cgroup_dir") = "", py::arg("new_session") = true,
-      py::arg("ready_pipe") = true);
+      py::arg("ready_pipe") = true, py::arg("uid") = -1, py::arg("gid") = -1);
 
   m.def("open_pidfd", &OpenPidfd, py::arg("pid"));
   m.def("signal_process", &SignalProcess, py::arg("pid"), py::arg("sig"),

@@ -210,8 +210,15 @@ def prepare_deploy_params(
     else:
         requested_ports = get_requested_ports(pod)
 
+    # securityContext.runAsUser/runAsGroup: pod-level default, container-level
+    # override (the reference's containers run as their image's user; local
+    # processes honor the same kubectl-facing fields).
+    pod_sc = pod.get("spec", {}).get("securityContext", {}) or {}
     containers: List[ContainerSpec] = []
     for i, c in enumerate(pod.get("spec", {}).get("containers", []) or []):
+        c_sc = c.get("securityContext", {}) or {}
+        uid = c_sc.get("runAsUser", pod_sc.get("runAsUser"))
+        gid = c_sc.get("runAsGroup", pod_sc.get("runAsGroup"))
         containers.append(
             ContainerSpec(
                 name=c.get("name", f"c{i}"),
@@ -221,6 +228,8 @@ def prepare_deploy_params(
                 env=per_container_env[i] if i < len(per_container_env) else {},
                 working_dir=c.get("workingDir", ""),
                 tcp_ports=tcp_ports_of_container(c),
+                run_as_uid=int(uid) if uid is not None else -1,
+                run_as_gid=int(gid) if gid is not None else -1,
             )
         )
     if not containers:

@@ -204,6 +204,13 @@ class ProcessRuntime(Runtime):
                     argv += ["--listen-port", str(port)]
             elif argv[0] in ("podworker", "amdvk-podworker"):
                 argv[0] = self.podworker_path()
+            if (cspec.run_as_uid >= 0 or cspec.run_as_gid >= 0) and \
+                    "/" not in argv[0]:
+                # Credential dropping happens on the execve fast path, which
+                # does no PATH search — resolve here instead.
+                resolved = shutil.which(argv[0])
+                if resolved:
+                    argv[0] = resolved
 
             env = dict(base_env)
             env.update(cspec.env)
@@ -215,6 +222,7 @@ class ProcessRuntime(Runtime):
                 cspec.working_dir or "",
                 stdout_path, stdout_path,
                 inst.cgroup_dir, True, True,
+                cspec.run_as_uid, cspec.run_as_gid,
             )
             spawn_total_s += spawn_s
             if inst.cgroup_dir:

@@ -93,6 +93,10 @@ class ContainerSpec:
     env: Dict[str, str] = field(default_factory=dict)
     working_dir: str = ""
     tcp_ports: List[int] = field(default_factory=list)
+    # securityContext.runAsUser/runAsGroup (container overrides pod; -1 =
+    # inherit the kubelet's credentials)
+    run_as_uid: int = -1
+    run_as_gid: int = -1
 
 
 @dataclass

@@ -261,3 +261,30 @@ def test_event_driven_pending_placement(tmp_state_dir):
         assert time.time() - t0 < 5  # placed via exit event, no 999 s tick
     finally:
         s.stop()
+
+
+def test_security_context_run_as_user(stack):
+    """securityContext.runAsUser/runAsGroup flow end to end: the pod's
+    process runs with dropped credentials (reference parity: containers run
+    as their image's user; kubectl-facing fields honored here)."""
+    import os
+
+    if os.geteuid() != 0:
+        pytest.skip("needs root to drop to another uid")
+    s, kube = stack
+    pod = make_pod("dropper", command=["/usr/bin/id"], args=[])
+    pod["spec"]["securityContext"] = {"runAsUser": 65534, "runAsGroup": 65534}
+    kube.create_pod("default", pod)
+
+    def done():
+        try:
+            p = kube.get_pod("default", "dropper")
+        except NotFoundError:
+            return None
+        return p if p.get("status", {}).get("phase") == "Succeeded" else None
+
+    assert wait_until(done, timeout_s=15) is not None
+    logs = s.provider.get_container_logs("default", "dropper")
+    assert "uid=65534" in logs and "gid=65534" in logs, logs
+    kube.delete_pod("default", "dropper")
+    assert wait_until(lambda: gone(kube, "dropper"), timeout_s=15)

@@ -236,3 +236,32 @@ def test_cgroup_device_filter_enforced(native, tmp_path):
                 break
             except OSError:
                 time.sleep(0.05)
+
+
+def test_run_as_user_drops_credentials(native, tmp_path):
+    """securityContext.runAsUser: the clone3 child setgid/setuids before
+    exec; the workload must observe the dropped identity."""
+    if os.geteuid() != 0:
+        pytest.skip("needs root to drop to another uid")
+    out = tmp_path / "id.log"
+    pid, pidfd, _, _, _ = native.launch_process(
+        ["/usr/bin/id"], ["PATH=/usr/bin:/bin"], "",
+        str(out), str(out), "", True, False, 65534, 65534,
+    )
+    loop = native.EventLoop()
+    loop.add_process(pid, pidfd, -1, pid)
+    events = drain(loop, "exited", timeout_s=10)
+    assert any(e.type == "exited" and e.exit_code == 0 for e in events)
+    text = out.read_text()
+    assert "uid=65534" in text and "gid=65534" in text, text
+
+
+def test_run_as_user_never_falls_back_to_root(native, tmp_path):
+    """A pod that asked for dropped credentials must fail loudly rather
+    than silently run as the kubelet's user when the fast path can't be
+    used (argv[0] without '/': no PATH search on the execve path)."""
+    with pytest.raises(RuntimeError, match="runAsUser"):
+        native.launch_process(
+            ["id"], ["PATH=/usr/bin:/bin"], "", "", "", "",
+            True, False, 65534, -1,
+        )
