@@ -112,6 +112,7 @@ def build_stack(
                 state_dir=config.state_dir,
                 cgroup_root=config.cgroup_root,
                 cgroup_parent=config.cgroup_parent,
+                pod_namespaces=config.pod_namespaces,
             )
 
     provider = Provider(client, config, runtime, ledger=ledger, inventory=inventory)

@@ -69,6 +69,8 @@ class Config:
     sysfs_root: str = "/sys"
     cgroup_root: str = "/sys/fs/cgroup"
     cgroup_parent: str = "amdvk.slice"
+    # pod PID/UTS namespace isolation (auto-degrades without CAP_SYS_ADMIN)
+    pod_namespaces: bool = True
     runtime: str = "process"  # process | fake
     pod_log_dir: str = ""  # defaults to <state_dir>/logs
     pod_controller_workers: int = 4  # reference uses 1 (main.go:263)

@@ -132,6 +132,8 @@ struct ChildCtx {
   bool new_session;
   int64_t uid;             // -1 = inherit
   int64_t gid;
+  const char* hostname;    // set in the new UTS ns (empty = none)
+  bool new_pid_ns;
   const sigset_t* parent_mask;  // restored just before exec
   volatile int* exec_errno;     // shared (CLONE_VM): child reports failure
 };
@@ -143,6 +145,10 @@ struct ChildCtx {
 int ChildMain(void* p) {
   ChildCtx* c = static_cast<ChildCtx*>(p);
   if (c->new_session) syscall(SYS_setsid);
+  if (c->hostname && c->hostname[0])
+    // We are in the pod's own UTS namespace (CLONE_NEWUTS): affects only
+    // this pod. Must run before credentials are dropped.
+    syscall(SYS_sethostname, c->hostname, strlen(c->hostname));
   const char* errp =
       (c->stderr_path && c->stderr_path[0]) ? c->stderr_path : c->stdout_path;
   if (c->stdout_path && c->stdout_path[0]) {
@@ -209,6 +215,10 @@ long SpawnIntoCgroup(const char* cgroup_dir, ChildCtx* ctx, int* pidfd_out) {
     ca.flags |= CLONE_INTO_CGROUP;
     ca.cgroup = static_cast<uint64_t>(cgfd);
   }
+  uint64_t ns_flags = 0;
+  if (ctx->new_pid_ns) ns_flags |= CLONE_NEWPID;
+  if (ctx->hostname && ctx->hostname[0]) ns_flags |= CLONE_NEWUTS;
+  ca.flags |= ns_flags;
   ca.pidfd = reinterpret_cast<uint64_t>(&pidfd);
   ca.exit_signal = SIGCHLD;
   ca.stack = reinterpret_cast<uint64_t>(stack);
@@ -222,6 +232,14 @@ long SpawnIntoCgroup(const char* cgroup_dir, ChildCtx* ctx, int* pidfd_out) {
   pthread_sigmask(SIG_SETMASK, &all, &saved);
   ctx->parent_mask = &saved;
   long rv = amdvk_clone3_run(&ca, sizeof(ca), ChildMain, ctx);
+  if (rv == -EPERM && ns_flags != 0) {
+    // Namespaces need CAP_SYS_ADMIN (the gpurun sandbox drops it):
+    // degrade to no-namespace isolation, keeping cgroup + credentials.
+    ctx->new_pid_ns = false;
+    ctx->hostname = "";
+    ca.flags &= ~ns_flags;
+    rv = amdvk_clone3_run(&ca, sizeof(ca), ChildMain, ctx);
+  }
   pthread_sigmask(SIG_SETMASK, &saved, nullptr);
 
   munmap(stack, kStackSize);  // child has execed (or exited): mapping is ours
@@ -287,7 +305,8 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
   // migration; see comment above SpawnIntoCgroup) and/or with dropped
   // credentials (posix_spawn cannot setuid). Needs an absolute/relative
   // path in argv[0] — execve does no PATH search.
-  if ((!spec.cgroup_dir.empty() || spec.uid >= 0 || spec.gid >= 0) &&
+  if ((!spec.cgroup_dir.empty() || spec.uid >= 0 || spec.gid >= 0 ||
+       spec.new_pid_ns || !spec.hostname.empty()) &&
       spec.argv[0].find('/') != std::string::npos) {
     volatile int exec_errno = 0;
     ChildCtx ctx{};
@@ -301,6 +320,8 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
     ctx.new_session = spec.new_session;
     ctx.uid = spec.uid;
     ctx.gid = spec.gid;
+    ctx.hostname = spec.hostname.c_str();
+    ctx.new_pid_ns = spec.new_pid_ns;
     ctx.exec_errno = &exec_errno;
     auto t0 = std::chrono::steady_clock::now();
     int pidfd = -1;

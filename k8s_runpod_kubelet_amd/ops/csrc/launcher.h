@@ -22,6 +22,12 @@ struct LaunchSpec {
   // (argv[0] with a '/'): posix_spawn has no setuid file action.
   int64_t uid = -1;
   int64_t gid = -1;
+  // Container-like isolation (clone3 path; needs CAP_SYS_ADMIN — retried
+  // without when the kernel refuses): own PID namespace (workload is pid 1,
+  // descendants die with it) and, when hostname is non-empty, own UTS
+  // namespace with that hostname (k8s pod-hostname semantics).
+  bool new_pid_ns = false;
+  std::string hostname;
 };
 
 struct LaunchResult {

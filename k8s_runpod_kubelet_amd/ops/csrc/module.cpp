@@ -55,7 +55,8 @@ PYBIND11_MODULE(_native, m) {
       [](const std::vector<std::string>& argv, const std::vector<std::string>& env,
          const std::string& cwd, const std::string& stdout_path,
          const std::string& stderr_path, const std::string& cgroup_dir,
-         bool new_session, bool ready_pipe, int64_t uid, int64_t gid) {
+         bool new_session, bool ready_pipe, int64_t uid, int64_t gid,
+         bool new_pid_ns, const std::string& hostname) {
         LaunchSpec spec;
         spec.argv = argv;
         spec.env = env;
@@ -67,6 +68,8 @@ PYBIND11_MODULE(_native, m) {
         spec.ready_pipe = ready_pipe;
         spec.uid = uid;
         spec.gid = gid;
+        spec.new_pid_ns = new_pid_ns;
+        spec.hostname = hostname;
         LaunchResult res;
         {
           py::gil_scoped_release release;
@@ -79,7 +82,8 @@ PYBIND11_MODULE(_native, m) {
       py::arg("argv"), py::arg("env"), py::arg("cwd") = "",
       py::arg("stdout_path") = "", py::arg("stderr_path") = "",
       py::arg("cgroup_dir") = "", py::arg("new_session") = true,
-      py::arg("ready_pipe") = true, py::arg("uid") = -1, py::arg("gid") = -1);
+      py::arg("ready_pipe") = true, py::arg("uid") = -1, py::arg("gid") = -1,
+      py::arg("new_pid_ns") = false, py::arg("hostname") = "");
 
   m.def("open_pidfd", &OpenPidfd, py::arg("pid"));
   m.def("signal_process", &SignalProcess, py::arg("pid"), py::arg("sig"),

@@ -254,4 +254,6 @@ def prepare_deploy_params(
         cpu_limit=cpu_max,
         memory_limit=memory_max,
         labels=dict(metadata.get("labels", {}) or {}),
+        hostname=pod.get("spec", {}).get("hostname")
+        or metadata.get("name", ""),
     )

@@ -119,6 +119,9 @@ class DeployParams:
     cpu_limit: str = ""      # cgroup cpu.max, e.g. "200000 100000"
     memory_limit: str = ""   # cgroup memory.max bytes or "max"
     labels: Dict[str, str] = field(default_factory=dict)
+    # k8s pod-hostname semantics (spec.hostname, else pod name); applied in
+    # the pod's own UTS namespace when namespace isolation is available
+    hostname: str = ""
 
 
 @dataclass
