@@ -116,6 +116,8 @@ def build_stack(
             )
 
     provider = Provider(client, config, runtime, ledger=ledger, inventory=inventory)
+    recorder = EventRecorder(client)
+    provider.recorder = recorder
     informer = PodInformer(
         client, config.node_name, resync_interval_s=config.reconcile_interval_s
     )
@@ -125,7 +127,6 @@ def build_stack(
     node_controller = NodeController(
         client, provider, status_interval_s=config.reconcile_interval_s
     )
-    recorder = EventRecorder(client)
     return Stack(
         config=config,
         client=client,

@@ -62,7 +62,8 @@ class KubeletApiServer:
                     ).encode()
                     self._respond(200, body, "application/json")
                 elif parts and parts[0] == "stats" and len(parts) > 1 and parts[1] == "summary":
-                    self._respond(200, json.dumps({"node": {}, "pods": []}).encode(),
+                    summary = outer.provider.get_stats_summary()
+                    self._respond(200, json.dumps(summary).encode(),
                                   "application/json")
                 elif parts and parts[0] == "healthz":
                     self._respond(200, b"ok")

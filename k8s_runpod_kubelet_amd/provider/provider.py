@@ -94,6 +94,10 @@ class Provider:
         self.backend_available = True
         self._tickers: List[Ticker] = []
         self._started = False
+        # Optional EventRecorder (set by app.build_stack): lifecycle events
+        # the reference gets from the virtual-kubelet library's recorder
+        # (main.go:172-177).
+        self.recorder = None
 
         self.registrar = Registrar(
             config.registration_endpoint,
@@ -213,9 +217,11 @@ class Provider:
             # semantics: CreatePod returns nil, retry loop re-deploys).
             log.info("pod queued: no placement yet",
                      extra={"pod": key, "reason": str(exc)})
+            self._emit(pod, "Normal", "PlacementPending", str(exc))
         except Exception as exc:
             log.warning("deploy failed; pod stays pending",
                         extra={"pod": key, "err": str(exc)})
+            self._emit(pod, "Warning", "DeployError", str(exc))
 
     def update_pod(self, pod: Dict[str, Any]) -> None:
         """UpdatePod (reference kubelet.go:421-432): refresh the cached copy."""
@@ -344,6 +350,9 @@ class Provider:
         )
         detailed = self.runtime.deploy(params)
         metrics.deploy_seconds.observe(time.monotonic() - t0)
+        self._emit(pod, "Normal", "Started",
+                   f"instance {detailed.id} on GPUs {detailed.gpu_indices}"
+                   if detailed.gpu_indices else f"instance {detailed.id}")
 
         with self._pods_lock:
             info = self._pod_status.get(key)
@@ -407,6 +416,14 @@ class Provider:
             status = PodStatus.NOT_FOUND
         if status in (PodStatus.EXITED, PodStatus.TERMINATED, PodStatus.NOT_FOUND):
             self._nudge_pending()
+
+    def _emit(self, pod: Dict[str, Any], etype: str, reason: str,
+              message: str) -> None:
+        if self.recorder is not None:
+            try:
+                self.recorder.event(pod, etype, reason, message)
+            except Exception:
+                log.debug("event emit failed", extra={"reason": reason})
 
     def _nudge_pending(self) -> None:
         with self._pods_lock:
@@ -507,6 +524,9 @@ class Provider:
         """Reference handlePodCompletion (kubelet.go:998-1065)."""
         status = translate_status(pod, info, detailed, self.config.internal_ip)
         self._push_status(pod, status)
+        if status.get("phase") == "Failed":
+            self._emit(pod, "Warning", "Failed",
+                       detailed.completion_message or "workload failed")
         log.info(
             "pod completed",
             extra={
@@ -556,6 +576,8 @@ class Provider:
                 # 15 min cutoff → PodFailed/RunPodDeploymentFailed
                 # (kubelet.go:788-806) — the timeout is configurable here.
                 log.warning("pending pod timed out", extra={"pod": key, "age_s": int(age)})
+                self._emit(pod, "Warning", "FailedDeployment",
+                           f"not placed within {int(age)}s")
                 info.status = PodStatus.EXITED
                 status = {
                     "phase": "Failed",
@@ -924,3 +946,62 @@ class Provider:
     def instance_info(self, namespace: str, name: str) -> Optional[InstanceInfo]:
         with self._pods_lock:
             return self._pod_status.get(f"{namespace}-{name}")
+
+    def get_stats_summary(self) -> Dict[str, Any]:
+        """Kubelet Summary-API-shaped stats (the reference stubs these hooks
+        off, main.go:233-235): node CPU/memory, per-pod usage from the
+        runtime, and an MI355X extension block with per-GPU HBM/busy/temp."""
+        mem_total_kb = mem_avail_kb = 0
+        try:
+            with open("/proc/meminfo", "r", encoding="ascii") as fh:
+                for line in fh:
+                    if line.startswith("MemTotal:"):
+                        mem_total_kb = int(line.split()[1])
+                    elif line.startswith("MemAvailable:"):
+                        mem_avail_kb = int(line.split()[1])
+        except OSError:
+            pass
+        summary: Dict[str, Any] = {
+            "node": {
+                "nodeName": self.node_name,
+                "cpu": {"numCores": os.cpu_count() or 1},
+                "memory": {
+                    "availableBytes": mem_avail_kb * 1024,
+                    "usageBytes": max(0, (mem_total_kb - mem_avail_kb) * 1024),
+                },
+            },
+            "pods": [],
+        }
+        stats_fn = getattr(self.runtime, "get_stats", None)
+        with self._pods_lock:
+            items = []
+            for k, p in self._pods.items():
+                info = self._pod_status.get(k)
+                items.append((
+                    k, dict(meta(p)),
+                    info.instance_id if info else "",
+                    list(info.gpu_indices) if info else [],
+                ))
+        for _key, md, instance_id, gpu_indices in items:
+            entry: Dict[str, Any] = {
+                "podRef": {"name": md.get("name", ""),
+                           "namespace": md.get("namespace", "default")},
+            }
+            if instance_id and callable(stats_fn):
+                entry.update(stats_fn(instance_id))
+            if gpu_indices:
+                entry["gpus"] = gpu_indices
+            summary["pods"].append(entry)
+        if self.ledger is not None:
+            summary["gpus"] = [
+                {
+                    "index": s.gpu.index,
+                    "hbmTotalBytes": s.gpu.vram_total_bytes,
+                    "hbmUsedBytes": s.gpu.vram_used_bytes,
+                    "busyPercent": s.gpu.busy_percent,
+                    "temperatureCelsius": (s.gpu.temperature_mc / 1000.0
+                                           if s.gpu.temperature_mc >= 0 else None),
+                }
+                for s in self.ledger.snapshot()
+            ]
+        return summary

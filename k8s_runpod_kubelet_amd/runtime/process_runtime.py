@@ -477,6 +477,52 @@ class ProcessRuntime(Runtime):
         """Processes the native event loop still watches (leak check)."""
         return int(self._loop.tracked_count())
 
+    def get_stats(self, instance_id: str) -> Dict[str, object]:
+        """Per-pod resource usage for /stats/summary: cgroup v2 counters
+        when the pod has a slot, /proc per-process fallback otherwise.
+        (The reference stubs its kubelet stats hooks off entirely,
+        cmd/virtual_kubelet/main.go:233-235 — local pods make them real.)"""
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        if inst is None:
+            return {}
+        stats: Dict[str, object] = {"containers": []}
+        if inst.cgroup_dir:
+            try:
+                with open(inst.cgroup_dir + "/cpu.stat", encoding="ascii") as fh:
+                    for line in fh:
+                        if line.startswith("usage_usec"):
+                            stats["cpuUsageCoreNanoSeconds"] = (
+                                int(line.split()[1]) * 1000)
+                            break
+                with open(inst.cgroup_dir + "/memory.current",
+                          encoding="ascii") as fh:
+                    stats["memoryUsageBytes"] = int(fh.read().strip())
+            except OSError:
+                pass
+        tick_ns = 1_000_000_000 // os.sysconf("SC_CLK_TCK")
+        page = os.sysconf("SC_PAGE_SIZE")
+        cpu_total = 0
+        mem_total = 0
+        for c in inst.containers:
+            entry: Dict[str, object] = {"name": c.name}
+            try:
+                with open(f"/proc/{c.pid}/stat", encoding="ascii") as fh:
+                    fields = fh.read().rsplit(") ", 1)[-1].split()
+                    # fields[11]=utime fields[12]=stime (post-comm offsets)
+                    entry["cpuUsageCoreNanoSeconds"] = (
+                        (int(fields[11]) + int(fields[12])) * tick_ns)
+                with open(f"/proc/{c.pid}/statm", encoding="ascii") as fh:
+                    entry["memoryRssBytes"] = int(fh.read().split()[1]) * page
+                cpu_total += entry.get("cpuUsageCoreNanoSeconds", 0)
+                mem_total += entry.get("memoryRssBytes", 0)
+            except (OSError, IndexError, ValueError):
+                pass  # container already exited
+            stats["containers"].append(entry)
+        stats.setdefault("cpuUsageCoreNanoSeconds", cpu_total)
+        stats.setdefault("memoryUsageBytes", mem_total)
+        return stats
+
     # ------------- logs -------------
 
     def get_logs(self, instance_id: str, container: str = "", tail: int = -1) -> str:

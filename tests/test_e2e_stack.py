@@ -288,3 +288,32 @@ def test_security_context_run_as_user(stack):
     assert "uid=65534" in logs and "gid=65534" in logs, logs
     kube.delete_pod("default", "dropper")
     assert wait_until(lambda: gone(kube, "dropper"), timeout_s=15)
+
+
+def test_lifecycle_events_emitted(stack):
+    """The provider emits kubectl-visible Events (Started on deploy,
+    Warning/Failed on workload failure) through the EventRecorder — the
+    reference relies on the virtual-kubelet lib's recorder for these."""
+    s, kube = stack
+    kube.create_pod("default", make_pod(
+        "eventful", gpus=1, command=["podworker"],
+        args=["--run-for", "0.05", "--exit-code", "5"]))
+
+    def failed():
+        try:
+            p = kube.get_pod("default", "eventful")
+        except NotFoundError:
+            return None
+        return p if p.get("status", {}).get("phase") == "Failed" else None
+
+    assert wait_until(failed, timeout_s=15) is not None
+
+    def events():
+        evs = [e for e in kube.events.objects.values()
+               if e["involvedObject"]["name"] == "eventful"]
+        reasons = {e["reason"] for e in evs}
+        return evs if {"Started", "Failed"} <= reasons else None
+
+    assert wait_until(events, timeout_s=10) is not None
+    kube.delete_pod("default", "eventful")
+    assert wait_until(lambda: gone(kube, "eventful"), timeout_s=15)

@@ -94,6 +94,14 @@ def test_kubelet_api(provider, process_runtime):
         assert get(f"{base}/healthz") == (200, "ok")
         code, body = get(f"{base}/stats/summary")
         assert code == 200
+        summary = json.loads(body)
+        assert summary["node"]["cpu"]["numCores"] >= 1
+        assert summary["node"]["memory"]["usageBytes"] > 0
+        api1 = [p for p in summary["pods"]
+                if p["podRef"]["name"] == "api1"][0]
+        # /proc-based fallback stats for the live pod process
+        assert api1["memoryUsageBytes"] > 0
+        assert api1["containers"][0]["name"] == "main"
     finally:
         srv.stop()
         pod_obj = prov.get_pod("default", "api1")
