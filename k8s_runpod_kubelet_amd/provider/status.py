@@ -112,6 +112,38 @@ def _container_statuses(
     return out
 
 
+def _init_container_statuses(
+    detailed: Optional[DetailedStatus],
+) -> List[Dict[str, Any]]:
+    """spec.initContainers status projection (running / terminated)."""
+    out = []
+    if detailed is None:
+        return out
+    for c in detailed.init_containers:
+        cs: Dict[str, Any] = {
+            "name": c.name, "image": "", "imageID": "",
+            "restartCount": 0, "ready": c.exit_code == 0,
+        }
+        if c.pid > 0:
+            cs["containerID"] = f"amdvk://{c.pid}"
+        if c.exit_code is None:
+            cs["state"] = {"running": {
+                "startedAt": ts_rfc3339(c.started_at) if c.started_at
+                else now_rfc3339()}}
+        else:
+            cs["state"] = {"terminated": {
+                "exitCode": c.exit_code,
+                "reason": "Completed" if c.exit_code == 0 else "Error",
+                "message": c.message,
+                "startedAt": ts_rfc3339(c.started_at) if c.started_at
+                else now_rfc3339(),
+                "finishedAt": ts_rfc3339(c.finished_at) if c.finished_at
+                else now_rfc3339(),
+            }}
+        out.append(cs)
+    return out
+
+
 def translate_status(
     pod: Dict[str, Any],
     info: InstanceInfo,
@@ -125,6 +157,9 @@ def translate_status(
         "podIP": node_ip,
         "startTime": start_time,
     }
+    inits = _init_container_statuses(detailed)
+    if inits:
+        base["initContainerStatuses"] = inits
     if status == PodStatus.RUNNING:
         ready = info.ports_exposed
         if ready:

@@ -235,6 +235,27 @@ def prepare_deploy_params(
     if not containers:
         raise ValidationError("pod has no containers")
 
+    # spec.initContainers (sequential, to completion, before mains) — real
+    # kubelet behavior the reference ignores entirely (it reads only
+    # Containers[0], runpod_client.go:1028-1030).
+    init_containers: List[ContainerSpec] = []
+    for i, c in enumerate(pod.get("spec", {}).get("initContainers", []) or []):
+        c_sc = c.get("securityContext", {}) or {}
+        uid = c_sc.get("runAsUser", pod_sc.get("runAsUser"))
+        gid = c_sc.get("runAsGroup", pod_sc.get("runAsGroup"))
+        init_containers.append(
+            ContainerSpec(
+                name=c.get("name", f"init{i}"),
+                image=c.get("image", ""),
+                command=list(c.get("command", []) or []),
+                args=list(c.get("args", []) or []),
+                env=dict(pod_env),
+                working_dir=c.get("workingDir", ""),
+                run_as_uid=int(uid) if uid is not None else -1,
+                run_as_gid=int(gid) if gid is not None else -1,
+            )
+        )
+
     cpu_max, memory_max = _cgroup_limits(pod)
 
     return DeployParams(
@@ -242,6 +263,7 @@ def prepare_deploy_params(
         name=metadata.get("name", ""),
         namespace=metadata.get("namespace", "default"),
         containers=containers,
+        init_containers=init_containers,
         env=pod_env,
         gpu_count=gpu_count,
         gpu_memory_bytes=gpu_memory_gb * GIB,

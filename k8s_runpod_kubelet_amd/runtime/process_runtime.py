@@ -130,7 +130,15 @@ class ProcessRuntime(Runtime):
         self._attach_device_filter(inst)
 
         try:
-            spawn_s = self._launch_containers(inst)
+            if params.init_containers:
+                # spec.initContainers: sequential, each to completion, before
+                # the main containers (progression driven by exit events in
+                # _on_init_exit).
+                spawn_s = self._launch_one(
+                    inst, params.init_containers[0], inst.init_containers)
+                inst.init_index = 1
+            else:
+                spawn_s = self._launch_containers(inst)
         except Exception:
             if gpu_indices:
                 self.binder.unbind(params.pod_key)
@@ -181,14 +189,14 @@ class ProcessRuntime(Runtime):
             log.debug("device filter unavailable (unprivileged?)",
                       extra={"pod": inst.pod_key})
 
-    def _launch_containers(self, inst: Instance) -> float:
-        """Spawn every container of the pod; returns the summed native
-        posix_spawnp time in seconds. The cgroup.procs migration is timed
-        separately (cgroup_migrate_seconds) — it serializes on the kernel's
+    def _launch_one(self, inst: Instance, cspec, into: List) -> float:
+        """Spawn one container of the pod into `into` (inst.containers or
+        inst.init_containers); returns the native posix_spawnp/clone3 time in
+        seconds. The cgroup.procs migration is timed separately
+        (cgroup_migrate_seconds) — it serializes on the kernel's
         cgroup_mutex and dominates launch cost under churn."""
         from ..server import metrics
 
-        spawn_total_s = 0.0
         params = inst.params
         base_env = dict(os.environ)
         # Drop our own GPU scoping so the pod's binding is authoritative.
@@ -199,47 +207,52 @@ class ProcessRuntime(Runtime):
         base_env["AMDVK_INSTANCE_ID"] = inst.id
         base_env["AMDVK_POD_KEY"] = params.pod_key
 
-        for cspec in params.containers:
-            argv = list(cspec.command) + list(cspec.args)
-            if not argv:
-                argv = [self.podworker_path(), "--hold"]
-                if inst.gpu_indices:
-                    argv += ["--expect-gpus", str(len(inst.gpu_indices))]
-                for port in cspec.tcp_ports:
-                    argv += ["--listen-port", str(port)]
-            elif argv[0] in ("podworker", "amdvk-podworker"):
-                argv[0] = self.podworker_path()
-            needs_fast_path = (cspec.run_as_uid >= 0 or cspec.run_as_gid >= 0
-                               or self.pod_namespaces)
-            if needs_fast_path and "/" not in argv[0]:
-                # Credential dropping / namespaces happen on the execve fast
-                # path, which does no PATH search — resolve here instead.
-                resolved = shutil.which(argv[0])
-                if resolved:
-                    argv[0] = resolved
+        argv = list(cspec.command) + list(cspec.args)
+        if not argv:
+            argv = [self.podworker_path(), "--hold"]
+            if inst.gpu_indices:
+                argv += ["--expect-gpus", str(len(inst.gpu_indices))]
+            for port in cspec.tcp_ports:
+                argv += ["--listen-port", str(port)]
+        elif argv[0] in ("podworker", "amdvk-podworker"):
+            argv[0] = self.podworker_path()
+        needs_fast_path = (cspec.run_as_uid >= 0 or cspec.run_as_gid >= 0
+                           or self.pod_namespaces)
+        if needs_fast_path and "/" not in argv[0]:
+            # Credential dropping / namespaces happen on the execve fast
+            # path, which does no PATH search — resolve here instead.
+            resolved = shutil.which(argv[0])
+            if resolved:
+                argv[0] = resolved
 
-            env = dict(base_env)
-            env.update(cspec.env)
-            envp = [f"{k}={v}" for k, v in env.items()]
-            stdout_path = str(self.logs_dir / f"{inst.id}-{cspec.name}.log")
+        env = dict(base_env)
+        env.update(cspec.env)
+        envp = [f"{k}={v}" for k, v in env.items()]
+        stdout_path = str(self.logs_dir / f"{inst.id}-{cspec.name}.log")
 
-            pid, pidfd, ready_fd, spawn_s, cgroup_s = self._native.launch_process(
-                argv, envp,
-                cspec.working_dir or "",
-                stdout_path, stdout_path,
-                inst.cgroup_dir, True, True,
-                cspec.run_as_uid, cspec.run_as_gid,
-                self.pod_namespaces,
-                (params.hostname or params.name) if self.pod_namespaces else "",
-            )
-            spawn_total_s += spawn_s
-            if inst.cgroup_dir:
-                metrics.cgroup_migrate_seconds.observe(cgroup_s)
-            cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
-            inst.containers.append(cinfo)
-            with self._lock:
-                self._pid_to_instance[pid] = inst.id
-            self._loop.add_process(pid, pidfd, ready_fd, pid)
+        pid, pidfd, ready_fd, spawn_s, cgroup_s = self._native.launch_process(
+            argv, envp,
+            cspec.working_dir or "",
+            stdout_path, stdout_path,
+            inst.cgroup_dir, True, True,
+            cspec.run_as_uid, cspec.run_as_gid,
+            self.pod_namespaces,
+            (params.hostname or params.name) if self.pod_namespaces else "",
+        )
+        if inst.cgroup_dir:
+            metrics.cgroup_migrate_seconds.observe(cgroup_s)
+        cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
+        into.append(cinfo)
+        with self._lock:
+            self._pid_to_instance[pid] = inst.id
+        self._loop.add_process(pid, pidfd, ready_fd, pid)
+        return spawn_s
+
+    def _launch_containers(self, inst: Instance) -> float:
+        """Spawn every main container; returns summed native spawn time."""
+        spawn_total_s = 0.0
+        for cspec in inst.params.containers:
+            spawn_total_s += self._launch_one(inst, cspec, inst.containers)
         inst.desired_status = PodStatus.RUNNING
         return spawn_total_s
 
@@ -333,16 +346,22 @@ class ProcessRuntime(Runtime):
                 if inst is None:
                     continue
                 cinfo = next((c for c in inst.containers if c.pid == ev.pid), None)
+                is_init = False
+                if cinfo is None:
+                    cinfo = next(
+                        (c for c in inst.init_containers if c.pid == ev.pid), None)
+                    is_init = cinfo is not None
                 if cinfo is None:
                     continue
                 if ev.type == "ready":
-                    cinfo.ready = True
-                    touched.add(inst_id)
+                    if not is_init:  # init containers do not gate readiness
+                        cinfo.ready = True
+                        touched.add(inst_id)
                 elif ev.type == "ready_closed":
                     # Workload never wrote READY: process start is readiness
                     # (generic binaries without the pipe protocol). Only
                     # meaningful if it is still running.
-                    if cinfo.exit_code is None and not cinfo.ready:
+                    if not is_init and cinfo.exit_code is None and not cinfo.ready:
                         cinfo.ready = True
                         touched.add(inst_id)
                 elif ev.type == "exited":
@@ -351,7 +370,10 @@ class ProcessRuntime(Runtime):
                     cinfo.ready = False
                     if ev.exit_code:
                         cinfo.message = f"exit code {ev.exit_code}"
-                    self._on_container_exit(inst)
+                    if is_init:
+                        self._on_init_exit(inst, cinfo)
+                    else:
+                        self._on_container_exit(inst)
                     touched.add(inst_id)
 
     def _on_container_exit(self, inst: Instance) -> None:
@@ -359,6 +381,36 @@ class ProcessRuntime(Runtime):
             return
         terminating = inst.desired_status == PodStatus.TERMINATING
         inst.desired_status = PodStatus.TERMINATED if terminating else PodStatus.EXITED
+        self._teardown_resources(inst)
+
+    def _on_init_exit(self, inst: Instance, cinfo) -> None:
+        """Init-container progression (spec.initContainers semantics): exit 0
+        starts the next init or the main containers; nonzero fails the pod
+        (restartPolicy=Never model)."""
+        if inst.desired_status == PodStatus.TERMINATING:
+            inst.desired_status = PodStatus.TERMINATED
+            self._teardown_resources(inst)
+            return
+        if cinfo.exit_code != 0:
+            inst.last_error = (
+                f"init container {cinfo.name} exited with code {cinfo.exit_code}")
+            inst.desired_status = PodStatus.EXITED
+            self._teardown_resources(inst)
+            return
+        try:
+            if inst.init_index < len(inst.params.init_containers):
+                nxt = inst.params.init_containers[inst.init_index]
+                inst.init_index += 1
+                self._launch_one(inst, nxt, inst.init_containers)
+            else:
+                self._launch_containers(inst)
+        except Exception as exc:
+            log.exception("post-init launch failed", extra={"instance": inst.id})
+            inst.last_error = f"launch after init failed: {exc}"
+            inst.desired_status = PodStatus.EXITED
+            self._teardown_resources(inst)
+
+    def _teardown_resources(self, inst: Instance) -> None:
         timer = self._kill_timers.pop(inst.id, None)
         if timer:
             timer.cancel()
@@ -394,6 +446,8 @@ class ProcessRuntime(Runtime):
             desired_status=inst.desired_status,
             port_mappings=ports,
             containers=[ContainerRuntimeInfo(**vars(c)) for c in inst.containers],
+            init_containers=[ContainerRuntimeInfo(**vars(c))
+                             for c in inst.init_containers],
             gpu_indices=list(inst.gpu_indices),
             cost_per_hr=inst.cost_per_hr,
             last_error=inst.last_error,
@@ -448,12 +502,13 @@ class ProcessRuntime(Runtime):
             inst = self._instances.get(instance_id)
         if inst is None:
             return
-        if any(c.exit_code is None for c in inst.containers):
+        if any(c.exit_code is None
+               for c in list(inst.containers) + list(inst.init_containers)):
             log.warning("grace period expired; SIGKILL", extra={"instance": instance_id})
             self._signal_all(inst, 9)
 
     def _signal_all(self, inst: Instance, sig: int) -> None:
-        for c in inst.containers:
+        for c in list(inst.containers) + list(inst.init_containers):
             if c.exit_code is None and c.pid > 0:
                 # Whole process group: the container is a session leader.
                 rc = self._native.signal_process(c.pid, sig, True)
@@ -464,7 +519,7 @@ class ProcessRuntime(Runtime):
         with self._lock:
             inst = self._instances.pop(instance_id, None)
             if inst is not None:
-                for c in inst.containers:
+                for c in list(inst.containers) + list(inst.init_containers):
                     self._pid_to_instance.pop(c.pid, None)
         if inst is not None:
             self.binder.unbind(inst.pod_key)
@@ -605,6 +660,17 @@ class ProcessRuntime(Runtime):
                 }
                 for c in inst.containers
             ],
+            "init_containers": [
+                {
+                    "name": c.name,
+                    "pid": c.pid,
+                    "started_at": c.started_at,
+                    "finished_at": c.finished_at,
+                    "exit_code": c.exit_code,
+                }
+                for c in inst.init_containers
+            ],
+            "init_index": inst.init_index,
             "container_specs": [
                 {
                     "name": c.name,
@@ -614,6 +680,15 @@ class ProcessRuntime(Runtime):
                     "tcp_ports": c.tcp_ports,
                 }
                 for c in inst.params.containers
+            ],
+            "init_specs": [
+                {
+                    "name": c.name,
+                    "image": c.image,
+                    "command": c.command,
+                    "args": c.args,
+                }
+                for c in inst.params.init_containers
             ],
         }
         # Unique temp name: the event thread and API threads may persist the
@@ -651,6 +726,13 @@ class ProcessRuntime(Runtime):
                     )
                     for c in rec.get("container_specs", [])
                 ],
+                init_containers=[
+                    ContainerSpec(
+                        name=c["name"], image=c.get("image", ""),
+                        command=c.get("command", []), args=c.get("args", []),
+                    )
+                    for c in rec.get("init_specs", [])
+                ],
             )
             inst = Instance(
                 id=rec["id"], pod_key=rec["pod_key"], params=params,
@@ -659,7 +741,30 @@ class ProcessRuntime(Runtime):
                 cgroup_dir=rec.get("cgroup_dir", ""),
                 created_at=rec.get("created_at", time.time()),
                 cost_per_hr=rec.get("cost_per_hr", 0.0),
+                init_index=rec.get("init_index", 0),
             )
+            for c in rec.get("init_containers", []):
+                icinfo = ContainerRuntimeInfo(
+                    name=c["name"], pid=c["pid"],
+                    started_at=c.get("started_at", 0.0),
+                    finished_at=c.get("finished_at", 0.0),
+                    exit_code=c.get("exit_code"),
+                )
+                if icinfo.exit_code is None:
+                    # Init was mid-flight across the restart: watch it again
+                    # so progression resumes on its exit event.
+                    pidfd = self._native.open_pidfd(icinfo.pid)
+                    if pidfd >= 0:
+                        self._loop.add_process(icinfo.pid, pidfd, -1, icinfo.pid)
+                        with self._lock:
+                            self._pid_to_instance[icinfo.pid] = inst.id
+                    else:
+                        icinfo.exit_code = -1
+                        icinfo.finished_at = time.time()
+                        icinfo.message = "init vanished during kubelet restart"
+                inst.init_containers.append(icinfo)
+            if any(c.exit_code not in (None, 0) for c in inst.init_containers):
+                inst.desired_status = PodStatus.EXITED
             all_alive = True
             for c in rec.get("containers", []):
                 cinfo = ContainerRuntimeInfo(

@@ -45,6 +45,7 @@ class DetailedStatus:
     # (machine.portMappings in the reference, kubelet.go:566-605 gate).
     port_mappings: Dict[int, int] = field(default_factory=dict)
     containers: List[ContainerRuntimeInfo] = field(default_factory=list)
+    init_containers: List[ContainerRuntimeInfo] = field(default_factory=list)
     gpu_indices: List[int] = field(default_factory=list)
     cost_per_hr: float = 0.0
     last_error: str = ""
@@ -52,7 +53,11 @@ class DetailedStatus:
 
     @property
     def exit_code(self) -> Optional[int]:
-        """Aggregate exit code: first nonzero, else 0 once all finished."""
+        """Aggregate exit code: a failed init container's code, else first
+        nonzero main code, else 0 once all mains finished."""
+        for c in self.init_containers:
+            if c.exit_code:
+                return c.exit_code
         codes = [c.exit_code for c in self.containers]
         if any(c is None for c in codes) or not codes:
             return None
@@ -63,6 +68,10 @@ class DetailedStatus:
 
     @property
     def completion_message(self) -> str:
+        for c in self.init_containers:
+            if c.exit_code:
+                return (c.message
+                        or f"init container {c.name} exited with code {c.exit_code}")
         for c in self.containers:
             if c.message:
                 return c.message
@@ -107,6 +116,10 @@ class DeployParams:
     name: str
     namespace: str = "default"
     containers: List[ContainerSpec] = field(default_factory=list)
+    # spec.initContainers: run sequentially to completion before the main
+    # containers start; any nonzero exit fails the pod (restartPolicy=Never
+    # semantics, matching the run-to-completion model of the rest)
+    init_containers: List[ContainerSpec] = field(default_factory=list)
     env: Dict[str, str] = field(default_factory=dict)  # pod-level (merged into all)
     gpu_count: int = 0
     gpu_memory_bytes: int = 0  # total across the GPU set
@@ -133,6 +146,8 @@ class Instance:
     gpu_indices: List[int] = field(default_factory=list)
     desired_status: str = PodStatus.STARTING
     containers: List[ContainerRuntimeInfo] = field(default_factory=list)
+    init_containers: List[ContainerRuntimeInfo] = field(default_factory=list)
+    init_index: int = 0  # next init container to run
     cgroup_dir: str = ""
     created_at: float = field(default_factory=time.time)
     cost_per_hr: float = 0.0

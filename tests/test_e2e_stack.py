@@ -317,3 +317,60 @@ def test_lifecycle_events_emitted(stack):
     assert wait_until(events, timeout_s=10) is not None
     kube.delete_pod("default", "eventful")
     assert wait_until(lambda: gone(kube, "eventful"), timeout_s=15)
+
+
+def test_init_containers_run_before_main(stack):
+    """spec.initContainers: sequential, each to completion, before the main
+    container starts (the reference ignores initContainers entirely —
+    runpod_client.go:1028 reads Containers[0] only)."""
+    s, kube = stack
+    pod = make_pod("inited", command=["podworker"], args=["--hold"])
+    pod["spec"]["initContainers"] = [
+        {"name": "init-a", "command": ["/bin/sh"],
+         "args": ["-c", "echo A; sleep 0.2"]},
+        {"name": "init-b", "command": ["/bin/sh"],
+         "args": ["-c", "echo B; sleep 0.2"]},
+    ]
+    t0 = time.time()
+    kube.create_pod("default", pod)
+
+    def ready_pod():
+        return ready(kube, "inited")
+
+    got = wait_until(ready_pod, timeout_s=15)
+    assert got is not None
+    assert time.time() - t0 >= 0.4  # both inits ran to completion first
+    inits = got["status"]["initContainerStatuses"]
+    assert [c["name"] for c in inits] == ["init-a", "init-b"]
+    for c in inits:
+        assert c["state"]["terminated"]["exitCode"] == 0
+    kube.delete_pod("default", "inited")
+    assert wait_until(lambda: gone(kube, "inited"), timeout_s=15)
+
+
+def test_init_container_failure_fails_pod(stack):
+    """A nonzero init exit fails the pod; the main container never starts."""
+    s, kube = stack
+    pod = make_pod("initfail", command=["podworker"], args=["--hold"])
+    pod["spec"]["initContainers"] = [
+        {"name": "boom", "command": ["/bin/sh"], "args": ["-c", "exit 7"]},
+    ]
+    kube.create_pod("default", pod)
+
+    def failed():
+        try:
+            p = kube.get_pod("default", "initfail")
+        except NotFoundError:
+            return None
+        return p if p.get("status", {}).get("phase") == "Failed" else None
+
+    got = wait_until(failed, timeout_s=15)
+    assert got is not None
+    inits = got["status"]["initContainerStatuses"]
+    assert inits[0]["state"]["terminated"]["exitCode"] == 7
+    # main container never ran: no containerID assigned
+    mains = got["status"].get("containerStatuses", [])
+    assert all(not c.get("containerID") for c in mains)
+    assert not s.ledger.reservations  # GPU-less pod, but ledger clean anyway
+    kube.delete_pod("default", "initfail")
+    assert wait_until(lambda: gone(kube, "initfail"), timeout_s=15)
