@@ -22,3 +22,4 @@ class InstanceInfo:
     last_error: str = ""
     ready_time: Optional[float] = None  # first time the pod went Ready (metrics)
     deploying: bool = False  # claim flag: a deploy for this pod is in flight
+    restart_sig: int = 0  # last seen restart/backoff signature (change detection)

@@ -470,15 +470,22 @@ class Provider:
         if detailed.desired_status == PodStatus.RUNNING and not containers_ready:
             ports_exposed = False
 
+        # restartPolicy transitions (crash → backoff → restarted) keep the
+        # pod RUNNING; the signature makes them visible to change detection
+        # so CrashLoopBackOff/restartCount reach the apiserver.
+        restart_sig = sum(c.restart_count for c in detailed.containers) * 2 + \
+            sum(1 for c in detailed.containers if c.backoff_until)
         changed = (
             detailed.desired_status != info.status
             or ports_exposed != info.ports_exposed
+            or restart_sig != info.restart_sig
         )
         if not changed:
             return
 
         info.status = detailed.desired_status
         info.ports_exposed = ports_exposed
+        info.restart_sig = restart_sig
         if info.status == PodStatus.RUNNING and ports_exposed and info.ready_time is None:
             info.ready_time = time.time()
             metrics.pod_ready_seconds.observe(info.ready_time - info.creation_time)

@@ -86,12 +86,36 @@ def _container_statuses(
             "name": name,
             "image": container.get("image", ""),
             "imageID": "",
-            "restartCount": 0,
+            "restartCount": rt.restart_count if rt else 0,
             "ready": ready,
         }
         if rt and rt.pid > 0:
             cs["containerID"] = f"amdvk://{rt.pid}"
         if state == "running":
+            # Per-container fidelity under restartPolicy: a crashed
+            # container awaiting its backoff shows CrashLoopBackOff; one
+            # that exited without restart shows terminated.
+            if rt is not None and rt.exit_code is not None:
+                if rt.backoff_until:
+                    cs["ready"] = False
+                    cs["state"] = {"waiting": {
+                        "reason": "CrashLoopBackOff",
+                        "message": rt.message or
+                        f"back-off restarting failed container (exit "
+                        f"{rt.exit_code})",
+                    }}
+                else:
+                    cs["ready"] = False
+                    cs["state"] = {"terminated": {
+                        "exitCode": rt.exit_code,
+                        "reason": "Completed" if rt.exit_code == 0 else "Error",
+                        "message": rt.message,
+                        "startedAt": started_at,
+                        "finishedAt": ts_rfc3339(rt.finished_at)
+                        if rt.finished_at else now_rfc3339(),
+                    }}
+                out.append(cs)
+                continue
             cs["state"] = {"running": {"startedAt": started_at}}
             cs["started"] = True
         elif state == "waiting":
@@ -260,5 +284,7 @@ def merge_container_status(
             cs["imageID"] = old["imageID"]
         if "started" not in cs and "started" in old:
             cs["started"] = old["started"]
-        cs["restartCount"] = old.get("restartCount", cs.get("restartCount", 0))
+        # Monotonic: the runtime's live count wins once restarts happen.
+        cs["restartCount"] = max(old.get("restartCount", 0),
+                                 cs.get("restartCount", 0))
     return new_statuses

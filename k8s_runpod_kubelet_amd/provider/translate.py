@@ -278,4 +278,9 @@ def prepare_deploy_params(
         labels=dict(metadata.get("labels", {}) or {}),
         hostname=pod.get("spec", {}).get("hostname")
         or metadata.get("name", ""),
+        # Deliberate deviation from vanilla k8s (default Always): unset means
+        # Never — run-to-completion is the reference's model (a stopped
+        # RunPod instance is EXITED/Succeeded, kubelet.go:1906); an explicit
+        # spec.restartPolicy is honored with full semantics.
+        restart_policy=pod.get("spec", {}).get("restartPolicy", "Never"),
     )

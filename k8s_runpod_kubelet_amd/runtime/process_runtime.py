@@ -92,6 +92,8 @@ class ProcessRuntime(Runtime):
         self._pid_to_instance: Dict[int, str] = {}
         self._subscribers: List[Callable[[str], None]] = []
         self._kill_timers: Dict[str, threading.Timer] = {}
+        # restartPolicy backoff timers, keyed (instance_id, container_name)
+        self._restart_timers: Dict[tuple, threading.Timer] = {}
 
         self._loop = self._native.EventLoop()
         self._stop = threading.Event()
@@ -373,15 +375,81 @@ class ProcessRuntime(Runtime):
                     if is_init:
                         self._on_init_exit(inst, cinfo)
                     else:
-                        self._on_container_exit(inst)
+                        self._on_container_exit(inst, cinfo)
                     touched.add(inst_id)
 
-    def _on_container_exit(self, inst: Instance) -> None:
-        if any(c.exit_code is None for c in inst.containers):
+    def _on_container_exit(self, inst: Instance, cinfo=None) -> None:
+        # spec.restartPolicy (k8s semantics; the reference's cloud instances
+        # are run-to-completion = Never): Always restarts any exit, OnFailure
+        # restarts nonzero exits, with exponential crash backoff
+        # (CrashLoopBackOff surfaced in container status).
+        if cinfo is not None and inst.desired_status == PodStatus.RUNNING:
+            policy = inst.params.restart_policy
+            if policy == "Always" or (policy == "OnFailure" and cinfo.exit_code != 0):
+                self._schedule_restart(inst, cinfo)
+                return
+        if any(c.exit_code is None or c.backoff_until != 0
+               for c in inst.containers):
             return
         terminating = inst.desired_status == PodStatus.TERMINATING
         inst.desired_status = PodStatus.TERMINATED if terminating else PodStatus.EXITED
         self._teardown_resources(inst)
+
+    def _schedule_restart(self, inst: Instance, cinfo) -> None:
+        ran_s = max(0.0, (cinfo.finished_at or time.time())
+                    - (cinfo.started_at or time.time()))
+        if ran_s > 60.0:
+            # ran long enough to be considered healthy: reset the crash
+            # backoff (kubelet's 10-minute analogue, scaled down)
+            cinfo.crash_streak = 0
+        delay = min(60.0, 2.0 ** cinfo.crash_streak)
+        cinfo.crash_streak += 1
+        cinfo.backoff_until = time.time() + delay
+        timer = threading.Timer(
+            delay, self._restart_container, args=(inst.id, cinfo.name))
+        timer.daemon = True
+        self._restart_timers[(inst.id, cinfo.name)] = timer
+        timer.start()
+        log.info("container restart scheduled",
+                 extra={"instance": inst.id, "container": cinfo.name,
+                        "delay_s": delay, "restarts": cinfo.restart_count})
+
+    def _restart_container(self, instance_id: str, name: str) -> None:
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        self._restart_timers.pop((instance_id, name), None)
+        if inst is None or inst.desired_status != PodStatus.RUNNING:
+            return
+        cspec = next((s for s in inst.params.containers if s.name == name), None)
+        cinfo = next((c for c in inst.containers if c.name == name), None)
+        if cspec is None or cinfo is None:
+            return
+        old_pid = cinfo.pid
+        tmp: List = []
+        try:
+            self._launch_one(inst, cspec, tmp)
+        except Exception as exc:
+            log.exception("container restart failed",
+                          extra={"instance": instance_id, "container": name})
+            cinfo.message = f"restart failed: {exc}"
+            cinfo.backoff_until = 0
+            self._on_container_exit(inst, None)  # may complete the pod now
+            self._persist(inst)
+            self._notify(instance_id)
+            return
+        fresh = tmp[0]
+        with self._lock:
+            self._pid_to_instance.pop(old_pid, None)
+        cinfo.pid = fresh.pid
+        cinfo.started_at = fresh.started_at
+        cinfo.finished_at = 0.0
+        cinfo.exit_code = None
+        cinfo.ready = False
+        cinfo.message = ""
+        cinfo.restart_count += 1
+        cinfo.backoff_until = 0
+        self._persist(inst)
+        self._notify(instance_id)
 
     def _on_init_exit(self, inst: Instance, cinfo) -> None:
         """Init-container progression (spec.initContainers semantics): exit 0
@@ -414,6 +482,10 @@ class ProcessRuntime(Runtime):
         timer = self._kill_timers.pop(inst.id, None)
         if timer:
             timer.cancel()
+        for (iid, cname), t in list(self._restart_timers.items()):
+            if iid == inst.id:
+                t.cancel()
+                self._restart_timers.pop((iid, cname), None)
         # Release GPUs as soon as the workload is gone — HBM headroom returns
         # to the ledger without waiting for pod deletion.
         self.binder.unbind(inst.pod_key)
@@ -490,6 +562,21 @@ class ProcessRuntime(Runtime):
             self._persist(inst)
             return
         inst.desired_status = PodStatus.TERMINATING
+        # Cancel pending restartPolicy backoffs: a terminating pod must not
+        # relaunch containers, and all-in-backoff pods complete immediately.
+        for (iid, cname), t in list(self._restart_timers.items()):
+            if iid == instance_id:
+                t.cancel()
+                self._restart_timers.pop((iid, cname), None)
+        for c in inst.containers:
+            c.backoff_until = 0
+        if all(c.exit_code is not None for c in inst.containers) and \
+                all(c.exit_code is not None for c in inst.init_containers):
+            inst.desired_status = PodStatus.TERMINATED
+            self._teardown_resources(inst)
+            self._persist(inst)
+            self._notify(instance_id)
+            return
         self._persist(inst)
         self._signal_all(inst, 15)  # SIGTERM
         timer = threading.Timer(TERM_GRACE_S, self._force_kill, args=(instance_id,))
@@ -657,6 +744,7 @@ class ProcessRuntime(Runtime):
                     "finished_at": c.finished_at,
                     "exit_code": c.exit_code,
                     "ready": c.ready,
+                    "restart_count": c.restart_count,
                 }
                 for c in inst.containers
             ],
@@ -771,6 +859,7 @@ class ProcessRuntime(Runtime):
                     name=c["name"], pid=c["pid"], started_at=c.get("started_at", 0.0),
                     finished_at=c.get("finished_at", 0.0),
                     exit_code=c.get("exit_code"), ready=c.get("ready", False),
+                    restart_count=c.get("restart_count", 0),
                 )
                 if cinfo.exit_code is None:
                     pidfd = self._native.open_pidfd(cinfo.pid)
@@ -820,6 +909,8 @@ class ProcessRuntime(Runtime):
         self._loop.wake()
         self._watcher.join(timeout=2.0)
         for timer in self._kill_timers.values():
+            timer.cancel()
+        for timer in self._restart_timers.values():
             timer.cancel()
 
 

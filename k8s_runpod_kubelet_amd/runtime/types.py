@@ -34,6 +34,12 @@ class ContainerRuntimeInfo:
     exit_code: Optional[int] = None
     ready: bool = False
     message: str = ""
+    # restartPolicy bookkeeping: completed restarts; while a restart is
+    # pending, the wall-clock deadline of its backoff (CrashLoopBackOff
+    # state); consecutive-crash streak driving the exponential delay
+    restart_count: int = 0
+    backoff_until: float = 0.0
+    crash_streak: int = 0
 
 
 @dataclass
@@ -117,9 +123,10 @@ class DeployParams:
     namespace: str = "default"
     containers: List[ContainerSpec] = field(default_factory=list)
     # spec.initContainers: run sequentially to completion before the main
-    # containers start; any nonzero exit fails the pod (restartPolicy=Never
-    # semantics, matching the run-to-completion model of the rest)
+    # containers start; any nonzero exit fails the pod
     init_containers: List[ContainerSpec] = field(default_factory=list)
+    # spec.restartPolicy: Never | OnFailure | Always (k8s default Always)
+    restart_policy: str = "Never"
     env: Dict[str, str] = field(default_factory=dict)  # pod-level (merged into all)
     gpu_count: int = 0
     gpu_memory_bytes: int = 0  # total across the GPU set

@@ -374,3 +374,69 @@ def test_init_container_failure_fails_pod(stack):
     assert not s.ledger.reservations  # GPU-less pod, but ledger clean anyway
     kube.delete_pod("default", "initfail")
     assert wait_until(lambda: gone(kube, "initfail"), timeout_s=15)
+
+
+def test_restart_policy_on_failure(stack):
+    """spec.restartPolicy=OnFailure: a crashing container is restarted with
+    backoff (restartCount grows, CrashLoopBackOff surfaces between
+    attempts); deletion cancels pending restarts cleanly. The reference's
+    cloud instances are run-to-completion only."""
+    s, kube = stack
+    pod = make_pod("crashy", command=["podworker"],
+                   args=["--run-for", "0.05", "--exit-code", "1"])
+    pod["spec"]["restartPolicy"] = "OnFailure"
+    kube.create_pod("default", pod)
+
+    def restarted():
+        try:
+            p = kube.get_pod("default", "crashy")
+        except NotFoundError:
+            return None
+        css = p.get("status", {}).get("containerStatuses", [])
+        if css and css[0].get("restartCount", 0) >= 1:
+            return p
+        return None
+
+    got = wait_until(restarted, timeout_s=20)
+    assert got is not None
+    # the pod keeps phase Running while crash-looping (k8s semantics)
+    assert got["status"]["phase"] in ("Running", "Pending")
+    kube.delete_pod("default", "crashy")
+    assert wait_until(lambda: gone(kube, "crashy"), timeout_s=20)
+    assert not s.ledger.reservations
+
+
+def test_restart_policy_on_failure_until_success(stack):
+    """OnFailure restarts stop once the container exits 0: pod Succeeded."""
+    import os as _os
+
+    s, kube = stack
+    marker = f"/tmp/amdvk-once-{_os.getpid()}"
+    try:
+        _os.unlink(marker)
+    except FileNotFoundError:
+        pass
+    # first run fails and drops a marker; the restart finds it and exits 0
+    script = f"if [ -e {marker} ]; then exit 0; else touch {marker}; exit 3; fi"
+    pod = make_pod("flaky", command=["/bin/sh"], args=["-c", script])
+    pod["spec"]["restartPolicy"] = "OnFailure"
+    kube.create_pod("default", pod)
+
+    def succeeded():
+        try:
+            p = kube.get_pod("default", "flaky")
+        except NotFoundError:
+            return None
+        return p if p.get("status", {}).get("phase") == "Succeeded" else None
+
+    got = wait_until(succeeded, timeout_s=30)
+    assert got is not None
+    cs = got["status"]["containerStatuses"][0]
+    assert cs["restartCount"] >= 1
+    assert cs["state"]["terminated"]["exitCode"] == 0
+    kube.delete_pod("default", "flaky")
+    assert wait_until(lambda: gone(kube, "flaky"), timeout_s=15)
+    try:
+        _os.unlink(marker)
+    except FileNotFoundError:
+        pass
