@@ -339,44 +339,49 @@ class ProcessRuntime(Runtime):
                 self._notify(inst_id)
 
     def _handle_events(self, events, touched) -> None:
+            # Each event is processed under the runtime lock: container state
+            # transitions (exit → restart-schedule → terminate) are mutated
+            # from three threads (watcher, restart timers, API callers) and
+            # must be atomic — a terminate interleaving with a restart
+            # decision must never strand a reservation.
             for ev in events:
-                inst_id = self._pid_to_instance.get(ev.pid)
-                if inst_id is None:
-                    continue
                 with self._lock:
-                    inst = self._instances.get(inst_id)
-                if inst is None:
-                    continue
-                cinfo = next((c for c in inst.containers if c.pid == ev.pid), None)
-                is_init = False
-                if cinfo is None:
-                    cinfo = next(
-                        (c for c in inst.init_containers if c.pid == ev.pid), None)
-                    is_init = cinfo is not None
-                if cinfo is None:
-                    continue
-                if ev.type == "ready":
-                    if not is_init:  # init containers do not gate readiness
-                        cinfo.ready = True
-                        touched.add(inst_id)
-                elif ev.type == "ready_closed":
-                    # Workload never wrote READY: process start is readiness
-                    # (generic binaries without the pipe protocol). Only
-                    # meaningful if it is still running.
-                    if not is_init and cinfo.exit_code is None and not cinfo.ready:
-                        cinfo.ready = True
-                        touched.add(inst_id)
-                elif ev.type == "exited":
-                    cinfo.exit_code = ev.exit_code
-                    cinfo.finished_at = time.time()
-                    cinfo.ready = False
-                    if ev.exit_code:
-                        cinfo.message = f"exit code {ev.exit_code}"
-                    if is_init:
-                        self._on_init_exit(inst, cinfo)
-                    else:
-                        self._on_container_exit(inst, cinfo)
-                    touched.add(inst_id)
+                    inst_id = self._pid_to_instance.get(ev.pid)
+                    inst = self._instances.get(inst_id) if inst_id else None
+                    if inst is not None:
+                        self._handle_one_event(ev, inst, touched, inst_id)
+
+    def _handle_one_event(self, ev, inst, touched, inst_id) -> None:
+        cinfo = next((c for c in inst.containers if c.pid == ev.pid), None)
+        is_init = False
+        if cinfo is None:
+            cinfo = next(
+                (c for c in inst.init_containers if c.pid == ev.pid), None)
+            is_init = cinfo is not None
+        if cinfo is None:
+            return
+        if ev.type == "ready":
+            if not is_init:  # init containers do not gate readiness
+                cinfo.ready = True
+                touched.add(inst_id)
+        elif ev.type == "ready_closed":
+            # Workload never wrote READY: process start is readiness
+            # (generic binaries without the pipe protocol). Only
+            # meaningful if it is still running.
+            if not is_init and cinfo.exit_code is None and not cinfo.ready:
+                cinfo.ready = True
+                touched.add(inst_id)
+        elif ev.type == "exited":
+            cinfo.exit_code = ev.exit_code
+            cinfo.finished_at = time.time()
+            cinfo.ready = False
+            if ev.exit_code:
+                cinfo.message = f"exit code {ev.exit_code}"
+            if is_init:
+                self._on_init_exit(inst, cinfo)
+            else:
+                self._on_container_exit(inst, cinfo)
+            touched.add(inst_id)
 
     def _on_container_exit(self, inst: Instance, cinfo=None) -> None:
         # spec.restartPolicy (k8s semantics; the reference's cloud instances
@@ -415,39 +420,40 @@ class ProcessRuntime(Runtime):
                         "delay_s": delay, "restarts": cinfo.restart_count})
 
     def _restart_container(self, instance_id: str, name: str) -> None:
+        # Entire transition under the runtime lock: must not interleave with
+        # terminate() (which cancels restarts and tears resources down) or
+        # with the exit-event handler.
         with self._lock:
+            self._restart_timers.pop((instance_id, name), None)
             inst = self._instances.get(instance_id)
-        self._restart_timers.pop((instance_id, name), None)
-        if inst is None or inst.desired_status != PodStatus.RUNNING:
-            return
-        cspec = next((s for s in inst.params.containers if s.name == name), None)
-        cinfo = next((c for c in inst.containers if c.name == name), None)
-        if cspec is None or cinfo is None:
-            return
-        old_pid = cinfo.pid
-        tmp: List = []
-        try:
-            self._launch_one(inst, cspec, tmp)
-        except Exception as exc:
-            log.exception("container restart failed",
-                          extra={"instance": instance_id, "container": name})
-            cinfo.message = f"restart failed: {exc}"
-            cinfo.backoff_until = 0
-            self._on_container_exit(inst, None)  # may complete the pod now
-            self._persist(inst)
-            self._notify(instance_id)
-            return
-        fresh = tmp[0]
-        with self._lock:
-            self._pid_to_instance.pop(old_pid, None)
-        cinfo.pid = fresh.pid
-        cinfo.started_at = fresh.started_at
-        cinfo.finished_at = 0.0
-        cinfo.exit_code = None
-        cinfo.ready = False
-        cinfo.message = ""
-        cinfo.restart_count += 1
-        cinfo.backoff_until = 0
+            if inst is None or inst.desired_status != PodStatus.RUNNING:
+                return
+            cspec = next(
+                (s for s in inst.params.containers if s.name == name), None)
+            cinfo = next((c for c in inst.containers if c.name == name), None)
+            if cspec is None or cinfo is None:
+                return
+            old_pid = cinfo.pid
+            tmp: List = []
+            try:
+                self._launch_one(inst, cspec, tmp)
+            except Exception as exc:
+                log.exception("container restart failed",
+                              extra={"instance": instance_id, "container": name})
+                cinfo.message = f"restart failed: {exc}"
+                cinfo.backoff_until = 0
+                self._on_container_exit(inst, None)  # may complete the pod now
+            else:
+                fresh = tmp[0]
+                self._pid_to_instance.pop(old_pid, None)
+                cinfo.pid = fresh.pid
+                cinfo.started_at = fresh.started_at
+                cinfo.finished_at = 0.0
+                cinfo.exit_code = None
+                cinfo.ready = False
+                cinfo.message = ""
+                cinfo.restart_count += 1
+                cinfo.backoff_until = 0
         self._persist(inst)
         self._notify(instance_id)
 
@@ -553,36 +559,42 @@ class ProcessRuntime(Runtime):
     # ------------- terminate / GC -------------
 
     def terminate(self, instance_id: str) -> None:
+        notify_done = False
+        # Transition under the runtime lock — atomic against the exit-event
+        # handler and restart timers (a terminate interleaving with a
+        # restart decision must never strand a reservation).
         with self._lock:
             inst = self._instances.get(instance_id)
-        if inst is None:
-            return
-        if inst.desired_status in (PodStatus.EXITED, PodStatus.TERMINATED):
-            inst.desired_status = PodStatus.TERMINATED
-            self._persist(inst)
-            return
-        inst.desired_status = PodStatus.TERMINATING
-        # Cancel pending restartPolicy backoffs: a terminating pod must not
-        # relaunch containers, and all-in-backoff pods complete immediately.
-        for (iid, cname), t in list(self._restart_timers.items()):
-            if iid == instance_id:
-                t.cancel()
-                self._restart_timers.pop((iid, cname), None)
-        for c in inst.containers:
-            c.backoff_until = 0
-        if all(c.exit_code is not None for c in inst.containers) and \
-                all(c.exit_code is not None for c in inst.init_containers):
-            inst.desired_status = PodStatus.TERMINATED
-            self._teardown_resources(inst)
-            self._persist(inst)
-            self._notify(instance_id)
-            return
+            if inst is None:
+                return
+            if inst.desired_status in (PodStatus.EXITED, PodStatus.TERMINATED):
+                inst.desired_status = PodStatus.TERMINATED
+                self._persist(inst)
+                return
+            inst.desired_status = PodStatus.TERMINATING
+            # Cancel pending restartPolicy backoffs: a terminating pod must
+            # not relaunch containers; all-in-backoff pods complete now.
+            for (iid, cname), t in list(self._restart_timers.items()):
+                if iid == instance_id:
+                    t.cancel()
+                    self._restart_timers.pop((iid, cname), None)
+            for c in inst.containers:
+                c.backoff_until = 0
+            if all(c.exit_code is not None for c in inst.containers) and \
+                    all(c.exit_code is not None for c in inst.init_containers):
+                inst.desired_status = PodStatus.TERMINATED
+                self._teardown_resources(inst)
+                notify_done = True
+            else:
+                self._signal_all(inst, 15)  # SIGTERM
+                timer = threading.Timer(
+                    TERM_GRACE_S, self._force_kill, args=(instance_id,))
+                timer.daemon = True
+                self._kill_timers[instance_id] = timer
+                timer.start()
         self._persist(inst)
-        self._signal_all(inst, 15)  # SIGTERM
-        timer = threading.Timer(TERM_GRACE_S, self._force_kill, args=(instance_id,))
-        timer.daemon = True
-        self._kill_timers[instance_id] = timer
-        timer.start()
+        if notify_done:
+            self._notify(instance_id)
 
     def _force_kill(self, instance_id: str) -> None:
         with self._lock:

@@ -382,7 +382,7 @@ def test_restart_policy_on_failure(stack):
     attempts); deletion cancels pending restarts cleanly. The reference's
     cloud instances are run-to-completion only."""
     s, kube = stack
-    pod = make_pod("crashy", command=["podworker"],
+    pod = make_pod("crashy", gpus=1, command=["podworker"],
                    args=["--run-for", "0.05", "--exit-code", "1"])
     pod["spec"]["restartPolicy"] = "OnFailure"
     kube.create_pod("default", pod)
@@ -401,9 +401,13 @@ def test_restart_policy_on_failure(stack):
     assert got is not None
     # the pod keeps phase Running while crash-looping (k8s semantics)
     assert got["status"]["phase"] in ("Running", "Pending")
+    # ...and keeps OWNING its GPU across restarts (no rebind churn)
+    assert s.ledger.get_reservation("default-crashy") is not None
     kube.delete_pod("default", "crashy")
     assert wait_until(lambda: gone(kube, "crashy"), timeout_s=20)
-    assert not s.ledger.reservations
+    # GPU release rides the exit event, which lands moments after the API
+    # object disappears when deletion caught the container mid-run.
+    assert wait_until(lambda: not s.ledger.reservations, timeout_s=10)
 
 
 def test_restart_policy_on_failure_until_success(stack):
