@@ -74,28 +74,86 @@ class SecretCollector:
         return self._cache[name]
 
 
-def _container_env(container: Dict[str, Any], collector: SecretCollector) -> Dict[str, str]:
+class ConfigMapCollector:
+    """ConfigMap analogue of SecretCollector — parity-plus: the reference
+    supports only Secrets (runpod_client.go:866-1054); ConfigMap-sourced env
+    is just as common in real manifests."""
+
+    def __init__(self, client: K8sClient, namespace: str):
+        self.client = client
+        self.namespace = namespace
+        self._cache: Dict[str, Dict[str, str]] = {}
+
+    def get(self, name: str) -> Dict[str, str]:
+        if name not in self._cache:
+            try:
+                cm = self.client.get_configmap(self.namespace, name)
+                self._cache[name] = {
+                    k: str(v) for k, v in (cm.get("data") or {}).items()
+                }
+            except Exception as exc:
+                if is_not_found(exc):
+                    log.warning("configmap not found",
+                                extra={"configmap": name, "ns": self.namespace})
+                    self._cache[name] = {}
+                else:
+                    raise
+        return self._cache[name]
+
+
+def _field_ref(pod: Dict[str, Any], path: str) -> str:
+    """Downward-API fieldRef subset (metadata/name/namespace/uid, nodeName)."""
+    md = pod.get("metadata", {})
+    return {
+        "metadata.name": md.get("name", ""),
+        "metadata.namespace": md.get("namespace", "default"),
+        "metadata.uid": md.get("uid", ""),
+        "spec.nodeName": pod.get("spec", {}).get("nodeName", ""),
+    }.get(path, "")
+
+
+def _container_env(container: Dict[str, Any], collector: SecretCollector,
+                   cm_collector: ConfigMapCollector,
+                   pod: Dict[str, Any]) -> Dict[str, str]:
     env: Dict[str, str] = {}
-    # envFrom secretRef: whole-secret import (runpod_client.go:981-1000)
+    # envFrom secretRef/configMapRef: whole-object import
+    # (runpod_client.go:981-1000 covers the secret half)
     for ef in container.get("envFrom", []) or []:
         ref = ef.get("secretRef")
         if ref and ref.get("name"):
             for key, value in collector.get(ref["name"]).items():
                 env[key] = _escape(value)
-    # explicit env entries (literal + secretKeyRef), later wins
+        cref = ef.get("configMapRef")
+        if cref and cref.get("name"):
+            for key, value in cm_collector.get(cref["name"]).items():
+                env[key] = _escape(value)
+    # explicit env entries (literal + secretKeyRef + configMapKeyRef +
+    # fieldRef), later wins
     for item in container.get("env", []) or []:
         name = item.get("name", "")
         if not name:
             continue
         if "value" in item:
             env[name] = _escape(str(item["value"]))
-        else:
-            ref = (item.get("valueFrom") or {}).get("secretKeyRef")
-            if ref and ref.get("name"):
-                data = collector.get(ref["name"])
-                key = ref.get("key", "")
-                if key in data:
-                    env[name] = _escape(data[key])
+            continue
+        vf = item.get("valueFrom") or {}
+        ref = vf.get("secretKeyRef")
+        if ref and ref.get("name"):
+            data = collector.get(ref["name"])
+            key = ref.get("key", "")
+            if key in data:
+                env[name] = _escape(data[key])
+            continue
+        cref = vf.get("configMapKeyRef")
+        if cref and cref.get("name"):
+            data = cm_collector.get(cref["name"])
+            key = cref.get("key", "")
+            if key in data:
+                env[name] = _escape(data[key])
+            continue
+        fref = vf.get("fieldRef")
+        if fref and fref.get("fieldPath"):
+            env[name] = _escape(_field_ref(pod, fref["fieldPath"]))
     return {k: v for k, v in env.items() if not is_k8s_auto_injected(k)}
 
 
@@ -132,10 +190,11 @@ def extract_env_vars(
     """
     namespace = pod.get("metadata", {}).get("namespace", "default")
     collector = SecretCollector(client, namespace)
+    cm_collector = ConfigMapCollector(client, namespace)
     containers = pod.get("spec", {}).get("containers", []) or []
 
     per_container: List[Dict[str, str]] = [
-        _container_env(c, collector) for c in containers
+        _container_env(c, collector, cm_collector, pod) for c in containers
     ]
     vol_env = _volume_secret_env(pod, collector)
 

@@ -277,3 +277,39 @@ def test_prepare_fail_fast_when_no_offers(fake_kube, synthetic_ledger):
     ok = make_pod(gpus=8, annotations={ann.GPU_MEMORY: "2000"})  # 250/GPU
     params = prepare_deploy_params(ok, fake_kube, Config(), catalog)
     assert params.gpu_count == 8
+
+
+def test_configmap_and_fieldref_env(fake_kube, pod_factory):
+    """ConfigMap-sourced env + Downward-API fieldRef (parity-plus: the
+    reference supports Secret sources only, runpod_client.go:866-1054)."""
+    from k8s_runpod_kubelet_amd.provider.envvars import extract_env_vars
+
+    fake_kube.put_configmap("default", {
+        "metadata": {"name": "app-config", "namespace": "default"},
+        "data": {"MODE": "prod", "THREADS": "8"},
+    })
+    pod = pod_factory("cmenv", containers=[{
+        "name": "main", "image": "x",
+        "envFrom": [{"configMapRef": {"name": "app-config"}}],
+        "env": [
+            {"name": "ONE_KEY",
+             "valueFrom": {"configMapKeyRef": {"name": "app-config",
+                                               "key": "MODE"}}},
+            {"name": "MY_POD",
+             "valueFrom": {"fieldRef": {"fieldPath": "metadata.name"}}},
+            {"name": "MY_NODE",
+             "valueFrom": {"fieldRef": {"fieldPath": "spec.nodeName"}}},
+        ],
+    }])
+    pod_env, per = extract_env_vars(pod, fake_kube)
+    assert pod_env["MODE"] == "prod" and pod_env["THREADS"] == "8"
+    assert pod_env["ONE_KEY"] == "prod"
+    assert pod_env["MY_POD"] == "cmenv"
+    assert pod_env["MY_NODE"] == "virtual-runpod"
+    # missing configmap degrades to empty, not a crash
+    pod2 = pod_factory("cm2", containers=[{
+        "name": "main", "image": "x",
+        "envFrom": [{"configMapRef": {"name": "nope"}}],
+    }])
+    pod_env2, _ = extract_env_vars(pod2, fake_kube)
+    assert "MODE" not in pod_env2
