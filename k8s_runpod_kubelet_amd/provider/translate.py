@@ -34,6 +34,7 @@ from ..kube.objects import (
     resource_parse_bytes,
     resource_parse_cpu,
 )
+from ..runtime.probes import ProbeSpec
 from ..runtime.types import ContainerSpec, DeployParams
 from . import annotations as ann
 from .envvars import extract_env_vars
@@ -230,6 +231,8 @@ def prepare_deploy_params(
                 tcp_ports=tcp_ports_of_container(c),
                 run_as_uid=int(uid) if uid is not None else -1,
                 run_as_gid=int(gid) if gid is not None else -1,
+                liveness=ProbeSpec.parse(c.get("livenessProbe")),
+                readiness=ProbeSpec.parse(c.get("readinessProbe")),
             )
         )
     if not containers:

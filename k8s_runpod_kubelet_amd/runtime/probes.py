@@ -1,0 +1,118 @@
+"""Container liveness/readiness probes (tcpSocket / httpGet / exec).
+
+The reference has no probe support at all — its backend is a cloud API and
+readiness is inferred from port mappings (kubelet.go:566-605). Local pods
+make real kubelet probe semantics cheap:
+
+- readinessProbe: governs the container's Ready state (overriding the
+  AMDVK_READY_FD pipe signal once defined),
+- livenessProbe: failureThreshold consecutive failures kill the container;
+  spec.restartPolicy then decides whether it restarts (CrashLoopBackOff
+  machinery applies).
+
+Host-process pods share the host network namespace, so tcpSocket/httpGet
+probe 127.0.0.1:<port> directly.
+"""
+
+from __future__ import annotations
+
+import http.client
+import logging
+import socket
+import subprocess
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+log = logging.getLogger("runtime.probes")
+
+
+@dataclass
+class ProbeSpec:
+    kind: str                 # "tcp" | "http" | "exec"
+    port: int = 0
+    path: str = "/"
+    command: List[str] = field(default_factory=list)
+    initial_delay_s: float = 0.0
+    period_s: float = 10.0
+    timeout_s: float = 1.0
+    failure_threshold: int = 3
+    success_threshold: int = 1
+
+    @staticmethod
+    def parse(raw: Optional[Dict[str, Any]]) -> Optional["ProbeSpec"]:
+        """Parse a k8s probe object; None when absent or handler unknown."""
+        if not raw:
+            return None
+        common = dict(
+            initial_delay_s=float(raw.get("initialDelaySeconds", 0)),
+            period_s=max(1.0, float(raw.get("periodSeconds", 10))),
+            timeout_s=max(0.1, float(raw.get("timeoutSeconds", 1))),
+            failure_threshold=max(1, int(raw.get("failureThreshold", 3))),
+            success_threshold=max(1, int(raw.get("successThreshold", 1))),
+        )
+        if raw.get("tcpSocket"):
+            port = int(raw["tcpSocket"].get("port", 0))
+            return ProbeSpec(kind="tcp", port=port, **common)
+        if raw.get("httpGet"):
+            hg = raw["httpGet"]
+            return ProbeSpec(kind="http", port=int(hg.get("port", 80)),
+                             path=hg.get("path", "/"), **common)
+        if raw.get("exec"):
+            cmd = list(raw["exec"].get("command", []) or [])
+            if cmd:
+                return ProbeSpec(kind="exec", command=cmd, **common)
+        return None
+
+
+@dataclass
+class ProbeState:
+    """Per-(container, probe-type) counters."""
+    successes: int = 0
+    failures: int = 0
+    last_run: float = 0.0
+    result: Optional[bool] = None  # thresholded outcome (None = undecided)
+
+
+def run_probe(spec: ProbeSpec, env: Dict[str, str]) -> bool:
+    """Execute one probe attempt; True = success. Never raises."""
+    try:
+        if spec.kind == "tcp":
+            with socket.create_connection(("127.0.0.1", spec.port),
+                                          timeout=spec.timeout_s):
+                return True
+        if spec.kind == "http":
+            conn = http.client.HTTPConnection("127.0.0.1", spec.port,
+                                              timeout=spec.timeout_s)
+            try:
+                conn.request("GET", spec.path)
+                status = conn.getresponse().status
+                return 200 <= status < 400
+            finally:
+                conn.close()
+        if spec.kind == "exec":
+            proc = subprocess.run(
+                spec.command, env=env, timeout=spec.timeout_s,
+                stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+            )
+            return proc.returncode == 0
+    except Exception:
+        return False
+    return False
+
+
+def advance(state: ProbeState, spec: ProbeSpec, ok: bool) -> Optional[bool]:
+    """Fold one attempt into the state; returns the thresholded outcome
+    (True/False) or None while undecided. k8s semantics: successThreshold
+    consecutive successes flip to healthy, failureThreshold consecutive
+    failures flip to unhealthy."""
+    if ok:
+        state.successes += 1
+        state.failures = 0
+        if state.successes >= spec.success_threshold:
+            state.result = True
+    else:
+        state.failures += 1
+        state.successes = 0
+        if state.failures >= spec.failure_threshold:
+            state.result = False
+    return state.result

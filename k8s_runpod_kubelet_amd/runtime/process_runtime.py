@@ -101,6 +101,13 @@ class ProcessRuntime(Runtime):
             target=self._watch_events, name="runtime-events", daemon=True
         )
         self._watcher.start()
+        # liveness/readiness probe scheduler (1 s resolution; per-probe
+        # period/threshold state in _probe_states keyed
+        # (instance, container, type))
+        from ..utils.backoff import Ticker
+
+        self._probe_states: Dict[tuple, object] = {}
+        self._probe_ticker = Ticker(1.0, self._run_probes, "probes").start()
 
     # ------------- deploy -------------
 
@@ -360,15 +367,21 @@ class ProcessRuntime(Runtime):
             is_init = cinfo is not None
         if cinfo is None:
             return
+        # A defined readinessProbe owns the Ready state: the AMDVK_READY_FD
+        # pipe / process-start heuristics are ignored for that container.
+        probe_gated = not is_init and any(
+            s.readiness is not None
+            for s in inst.params.containers if s.name == cinfo.name)
         if ev.type == "ready":
-            if not is_init:  # init containers do not gate readiness
+            if not is_init and not probe_gated:
                 cinfo.ready = True
                 touched.add(inst_id)
         elif ev.type == "ready_closed":
             # Workload never wrote READY: process start is readiness
             # (generic binaries without the pipe protocol). Only
             # meaningful if it is still running.
-            if not is_init and cinfo.exit_code is None and not cinfo.ready:
+            if not is_init and not probe_gated and \
+                    cinfo.exit_code is None and not cinfo.ready:
                 cinfo.ready = True
                 touched.add(inst_id)
         elif ev.type == "exited":
@@ -454,6 +467,7 @@ class ProcessRuntime(Runtime):
                 cinfo.message = ""
                 cinfo.restart_count += 1
                 cinfo.backoff_until = 0
+                self._clear_probe_state(instance_id, name)
         self._persist(inst)
         self._notify(instance_id)
 
@@ -497,6 +511,64 @@ class ProcessRuntime(Runtime):
         self.binder.unbind(inst.pod_key)
         self._release_cgroup(inst.cgroup_dir)
         inst.cgroup_dir = ""
+
+    def _run_probes(self) -> None:
+        """One probe-scheduler tick: run due liveness/readiness probes on
+        RUNNING containers (see runtime/probes.py; the reference has no
+        probe support at all)."""
+        from .probes import ProbeState, advance, run_probe
+
+        now = time.time()
+        with self._lock:
+            insts = [i for i in self._instances.values()
+                     if i.desired_status == PodStatus.RUNNING]
+        for inst in insts:
+            by_name = {c.name: c for c in inst.containers}
+            for cspec in inst.params.containers:
+                if cspec.liveness is None and cspec.readiness is None:
+                    continue
+                cinfo = by_name.get(cspec.name)
+                if cinfo is None or cinfo.exit_code is not None:
+                    continue
+                env = None
+                for kind, spec in (("liveness", cspec.liveness),
+                                   ("readiness", cspec.readiness)):
+                    if spec is None:
+                        continue
+                    key = (inst.id, cspec.name, kind)
+                    st = self._probe_states.setdefault(key, ProbeState())
+                    if now - cinfo.started_at < spec.initial_delay_s:
+                        continue
+                    if now - st.last_run < spec.period_s:
+                        continue
+                    st.last_run = now
+                    if env is None and spec.kind == "exec":
+                        env = dict(os.environ)
+                        env.update(inst.params.env)
+                        env.update(device_env(inst.gpu_indices,
+                                              self.binder.ledger.inventory))
+                        env.update(cspec.env)
+                    ok = run_probe(spec, env or {})
+                    outcome = advance(st, spec, ok)
+                    if kind == "readiness" and outcome is not None:
+                        if cinfo.ready != outcome:
+                            cinfo.ready = outcome
+                            self._persist(inst)
+                            self._notify(inst.id)
+                    elif kind == "liveness" and outcome is False:
+                        log.warning(
+                            "liveness probe failed; killing container",
+                            extra={"instance": inst.id, "container": cspec.name,
+                                   "failures": st.failures})
+                        cinfo.message = "liveness probe failed"
+                        self._probe_states.pop(key, None)
+                        self._native.signal_process(cinfo.pid, 9, True)
+                        # exit event drives restartPolicy from here
+
+    def _clear_probe_state(self, instance_id: str, container: str = "") -> None:
+        for key in list(self._probe_states):
+            if key[0] == instance_id and (not container or key[1] == container):
+                self._probe_states.pop(key, None)
 
     def _notify(self, instance_id: str) -> None:
         for cb in list(self._subscribers):
@@ -622,6 +694,7 @@ class ProcessRuntime(Runtime):
                     self._pid_to_instance.pop(c.pid, None)
         if inst is not None:
             self.binder.unbind(inst.pod_key)
+            self._clear_probe_state(instance_id)
             (self.instances_dir / f"{instance_id}.json").unlink(missing_ok=True)
 
     def healthy(self) -> bool:
@@ -918,6 +991,7 @@ class ProcessRuntime(Runtime):
 
     def close(self) -> None:
         self._stop.set()
+        self._probe_ticker.stop()
         self._loop.wake()
         self._watcher.join(timeout=2.0)
         for timer in self._kill_timers.values():

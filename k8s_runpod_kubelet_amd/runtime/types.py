@@ -13,6 +13,8 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
+from .probes import ProbeSpec  # noqa: F401 (re-exported for ContainerSpec)
+
 
 class PodStatus:
     RUNNING = "RUNNING"
@@ -112,6 +114,9 @@ class ContainerSpec:
     # inherit the kubelet's credentials)
     run_as_uid: int = -1
     run_as_gid: int = -1
+    # livenessProbe / readinessProbe (None = absent)
+    liveness: Optional["ProbeSpec"] = None
+    readiness: Optional["ProbeSpec"] = None
 
 
 @dataclass
