@@ -286,4 +286,6 @@ def prepare_deploy_params(
         # RunPod instance is EXITED/Succeeded, kubelet.go:1906); an explicit
         # spec.restartPolicy is honored with full semantics.
         restart_policy=pod.get("spec", {}).get("restartPolicy", "Never"),
+        termination_grace_s=float(
+            pod.get("spec", {}).get("terminationGracePeriodSeconds", 10)),
     )

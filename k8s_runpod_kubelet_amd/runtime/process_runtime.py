@@ -660,7 +660,10 @@ class ProcessRuntime(Runtime):
             else:
                 self._signal_all(inst, 15)  # SIGTERM
                 timer = threading.Timer(
-                    TERM_GRACE_S, self._force_kill, args=(instance_id,))
+                    # spec.terminationGracePeriodSeconds (default 10 s here;
+                    # our pods are processes, not images pulling state)
+                    max(0.1, inst.params.termination_grace_s),
+                    self._force_kill, args=(instance_id,))
                 timer.daemon = True
                 self._kill_timers[instance_id] = timer
                 timer.start()

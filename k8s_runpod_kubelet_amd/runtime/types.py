@@ -132,6 +132,8 @@ class DeployParams:
     init_containers: List[ContainerSpec] = field(default_factory=list)
     # spec.restartPolicy: Never | OnFailure | Always (k8s default Always)
     restart_policy: str = "Never"
+    # spec.terminationGracePeriodSeconds: SIGTERM → SIGKILL ladder window
+    termination_grace_s: float = 10.0
     env: Dict[str, str] = field(default_factory=dict)  # pod-level (merged into all)
     gpu_count: int = 0
     gpu_memory_bytes: int = 0  # total across the GPU set

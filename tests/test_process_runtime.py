@@ -178,3 +178,34 @@ def test_adoption_of_vanished_process(synthetic_ledger, tmp_state_dir):
     finally:
         rt2.close()
         rt.close()
+
+
+def test_termination_grace_period(process_runtime):
+    """spec.terminationGracePeriodSeconds: a TERM-ignoring workload is
+    SIGKILLed after the pod's own grace window, not the global default."""
+    import time
+
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+
+    rt = process_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-stubborn", name="stubborn",
+        termination_grace_s=0.5,
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"],
+            args=["-c", "trap '' TERM; sleep 60"])],
+    ))
+    time.sleep(0.2)  # shell up with TERM ignored
+    t0 = time.time()
+    rt.terminate(st.id)
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        if s.desired_status == PodStatus.TERMINATED:
+            break
+        time.sleep(0.02)
+    assert s.desired_status == PodStatus.TERMINATED
+    took = time.time() - t0
+    assert 0.4 < took < 5, took  # killed at ~0.5 s grace, not 10 s
+    assert s.containers[0].exit_code == 128 + 9
