@@ -59,24 +59,32 @@ def main() -> int:
     pw_args = (["--expect-gpus", "1"] if gpus else [])
 
     def make(name, mode):
-        if mode == "hold" or mode == "crash":
+        if mode in ("hold", "crash", "probed"):
             a = pw_args + ["--hold"]
         elif mode == "ok":
             a = pw_args + ["--run-for", f"{rng.uniform(0.05, 0.4):.2f}"]
+        elif mode == "restarting":
+            a = pw_args + ["--run-for", "0.05", "--exit-code", "9"]
         else:  # fail
             a = pw_args + ["--run-for", "0.05", "--exit-code", "7"]
+        container = {
+            "name": "main", "image": "amdvk/podworker:soak",
+            "command": ["podworker"], "args": a,
+            **({"resources": {"limits": {"amd.com/gpu": str(gpus)}}}
+               if gpus else {}),
+        }
+        spec = {"nodeName": cfg.node_name, "containers": [container]}
+        if mode == "probed":
+            container["readinessProbe"] = {
+                "exec": {"command": ["/bin/true"]},
+                "periodSeconds": 1, "failureThreshold": 1,
+            }
+        elif mode == "restarting":
+            spec["restartPolicy"] = "OnFailure"
         return {
             "apiVersion": "v1", "kind": "Pod",
             "metadata": {"name": name, "namespace": "default"},
-            "spec": {
-                "nodeName": cfg.node_name,
-                "containers": [{
-                    "name": "main", "image": "amdvk/podworker:soak",
-                    "command": ["podworker"], "args": a,
-                    **({"resources": {"limits": {"amd.com/gpu": str(gpus)}}}
-                       if gpus else {}),
-                }],
-            },
+            "spec": spec,
         }
 
     def pod_state(name):
@@ -97,7 +105,7 @@ def main() -> int:
     fd0 = fd_count()
     active = {}  # name -> dict(mode, created, state)
     counters = {"created": 0, "succeeded": 0, "failed": 0, "crashed": 0,
-                "deleted_holds": 0, "timeouts": 0}
+                "deleted_holds": 0, "restarted": 0, "timeouts": 0}
     seq = 0
     deadline = time.time() + args.duration
     try:
@@ -105,8 +113,9 @@ def main() -> int:
             now = time.time()
             # top up population (only while inside the window)
             while now < deadline and len(active) < args.max_active:
-                mode = rng.choices(["hold", "ok", "fail", "crash"],
-                                   weights=[3, 4, 2, 1])[0]
+                mode = rng.choices(
+                    ["hold", "ok", "fail", "crash", "probed", "restarting"],
+                    weights=[3, 4, 2, 1, 2, 1])[0]
                 name = f"soak-{seq:05d}"
                 seq += 1
                 client.create_pod("default", make(name, mode))
@@ -148,7 +157,15 @@ def main() -> int:
                         counters["crashed"] += 1
                         st["deleted"] = True
                         client.delete_pod("default", name)
-                else:  # hold
+                elif st["mode"] == "restarting":
+                    css = (pod or {}).get("status", {}).get(
+                        "containerStatuses", [])
+                    if css and css[0].get("restartCount", 0) >= 1 \
+                            and not st["deleted"]:
+                        counters["restarted"] += 1
+                        st["deleted"] = True
+                        client.delete_pod("default", name)
+                else:  # hold / probed
                     if state == "Ready" and age > st["dwell"] and not st["deleted"]:
                         counters["deleted_holds"] += 1
                         st["deleted"] = True
@@ -162,7 +179,8 @@ def main() -> int:
             "mode": "cpu" if args.cpu else "gpu",
             **counters,
             "pods_done": counters["succeeded"] + counters["failed"]
-            + counters["crashed"] + counters["deleted_holds"],
+            + counters["crashed"] + counters["deleted_holds"]
+            + counters["restarted"],
             "mean_ready_ms": round(m.hist_mean_ms(m.pod_ready_seconds), 1),
             "leaks": {
                 "reservations": len(stack.ledger.reservations),
