@@ -274,10 +274,16 @@ class Provider:
         detailed = None
         if info.instance_id:
             detailed = self.runtime.get_detailed_status(info.instance_id)
-            if info.status == PodStatus.RUNNING and info.requested_ports:
-                info.ports_exposed = check_ports_exposed(
+            if info.status == PodStatus.RUNNING:
+                # Same readiness rule as the sync path: ports exposed AND
+                # every container ready (pipe signal or readinessProbe).
+                ports_ok = check_ports_exposed(
                     info.requested_ports, detailed.port_mappings
                 )
+                containers_ready = bool(detailed.containers) and all(
+                    c.ready for c in detailed.containers
+                )
+                info.ports_exposed = ports_ok and containers_ready
         return translate_status(pod, info, detailed, self.config.internal_ip)
 
     def notify_pods(self, fn: NotifyFunc) -> None:
