@@ -824,6 +824,7 @@ class ProcessRuntime(Runtime):
             "gpu_count": inst.params.gpu_count,
             "namespace": inst.params.namespace,
             "name": inst.params.name,
+            "restart_policy": inst.params.restart_policy,
             "containers": [
                 {
                     "name": c.name,
@@ -894,6 +895,7 @@ class ProcessRuntime(Runtime):
                 namespace=rec.get("namespace", "default"),
                 gpu_count=rec.get("gpu_count", 0),
                 gpu_memory_bytes=rec.get("gpu_memory_bytes", 0),
+                restart_policy=rec.get("restart_policy", "Never"),
                 containers=[
                     ContainerSpec(
                         name=c["name"], image=c.get("image", ""),
@@ -962,12 +964,26 @@ class ProcessRuntime(Runtime):
                         cinfo.message = "process vanished during kubelet restart"
                         all_alive = False
                 inst.containers.append(cinfo)
-            if inst.containers and all(c.exit_code is not None for c in inst.containers):
+            # Containers that crashed across the restart window resume their
+            # restartPolicy loop instead of failing the pod: re-reserve the
+            # GPUs and schedule restarts with fresh backoff.
+            def wants_restart(c) -> bool:
+                if inst.desired_status != PodStatus.RUNNING:
+                    return False
+                if params.restart_policy == "Always":
+                    return c.exit_code is not None
+                if params.restart_policy == "OnFailure":
+                    return c.exit_code is not None and c.exit_code != 0
+                return False
+
+            restartable = [c for c in inst.containers if wants_restart(c)]
+            all_exited = inst.containers and all(
+                c.exit_code is not None for c in inst.containers)
+            if all_exited and not restartable:
                 if inst.desired_status not in (PodStatus.TERMINATED,):
-                    inst.desired_status = (
-                        PodStatus.EXITED if all_alive else PodStatus.EXITED
-                    )
+                    inst.desired_status = PodStatus.EXITED
             elif inst.desired_status == PodStatus.RUNNING and inst.gpu_indices:
+                # Live (or about-to-restart) pod keeps owning its GPUs.
                 self.ledger.adopt(
                     inst.pod_key, inst.gpu_indices,
                     BindRequest(inst.pod_key, len(inst.gpu_indices),
@@ -987,6 +1003,8 @@ class ProcessRuntime(Runtime):
                         pass
             with self._lock:
                 self._instances[inst.id] = inst
+                for c in restartable:
+                    self._schedule_restart(inst, c)
             adopted.append(inst.id)
         if adopted:
             log.info("adopted persisted instances", extra={"count": len(adopted)})

@@ -444,3 +444,51 @@ def test_restart_policy_on_failure_until_success(stack):
         _os.unlink(marker)
     except FileNotFoundError:
         pass
+
+
+def test_restart_adoption_resumes_crashloop(tmp_state_dir):
+    """A crash-looping OnFailure pod caught mid-backoff by a kubelet crash
+    resumes its restart loop after adoption (restartCount keeps growing)
+    instead of being marked Failed."""
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=8,
+                 notify_interval_s=0, pending_retry_interval_s=0.2)
+    kube = FakeKube()
+    s1 = build_stack(cfg, client=kube)
+    s1.runtime.enable_cgroups = False
+    s1.start(serve_http=False)
+    pod = make_pod("loopy", gpus=1, command=["podworker"],
+                   args=["--run-for", "0.05", "--exit-code", "1"])
+    pod["spec"]["restartPolicy"] = "OnFailure"
+    kube.create_pod("default", pod)
+
+    def count_at_least(n):
+        def check():
+            try:
+                p = kube.get_pod("default", "loopy")
+            except NotFoundError:
+                return None
+            css = p.get("status", {}).get("containerStatuses", [])
+            return p if css and css[0].get("restartCount", 0) >= n else None
+        return check
+
+    assert wait_until(count_at_least(1), timeout_s=20)
+    # kubelet crash (pods untouched, restart timers lost)
+    s1.pod_controller.stop(); s1.node_controller.stop(); s1.provider.stop()
+    s1.runtime._stop.set(); s1.runtime._loop.wake()
+    for t in s1.runtime._restart_timers.values():
+        t.cancel()
+
+    s2 = build_stack(cfg, client=kube)
+    s2.runtime.enable_cgroups = False
+    s2.start(serve_http=False)
+    try:
+        # the loop resumes: restartCount must grow beyond the pre-crash value
+        p = kube.get_pod("default", "loopy")
+        before = p["status"]["containerStatuses"][0]["restartCount"]
+        assert wait_until(count_at_least(before + 1), timeout_s=30)
+        assert s2.ledger.get_reservation("default-loopy") is not None
+        kube.delete_pod("default", "loopy")
+        assert wait_until(lambda: gone(kube, "loopy"), timeout_s=20)
+        assert wait_until(lambda: not s2.ledger.reservations, timeout_s=10)
+    finally:
+        s2.stop()
