@@ -616,7 +616,35 @@ class Provider:
 
     def cleanup_deleted_pods(self) -> None:
         """Reference cleanupDeletedPods (kubelet.go:1190-1227): when the K8s
-        pod is really gone, make sure the backend instance is too."""
+        pod is really gone, make sure the backend instance is too — plus an
+        orphan sweep for terminal instance records whose pod no longer
+        exists anywhere (accumulates otherwise after kubelet restarts: the
+        in-memory deletedPods map dies with the old process)."""
+        try:
+            terminal = self.runtime.list_instances(
+                [PodStatus.EXITED, PodStatus.TERMINATED])
+        except Exception:
+            terminal = []
+        with self._pods_lock:
+            referenced = {info.instance_id
+                          for info in self._pod_status.values()}
+        for st in terminal:
+            if st.id in referenced or not st.name:
+                continue
+            try:
+                self.client.get_pod(st.namespace or "default", st.name)
+                continue  # pod still exists; normal paths own this record
+            except Exception as exc:
+                if not is_not_found(exc):
+                    continue
+            try:
+                self.runtime.remove(st.id)
+                log.info("removed orphan terminal instance",
+                         extra={"instance": st.id,
+                                "pod": f"{st.namespace}/{st.name}"})
+            except Exception:
+                log.exception("orphan instance removal failed",
+                              extra={"instance": st.id})
         with self._deleted_lock:
             items = list(self._deleted_pods.items())
         for fkey, instance_id in items:

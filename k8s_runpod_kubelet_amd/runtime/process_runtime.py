@@ -594,6 +594,8 @@ class ProcessRuntime(Runtime):
         return DetailedStatus(
             id=inst.id,
             desired_status=inst.desired_status,
+            namespace=inst.params.namespace,
+            name=inst.params.name,
             port_mappings=ports,
             containers=[ContainerRuntimeInfo(**vars(c)) for c in inst.containers],
             init_containers=[ContainerRuntimeInfo(**vars(c))

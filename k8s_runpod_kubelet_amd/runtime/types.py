@@ -52,6 +52,9 @@ class DetailedStatus:
     # "exposed" TCP ports actually listening — the portMappings analogue
     # (machine.portMappings in the reference, kubelet.go:566-605 gate).
     port_mappings: Dict[int, int] = field(default_factory=dict)
+    # pod identity (for orphan GC after kubelet restarts)
+    namespace: str = ""
+    name: str = ""
     containers: List[ContainerRuntimeInfo] = field(default_factory=list)
     init_containers: List[ContainerRuntimeInfo] = field(default_factory=list)
     gpu_indices: List[int] = field(default_factory=list)

@@ -32,6 +32,9 @@ def main() -> int:
     ap.add_argument("--cpu", action="store_true",
                     help="synthetic inventory + CPU podworker")
     ap.add_argument("--seed", type=int, default=12345)
+    ap.add_argument("--restart-kubelet-at", type=float, default=0.0,
+                    help="simulate a kubelet crash+restart T seconds in "
+                         "(pods untouched; adoption must resume everything)")
     args = ap.parse_args()
     rng = random.Random(args.seed)
 
@@ -107,10 +110,25 @@ def main() -> int:
     counters = {"created": 0, "succeeded": 0, "failed": 0, "crashed": 0,
                 "deleted_holds": 0, "restarted": 0, "timeouts": 0}
     seq = 0
-    deadline = time.time() + args.duration
+    start_ts = time.time()
+    deadline = start_ts + args.duration
+    kubelet_restarts = 0
     try:
         while time.time() < deadline or active:
             now = time.time()
+            if (args.restart_kubelet_at > 0 and kubelet_restarts == 0
+                    and now - start_ts >= args.restart_kubelet_at):
+                # Simulated kubelet crash: stop every control-plane thread,
+                # leave pod processes running, then build a fresh stack on
+                # the same state dir — adoption must pick everything up.
+                print("KUBELET RESTART", file=sys.stderr)
+                stack.pod_controller.stop()
+                stack.node_controller.stop()
+                stack.provider.stop()
+                stack.runtime.close()
+                stack = build_stack(cfg, client=client)
+                stack.start(serve_http=False)
+                kubelet_restarts += 1
             # top up population (only while inside the window)
             while now < deadline and len(active) < args.max_active:
                 mode = rng.choices(
@@ -177,6 +195,7 @@ def main() -> int:
         summary = {
             "duration_s": args.duration,
             "mode": "cpu" if args.cpu else "gpu",
+            "kubelet_restarts": kubelet_restarts,
             **counters,
             "pods_done": counters["succeeded"] + counters["failed"]
             + counters["crashed"] + counters["deleted_holds"]
