@@ -126,11 +126,33 @@ class Inventory:
         return gpus
 
     def refresh_dynamic(self) -> None:
-        """Cheap per-tick refresh of live VRAM/busy/temp counters."""
+        """Per-tick refresh of live VRAM/busy/temp counters AND health: RAS
+        uncorrectable errors appearing at runtime must cordon the GPU (the
+        reference re-probes backend health every tick, kubelet.go:320-331 —
+        here that check is per-GPU)."""
         if self.synthetic:
             return
         native = self._load_native()
+        by_index = {}
+        try:
+            by_index = {r.index: r for r in native.enumerate_gpus(self.sysfs_root)}
+        except Exception:
+            log.exception("health re-probe failed; refreshing counters only")
         for g in self.gpus:
+            r = by_index.get(g.index)
+            if r is not None:
+                g.vram_total_bytes = int(r.vram_total_bytes) or g.vram_total_bytes
+                g.vram_used_bytes = int(r.vram_used_bytes)
+                g.busy_percent = int(r.busy_percent)
+                g.temperature_mc = int(r.temperature_mc)
+                was_healthy = g.healthy
+                g.ras_uncorrectable = int(r.ras_uncorrectable)
+                g.healthy = bool(r.healthy)
+                if was_healthy and not g.healthy:
+                    log.warning("GPU became unhealthy (RAS)",
+                                extra={"gpu": g.index,
+                                       "uncorrectable": g.ras_uncorrectable})
+                continue
             if g.render_minor < 0:
                 continue
             d = native.read_gpu_dynamic(self.sysfs_root, g.render_minor)
