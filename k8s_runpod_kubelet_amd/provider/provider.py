@@ -579,6 +579,14 @@ class Provider:
                 (key, pod, self._pod_status.get(key))
                 for key, pod in self._pods.items()
             ]
+        # Placement order: spec.priority descending (scheduler-resolved
+        # priorityClass), FIFO within a class — the reference retries in map
+        # order (kubelet.go:760), which starves nothing but also respects
+        # nothing.
+        items.sort(key=lambda t: (
+            -int(t[1].get("spec", {}).get("priority", 0) or 0),
+            t[2].creation_time if t[2] is not None else now,
+        ))
         for key, pod, info in items:
             if info is None or info.instance_id:
                 continue

@@ -492,3 +492,34 @@ def test_restart_adoption_resumes_crashloop(tmp_state_dir):
         assert wait_until(lambda: not s2.ledger.reservations, timeout_s=10)
     finally:
         s2.stop()
+
+
+def test_priority_ordering_in_pending_queue(tmp_state_dir):
+    """spec.priority orders the pending queue: with one free GPU and three
+    queued pods, the high-priority pod places first (FIFO within a class)."""
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=1,
+                 notify_interval_s=0, pending_retry_interval_s=999)
+    kube = FakeKube()
+    s = build_stack(cfg, client=kube)
+    s.runtime.enable_cgroups = False
+    s.start(serve_http=False)
+    try:
+        kube.create_pod("default", make_pod("holder", gpus=1,
+                                            command=["podworker"],
+                                            args=["--hold"]))
+        assert wait_until(lambda: ready(kube, "holder"), timeout_s=10)
+        for name, prio in (("low1", 0), ("high", 100), ("low2", 0)):
+            p = make_pod(name, gpus=1, command=["podworker"], args=["--hold"])
+            if prio:
+                p["spec"]["priority"] = prio
+            kube.create_pod("default", p)
+        time.sleep(0.3)
+        kube.delete_pod("default", "holder")
+        assert wait_until(lambda: ready(kube, "high"), timeout_s=10)
+        assert ready(kube, "low1") is None and ready(kube, "low2") is None
+        kube.delete_pod("default", "high")
+        assert wait_until(lambda: ready(kube, "low1"), timeout_s=10)  # FIFO
+        for n in ("low1", "low2"):
+            kube.delete_pod("default", n)
+    finally:
+        s.stop()
