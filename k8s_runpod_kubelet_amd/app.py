@@ -52,7 +52,8 @@ class Stack:
         self.pod_controller.start()
         if serve_http:
             self.health = HealthServer(
-                self.config.health_server_address, self.provider.ping
+                self.config.health_server_address, self.provider.ping,
+                ledger=self.ledger,
             )
             self.health.start()
             self.api_server = KubeletApiServer(

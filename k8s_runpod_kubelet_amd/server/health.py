@@ -20,11 +20,13 @@ log = logging.getLogger("server.health")
 
 
 class HealthServer:
-    def __init__(self, address: str, ready_fn: Optional[Callable[[], None]] = None):
+    def __init__(self, address: str, ready_fn: Optional[Callable[[], None]] = None,
+                 ledger=None):
         host, _, port = address.rpartition(":")
         self.host = host or "0.0.0.0"
         self.port = int(port)
         self.ready_fn = ready_fn
+        self.ledger = ledger  # enables the cordon/uncordon admin endpoints
         self._alive = True
         self._server: Optional[ThreadingHTTPServer] = None
         self._thread: Optional[threading.Thread] = None
@@ -79,6 +81,28 @@ class HealthServer:
                                 for line in traceback.format_stack(frame)
                             )
                     self._respond(200, "\n".join(lines).encode())
+                else:
+                    self._respond(404, b"not found")
+
+            def do_POST(self):
+                # Operator cordon/uncordon (kubectl-cordon analogue at GPU
+                # granularity): POST /cordon/<idx> | /uncordon/<idx>.
+                parts = [p for p in self.path.strip("/").split("/") if p]
+                if (len(parts) == 2 and parts[0] in ("cordon", "uncordon")
+                        and outer.ledger is not None):
+                    try:
+                        idx = int(parts[1])
+                    except ValueError:
+                        self._respond(400, b"bad gpu index")
+                        return
+                    state = outer.ledger.states.get(idx)
+                    if state is None:
+                        self._respond(404, f"no GPU {idx}".encode())
+                        return
+                    state.cordoned = parts[0] == "cordon"
+                    log.info("gpu cordon state changed",
+                             extra={"gpu": idx, "cordoned": state.cordoned})
+                    self._respond(200, b"ok")
                 else:
                     self._respond(404, b"not found")
 

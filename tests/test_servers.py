@@ -107,3 +107,31 @@ def test_kubelet_api(provider, process_runtime):
         pod_obj = prov.get_pod("default", "api1")
         if pod_obj:
             prov.delete_pod(pod_obj)
+
+
+def test_cordon_endpoint(synthetic_ledger):
+    """POST /cordon/<i> removes a GPU from scheduling (kubectl-cordon at
+    GPU granularity); /uncordon restores it."""
+    from k8s_runpod_kubelet_amd.server.health import HealthServer
+
+    hs = HealthServer("127.0.0.1:0", None, ledger=synthetic_ledger)
+    hs.start()
+    base = f"http://127.0.0.1:{hs.port}"
+    try:
+        assert synthetic_ledger.schedulable_count() == 8
+        req = urllib.request.Request(f"{base}/cordon/3", method="POST")
+        assert urllib.request.urlopen(req, timeout=5).status == 200
+        assert synthetic_ledger.schedulable_count() == 7
+        assert not synthetic_ledger.states[3].schedulable
+        req = urllib.request.Request(f"{base}/uncordon/3", method="POST")
+        assert urllib.request.urlopen(req, timeout=5).status == 200
+        assert synthetic_ledger.schedulable_count() == 8
+        # bad index -> 404
+        req = urllib.request.Request(f"{base}/cordon/99", method="POST")
+        try:
+            urllib.request.urlopen(req, timeout=5)
+            assert False, "expected 404"
+        except urllib.error.HTTPError as e:
+            assert e.code == 404
+    finally:
+        hs.stop()
