@@ -22,7 +22,10 @@ def extract_ports_from_pod(pod: Dict[str, Any]) -> List[str]:
             proto = (port.get("protocol") or "TCP").upper()
             if proto != "TCP":
                 continue
-            number = int(port.get("containerPort", 0))
+            try:
+                number = int(port.get("containerPort", 0) or 0)
+            except (TypeError, ValueError):
+                continue  # malformed containerPort: skip, don't crash
             if not number or number in seen:
                 continue
             seen.add(number)
@@ -64,7 +67,10 @@ def tcp_ports_of_container(container: Dict[str, Any]) -> List[int]:
     out = []
     for port in container.get("ports", []) or []:
         if (port.get("protocol") or "TCP").upper() == "TCP":
-            number = int(port.get("containerPort", 0))
+            try:
+                number = int(port.get("containerPort", 0) or 0)
+            except (TypeError, ValueError):
+                continue  # malformed containerPort: skip, don't crash
             if number:
                 out.append(number)
     return out

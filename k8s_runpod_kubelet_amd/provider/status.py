@@ -239,6 +239,8 @@ def translate_status(
                     exit_code=code if code else 1, reason="Error", message=message,
                 ),
             )
+            if detailed is not None and detailed.last_error == "DeadlineExceeded":
+                base["reason"] = "DeadlineExceeded"  # activeDeadlineSeconds
     elif status == PodStatus.TERMINATING:
         base.update(
             phase="Running",

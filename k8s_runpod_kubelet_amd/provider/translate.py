@@ -139,21 +139,45 @@ def gpu_count_of(pod: Dict[str, Any]) -> int:
         for section in ("limits", "requests"):
             value = (res.get(section) or {}).get(ann.GPU_RESOURCE)
             if value is not None:
-                total += int(value)
+                try:
+                    total += int(value)
+                except (TypeError, ValueError):
+                    raise ValidationError(
+                        f"invalid {ann.GPU_RESOURCE} quantity: {value!r}")
                 break
     return total
 
 
+def _int_or_error(value, what: str) -> int:
+    try:
+        return int(value)
+    except (TypeError, ValueError):
+        raise ValidationError(f"invalid {what}: {value!r}")
+
+
+def _float_or(value, default: float) -> float:
+    try:
+        return float(value)
+    except (TypeError, ValueError):
+        return default
+
+
 def _cgroup_limits(pod: Dict[str, Any]) -> (str, str):
-    """Aggregate container CPU/memory limits into cgroup v2 strings."""
+    """Aggregate container CPU/memory limits into cgroup v2 strings; a
+    malformed quantity is a ValidationError (the apiserver would normally
+    reject it; a fake/relaxed one must not crash translation)."""
     cpu_cores = 0.0
     mem_bytes = 0
     for container in pod.get("spec", {}).get("containers", []) or []:
         limits = (container.get("resources", {}) or {}).get("limits") or {}
-        if "cpu" in limits:
-            cpu_cores += resource_parse_cpu(limits["cpu"])
-        if "memory" in limits:
-            mem_bytes += resource_parse_bytes(limits["memory"])
+        try:
+            if "cpu" in limits:
+                cpu_cores += resource_parse_cpu(limits["cpu"])
+            if "memory" in limits:
+                mem_bytes += resource_parse_bytes(limits["memory"])
+        except (TypeError, ValueError):
+            raise ValidationError(
+                f"invalid resource quantity in limits: {limits!r}")
     cpu_max = f"{int(cpu_cores * 100000)} 100000" if cpu_cores > 0 else ""
     memory_max = str(mem_bytes) if mem_bytes > 0 else ""
     return cpu_max, memory_max
@@ -229,8 +253,10 @@ def prepare_deploy_params(
                 env=per_container_env[i] if i < len(per_container_env) else {},
                 working_dir=c.get("workingDir", ""),
                 tcp_ports=tcp_ports_of_container(c),
-                run_as_uid=int(uid) if uid is not None else -1,
-                run_as_gid=int(gid) if gid is not None else -1,
+                run_as_uid=(_int_or_error(uid, "runAsUser")
+                            if uid not in (None, "") else -1),
+                run_as_gid=(_int_or_error(gid, "runAsGroup")
+                            if gid not in (None, "") else -1),
                 liveness=ProbeSpec.parse(c.get("livenessProbe")),
                 readiness=ProbeSpec.parse(c.get("readinessProbe")),
             )
@@ -254,8 +280,10 @@ def prepare_deploy_params(
                 args=list(c.get("args", []) or []),
                 env=dict(pod_env),
                 working_dir=c.get("workingDir", ""),
-                run_as_uid=int(uid) if uid is not None else -1,
-                run_as_gid=int(gid) if gid is not None else -1,
+                run_as_uid=(_int_or_error(uid, "runAsUser")
+                            if uid not in (None, "") else -1),
+                run_as_gid=(_int_or_error(gid, "runAsGroup")
+                            if gid not in (None, "") else -1),
             )
         )
 
@@ -286,6 +314,8 @@ def prepare_deploy_params(
         # RunPod instance is EXITED/Succeeded, kubelet.go:1906); an explicit
         # spec.restartPolicy is honored with full semantics.
         restart_policy=pod.get("spec", {}).get("restartPolicy", "Never"),
-        termination_grace_s=float(
-            pod.get("spec", {}).get("terminationGracePeriodSeconds", 10)),
+        termination_grace_s=_float_or(
+            pod.get("spec", {}).get("terminationGracePeriodSeconds", 10), 10.0),
+        active_deadline_s=_float_or(
+            pod.get("spec", {}).get("activeDeadlineSeconds", 0) or 0, 0.0),
     )

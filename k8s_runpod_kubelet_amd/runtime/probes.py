@@ -43,19 +43,32 @@ class ProbeSpec:
         """Parse a k8s probe object; None when absent or handler unknown."""
         if not raw:
             return None
+        def num(key, default, cast=float):
+            try:
+                return cast(raw.get(key, default))
+            except (TypeError, ValueError):
+                return default
+
         common = dict(
-            initial_delay_s=float(raw.get("initialDelaySeconds", 0)),
-            period_s=max(1.0, float(raw.get("periodSeconds", 10))),
-            timeout_s=max(0.1, float(raw.get("timeoutSeconds", 1))),
-            failure_threshold=max(1, int(raw.get("failureThreshold", 3))),
-            success_threshold=max(1, int(raw.get("successThreshold", 1))),
+            initial_delay_s=num("initialDelaySeconds", 0.0),
+            period_s=max(1.0, num("periodSeconds", 10.0)),
+            timeout_s=max(0.1, num("timeoutSeconds", 1.0)),
+            failure_threshold=max(1, num("failureThreshold", 3, int)),
+            success_threshold=max(1, num("successThreshold", 1, int)),
         )
         if raw.get("tcpSocket"):
-            port = int(raw["tcpSocket"].get("port", 0))
+            try:
+                port = int(raw["tcpSocket"].get("port", 0))
+            except (TypeError, ValueError):
+                return None
             return ProbeSpec(kind="tcp", port=port, **common)
         if raw.get("httpGet"):
             hg = raw["httpGet"]
-            return ProbeSpec(kind="http", port=int(hg.get("port", 80)),
+            try:
+                port = int(hg.get("port", 80))
+            except (TypeError, ValueError):
+                return None
+            return ProbeSpec(kind="http", port=port,
                              path=hg.get("path", "/"), **common)
         if raw.get("exec"):
             cmd = list(raw["exec"].get("command", []) or [])

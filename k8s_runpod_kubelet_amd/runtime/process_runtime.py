@@ -94,6 +94,8 @@ class ProcessRuntime(Runtime):
         self._kill_timers: Dict[str, threading.Timer] = {}
         # restartPolicy backoff timers, keyed (instance_id, container_name)
         self._restart_timers: Dict[tuple, threading.Timer] = {}
+        # spec.activeDeadlineSeconds timers, keyed instance_id
+        self._deadline_timers: Dict[str, threading.Timer] = {}
 
         self._loop = self._native.EventLoop()
         self._stop = threading.Event()
@@ -160,6 +162,12 @@ class ProcessRuntime(Runtime):
         metrics.launch_seconds.observe(spawn_s)
         with self._lock:
             self._instances[instance_id] = inst
+            if params.active_deadline_s > 0:
+                t = threading.Timer(params.active_deadline_s,
+                                    self._deadline_exceeded, args=(instance_id,))
+                t.daemon = True
+                self._deadline_timers[instance_id] = t
+                t.start()
         self._persist(inst)
         log.info(
             "deployed instance",
@@ -401,7 +409,8 @@ class ProcessRuntime(Runtime):
         # are run-to-completion = Never): Always restarts any exit, OnFailure
         # restarts nonzero exits, with exponential crash backoff
         # (CrashLoopBackOff surfaced in container status).
-        if cinfo is not None and inst.desired_status == PodStatus.RUNNING:
+        if cinfo is not None and inst.desired_status == PodStatus.RUNNING \
+                and not inst.deadline_exceeded:
             policy = inst.params.restart_policy
             if policy == "Always" or (policy == "OnFailure" and cinfo.exit_code != 0):
                 self._schedule_restart(inst, cinfo)
@@ -498,10 +507,47 @@ class ProcessRuntime(Runtime):
             inst.desired_status = PodStatus.EXITED
             self._teardown_resources(inst)
 
+    def _deadline_exceeded(self, instance_id: str) -> None:
+        """spec.activeDeadlineSeconds fired: kill the pod; it completes as
+        Failed/DeadlineExceeded (restartPolicy no longer applies)."""
+        notify = False
+        with self._lock:
+            self._deadline_timers.pop(instance_id, None)
+            inst = self._instances.get(instance_id)
+            if inst is None or inst.desired_status not in (
+                    PodStatus.RUNNING, PodStatus.STARTING):
+                return
+            log.warning("activeDeadlineSeconds exceeded",
+                        extra={"instance": instance_id, "pod": inst.pod_key})
+            inst.deadline_exceeded = True
+            inst.last_error = "DeadlineExceeded"
+            for (iid, cname), t in list(self._restart_timers.items()):
+                if iid == instance_id:
+                    t.cancel()
+                    self._restart_timers.pop((iid, cname), None)
+            live = False
+            for c in list(inst.containers) + list(inst.init_containers):
+                c.backoff_until = 0
+                if c.exit_code is None:
+                    c.message = "deadline exceeded"
+                    live = True
+            if live:
+                self._signal_all(inst, 9)  # exits drive completion
+            else:
+                inst.desired_status = PodStatus.EXITED
+                self._teardown_resources(inst)
+                notify = True
+        if notify:
+            self._persist(inst)
+            self._notify(instance_id)
+
     def _teardown_resources(self, inst: Instance) -> None:
         timer = self._kill_timers.pop(inst.id, None)
         if timer:
             timer.cancel()
+        dtimer = self._deadline_timers.pop(inst.id, None)
+        if dtimer:
+            dtimer.cancel()
         for (iid, cname), t in list(self._restart_timers.items()):
             if iid == inst.id:
                 t.cancel()
@@ -1020,6 +1066,8 @@ class ProcessRuntime(Runtime):
         for timer in self._kill_timers.values():
             timer.cancel()
         for timer in self._restart_timers.values():
+            timer.cancel()
+        for timer in self._deadline_timers.values():
             timer.cancel()
 
 

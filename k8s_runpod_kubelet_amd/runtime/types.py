@@ -137,6 +137,9 @@ class DeployParams:
     restart_policy: str = "Never"
     # spec.terminationGracePeriodSeconds: SIGTERM → SIGKILL ladder window
     termination_grace_s: float = 10.0
+    # spec.activeDeadlineSeconds: pod killed + Failed/DeadlineExceeded after
+    # this long (0 = unlimited)
+    active_deadline_s: float = 0.0
     env: Dict[str, str] = field(default_factory=dict)  # pod-level (merged into all)
     gpu_count: int = 0
     gpu_memory_bytes: int = 0  # total across the GPU set
@@ -169,3 +172,4 @@ class Instance:
     created_at: float = field(default_factory=time.time)
     cost_per_hr: float = 0.0
     last_error: str = ""
+    deadline_exceeded: bool = False  # activeDeadlineSeconds fired

@@ -523,3 +523,28 @@ def test_priority_ordering_in_pending_queue(tmp_state_dir):
             kube.delete_pod("default", n)
     finally:
         s.stop()
+
+
+def test_active_deadline_seconds(stack):
+    """spec.activeDeadlineSeconds: the pod is killed and marked
+    Failed/DeadlineExceeded once the deadline passes — even under
+    restartPolicy=Always (which would otherwise keep it alive forever)."""
+    s, kube = stack
+    pod = make_pod("deadliner", command=["podworker"], args=["--hold"])
+    pod["spec"]["activeDeadlineSeconds"] = 1
+    pod["spec"]["restartPolicy"] = "Always"
+    kube.create_pod("default", pod)
+    assert wait_until(lambda: ready(kube, "deadliner"), timeout_s=10)
+
+    def failed():
+        try:
+            p = kube.get_pod("default", "deadliner")
+        except NotFoundError:
+            return None
+        return p if p.get("status", {}).get("phase") == "Failed" else None
+
+    got = wait_until(failed, timeout_s=15)
+    assert got is not None
+    assert got["status"].get("reason") == "DeadlineExceeded"
+    kube.delete_pod("default", "deadliner")
+    assert wait_until(lambda: gone(kube, "deadliner"), timeout_s=15)

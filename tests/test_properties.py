@@ -176,3 +176,91 @@ def test_tcp_ports_gate_readiness(requested, listening):
 def test_http_ports_always_ready(requested):
     reqs = [f"{p}/http" for p in requested]
     assert check_ports_exposed(reqs, {})
+
+
+# -- spec-translation fuzz ---------------------------------------------------
+
+_names = st.text(string.ascii_lowercase + "-", min_size=1, max_size=10)
+_maybe_int = st.one_of(st.none(), st.integers(-5, 500),
+                       st.text(string.digits + "GiB", max_size=6))
+_container = st.fixed_dictionaries({}, optional={
+    "name": _names,
+    "image": st.text(max_size=12),
+    "command": st.lists(st.text(max_size=8), max_size=3),
+    "args": st.lists(st.text(max_size=8), max_size=3),
+    "workingDir": st.text(max_size=10),
+    "ports": st.lists(st.fixed_dictionaries({}, optional={
+        "containerPort": _maybe_int,
+        "protocol": st.sampled_from(["TCP", "UDP", "SCTP", ""]),
+    }), max_size=3),
+    "env": st.lists(st.fixed_dictionaries({}, optional={
+        "name": st.text(max_size=10),
+        "value": st.text(max_size=10),
+        "valueFrom": st.fixed_dictionaries({}, optional={
+            "secretKeyRef": st.fixed_dictionaries(
+                {}, optional={"name": _names, "key": st.text(max_size=6)}),
+            "configMapKeyRef": st.fixed_dictionaries(
+                {}, optional={"name": _names, "key": st.text(max_size=6)}),
+            "fieldRef": st.fixed_dictionaries(
+                {}, optional={"fieldPath": st.text(max_size=20)}),
+        }),
+    }), max_size=3),
+    "resources": st.fixed_dictionaries({}, optional={
+        "limits": st.dictionaries(
+            st.sampled_from(["amd.com/gpu", "cpu", "memory", "junk"]),
+            st.one_of(st.text(max_size=6), st.integers(0, 16)), max_size=3),
+    }),
+    "securityContext": st.fixed_dictionaries({}, optional={
+        "runAsUser": _maybe_int, "runAsGroup": _maybe_int}),
+    "livenessProbe": st.fixed_dictionaries({}, optional={
+        "tcpSocket": st.fixed_dictionaries({}, optional={"port": _maybe_int}),
+        "exec": st.fixed_dictionaries({}, optional={
+            "command": st.lists(st.text(max_size=6), max_size=2)}),
+        "periodSeconds": _maybe_int,
+    }),
+})
+_podspec = st.fixed_dictionaries({
+    "containers": st.lists(_container, max_size=3),
+}, optional={
+    "initContainers": st.lists(_container, max_size=2),
+    "restartPolicy": st.sampled_from(["Always", "OnFailure", "Never", "??"]),
+    "hostname": st.text(max_size=10),
+    "terminationGracePeriodSeconds": _maybe_int,
+    "activeDeadlineSeconds": _maybe_int,
+    "securityContext": st.fixed_dictionaries({}, optional={
+        "runAsUser": _maybe_int}),
+})
+_fuzz_pod = st.fixed_dictionaries({
+    "metadata": st.fixed_dictionaries({
+        "name": _names,
+        "namespace": _names,
+    }, optional={
+        "annotations": st.dictionaries(
+            st.sampled_from(["runpod.io/required-gpu-memory",
+                             "runpod.io/gpu-memory", "runpod.io/cloud-type",
+                             "runpod.io/ports", "runpod.io/datacenter-ids",
+                             "other"]),
+            st.text(max_size=10), max_size=3),
+    }),
+    "spec": _podspec,
+})
+
+
+@settings(max_examples=150, deadline=None)
+@given(pod=_fuzz_pod)
+def test_prepare_deploy_params_never_crashes(pod):
+    """Fuzz: arbitrary (semi-structured) pod specs either translate into
+    DeployParams or raise ValidationError — never TypeError/ValueError/
+    KeyError from deep inside the translation pipeline."""
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.provider.translate import (
+        ValidationError, prepare_deploy_params)
+    from k8s_runpod_kubelet_amd.runtime.types import DeployParams
+
+    try:
+        out = prepare_deploy_params(pod, FakeKube(), Config())
+    except ValidationError:
+        return
+    assert isinstance(out, DeployParams)
+    assert out.gpu_count >= 0
