@@ -51,6 +51,11 @@ def build_parser() -> argparse.ArgumentParser:
                    help="pod controller workers (reference uses 1)")
     p.add_argument("--gpu-count-override", type=int, default=-1,
                    help="force synthetic GPU inventory (CPU-only dev)")
+    p.add_argument("--fake-apiserver", action="store_true",
+                   help="start an in-process fake Kubernetes apiserver and "
+                        "connect to it (self-contained demo; no cluster "
+                        "needed — create pods with curl/kubectl against the "
+                        "printed URL)")
     return p
 
 
@@ -87,7 +92,19 @@ def main(argv=None) -> int:
     cfg.gpu_count_override = args.gpu_count_override
 
     validate_environment()
-    stack = build_stack(cfg)
+    fake_srv = None
+    client = None
+    if args.fake_apiserver:
+        from .kube.fake_apiserver import FakeApiServer
+        from .kube.real import ClusterConfig, HttpK8sClient
+
+        fake_srv = FakeApiServer().start()
+        client = HttpK8sClient(ClusterConfig(server=fake_srv.url))
+        print(f"fake apiserver at {fake_srv.url} — e.g.\n"
+              f"  curl -s {fake_srv.url}/api/v1/nodes/{cfg.node_name} | head\n"
+              f"  curl -s -XPOST {fake_srv.url}/api/v1/namespaces/default/pods"
+              f" -d @pod.json")
+    stack = build_stack(cfg, client=client)
     # Auth introspection (reference logAuthInfo, main.go:92-108).
     try:
         review = stack.client.self_subject_review()
