@@ -257,6 +257,12 @@ def test_gpu_port_readiness_and_exec(tmp_path):
         assert code == 0, out
         assert "ROCR_VISIBLE_DEVICES=" in out
         assert "AMDVK_GPU_IDS=" in out
+        # Clean up the held pod: stack.stop() does NOT kill pods (kubelet
+        # restarts must adopt them), and a leaked listener on 18081 makes a
+        # second suite run on the same box fail with EADDRINUSE.
+        kube.delete_pod("default", "gpuport")
+        assert wait_until(
+            lambda: stack.runtime.tracked_process_count() == 0, timeout_s=30)
     finally:
         stack.stop()
 
@@ -420,6 +426,9 @@ def test_gpu_crash_detection_and_replacement(tmp_path):
 
         assert wait_until(heir_ready, timeout_s=60) is not None
         assert time.time() - t0 < 30  # event path, not the 999 s tick
+        kube.delete_pod("default", "heir")
+        assert wait_until(
+            lambda: stack.runtime.tracked_process_count() == 0, timeout_s=30)
     finally:
         stack.stop()
 
