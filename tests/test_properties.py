@@ -264,3 +264,50 @@ def test_prepare_deploy_params_never_crashes(pod):
         return
     assert isinstance(out, DeployParams)
     assert out.gpu_count >= 0
+
+
+# -- status translation invariants -------------------------------------------
+
+_statuses = st.sampled_from(
+    ["RUNNING", "STARTING", "EXITED", "TERMINATING", "TERMINATED", "NOT_FOUND"])
+
+
+@settings(max_examples=200, deadline=None)
+@given(status=_statuses, ports_exposed=st.booleans(),
+       exit_code=st.one_of(st.none(), st.integers(0, 255)),
+       ready=st.booleans(), n_containers=st.integers(1, 3))
+def test_translate_status_invariants(status, ports_exposed, exit_code,
+                                     ready, n_containers):
+    """Every translation yields a kubectl-valid status: a known phase, all
+    four conditions exactly once, one containerStatus per spec container,
+    each with exactly one state key."""
+    from k8s_runpod_kubelet_amd.provider.instance import InstanceInfo
+    from k8s_runpod_kubelet_amd.provider.status import translate_status
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerRuntimeInfo, DetailedStatus)
+
+    pod = {"metadata": {"name": "p", "namespace": "d"},
+           "spec": {"containers": [{"name": f"c{i}", "image": "x"}
+                                   for i in range(n_containers)]}}
+    info = InstanceInfo(instance_id="i1", status=status,
+                        ports_exposed=ports_exposed)
+    detailed = DetailedStatus(
+        id="i1", desired_status=status,
+        containers=[ContainerRuntimeInfo(name=f"c{i}", pid=100 + i,
+                                         exit_code=exit_code, ready=ready)
+                    for i in range(n_containers)])
+    out = translate_status(pod, info, detailed)
+    assert out["phase"] in ("Pending", "Running", "Succeeded", "Failed",
+                            "Unknown")
+    conds = [c["type"] for c in out["conditions"]]
+    assert sorted(conds) == sorted(
+        ["PodScheduled", "Initialized", "Ready", "ContainersReady"])
+    css = out["containerStatuses"]
+    assert [c["name"] for c in css] == [f"c{i}" for i in range(n_containers)]
+    for cs in css:
+        assert len(cs["state"]) == 1
+        assert next(iter(cs["state"])) in ("running", "waiting", "terminated")
+    # Ready condition True only when the pod is actually Running+exposed
+    ready_cond = {c["type"]: c["status"] for c in out["conditions"]}["Ready"]
+    if ready_cond == "True":
+        assert status == "RUNNING" and ports_exposed
