@@ -129,3 +129,45 @@ def test_http_stack_watch_reconnect(http_stack):
     assert wait_until(
         lambda: (_pod_via_http(client, "w2") or {}).get("status", {}).get(
             "phase") == "Succeeded", timeout_s=15)
+
+
+def test_lease_renewed_over_time(http_stack, monkeypatch):
+    """Lease renewTime advances at the renewal cadence (reference
+    main.go:193-213 semantics)."""
+    import time as _time
+
+    stack, srv, client, cfg = http_stack
+    lease1 = srv.kube.leases.objects.get(f"kube-node-lease/{cfg.node_name}")
+    assert lease1 is not None
+    t1 = lease1["spec"]["renewTime"]
+    # force an immediate renewal instead of waiting 30 s
+    stack.node_controller._renew_lease()
+    _time.sleep(1.1)  # rfc3339 second resolution
+    stack.node_controller._renew_lease()
+    lease2 = srv.kube.leases.objects.get(f"kube-node-lease/{cfg.node_name}")
+    assert lease2["spec"]["renewTime"] >= t1
+    assert lease2["spec"]["holderIdentity"] == cfg.node_name
+
+
+def test_apiserver_survives_bad_requests(http_stack):
+    """Malformed bodies and unknown routes return errors without killing
+    the server (the informer keeps streaming afterwards)."""
+    import urllib.error
+    import urllib.request
+
+    stack, srv, client, cfg = http_stack
+    for path, data in (("/api/v1/namespaces/default/pods", b"{not json"),
+                       ("/api/v1/nonsense", None)):
+        req = urllib.request.Request(srv.url + path, data=data,
+                                     method="POST" if data else "GET")
+        try:
+            urllib.request.urlopen(req, timeout=5)
+        except urllib.error.HTTPError as e:
+            assert e.code in (404, 500)
+    # server still healthy: full pod lifecycle works after the abuse
+    client.create_pod("default", make_pod(
+        "afterabuse", gpus=0, command=["podworker"],
+        args=["--run-for", "0.05"], node=cfg.node_name))
+    assert wait_until(
+        lambda: (_pod_via_http(client, "afterabuse") or {}).get(
+            "status", {}).get("phase") == "Succeeded", timeout_s=15)
