@@ -175,3 +175,26 @@ def test_create_client_no_config(tmp_path, monkeypatch):
     monkeypatch.delenv("KUBERNETES_SERVICE_HOST", raising=False)
     with pytest.raises(RuntimeError, match="no Kubernetes config"):
         create_k8s_client("")
+
+
+def test_in_cluster_config(tmp_path, monkeypatch):
+    """In-cluster detection: serviceaccount token + env vars → https server
+    with Bearer token and the mounted CA (reference createK8sClient order,
+    main.go:464-502)."""
+    import k8s_runpod_kubelet_amd.kube.real as real
+
+    sa = tmp_path / "serviceaccount"
+    sa.mkdir()
+    (sa / "token").write_text("sekrit-token\n")
+    (sa / "ca.crt").write_text("CERT")
+    monkeypatch.setattr(real, "SA_DIR", str(sa))
+    monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+    monkeypatch.setenv("KUBERNETES_SERVICE_PORT", "6443")
+    cc = real.load_in_cluster()
+    assert cc is not None
+    assert cc.server == "https://10.0.0.1:6443"
+    assert cc.token == "sekrit-token"
+    assert cc.ca_path.endswith("ca.crt")
+    # flag beats in-cluster
+    monkeypatch.delenv("KUBERNETES_SERVICE_HOST")
+    assert real.load_in_cluster() is None
