@@ -20,6 +20,7 @@ def run_bench(args, timeout=240):
 
 def check_contract(out, n_gpus, steps, warmup):
     assert out["metric"] == "gpu_pods_scheduled_per_sec"
+    assert "p50 pod-Ready latency" in out["baseline_metric"]
     assert out["unit"] == "pods/s"
     assert out["n_gpus"] == n_gpus
     assert out["steps"] == steps
