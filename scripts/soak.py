@@ -149,7 +149,13 @@ def main() -> int:
                     continue
                 if age > 120:
                     counters["timeouts"] += 1
-                    print(f"TIMEOUT: {name} mode={st['mode']} state={state}",
+                    print(f"TIMEOUT: {name} mode={st['mode']} state={state} "
+                          f"pod={json.dumps((pod or {}).get('status', {}))}",
+                          file=sys.stderr)
+                    for inst in stack.runtime.list_instances():
+                        print(f"  inst {inst.id} {inst.desired_status} "
+                              f"{inst.namespace}/{inst.name}", file=sys.stderr)
+                    print(f"  reservations={list(stack.ledger.reservations)}",
                           file=sys.stderr)
                     try:
                         client.delete_pod("default", name)
