@@ -126,9 +126,16 @@ def main() -> int:
                 stack.node_controller.stop()
                 stack.provider.stop()
                 stack.runtime.close()
+                old = stack
                 stack = build_stack(cfg, client=client)
                 stack.start(serve_http=False)
                 kubelet_restarts += 1
+                # A real crash ends the process and the kernel reaps its
+                # fds; emulate that for the fd-leak check by collecting the
+                # dead stack's objects (EventLoop epfd/pidfds close on GC).
+                del old
+                import gc
+                gc.collect()
             # top up population (only while inside the window)
             while now < deadline and len(active) < args.max_active:
                 mode = rng.choices(
