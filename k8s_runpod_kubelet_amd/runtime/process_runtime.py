@@ -29,6 +29,7 @@ kubelet.go:1380-1535) without double-binding GPUs.
 
 from __future__ import annotations
 
+import dataclasses
 import json
 import logging
 import os
@@ -903,6 +904,13 @@ class ProcessRuntime(Runtime):
                     "command": c.command,
                     "args": c.args,
                     "tcp_ports": c.tcp_ports,
+                    # probes must survive kubelet restarts: an adopted pod
+                    # whose readinessProbe was lost could never become Ready
+                    "readiness": dataclasses.asdict(c.readiness)
+                    if c.readiness else None,
+                    "liveness": dataclasses.asdict(c.liveness)
+                    if c.liveness else None,
+                    "env": c.env,  # exec probes run in the container env
                 }
                 for c in inst.params.containers
             ],
@@ -935,6 +943,7 @@ class ProcessRuntime(Runtime):
                 rec = json.loads(path.read_text())
             except (OSError, json.JSONDecodeError):
                 continue
+            from .probes import ProbeSpec
             from .types import ContainerSpec
 
             params = DeployParams(
@@ -949,6 +958,11 @@ class ProcessRuntime(Runtime):
                         name=c["name"], image=c.get("image", ""),
                         command=c.get("command", []), args=c.get("args", []),
                         tcp_ports=c.get("tcp_ports", []),
+                        env=c.get("env", {}) or {},
+                        readiness=ProbeSpec(**c["readiness"])
+                        if c.get("readiness") else None,
+                        liveness=ProbeSpec(**c["liveness"])
+                        if c.get("liveness") else None,
                     )
                     for c in rec.get("container_specs", [])
                 ],
@@ -1005,6 +1019,20 @@ class ProcessRuntime(Runtime):
                         self._loop.add_process(cinfo.pid, pidfd, -1, cinfo.pid)
                         with self._lock:
                             self._pid_to_instance[cinfo.pid] = inst.id
+                        if not cinfo.ready:
+                            # The AMDVK_READY_FD pipe died with the old
+                            # kubelet: its signal can never arrive now. A
+                            # container WITH a readinessProbe re-proves
+                            # readiness via the probe loop; without one,
+                            # k8s semantics apply — a running container is
+                            # ready (found by soak race hunting: a pod
+                            # deployed moments before a kubelet crash was
+                            # stuck NotReady forever).
+                            spec = next(
+                                (s for s in params.containers
+                                 if s.name == cinfo.name), None)
+                            if spec is None or spec.readiness is None:
+                                cinfo.ready = True
                     else:
                         # Process died while we were away; exact code unknown.
                         cinfo.exit_code = -1

@@ -548,3 +548,54 @@ def test_active_deadline_seconds(stack):
     assert got["status"].get("reason") == "DeadlineExceeded"
     kube.delete_pod("default", "deadliner")
     assert wait_until(lambda: gone(kube, "deadliner"), timeout_s=15)
+
+
+def test_adoption_before_ready_still_becomes_ready(tmp_state_dir):
+    """A pod deployed moments before a kubelet crash — not yet Ready — must
+    still become Ready after adoption: probe specs are persisted (probed
+    pods resume probing) and pipe-based pods fall back to running=ready
+    (the READY pipe died with the old kubelet). Found by soak race
+    hunting (seed 9115)."""
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=8,
+                 notify_interval_s=0, pending_retry_interval_s=0.2)
+    kube = FakeKube()
+    s1 = build_stack(cfg, client=kube)
+    s1.runtime.enable_cgroups = False
+    s1.start(serve_http=False)
+    # Pipe-based pod whose READY write comes after the crash:
+    kube.create_pod("default", make_pod(
+        "latepipe", command=["podworker"],
+        args=["--startup-delay", "1.5", "--hold"]))
+    # Probed pod (pipe signal suppressed by the probe gate):
+    probed = make_pod("lateprobe", command=["podworker"], args=["--hold"])
+    probed["spec"]["containers"][0]["readinessProbe"] = {
+        "exec": {"command": ["/bin/true"]}, "periodSeconds": 1,
+        "failureThreshold": 1, "initialDelaySeconds": 2,
+    }
+    kube.create_pod("default", probed)
+
+    # Wait only for deployment (annotations), NOT readiness, then crash.
+    def deployed(name):
+        try:
+            p = kube.get_pod("default", name)
+        except NotFoundError:
+            return None
+        return p if p["metadata"]["annotations"].get(ann.POD_ID) else None
+
+    assert wait_until(lambda: deployed("latepipe"), timeout_s=10)
+    assert wait_until(lambda: deployed("lateprobe"), timeout_s=10)
+    assert ready(kube, "latepipe") is None  # still pre-Ready
+    s1.pod_controller.stop(); s1.node_controller.stop(); s1.provider.stop()
+    s1.runtime.close()
+
+    s2 = build_stack(cfg, client=kube)
+    s2.runtime.enable_cgroups = False
+    s2.start(serve_http=False)
+    try:
+        assert wait_until(lambda: ready(kube, "latepipe"), timeout_s=20)
+        assert wait_until(lambda: ready(kube, "lateprobe"), timeout_s=20)
+        for n in ("latepipe", "lateprobe"):
+            kube.delete_pod("default", n)
+            assert wait_until(lambda n=n: gone(kube, n), timeout_s=15)
+    finally:
+        s2.stop()
