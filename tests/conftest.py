@@ -18,10 +18,12 @@ def pytest_sessionfinish(session, exitstatus):
     (a joinable library thread torn down out of order), turning a fully
     green GPU run into SIGABRT/rc=134 — observed intermittently on MI355X
     boxes AFTER '14 passed' was printed. When HIP was initialized in this
-    process, register an atexit hook that exits with pytest's own status
-    before those destructors run. Test results are unaffected: this runs
-    only after the session (and its reporting) is complete, and propagates
-    the real exit status."""
+    process, install the exit guard (flush + drain earlier atexit hooks +
+    _exit) with pytest's own status. Test results are unaffected: this runs
+    only after the session (and its reporting) is complete, propagates the
+    real exit status, and — unlike a bare os._exit — still runs any
+    instrumentation hooks the harness registered at process start
+    (utils/exit_guard.py)."""
     try:
         import torch
 
@@ -32,9 +34,9 @@ def pytest_sessionfinish(session, exitstatus):
     except Exception:
         gpu_touched = False
     if gpu_touched:
-        import atexit
+        from k8s_runpod_kubelet_amd.utils.exit_guard import install
 
-        atexit.register(os._exit, int(exitstatus))
+        install(int(exitstatus))
 
 
 @pytest.fixture
