@@ -9,6 +9,7 @@ tests (with fake client/runtime injected)."""
 from __future__ import annotations
 
 import logging
+import os
 import signal
 import threading
 from dataclasses import dataclass, field
@@ -54,6 +55,8 @@ class Stack:
             self.health = HealthServer(
                 self.config.health_server_address, self.provider.ping,
                 ledger=self.ledger,
+                admin_token=self.config.admin_token
+                or os.environ.get("AMDVK_ADMIN_TOKEN", ""),
             )
             self.health.start()
             self.api_server = KubeletApiServer(

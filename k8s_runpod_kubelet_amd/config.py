@@ -60,6 +60,10 @@ class Config:
 
     # Ops
     health_server_address: str = ":8080"
+    # Bearer token for the state-mutating admin endpoints (cordon/uncordon)
+    # when called from non-loopback peers; empty = loopback-only. Env
+    # AMDVK_ADMIN_TOKEN overrides when this is unset.
+    admin_token: str = ""
     log_level: str = "info"
     heartbeat_interval_s: float = 300.0  # 0 disables (reference kubelet.go:73)
     registration_endpoint: str = ""  # optional registration hook, off by default

@@ -153,3 +153,14 @@ class Ledger:
     def snapshot(self) -> List[GpuState]:
         with self._lock:
             return list(self.states.values())
+
+    def set_cordoned(self, idx: int, cordoned: bool) -> bool:
+        """Operator cordon/uncordon under the ledger lock (admin endpoints
+        must not poke GpuState fields directly — unlocked writes raced the
+        placement path). Returns False for an unknown GPU index."""
+        with self._lock:
+            state = self.states.get(idx)
+            if state is None:
+                return False
+            state.cordoned = cordoned
+            return True

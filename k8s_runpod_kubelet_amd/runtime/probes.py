@@ -86,8 +86,15 @@ class ProbeState:
     result: Optional[bool] = None  # thresholded outcome (None = undecided)
 
 
-def run_probe(spec: ProbeSpec, env: Dict[str, str]) -> bool:
-    """Execute one probe attempt; True = success. Never raises."""
+def run_probe(spec: ProbeSpec, env: Dict[str, str],
+              exec_runner=None) -> bool:
+    """Execute one probe attempt; True = success. Never raises.
+
+    ``exec_runner(command, env, timeout_s) -> exit_code`` runs exec probes
+    inside the container's confinement (cgroup/credentials via the native
+    launcher — k8s runs probes *inside* the container, so they must never
+    execute as the kubelet's root user when runAsUser is set). The
+    subprocess fallback exists only for unit tests of this module."""
     try:
         if spec.kind == "tcp":
             with socket.create_connection(("127.0.0.1", spec.port),
@@ -103,6 +110,8 @@ def run_probe(spec: ProbeSpec, env: Dict[str, str]) -> bool:
             finally:
                 conn.close()
         if spec.kind == "exec":
+            if exec_runner is not None:
+                return exec_runner(spec.command, env, spec.timeout_s) == 0
             proc = subprocess.run(
                 spec.command, env=env, timeout=spec.timeout_s,
                 stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,

@@ -489,6 +489,13 @@ class ProcessRuntime(Runtime):
             inst.desired_status = PodStatus.TERMINATED
             self._teardown_resources(inst)
             return
+        if inst.deadline_exceeded:
+            # Pod was killed by activeDeadlineSeconds while init was running:
+            # keep last_error="DeadlineExceeded" so status reports the real
+            # reason, not the init kill's exit message.
+            inst.desired_status = PodStatus.EXITED
+            self._teardown_resources(inst)
+            return
         if cinfo.exit_code != 0:
             inst.last_error = (
                 f"init container {cinfo.name} exited with code {cinfo.exit_code}")
@@ -589,13 +596,25 @@ class ProcessRuntime(Runtime):
                     if now - st.last_run < spec.period_s:
                         continue
                     st.last_run = now
-                    if env is None and spec.kind == "exec":
-                        env = dict(os.environ)
-                        env.update(inst.params.env)
-                        env.update(device_env(inst.gpu_indices,
-                                              self.binder.ledger.inventory))
-                        env.update(cspec.env)
-                    ok = run_probe(spec, env or {})
+                    exec_runner = None
+                    if spec.kind == "exec":
+                        if env is None:
+                            env = dict(os.environ)
+                            env.update(inst.params.env)
+                            env.update(device_env(inst.gpu_indices,
+                                                  self.binder.ledger.inventory))
+                            env.update(cspec.env)
+                        # confine the probe like the container itself:
+                        # same cgroup (device filter included) and the
+                        # container's runAsUser/runAsGroup credentials
+                        cgroup_dir = inst.cgroup_dir
+                        uid, gid = cspec.run_as_uid, cspec.run_as_gid
+
+                        def exec_runner(command, penv, timeout_s,
+                                        _cg=cgroup_dir, _uid=uid, _gid=gid):
+                            return self._run_confined(
+                                command, penv, _cg, _uid, _gid, timeout_s)
+                    ok = run_probe(spec, env or {}, exec_runner=exec_runner)
                     outcome = advance(st, spec, ok)
                     if kind == "readiness" and outcome is not None:
                         if cinfo.ready != outcome:
@@ -822,27 +841,28 @@ class ProcessRuntime(Runtime):
             text = "\n".join(text.splitlines()[-tail:]) + "\n"
         return text
 
-    def exec_in_instance(self, instance_id: str, command: List[str],
-                         timeout_s: float = 30.0) -> tuple:
-        """Non-interactive exec with the instance's environment (same GPU
-        binding, same cgroup). Returns (exit_code, combined_output)."""
-        with self._lock:
-            inst = self._instances.get(instance_id)
-        if inst is None:
-            return 127, f"instance {instance_id} not found"
-        env = dict(os.environ)
-        env.pop("ROCR_VISIBLE_DEVICES", None)
-        env.pop("HIP_VISIBLE_DEVICES", None)
-        env.update(inst.params.env)
-        env.update(device_env(inst.gpu_indices, self.binder.ledger.inventory))
-        out_path = self.logs_dir / f".exec-{inst.id}-{secrets.token_hex(4)}.log"
+    def _run_confined(self, command: List[str], env: Dict[str, str],
+                      cgroup_dir: str, uid: int, gid: int,
+                      timeout_s: float, out_path: str = "") -> int:
+        """Run one command to completion inside the pod's confinement
+        (cgroup + device filter + credentials) via the native launcher.
+        Returns the exit code (127 = spawn failure, 124 = timeout killed).
+        Used by kubectl-exec and by exec probes — probes must NOT run as
+        the kubelet (root) when the container has runAsUser."""
+        argv = list(command)
+        if "/" not in argv[0]:
+            # credential-drop/exec fast path does no PATH search
+            resolved = shutil.which(argv[0])
+            if resolved:
+                argv[0] = resolved
+        sink = out_path or "/dev/null"
         try:
             pid, pidfd, _, _, _ = self._native.launch_process(
-                list(command), [f"{k}={v}" for k, v in env.items()],
-                "", str(out_path), str(out_path), inst.cgroup_dir, True, False,
+                argv, [f"{k}={v}" for k, v in env.items()],
+                "", sink, sink, cgroup_dir, True, False, uid, gid,
             )
-        except RuntimeError as exc:
-            return 127, str(exc)
+        except RuntimeError:
+            return 127
         loop = self._native.EventLoop()
         loop.add_process(pid, pidfd, -1, 0)
         deadline = time.time() + timeout_s
@@ -854,6 +874,31 @@ class ProcessRuntime(Runtime):
         if exit_code is None:
             self._native.signal_process(pid, 9, True)
             exit_code = 124
+        return exit_code
+
+    def exec_in_instance(self, instance_id: str, command: List[str],
+                         timeout_s: float = 30.0) -> tuple:
+        """Non-interactive exec with the instance's environment (same GPU
+        binding, same cgroup, same credentials as the first container).
+        Returns (exit_code, combined_output)."""
+        with self._lock:
+            inst = self._instances.get(instance_id)
+        if inst is None:
+            return 127, f"instance {instance_id} not found"
+        env = dict(os.environ)
+        env.pop("ROCR_VISIBLE_DEVICES", None)
+        env.pop("HIP_VISIBLE_DEVICES", None)
+        env.update(inst.params.env)
+        env.update(device_env(inst.gpu_indices, self.binder.ledger.inventory))
+        cspec = inst.params.containers[0] if inst.params.containers else None
+        uid = cspec.run_as_uid if cspec else -1
+        gid = cspec.run_as_gid if cspec else -1
+        out_path = self.logs_dir / f".exec-{inst.id}-{secrets.token_hex(4)}.log"
+        exit_code = self._run_confined(
+            list(command), env, inst.cgroup_dir, uid, gid, timeout_s,
+            out_path=str(out_path))
+        if exit_code == 127 and not out_path.exists():
+            return 127, f"exec spawn failed: {command[0]!r}"
         output = out_path.read_text(errors="replace") if out_path.exists() else ""
         out_path.unlink(missing_ok=True)
         return exit_code, output
@@ -874,6 +919,12 @@ class ProcessRuntime(Runtime):
             "namespace": inst.params.namespace,
             "name": inst.params.name,
             "restart_policy": inst.params.restart_policy,
+            # deadline/grace must survive restarts: an adopted RUNNING pod
+            # keeps its activeDeadlineSeconds budget (timer re-armed with the
+            # remainder in adopt_persisted) and its grace window
+            "active_deadline_s": inst.params.active_deadline_s,
+            "termination_grace_s": inst.params.termination_grace_s,
+            "deadline_exceeded": inst.deadline_exceeded,
             "containers": [
                 {
                     "name": c.name,
@@ -911,6 +962,13 @@ class ProcessRuntime(Runtime):
                     "liveness": dataclasses.asdict(c.liveness)
                     if c.liveness else None,
                     "env": c.env,  # exec probes run in the container env
+                    # credentials/cwd must survive restarts: an adopted pod
+                    # whose container crash-restarts would otherwise relaunch
+                    # as the kubelet's user (root) — silently dropping its
+                    # securityContext.runAsUser
+                    "run_as_uid": c.run_as_uid,
+                    "run_as_gid": c.run_as_gid,
+                    "working_dir": c.working_dir,
                 }
                 for c in inst.params.containers
             ],
@@ -920,6 +978,9 @@ class ProcessRuntime(Runtime):
                     "image": c.image,
                     "command": c.command,
                     "args": c.args,
+                    "run_as_uid": c.run_as_uid,
+                    "run_as_gid": c.run_as_gid,
+                    "working_dir": c.working_dir,
                 }
                 for c in inst.params.init_containers
             ],
@@ -953,6 +1014,9 @@ class ProcessRuntime(Runtime):
                 gpu_count=rec.get("gpu_count", 0),
                 gpu_memory_bytes=rec.get("gpu_memory_bytes", 0),
                 restart_policy=rec.get("restart_policy", "Never"),
+                active_deadline_s=rec.get("active_deadline_s", 0.0),
+                termination_grace_s=rec.get("termination_grace_s",
+                                            TERM_GRACE_S),
                 containers=[
                     ContainerSpec(
                         name=c["name"], image=c.get("image", ""),
@@ -963,6 +1027,9 @@ class ProcessRuntime(Runtime):
                         if c.get("readiness") else None,
                         liveness=ProbeSpec(**c["liveness"])
                         if c.get("liveness") else None,
+                        run_as_uid=c.get("run_as_uid", -1),
+                        run_as_gid=c.get("run_as_gid", -1),
+                        working_dir=c.get("working_dir", ""),
                     )
                     for c in rec.get("container_specs", [])
                 ],
@@ -970,6 +1037,9 @@ class ProcessRuntime(Runtime):
                     ContainerSpec(
                         name=c["name"], image=c.get("image", ""),
                         command=c.get("command", []), args=c.get("args", []),
+                        run_as_uid=c.get("run_as_uid", -1),
+                        run_as_gid=c.get("run_as_gid", -1),
+                        working_dir=c.get("working_dir", ""),
                     )
                     for c in rec.get("init_specs", [])
                 ],
@@ -982,6 +1052,7 @@ class ProcessRuntime(Runtime):
                 created_at=rec.get("created_at", time.time()),
                 cost_per_hr=rec.get("cost_per_hr", 0.0),
                 init_index=rec.get("init_index", 0),
+                deadline_exceeded=rec.get("deadline_exceeded", False),
             )
             for c in rec.get("init_containers", []):
                 icinfo = ContainerRuntimeInfo(
@@ -1081,6 +1152,20 @@ class ProcessRuntime(Runtime):
                 self._instances[inst.id] = inst
                 for c in restartable:
                     self._schedule_restart(inst, c)
+                if (params.active_deadline_s > 0 and not inst.deadline_exceeded
+                        and inst.desired_status in (PodStatus.RUNNING,
+                                                    PodStatus.STARTING)):
+                    # Re-arm activeDeadlineSeconds with the *remaining*
+                    # budget (deadline anchored at deploy time, not at
+                    # adoption); an already-blown budget fires immediately.
+                    remaining = (inst.created_at + params.active_deadline_s
+                                 - time.time())
+                    t = threading.Timer(max(0.0, remaining),
+                                        self._deadline_exceeded,
+                                        args=(inst.id,))
+                    t.daemon = True
+                    self._deadline_timers[inst.id] = t
+                    t.start()
             adopted.append(inst.id)
         if adopted:
             log.info("adopted persisted instances", extra={"count": len(adopted)})

@@ -21,12 +21,18 @@ log = logging.getLogger("server.health")
 
 class HealthServer:
     def __init__(self, address: str, ready_fn: Optional[Callable[[], None]] = None,
-                 ledger=None):
+                 ledger=None, admin_token: str = ""):
         host, _, port = address.rpartition(":")
         self.host = host or "0.0.0.0"
         self.port = int(port)
         self.ready_fn = ready_fn
         self.ledger = ledger  # enables the cordon/uncordon admin endpoints
+        # Admin (state-mutating) endpoints are restricted: loopback peers
+        # always allowed; non-loopback needs this bearer token (set via
+        # config/AMDVK_ADMIN_TOKEN). Empty token = loopback-only. The
+        # health server binds 0.0.0.0 by default for probes — without this
+        # gate any network peer could cordon every GPU.
+        self.admin_token = admin_token
         self._alive = True
         self._server: Optional[ThreadingHTTPServer] = None
         self._thread: Optional[threading.Thread] = None
@@ -84,24 +90,38 @@ class HealthServer:
                 else:
                     self._respond(404, b"not found")
 
+            def _admin_authorized(self) -> bool:
+                peer = self.client_address[0] if self.client_address else ""
+                if peer in ("127.0.0.1", "::1", "::ffff:127.0.0.1"):
+                    return True
+                if outer.admin_token:
+                    auth = self.headers.get("Authorization", "")
+                    return auth == f"Bearer {outer.admin_token}"
+                return False
+
             def do_POST(self):
                 # Operator cordon/uncordon (kubectl-cordon analogue at GPU
                 # granularity): POST /cordon/<idx> | /uncordon/<idx>.
+                # Loopback or bearer-token only — these mutate scheduling
+                # state and must not be open to arbitrary network peers.
                 parts = [p for p in self.path.strip("/").split("/") if p]
                 if (len(parts) == 2 and parts[0] in ("cordon", "uncordon")
                         and outer.ledger is not None):
+                    if not self._admin_authorized():
+                        self._respond(403, b"admin endpoint: loopback or "
+                                           b"bearer token required")
+                        return
                     try:
                         idx = int(parts[1])
                     except ValueError:
                         self._respond(400, b"bad gpu index")
                         return
-                    state = outer.ledger.states.get(idx)
-                    if state is None:
+                    cordoned = parts[0] == "cordon"
+                    if not outer.ledger.set_cordoned(idx, cordoned):
                         self._respond(404, f"no GPU {idx}".encode())
                         return
-                    state.cordoned = parts[0] == "cordon"
                     log.info("gpu cordon state changed",
-                             extra={"gpu": idx, "cordoned": state.cordoned})
+                             extra={"gpu": idx, "cordoned": cordoned})
                     self._respond(200, b"ok")
                 else:
                     self._respond(404, b"not found")

@@ -209,3 +209,121 @@ def test_termination_grace_period(process_runtime):
     took = time.time() - t0
     assert 0.4 < took < 5, took  # killed at ~0.5 s grace, not 10 s
     assert s.containers[0].exit_code == 128 + 9
+
+
+def test_adoption_preserves_credentials_and_deadline(synthetic_ledger,
+                                                     tmp_state_dir):
+    """Round-1 advisory (high): securityContext credentials, workingDir,
+    activeDeadlineSeconds and terminationGracePeriodSeconds must survive a
+    kubelet restart — a crash-restarting container was being relaunched as
+    root with default grace/deadline."""
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    st = rt.deploy(DeployParams(
+        pod_key="default-cred", name="cred",
+        restart_policy="Always",
+        active_deadline_s=3600.0,
+        termination_grace_s=2.5,
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"], args=["-c", "sleep 60"],
+            run_as_uid=12345, run_as_gid=54321, working_dir="/tmp")],
+        init_containers=[ContainerSpec(
+            name="setup", command=["/bin/true"],
+            run_as_uid=12345, run_as_gid=54321, working_dir="/tmp")],
+    ))
+    wait_status(rt, st.id, PodStatus.RUNNING)
+    rt._stop.set(); rt._loop.wake(); rt._watcher.join(timeout=2)
+
+    rt2 = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    try:
+        rt2.adopt_persisted()
+        with rt2._lock:
+            inst = rt2._instances[st.id]
+        p = inst.params
+        assert p.active_deadline_s == 3600.0
+        assert p.termination_grace_s == 2.5
+        c = p.containers[0]
+        assert (c.run_as_uid, c.run_as_gid, c.working_dir) == (
+            12345, 54321, "/tmp")
+        ic = p.init_containers[0]
+        assert (ic.run_as_uid, ic.run_as_gid, ic.working_dir) == (
+            12345, 54321, "/tmp")
+        # the re-armed deadline timer exists for the adopted RUNNING pod
+        assert st.id in rt2._deadline_timers
+        rt2.terminate(st.id)
+        wait_status(rt2, st.id, PodStatus.TERMINATED)
+    finally:
+        rt2.close()
+        rt.close()
+
+
+def test_adoption_rearms_expired_deadline(synthetic_ledger, tmp_state_dir):
+    """activeDeadlineSeconds whose budget expired while the kubelet was
+    down fires immediately on adoption (remaining ≤ 0)."""
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    st = rt.deploy(params(pod_key="default-dl", args=["--hold"],
+                          active_deadline_s=0.3))
+    wait_status(rt, st.id, PodStatus.RUNNING)
+    rt._stop.set(); rt._loop.wake(); rt._watcher.join(timeout=2)
+    # cancel the first runtime's own deadline timer so IT doesn't kill the
+    # process — the restart must enforce the deadline itself
+    with rt._lock:
+        t = rt._deadline_timers.pop(st.id, None)
+    if t:
+        t.cancel()
+    time.sleep(0.4)  # budget now blown
+
+    rt2 = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
+    try:
+        rt2.adopt_persisted()
+        s = wait_status(rt2, st.id, PodStatus.EXITED, timeout=5)
+        assert s.desired_status == PodStatus.EXITED
+        assert s.last_error == "DeadlineExceeded"
+    finally:
+        rt2.close()
+        rt.close()
+
+
+def test_deadline_during_init_keeps_reason(process_runtime):
+    """Round-1 advisory (low): a pod killed by activeDeadlineSeconds while
+    an init container runs must still report DeadlineExceeded, not the init
+    kill's exit message."""
+    rt = process_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-initdl", name="initdl",
+        active_deadline_s=0.2,
+        init_containers=[ContainerSpec(
+            name="slow-init", command=["/bin/sh"], args=["-c", "sleep 30"])],
+        containers=[ContainerSpec(
+            name="main", command=["/bin/true"])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED, timeout=5)
+    assert s.desired_status == PodStatus.EXITED
+    assert s.last_error == "DeadlineExceeded"
+
+
+def test_exec_probe_runs_with_container_credentials(process_runtime):
+    """Round-1 advisory (medium): exec probes must run with the container's
+    runAsUser, not as the kubelet (root). The readiness probe here succeeds
+    only when its uid is the container's 65534."""
+    import os
+
+    if os.geteuid() != 0:
+        pytest.skip("needs root to exercise credential drop")
+    rt = process_runtime
+    from k8s_runpod_kubelet_amd.runtime.probes import ProbeSpec
+
+    st = rt.deploy(DeployParams(
+        pod_key="default-probeuid", name="probeuid",
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"], args=["-c", "sleep 30"],
+            run_as_uid=65534, run_as_gid=65534,
+            readiness=ProbeSpec(kind="exec",
+                                command=["/bin/sh", "-c",
+                                         '[ "$(id -u)" = 65534 ]'],
+                                period_s=1.0, timeout_s=5.0))],
+    ))
+    s = wait_ready(rt, st.id, timeout=10.0)
+    assert s.containers[0].ready, \
+        "exec probe did not run as the container's uid"
+    rt.terminate(st.id)
+    wait_status(rt, st.id, PodStatus.TERMINATED)

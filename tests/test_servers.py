@@ -135,3 +135,66 @@ def test_cordon_endpoint(synthetic_ledger):
             assert e.code == 404
     finally:
         hs.stop()
+
+
+def _nonloopback_ip():
+    """A local non-loopback IPv4, or None (UDP connect sends no packets)."""
+    import socket
+
+    s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+    try:
+        s.connect(("10.255.255.255", 1))
+        ip = s.getsockname()[0]
+        return None if ip.startswith("127.") else ip
+    except OSError:
+        return None
+    finally:
+        s.close()
+
+
+def test_cordon_admin_requires_loopback_or_token(synthetic_ledger):
+    """Admin endpoints mutate scheduling state: non-loopback peers are
+    rejected without the bearer token (round-1 advisory: anyone on the
+    network could cordon every GPU)."""
+    import urllib.error
+    import urllib.request
+
+    from k8s_runpod_kubelet_amd.server.health import HealthServer
+
+    ip = _nonloopback_ip()
+    if ip is None:
+        import pytest
+
+        pytest.skip("no non-loopback interface")
+    hs = HealthServer("0.0.0.0:0", None, ledger=synthetic_ledger,
+                      admin_token="sekrit")
+    hs.start()
+    try:
+        base = f"http://{ip}:{hs.port}"
+        req = urllib.request.Request(f"{base}/cordon/3", method="POST")
+        try:
+            urllib.request.urlopen(req, timeout=5)
+            raise AssertionError("unauthenticated non-loopback cordon allowed")
+        except urllib.error.HTTPError as exc:
+            assert exc.code == 403
+        assert not synthetic_ledger.states[3].cordoned
+        # wrong token still rejected
+        req = urllib.request.Request(f"{base}/cordon/3", method="POST",
+                                     headers={"Authorization": "Bearer nope"})
+        try:
+            urllib.request.urlopen(req, timeout=5)
+            raise AssertionError("wrong token accepted")
+        except urllib.error.HTTPError as exc:
+            assert exc.code == 403
+        # correct token accepted
+        req = urllib.request.Request(f"{base}/cordon/3", method="POST",
+                                     headers={"Authorization": "Bearer sekrit"})
+        assert urllib.request.urlopen(req, timeout=5).status == 200
+        assert synthetic_ledger.states[3].cordoned
+        # loopback works without a token (readiness for kubectl-exec'd ops)
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{hs.port}/uncordon/3", method="POST")
+        assert urllib.request.urlopen(req, timeout=5).status == 200
+        assert not synthetic_ledger.states[3].cordoned
+    finally:
+        hs.stop()
