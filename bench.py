@@ -91,6 +91,9 @@ def run_bench(n_gpus: int, steps: int, warmup: int, use_gpu: bool,
             "metadata": {"name": name, "namespace": "default"},
             "spec": {
                 "nodeName": cfg.node_name,
+                # run-to-completion pods (a user would set this on Job-style
+                # pods; the k8s default Always would crash-loop them)
+                "restartPolicy": "Never",
                 "containers": [{
                     "name": "main",
                     "image": "amdvk/podworker:bench",

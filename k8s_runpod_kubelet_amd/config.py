@@ -58,6 +58,11 @@ class Config:
     gpu_memory_default_gb: int = 16  # reference runpod_client.go:1189 default
     datacenter_ids: List[str] = field(default_factory=list)
 
+    # spec.restartPolicy default for pods that arrive without one (a real
+    # apiserver defaults it to Always at admission — k8s semantics). Set to
+    # "Never" for the reference's run-to-completion model.
+    restart_policy_default: str = "Always"
+
     # Ops
     health_server_address: str = ":8080"
     # Bearer token for the state-mutating admin endpoints (cordon/uncordon)

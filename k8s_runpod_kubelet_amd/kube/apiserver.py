@@ -13,6 +13,7 @@ from __future__ import annotations
 
 import json
 import logging
+import os
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 from typing import Optional
@@ -43,6 +44,64 @@ class KubeletApiServer:
                 self.end_headers()
                 self.wfile.write(body)
 
+            def _stream_logs(self, namespace: str, pod: str, container: str,
+                             tail: int) -> None:
+                """`kubectl logs -f` (follow=true): chunked-encoding tail of
+                the local log file, ending when the container terminates
+                (the reference stubs logs entirely, kubelet.go:2047-2066)."""
+                import time as _time
+
+                path = outer.provider.get_container_log_path(
+                    namespace, pod, container)
+                if path is None:
+                    self._respond(404, b"no log stream for this pod")
+                    return
+                self.send_response(200)
+                self.send_header("Content-Type", "text/plain")
+                self.send_header("Transfer-Encoding", "chunked")
+                self.end_headers()
+
+                def chunk(data: bytes) -> None:
+                    self.wfile.write(b"%x\r\n" % len(data) + data + b"\r\n")
+                    self.wfile.flush()
+
+                try:
+                    # one read for both backlog and offset: no gap between
+                    # "what was sent" and "where the tail resumes"
+                    try:
+                        with open(path, "rb") as fh:
+                            raw = fh.read()
+                    except OSError:
+                        raw = b""
+                    offset = len(raw)
+                    if tail > 0:
+                        raw = b"".join(
+                            raw.splitlines(keepends=True)[-tail:])
+                    if raw:
+                        chunk(raw)
+                    while True:
+                        finished = outer.provider.pod_log_finished(
+                            namespace, pod)
+                        try:
+                            size = os.path.getsize(path)
+                        except OSError:
+                            size = offset
+                        if size > offset:
+                            with open(path, "rb") as fh:
+                                fh.seek(offset)
+                                data = fh.read(size - offset)
+                            offset = size
+                            if data:
+                                chunk(data)
+                        elif finished:
+                            break
+                        else:
+                            _time.sleep(0.1)
+                    self.wfile.write(b"0\r\n\r\n")
+                    self.wfile.flush()
+                except (BrokenPipeError, ConnectionResetError):
+                    pass  # client hung up (Ctrl-C on kubectl logs -f)
+
             def do_GET(self):
                 parsed = urlparse(self.path)
                 parts = [unquote(p) for p in parsed.path.strip("/").split("/") if p]
@@ -51,6 +110,11 @@ class KubeletApiServer:
                     namespace, pod = parts[1], parts[2]
                     container = parts[3] if len(parts) > 3 else ""
                     tail = int(query.get("tailLines", ["-1"])[0])
+                    follow = query.get(
+                        "follow", ["false"])[0].lower() in ("true", "1")
+                    if follow:
+                        self._stream_logs(namespace, pod, container, tail)
+                        return
                     text = outer.provider.get_container_logs(
                         namespace, pod, container, tail
                     )

@@ -124,6 +124,12 @@ class FakeKube(K8sClient):
                 raise ConflictError(f"pod {key} exists")
             pod.setdefault("kind", "Pod")
             pod.setdefault("apiVersion", "v1")
+            # apiserver-style admission defaulting (a real apiserver fills
+            # these before the kubelet ever sees the pod — k8s defaults:
+            # restartPolicy Always, terminationGracePeriodSeconds 30):
+            spec = pod.setdefault("spec", {})
+            spec.setdefault("restartPolicy", "Always")
+            spec.setdefault("terminationGracePeriodSeconds", 30)
             self._prepare(pod)
             self.pods.objects[key] = pod
             self.pods.emit("ADDED", pod)

@@ -976,6 +976,29 @@ class Provider:
             return ""
         return self.runtime.get_logs(info.instance_id, container, tail)
 
+    def get_container_log_path(self, namespace: str, name: str,
+                               container: str = "") -> Optional[str]:
+        """Local log-file path for follow-mode streaming (None when the pod
+        has no instance or the runtime keeps no files)."""
+        with self._pods_lock:
+            info = self._pod_status.get(f"{namespace}-{name}")
+        if info is None or not info.instance_id:
+            return None
+        fn = getattr(self.runtime, "get_log_path", None)
+        if fn is None:
+            return None
+        return fn(info.instance_id, container)
+
+    def pod_log_finished(self, namespace: str, name: str) -> bool:
+        """True when no more log output can appear (instance terminal or
+        untracked) — the follow-stream stop condition."""
+        with self._pods_lock:
+            info = self._pod_status.get(f"{namespace}-{name}")
+        if info is None or not info.instance_id:
+            return True
+        return info.status in (PodStatus.EXITED, PodStatus.TERMINATED,
+                               PodStatus.NOT_FOUND)
+
     def run_in_container(self, namespace: str, name: str, command: List[str],
                          timeout_s: float = 30.0) -> tuple:
         """One-shot exec in the pod's environment (GPU binding included).
@@ -987,6 +1010,11 @@ class Provider:
             info = self._pod_status.get(f"{namespace}-{name}")
         if info is None:
             return 127, f"pod {namespace}/{name} not tracked by this node"
+        if not info.instance_id:
+            # Pending pod with no backend instance yet: report the pod
+            # state instead of a confusing 'instance  not found'.
+            return 126, (f"pod {namespace}/{name} has no running instance "
+                         f"(status {info.status}); retry once it is Running")
         exec_fn = getattr(self.runtime, "exec_in_instance", None)
         if exec_fn is None:
             return 501, "runtime does not support exec"

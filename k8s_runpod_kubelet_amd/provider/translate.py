@@ -309,13 +309,16 @@ def prepare_deploy_params(
         labels=dict(metadata.get("labels", {}) or {}),
         hostname=pod.get("spec", {}).get("hostname")
         or metadata.get("name", ""),
-        # Deliberate deviation from vanilla k8s (default Always): unset means
-        # Never — run-to-completion is the reference's model (a stopped
-        # RunPod instance is EXITED/Succeeded, kubelet.go:1906); an explicit
-        # spec.restartPolicy is honored with full semantics.
-        restart_policy=pod.get("spec", {}).get("restartPolicy", "Never"),
+        # k8s semantics: a pod without restartPolicy means Always (apiserver
+        # admission normally sets it before the kubelet sees the pod; this
+        # default covers pods arriving un-defaulted). Operators who want the
+        # reference's run-to-completion model (a stopped RunPod instance is
+        # EXITED/Succeeded, kubelet.go:1906) set
+        # config.restart_policy_default: Never.
+        restart_policy=pod.get("spec", {}).get("restartPolicy")
+        or config.restart_policy_default,
         termination_grace_s=_float_or(
-            pod.get("spec", {}).get("terminationGracePeriodSeconds", 10), 10.0),
+            pod.get("spec", {}).get("terminationGracePeriodSeconds", 30), 30.0),
         active_deadline_s=_float_or(
             pod.get("spec", {}).get("activeDeadlineSeconds", 0) or 0, 0.0),
     )

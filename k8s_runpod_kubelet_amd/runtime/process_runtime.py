@@ -648,6 +648,64 @@ class ProcessRuntime(Runtime):
 
     # ------------- status -------------
 
+    def _pod_pids(self, inst: Instance) -> set:
+        """Every live pid belonging to the pod: the cgroup's procs when the
+        pod has a slot, else the container pids plus their descendants
+        (via /proc/<pid>/task/*/children)."""
+        if inst.cgroup_dir:
+            try:
+                with open(inst.cgroup_dir + "/cgroup.procs",
+                          encoding="ascii") as fh:
+                    pids = {int(line) for line in fh if line.strip()}
+                if pids:
+                    return pids
+            except OSError:
+                pass
+        pids = {c.pid for c in inst.containers
+                if c.exit_code is None and c.pid > 0}
+        frontier = list(pids)
+        while frontier:
+            pid = frontier.pop()
+            try:
+                for task in os.listdir(f"/proc/{pid}/task"):
+                    with open(f"/proc/{pid}/task/{task}/children",
+                              encoding="ascii") as fh:
+                        for child in fh.read().split():
+                            c = int(child)
+                            if c not in pids:
+                                pids.add(c)
+                                frontier.append(c)
+            except OSError:
+                continue
+        return pids
+
+    def _pod_listening_ports(self, inst: Instance) -> set:
+        """TCP ports in LISTEN state owned by *this pod's* processes —
+        socket inodes from /proc/net/tcp{,6} attributed via
+        /proc/<pid>/fd. A host-wide scan (round-1 weak #3) marked a pod's
+        port exposed when any unrelated process listened on it; the
+        reference gates readiness on the backend's per-pod portMappings
+        (kubelet.go:566-605), i.e. per-instance."""
+        by_inode = _listening_tcp_inodes()
+        if not by_inode:
+            return set()
+        ports = set()
+        for pid in self._pod_pids(inst):
+            try:
+                for fd in os.listdir(f"/proc/{pid}/fd"):
+                    try:
+                        link = os.readlink(f"/proc/{pid}/fd/{fd}")
+                    except OSError:
+                        continue
+                    if link.startswith("socket:["):
+                        ino = int(link[8:-1])
+                        port = by_inode.get(ino)
+                        if port is not None:
+                            ports.add(port)
+            except OSError:
+                continue
+        return ports
+
     def _status_of(self, inst: Instance) -> DetailedStatus:
         ports: Dict[int, int] = {}
         if inst.desired_status == PodStatus.RUNNING:
@@ -655,7 +713,7 @@ class ProcessRuntime(Runtime):
             for c in inst.params.containers:
                 want.extend(c.tcp_ports)
             if want:
-                listening = _listening_tcp_ports()
+                listening = self._pod_listening_ports(inst)
                 ports = {p: p for p in want if p in listening}
         return DetailedStatus(
             id=inst.id,
@@ -823,17 +881,25 @@ class ProcessRuntime(Runtime):
 
     # ------------- logs -------------
 
-    def get_logs(self, instance_id: str, container: str = "", tail: int = -1) -> str:
+    def get_log_path(self, instance_id: str, container: str = "") -> Optional[str]:
+        """Filesystem path of a container's log (for `kubectl logs -f`
+        streaming — the file is local, so follow is a tail)."""
         with self._lock:
             inst = self._instances.get(instance_id)
         if inst is None:
-            return ""
+            return None
         names = [c.name for c in inst.params.containers]
         if container and container in names:
             name = container
         else:
             name = names[0] if names else ""
-        path = self.logs_dir / f"{instance_id}-{name}.log"
+        return str(self.logs_dir / f"{instance_id}-{name}.log")
+
+    def get_logs(self, instance_id: str, container: str = "", tail: int = -1) -> str:
+        path = self.get_log_path(instance_id, container)
+        if path is None:
+            return ""
+        path = Path(path)
         if not path.exists():
             return ""
         text = path.read_text(errors="replace")
@@ -1184,17 +1250,22 @@ class ProcessRuntime(Runtime):
             timer.cancel()
 
 
-def _listening_tcp_ports() -> set:
-    """LISTEN-state local TCP ports from /proc/net/tcp{,6} (state 0A)."""
-    ports = set()
+def _listening_tcp_inodes() -> dict:
+    """socket-inode → local port for LISTEN-state TCP sockets from
+    /proc/net/tcp{,6} (state 0A, inode field 9)."""
+    by_inode = {}
     for path in ("/proc/net/tcp", "/proc/net/tcp6"):
         try:
             with open(path, "r", encoding="ascii") as fh:
                 next(fh, None)
                 for line in fh:
                     parts = line.split()
-                    if len(parts) > 3 and parts[3] == "0A":
-                        ports.add(int(parts[1].rsplit(":", 1)[1], 16))
+                    if len(parts) > 9 and parts[3] == "0A":
+                        try:
+                            port = int(parts[1].rsplit(":", 1)[1], 16)
+                            by_inode[int(parts[9])] = port
+                        except ValueError:
+                            continue
         except OSError:
             continue
-    return ports
+    return by_inode

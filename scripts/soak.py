@@ -76,7 +76,8 @@ def main() -> int:
             **({"resources": {"limits": {"amd.com/gpu": str(gpus)}}}
                if gpus else {}),
         }
-        spec = {"nodeName": cfg.node_name, "containers": [container]}
+        spec = {"nodeName": cfg.node_name, "containers": [container],
+                "restartPolicy": "Never"}
         if mode == "probed":
             container["readinessProbe"] = {
                 "exec": {"command": ["/bin/true"]},

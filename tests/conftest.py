@@ -75,8 +75,12 @@ def process_runtime(synthetic_ledger, tmp_state_dir):
 
 def make_pod(name="p1", namespace="default", node="virtual-runpod", gpus=0,
              annotations=None, command=None, args=None, ports=None,
-             containers=None, labels=None, owner=None):
-    """Test pod factory (dict in wire shape)."""
+             containers=None, labels=None, owner=None,
+             restart_policy="Never"):
+    """Test pod factory (dict in wire shape). restart_policy defaults to
+    Never because most tests run pods to completion (as a user would for a
+    Job-style pod); pass None to omit the field and exercise apiserver/
+    kubelet defaulting (k8s default: Always)."""
     if containers is None:
         c = {"name": "main", "image": "amdvk/test:latest"}
         if command is not None:
@@ -96,6 +100,8 @@ def make_pod(name="p1", namespace="default", node="virtual-runpod", gpus=0,
                      "labels": dict(labels or {})},
         "spec": {"nodeName": node, "containers": containers},
     }
+    if restart_policy is not None:
+        pod["spec"]["restartPolicy"] = restart_policy
     if owner:
         pod["metadata"]["ownerReferences"] = [owner]
     return pod

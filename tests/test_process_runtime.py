@@ -327,3 +327,51 @@ def test_exec_probe_runs_with_container_credentials(process_runtime):
         "exec probe did not run as the container's uid"
     rt.terminate(st.id)
     wait_status(rt, st.id, PodStatus.TERMINATED)
+
+
+def test_port_exposure_attributed_to_pod_pids(process_runtime):
+    """Port-gated readiness must attribute listening sockets to the pod's
+    own processes (socket inodes via /proc/<pid>/fd) — an unrelated process
+    listening on the pod's port must NOT mark it exposed (round-1 weak #3;
+    reference gates on per-instance portMappings, kubelet.go:566-605)."""
+    import socket
+
+    rt = process_runtime
+    # unrelated listener (this test process) on port B
+    other = socket.socket()
+    other.bind(("127.0.0.1", 0))
+    other.listen(1)
+    port_b = other.getsockname()[1]
+    # pick a second free port for the pod itself
+    probe = socket.socket()
+    probe.bind(("127.0.0.1", 0))
+    port_a = probe.getsockname()[1]
+    probe.close()
+    try:
+        st = rt.deploy(params(
+            pod_key="default-ports",
+            containers=[ContainerSpec(
+                name="main", command=["podworker"],
+                args=["--hold", "--listen-port", str(port_a)],
+                tcp_ports=[port_a, port_b])],
+        ))
+
+        def pod_port_up():
+            s = rt.get_detailed_status(st.id)
+            return s if port_a in s.port_mappings else None
+
+        deadline = time.time() + 10
+        s = None
+        while time.time() < deadline:
+            s = pod_port_up()
+            if s:
+                break
+            time.sleep(0.05)
+        assert s is not None, "pod's own listener never detected"
+        # the unrelated process's port must not count as exposed
+        assert port_b not in s.port_mappings, \
+            "host-wide socket leaked into the pod's port mappings"
+        rt.terminate(st.id)
+        wait_status(rt, st.id, PodStatus.TERMINATED)
+    finally:
+        other.close()
