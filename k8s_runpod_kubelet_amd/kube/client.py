@@ -32,6 +32,15 @@ class ConflictError(ApiError):
         super().__init__(409, message)
 
 
+class GoneError(ApiError):
+    """HTTP 410 — the requested resourceVersion has been compacted away
+    (etcd compaction). Watchers must relist and re-watch from the fresh
+    list's resourceVersion."""
+
+    def __init__(self, message: str = "resourceVersion too old"):
+        super().__init__(410, message)
+
+
 def is_not_found(exc: BaseException) -> bool:
     return isinstance(exc, ApiError) and exc.status_code == 404
 
@@ -44,6 +53,16 @@ class K8sClient(ABC):
     @abstractmethod
     def list_pods(self, namespace: Optional[str] = None,
                   field_selector: str = "", label_selector: str = "") -> List[Dict[str, Any]]: ...
+
+    def list_pods_with_rv(self, namespace: Optional[str] = None,
+                          field_selector: str = "", label_selector: str = ""
+                          ) -> Tuple[List[Dict[str, Any]], str]:
+        """List plus the PodList's resourceVersion — what a correct informer
+        threads into its first watch (list→watch continuity). Default falls
+        back to a plain list with no RV (watch then starts from 'now')."""
+        return (self.list_pods(namespace=namespace,
+                               field_selector=field_selector,
+                               label_selector=label_selector), "")
 
     @abstractmethod
     def get_pod(self, namespace: str, name: str) -> Dict[str, Any]: ...

@@ -16,7 +16,7 @@ import time
 import uuid
 from typing import Any, Dict, Iterator, List, Optional
 
-from .client import ConflictError, K8sClient, NotFoundError, WatchEvent
+from .client import ConflictError, GoneError, K8sClient, NotFoundError, WatchEvent
 from .objects import (
     deletion_timestamp,
     meta,
@@ -47,15 +47,45 @@ def _match_field_selector(pod: Dict[str, Any], selector: str) -> bool:
 
 
 class _Store:
-    """One resource kind's objects + watchers."""
+    """One resource kind's objects + watchers + bounded event history.
+
+    The history makes watches resumable from a resourceVersion (like the
+    apiserver's watch cache): a watch with ``resourceVersion=N`` replays
+    retained events with rv>N, and an rv older than the retained window is
+    a 410 Gone — so informer RV-continuity/410 handling is actually
+    exercised by the fake, not just code-reviewed."""
+
+    HISTORY_LIMIT = 2048
 
     def __init__(self):
         self.objects: Dict[str, Dict[str, Any]] = {}  # "ns/name" or "name"
         self.watchers: List[queue.Queue] = []
+        self.history: List[tuple] = []  # (rv:int, ev_type, obj)
+        # highest rv ever dropped from history: a watch resuming from
+        # rv < trimmed_max_rv has missed unreplayable events -> 410 Gone
+        self.trimmed_max_rv = 0
 
     def emit(self, ev_type: str, obj: Dict[str, Any]) -> None:
+        try:
+            rv = int(meta(obj).get("resourceVersion", 0))
+        except (TypeError, ValueError):
+            rv = 0
+        self.history.append((rv, ev_type, copy.deepcopy(obj)))
+        if len(self.history) > self.HISTORY_LIMIT:
+            drop = len(self.history) - self.HISTORY_LIMIT
+            self.trimmed_max_rv = max(self.trimmed_max_rv,
+                                      self.history[drop - 1][0])
+            del self.history[:drop]
         for q in list(self.watchers):
             q.put((ev_type, copy.deepcopy(obj)))
+
+    def compact(self, below_rv: int) -> None:
+        """Drop history older than below_rv (etcd-compaction analogue —
+        watches resuming from an older rv get 410 Gone)."""
+        dropped = [h[0] for h in self.history if h[0] < below_rv]
+        if dropped:
+            self.trimmed_max_rv = max(self.trimmed_max_rv, max(dropped))
+        self.history = [h for h in self.history if h[0] >= below_rv]
 
 
 class FakeKube(K8sClient):
@@ -188,13 +218,47 @@ class FakeKube(K8sClient):
                 meta(pod)["resourceVersion"] = self._next_rv()
                 self.pods.emit("MODIFIED", pod)
 
+    def list_pods_with_rv(self, namespace=None, field_selector="",
+                          label_selector=""):
+        with self._lock:
+            return (self.list_pods(namespace=namespace,
+                                   field_selector=field_selector,
+                                   label_selector=label_selector),
+                    str(self._rv))
+
+    def compact_watch_history(self, below_rv: Optional[int] = None) -> None:
+        """Test hook: emulate etcd compaction — watches resuming from an rv
+        older than the retained window will get 410 Gone."""
+        with self._lock:
+            self.pods.compact(self._rv + 1 if below_rv is None else below_rv)
+
     def watch_pods(self, namespace=None, field_selector="", resource_version="",
                    timeout_s: float = 60.0) -> Iterator[WatchEvent]:
         q: queue.Queue = queue.Queue()
+        replay: List[WatchEvent] = []
         with self._lock:
+            if resource_version:
+                try:
+                    from_rv = int(resource_version)
+                except ValueError:
+                    raise GoneError(f"bad resourceVersion {resource_version!r}")
+                if from_rv < self.pods.trimmed_max_rv:
+                    # pod events after from_rv were dropped from the
+                    # retained window: unreplayable -> 410 Gone
+                    raise GoneError(
+                        f"resourceVersion {from_rv} compacted "
+                        f"(trimmed through {self.pods.trimmed_max_rv})")
+                replay = [(t, copy.deepcopy(o))
+                          for rv, t, o in self.pods.history if rv > from_rv]
             self.pods.watchers.append(q)
         deadline = time.time() + timeout_s
         try:
+            for ev_type, obj in replay:
+                if namespace and namespace_of(obj) != namespace:
+                    continue
+                if not _match_field_selector(obj, field_selector):
+                    continue
+                yield ev_type, obj
             while True:
                 remaining = deadline - time.time()
                 if remaining <= 0:

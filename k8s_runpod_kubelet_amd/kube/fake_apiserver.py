@@ -25,7 +25,7 @@ from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 from typing import Any, Dict, Optional, Tuple
 from urllib.parse import parse_qs, urlparse
 
-from .client import ApiError, ConflictError, NotFoundError
+from .client import ApiError, ConflictError, GoneError, NotFoundError
 from .fake import FakeKube
 
 log = logging.getLogger("kube.fake_apiserver")
@@ -112,12 +112,15 @@ class FakeApiServer:
                         namespace = ns_m.group(1) if ns_m else None
                         if params.get("watch") == "true":
                             return self._stream_watch(namespace, params)
-                        items = kube.list_pods(
+                        items, rv = kube.list_pods_with_rv(
                             namespace=namespace,
                             field_selector=params.get("fieldSelector", ""),
                             label_selector=params.get("labelSelector", ""),
                         )
-                        return self._json(200, {"kind": "PodList", "items": items})
+                        return self._json(200, {
+                            "kind": "PodList",
+                            "metadata": {"resourceVersion": rv},
+                            "items": items})
                     if m := _POD_RE.match(path):
                         return self._json(200, kube.get_pod(m.group(1), m.group(2)))
                     if m := _NODE_RE.match(path):
@@ -152,13 +155,26 @@ class FakeApiServer:
                     self.wfile.flush()
 
                 try:
-                    for ev_type, obj in kube.watch_pods(
-                        namespace=namespace,
-                        field_selector=params.get("fieldSelector", ""),
-                        timeout_s=timeout_s,
-                    ):
-                        chunk(json.dumps(
-                            {"type": ev_type, "object": obj}).encode() + b"\n")
+                    try:
+                        for ev_type, obj in kube.watch_pods(
+                            namespace=namespace,
+                            field_selector=params.get("fieldSelector", ""),
+                            resource_version=params.get(
+                                "resourceVersion", ""),
+                            timeout_s=timeout_s,
+                        ):
+                            chunk(json.dumps(
+                                {"type": ev_type,
+                                 "object": obj}).encode() + b"\n")
+                    except GoneError as exc:
+                        # compacted RV: in-stream Status(410), the same
+                        # shape a real apiserver emits mid-watch
+                        chunk(json.dumps({
+                            "type": "ERROR",
+                            "object": {"kind": "Status", "code": 410,
+                                       "reason": "Expired",
+                                       "message": str(exc)},
+                        }).encode() + b"\n")
                     chunk(b"")  # terminating chunk
                 except (BrokenPipeError, ConnectionResetError):
                     pass  # client hung up — normal for watches

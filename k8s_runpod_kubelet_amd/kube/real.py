@@ -22,7 +22,8 @@ import httpx
 import yaml
 
 from ..utils.backoff import retry
-from .client import ApiError, ConflictError, K8sClient, NotFoundError, WatchEvent
+from .client import (ApiError, ConflictError, GoneError, K8sClient,
+                     NotFoundError, WatchEvent)
 
 log = logging.getLogger("kube.real")
 
@@ -147,6 +148,8 @@ class HttpK8sClient(K8sClient):
             raise NotFoundError(resp.text[:200])
         if resp.status_code == 409:
             raise ConflictError(resp.text[:200])
+        if resp.status_code == 410:
+            raise GoneError(resp.text[:200])
         if resp.status_code >= 400:
             raise ApiError(resp.status_code, resp.text[:500])
         if resp.status_code == 204 or not resp.content:
@@ -178,6 +181,18 @@ class HttpK8sClient(K8sClient):
         if label_selector:
             params["labelSelector"] = label_selector
         return self._request("GET", path, params=params).get("items", [])
+
+    def list_pods_with_rv(self, namespace=None, field_selector="",
+                          label_selector=""):
+        path = f"/api/v1/namespaces/{namespace}/pods" if namespace else "/api/v1/pods"
+        params = {}
+        if field_selector:
+            params["fieldSelector"] = field_selector
+        if label_selector:
+            params["labelSelector"] = label_selector
+        body = self._request("GET", path, params=params)
+        return (body.get("items", []),
+                body.get("metadata", {}).get("resourceVersion", ""))
 
     def get_pod(self, namespace: str, name: str) -> Dict[str, Any]:
         return self._request("GET", f"/api/v1/namespaces/{namespace}/pods/{name}")
@@ -213,7 +228,8 @@ class HttpK8sClient(K8sClient):
     def watch_pods(self, namespace=None, field_selector="", resource_version="",
                    timeout_s: float = 60.0) -> Iterator[WatchEvent]:
         path = f"/api/v1/namespaces/{namespace}/pods" if namespace else "/api/v1/pods"
-        params = {"watch": "true", "timeoutSeconds": str(int(timeout_s))}
+        params = {"watch": "true", "timeoutSeconds": str(int(timeout_s)),
+                  "allowWatchBookmarks": "true"}
         if field_selector:
             params["fieldSelector"] = field_selector
         if resource_version:
@@ -232,8 +248,15 @@ class HttpK8sClient(K8sClient):
                     continue
                 ev_type = event.get("type", "")
                 obj = event.get("object", {})
-                if ev_type in ("ADDED", "MODIFIED", "DELETED"):
+                if ev_type in ("ADDED", "MODIFIED", "DELETED", "BOOKMARK"):
                     yield ev_type, obj
+                elif ev_type == "ERROR":
+                    # apiserver reports expired RVs as an in-stream Status
+                    # with code 410 (Gone) — the watcher must relist
+                    if obj.get("code") == 410:
+                        raise GoneError(obj.get("message", ""))
+                    raise ApiError(int(obj.get("code", 500)),
+                                   obj.get("message", "watch error"))
 
     # ---- nodes ----
 

@@ -171,3 +171,32 @@ def test_apiserver_survives_bad_requests(http_stack):
     assert wait_until(
         lambda: (_pod_via_http(client, "afterabuse") or {}).get(
             "status", {}).get("phase") == "Succeeded", timeout_s=15)
+
+
+def test_http_watch_rv_continuity_and_410():
+    """Production client path for informer correctness: list_pods_with_rv
+    returns the PodList RV; a watch from that RV replays only later events;
+    a compacted RV surfaces as GoneError (in-stream ERROR Status 410)."""
+    from k8s_runpod_kubelet_amd.kube.client import GoneError
+
+    srv = FakeApiServer().start()
+    client = HttpK8sClient(ClusterConfig(server=srv.url))
+    try:
+        client.create_pod("default", make_pod("a"))
+        items, rv = client.list_pods_with_rv()
+        assert [p["metadata"]["name"] for p in items] == ["a"]
+        assert rv and int(rv) > 0
+        client.create_pod("default", make_pod("b"))
+        events = [(t, p["metadata"]["name"])
+                  for t, p in client.watch_pods(resource_version=rv,
+                                                timeout_s=1)]
+        assert ("ADDED", "b") in events
+        assert all(name != "a" for _, name in events)
+
+        srv.kube.compact_watch_history()
+        with pytest.raises(GoneError):
+            for _ in client.watch_pods(resource_version=rv, timeout_s=1):
+                pass
+    finally:
+        client.close()
+        srv.stop()

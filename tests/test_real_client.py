@@ -105,17 +105,41 @@ def test_delete_pod_grace():
 def test_watch_stream_parsing():
     lines = [
         json.dumps({"type": "ADDED", "object": {"metadata": {"name": "a"}}}),
-        json.dumps({"type": "BOOKMARK", "object": {}}),  # ignored
+        # BOOKMARK is yielded so the informer can thread its RV forward
+        json.dumps({"type": "BOOKMARK",
+                    "object": {"metadata": {"resourceVersion": "7"}}}),
         json.dumps({"type": "MODIFIED", "object": {"metadata": {"name": "a"}}}),
     ]
 
     def handler(request):
         assert request.url.params["watch"] == "true"
+        assert request.url.params["allowWatchBookmarks"] == "true"
         return httpx.Response(200, text="\n".join(lines) + "\n")
 
     c = make_client(handler)
     events = list(c.watch_pods(field_selector="spec.nodeName=n1", timeout_s=5))
-    assert [t for t, _ in events] == ["ADDED", "MODIFIED"]
+    assert [t for t, _ in events] == ["ADDED", "BOOKMARK", "MODIFIED"]
+
+
+def test_watch_stream_error_410_raises_gone():
+    from k8s_runpod_kubelet_amd.kube.client import GoneError
+
+    lines = [
+        json.dumps({"type": "ADDED", "object": {"metadata": {"name": "a"}}}),
+        json.dumps({"type": "ERROR",
+                    "object": {"kind": "Status", "code": 410,
+                               "message": "too old resource version"}}),
+    ]
+
+    def handler(request):
+        return httpx.Response(200, text="\n".join(lines) + "\n")
+
+    c = make_client(handler)
+    got = []
+    with pytest.raises(GoneError):
+        for t, o in c.watch_pods(timeout_s=5):
+            got.append(t)
+    assert got == ["ADDED"]  # events before the error still delivered
 
 
 def test_retry_on_transport_error():
