@@ -1,0 +1,198 @@
+"""OCI image store: layout build/resolve, layer unpacking with whiteouts,
+user resolution, ref normalization, archive import, traversal safety.
+Reference contract: Containers[0].Image actually runs
+(runpod_client.go:1304) — these are the store-layer pieces."""
+
+import io
+import json
+import os
+import tarfile
+from pathlib import Path
+
+import pytest
+
+from k8s_runpod_kubelet_amd.runtime.oci import (
+    ImageError,
+    ImageStore,
+    build_layout,
+    normalize_ref,
+)
+
+
+def test_normalize_ref():
+    assert normalize_ref("busybox") == "docker.io/library/busybox:latest"
+    assert normalize_ref("busybox:1.36") == "docker.io/library/busybox:1.36"
+    assert normalize_ref("rocm/pytorch") == "docker.io/rocm/pytorch:latest"
+    assert (normalize_ref("ghcr.io/org/app:v1")
+            == "ghcr.io/org/app:v1")
+    assert (normalize_ref("localhost:5000/x")
+            == "localhost:5000/x:latest")
+    assert normalize_ref("docker.io/busybox") == \
+        "docker.io/library/busybox:latest"
+
+
+def make_tree(base: Path, files: dict) -> None:
+    for rel, content in files.items():
+        p = base / rel
+        p.parent.mkdir(parents=True, exist_ok=True)
+        if content is None:
+            p.mkdir(exist_ok=True)
+        elif isinstance(content, tuple) and content[0] == "symlink":
+            p.symlink_to(content[1])
+        else:
+            p.write_text(content)
+
+
+@pytest.fixture
+def store(tmp_path):
+    return ImageStore(str(tmp_path / "store"))
+
+
+def build_and_add(store, tmp_path, name, ref, files, **kw):
+    tree = tmp_path / f"tree-{name}"
+    tree.mkdir()
+    make_tree(tree, files)
+    layout = tmp_path / f"layout-{name}"
+    layout.mkdir()
+    build_layout(str(layout), ref, str(tree), **kw)
+    store.add_layout(str(layout), ref)
+    return tree
+
+
+def test_build_resolve_unpack(store, tmp_path):
+    build_and_add(
+        store, tmp_path, "a", "example/app:v1",
+        {"bin/app": "#!/bin/true\n", "etc/conf": "x=1\n",
+         "lib/liby.so": "ELF", "link": ("symlink", "etc/conf")},
+        entrypoint=["/bin/app"], cmd=["--serve"],
+        env=["PATH=/bin", "MODE=prod"], working_dir="/etc", user="0:0")
+    img = store.resolve("example/app:v1")
+    assert img is not None
+    assert img.config.entrypoint == ["/bin/app"]
+    assert img.config.cmd == ["--serve"]
+    assert "MODE=prod" in img.config.env
+    assert img.config.working_dir == "/etc"
+    assert img.config.path_env == "/bin"
+    rootfs = store.rootfs_for(img)
+    assert (rootfs / "bin/app").read_text() == "#!/bin/true\n"
+    assert (rootfs / "link").is_symlink()
+    # cache: second call returns same dir without re-unpack
+    assert store.rootfs_for(img) == rootfs
+    # unknown image -> None (caller's fallback policy decides)
+    assert store.resolve("example/missing:v9") is None
+    assert "docker.io/example/app:v1" in store.list_refs()
+
+
+def test_multi_layer_whiteouts(store, tmp_path):
+    base = tmp_path / "l0"
+    upper = tmp_path / "l1"
+    make_tree(base, {"a.txt": "base", "drop.txt": "gone",
+                     "dir/keep": "k", "wipe/x": "1", "wipe/y": "2"})
+    make_tree(upper, {"a.txt": "upper", "new.txt": "n"})
+    layout = tmp_path / "layout-w"
+    layout.mkdir()
+    build_layout(str(layout), "example/wh:v1", str(base),
+                 extra_layer_dirs=[str(upper)],
+                 whiteouts=["drop.txt", "wipe"])
+    store.add_layout(str(layout), "example/wh:v1")
+    img = store.resolve("example/wh:v1")
+    rootfs = store.rootfs_for(img)
+    assert (rootfs / "a.txt").read_text() == "upper"  # upper layer wins
+    assert (rootfs / "new.txt").exists()
+    assert (rootfs / "dir/keep").exists()
+    assert not (rootfs / "drop.txt").exists()  # .wh. file delete
+    assert not (rootfs / "wipe").exists()      # .wh. directory delete
+
+
+def test_layer_traversal_rejected(store, tmp_path):
+    layout = tmp_path / "layout-evil"
+    blobs = layout / "blobs" / "sha256"
+    blobs.mkdir(parents=True)
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w") as tf:
+        info = tarfile.TarInfo("../../escape.txt")
+        data = b"pwn"
+        info.size = len(data)
+        tf.addfile(info, io.BytesIO(data))
+    import hashlib
+
+    raw = buf.getvalue()
+    ldig = hashlib.sha256(raw).hexdigest()
+    (blobs / ldig).write_bytes(raw)
+    cfg = json.dumps({"architecture": "amd64", "os": "linux",
+                      "config": {}}).encode()
+    cdig = hashlib.sha256(cfg).hexdigest()
+    (blobs / cdig).write_bytes(cfg)
+    man = json.dumps({
+        "schemaVersion": 2,
+        "mediaType": "application/vnd.oci.image.manifest.v1+json",
+        "config": {"mediaType": "application/vnd.oci.image.config.v1+json",
+                   "digest": f"sha256:{cdig}", "size": len(cfg)},
+        "layers": [{"mediaType": "application/vnd.oci.image.layer.v1.tar",
+                    "digest": f"sha256:{ldig}", "size": len(raw)}],
+    }).encode()
+    mdig = hashlib.sha256(man).hexdigest()
+    (blobs / mdig).write_bytes(man)
+    (layout / "index.json").write_text(json.dumps({
+        "schemaVersion": 2,
+        "manifests": [{"mediaType":
+                       "application/vnd.oci.image.manifest.v1+json",
+                       "digest": f"sha256:{mdig}", "size": len(man)}]}))
+    store.add_layout(str(layout), "example/evil:v1")
+    img = store.resolve("example/evil:v1")
+    with pytest.raises(ImageError):
+        store.rootfs_for(img)
+
+
+def test_import_archive(store, tmp_path):
+    tree = tmp_path / "tree-arc"
+    make_tree(tree, {"hello.txt": "hi"})
+    layout = tmp_path / "layout-arc"
+    layout.mkdir()
+    build_layout(str(layout), "example/arc:v2", str(tree))
+    tar_path = tmp_path / "img.tar"
+    with tarfile.open(tar_path, "w") as tf:
+        for p in sorted(layout.rglob("*")):
+            tf.add(p, arcname=str(p.relative_to(layout)), recursive=False)
+    ref = store.import_archive(str(tar_path))  # ref from annotation
+    assert ref == "docker.io/example/arc:v2"
+    img = store.resolve("example/arc:v2")
+    rootfs = store.rootfs_for(img)
+    assert (rootfs / "hello.txt").read_text() == "hi"
+
+
+def test_resolve_user(store, tmp_path):
+    tree = build_and_add(
+        store, tmp_path, "u", "example/users:v1",
+        {"etc/passwd": "root:x:0:0:root:/root:/bin/sh\n"
+                       "app:x:1001:2002:app:/home/app:/bin/sh\n",
+         "etc/group": "root:x:0:\nappgrp:x:3003:\n"})
+    img = store.resolve("example/users:v1")
+    rootfs = store.rootfs_for(img)
+    assert store.resolve_user(rootfs, "") == (-1, -1)
+    assert store.resolve_user(rootfs, "1000") == (1000, 1000)
+    assert store.resolve_user(rootfs, "1000:1001") == (1000, 1001)
+    assert store.resolve_user(rootfs, "app") == (1001, 2002)
+    assert store.resolve_user(rootfs, "app:appgrp") == (1001, 3003)
+    with pytest.raises(ImageError):
+        store.resolve_user(rootfs, "ghost")
+    del tree
+
+
+def test_changed_type_replaces(store, tmp_path):
+    """A file in an upper layer replacing a lower-layer directory (and
+    vice versa) must apply cleanly."""
+    base = tmp_path / "t0"
+    upper = tmp_path / "t1"
+    make_tree(base, {"thing/inner.txt": "dir-content", "plain": "file"})
+    make_tree(upper, {"thing": "now-a-file", "plain/sub": "now-a-dir"})
+    layout = tmp_path / "layout-t"
+    layout.mkdir()
+    build_layout(str(layout), "example/types:v1", str(base),
+                 extra_layer_dirs=[str(upper)])
+    store.add_layout(str(layout), "example/types:v1")
+    rootfs = store.rootfs_for(store.resolve("example/types:v1"))
+    assert (rootfs / "thing").is_file()
+    assert (rootfs / "thing").read_text() == "now-a-file"
+    assert (rootfs / "plain").is_dir()
+    assert (rootfs / "plain/sub").read_text() == "now-a-dir"
