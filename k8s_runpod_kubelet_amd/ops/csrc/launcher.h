@@ -8,6 +8,19 @@
 
 namespace amdvk {
 
+// One mount performed by the child inside its own mount namespace, in
+// order, before pivot_root. dst is the absolute (rootfs-prefixed) target;
+// all target files/dirs are pre-created by the caller — the child only
+// issues mount(2) (it runs with CLONE_VM and may not allocate).
+struct MountSpec {
+  std::string src;
+  std::string dst;
+  std::string fstype;  // "" = bind mount (caller sets MS_BIND in flags)
+  std::string data;    // e.g. overlay "lowerdir=..,upperdir=..,workdir=.."
+  unsigned long flags = 0;
+  bool readonly = false;  // bind remount MS_RDONLY after mounting
+};
+
 struct LaunchSpec {
   std::vector<std::string> argv;
   std::vector<std::string> env;  // "KEY=VALUE" entries (full environment)
@@ -28,6 +41,15 @@ struct LaunchSpec {
   // namespace with that hostname (k8s pod-hostname semantics).
   bool new_pid_ns = false;
   std::string hostname;
+  // Container-image execution (OCI rootfs). Non-empty rootfs + mount mode:
+  // child gets CLONE_NEWNS, performs `mounts` in order, pivot_roots into
+  // rootfs. chroot_only is the degraded mode for hosts without
+  // CAP_SYS_ADMIN-in-sandbox: plain chroot(rootfs) (caller prepared a
+  // fully-populated private rootfs copy; no mounts possible). A failed
+  // rootfs setup NEVER execs on the host filesystem.
+  std::string rootfs;
+  std::vector<MountSpec> mounts;
+  bool chroot_only = false;
 };
 
 struct LaunchResult {
@@ -45,6 +67,11 @@ struct LaunchResult {
 };
 
 LaunchResult LaunchProcess(const LaunchSpec& spec);
+
+// Capability probe: can this process create a mount namespace and mount in
+// it? (The gpurun sandbox drops CAP_SYS_ADMIN.) Forks a throwaway child —
+// call once and cache.
+bool ProbeMountNamespace();
 
 // pidfd for an already-running process (adoption-on-restart path).
 int OpenPidfd(int64_t pid);
