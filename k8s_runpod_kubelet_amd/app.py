@@ -109,6 +109,7 @@ def build_stack(
 
             runtime = FakeRuntime(gpu_count=ledger.total_gpus())
         else:
+            from .runtime.oci import ImageStore
             from .runtime.process_runtime import ProcessRuntime
 
             runtime = ProcessRuntime(
@@ -117,6 +118,10 @@ def build_stack(
                 cgroup_root=config.cgroup_root,
                 cgroup_parent=config.cgroup_parent,
                 pod_namespaces=config.pod_namespaces,
+                image_store=ImageStore(config.resolved_image_store_dir()),
+                image_isolation=config.image_isolation,
+                image_gpu_binds=config.image_gpu_binds,
+                image_extra_binds=config.image_extra_binds,
             )
 
     provider = Provider(client, config, runtime, ledger=ledger, inventory=inventory)

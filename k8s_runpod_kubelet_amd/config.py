@@ -82,6 +82,16 @@ class Config:
     pod_namespaces: bool = True
     runtime: str = "process"  # process | fake
     pod_log_dir: str = ""  # defaults to <state_dir>/logs
+    # OCI image execution: pods whose image is present in the local store
+    # run inside it (mount-ns overlay + pivot_root, or chroot fallback).
+    # The store holds OCI image layouts/archives imported via
+    # `python -m k8s_runpod_kubelet_amd.runtime.imagetool`.
+    image_store_dir: str = ""       # defaults to <state_dir>/images
+    image_isolation: str = "auto"   # auto | mountns | chroot
+    # host paths bound read-only into GPU image pods (driver userspace —
+    # the thin-image + host-ROCm pattern); "src[:dst[:ro|rw]]"
+    image_gpu_binds: List[str] = field(default_factory=lambda: ["/opt/rocm"])
+    image_extra_binds: List[str] = field(default_factory=list)
     pod_controller_workers: int = 4  # reference uses 1 (main.go:263)
 
     # GPU inventory overrides (mostly for tests / CPU-only dev)
@@ -90,6 +100,9 @@ class Config:
 
     def resolved_pod_log_dir(self) -> str:
         return self.pod_log_dir or os.path.join(self.state_dir, "logs")
+
+    def resolved_image_store_dir(self) -> str:
+        return self.image_store_dir or os.path.join(self.state_dir, "images")
 
 
 def load_config(path: Optional[str]) -> Config:

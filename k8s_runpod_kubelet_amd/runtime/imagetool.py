@@ -1,0 +1,76 @@
+"""Operator CLI for the node-local OCI image store (no registry on this
+node — images arrive as OCI layouts / oci-archive tars, e.g. produced by
+``skopeo copy docker://rocm/pytorch oci-archive:img.tar``).
+
+    python -m k8s_runpod_kubelet_amd.runtime.imagetool list
+    python -m k8s_runpod_kubelet_amd.runtime.imagetool import img.tar [--ref R]
+    python -m k8s_runpod_kubelet_amd.runtime.imagetool add-layout DIR --ref R
+    python -m k8s_runpod_kubelet_amd.runtime.imagetool build DIR --ref R \
+        [--entrypoint CMD...] [--env K=V...]
+"""
+
+from __future__ import annotations
+
+import argparse
+import sys
+import tempfile
+
+from ..config import Config
+from .oci import ImageStore, build_layout
+
+
+def main(argv=None) -> int:
+    parser = argparse.ArgumentParser(prog="amdvk-image")
+    parser.add_argument("--store", default="",
+                        help="image store dir (default: config state_dir/images)")
+    sub = parser.add_subparsers(dest="cmd", required=True)
+
+    sub.add_parser("list", help="list stored image references")
+
+    p_imp = sub.add_parser("import", help="import an oci-archive tar")
+    p_imp.add_argument("tar")
+    p_imp.add_argument("--ref", default="",
+                       help="reference (default: from the archive annotation)")
+
+    p_add = sub.add_parser("add-layout", help="register an OCI layout dir")
+    p_add.add_argument("dir")
+    p_add.add_argument("--ref", required=True)
+
+    p_b = sub.add_parser("build", help="build an image from a rootfs dir")
+    p_b.add_argument("rootfs")
+    p_b.add_argument("--ref", required=True)
+    p_b.add_argument("--entrypoint", nargs="+", default=[])
+    p_b.add_argument("--cmd", nargs="+", default=[])
+    p_b.add_argument("--env", nargs="*", default=[])
+    p_b.add_argument("--workdir", default="")
+    p_b.add_argument("--user", default="")
+
+    args = parser.parse_args(argv)
+    store = ImageStore(args.store or Config().resolved_image_store_dir())
+
+    if args.cmd == "list":
+        for ref in store.list_refs():
+            print(ref)
+        return 0
+    if args.cmd == "import":
+        ref = store.import_archive(args.tar, args.ref)
+        print(f"imported {ref}")
+        return 0
+    if args.cmd == "add-layout":
+        ref = store.add_layout(args.dir, args.ref)
+        print(f"added {ref}")
+        return 0
+    if args.cmd == "build":
+        with tempfile.TemporaryDirectory(prefix="amdvk-build-") as td:
+            build_layout(td, args.ref, args.rootfs,
+                         entrypoint=args.entrypoint, cmd=args.cmd,
+                         env=args.env or None, working_dir=args.workdir,
+                         user=args.user)
+            ref = store.add_layout(td, args.ref)
+        print(f"built {ref}")
+        return 0
+    return 2
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -66,6 +66,10 @@ class ProcessRuntime(Runtime):
         podworker: Optional[str] = None,
         enable_cgroups: bool = True,
         pod_namespaces: bool = True,
+        image_store=None,
+        image_isolation: str = "auto",
+        image_gpu_binds: Optional[List[str]] = None,
+        image_extra_binds: Optional[List[str]] = None,
     ):
         from ..ops import load_native
 
@@ -84,6 +88,21 @@ class ProcessRuntime(Runtime):
         # Degrades automatically when CAP_SYS_ADMIN is absent.
         self.pod_namespaces = pod_namespaces
         self._podworker = podworker
+        # OCI-image execution (reference contract: Containers[0].Image
+        # actually runs, runpod_client.go:1304). image_store=None keeps the
+        # legacy host-binary behavior everywhere.
+        self.image_store = image_store
+        self._rootfs_mgr = None
+        if image_store is not None:
+            from .rootfs import RootfsManager
+
+            self._rootfs_mgr = RootfsManager(
+                image_store, str(self.state_dir / "containers"),
+                self._native,
+                gpu_binds=image_gpu_binds,
+                extra_binds=image_extra_binds,
+                isolation=image_isolation,
+            )
 
         self._lock = threading.RLock()
         self._cgroup_pool: List[str] = []
@@ -207,6 +226,27 @@ class ProcessRuntime(Runtime):
             log.debug("device filter unavailable (unprivileged?)",
                       extra={"pod": inst.pod_key})
 
+    def _resolve_image(self, cspec):
+        """ImageStore resolution for a container spec; None keeps the
+        legacy host-binary path (reserved amdvk/ refs, absent store, or an
+        image not present in the local store)."""
+        if (self._rootfs_mgr is None or not cspec.image
+                or cspec.image.startswith("amdvk/")):
+            return None
+        return self.image_store.resolve(cspec.image)
+
+    def _gpu_device_paths(self, inst: Instance) -> List[str]:
+        """Host device nodes an image pod needs for its bound GPUs."""
+        if not inst.gpu_indices or not os.path.exists("/dev/kfd"):
+            return []
+        paths = ["/dev/kfd"]
+        inv = self.binder.ledger.inventory
+        for idx in inst.gpu_indices:
+            gpu = inv.get(idx)
+            if gpu is not None and gpu.render_minor >= 0:
+                paths.append(f"/dev/dri/renderD{gpu.render_minor}")
+        return [p for p in paths if os.path.exists(p)]
+
     def _launch_one(self, inst: Instance, cspec, into: List) -> float:
         """Spawn one container of the pod into `into` (inst.containers or
         inst.init_containers); returns the native posix_spawnp/clone3 time in
@@ -216,7 +256,16 @@ class ProcessRuntime(Runtime):
         from ..server import metrics
 
         params = inst.params
-        base_env = dict(os.environ)
+        image = self._resolve_image(cspec)
+        if image is not None:
+            # container env starts from the IMAGE config, not the kubelet's
+            # environment — host env must not leak into containers
+            base_env: Dict[str, str] = {}
+            for e in image.config.env:
+                k, _, v = e.partition("=")
+                base_env[k] = v
+        else:
+            base_env = dict(os.environ)
         # Drop our own GPU scoping so the pod's binding is authoritative.
         base_env.pop("ROCR_VISIBLE_DEVICES", None)
         base_env.pop("HIP_VISIBLE_DEVICES", None)
@@ -226,7 +275,34 @@ class ProcessRuntime(Runtime):
         base_env["AMDVK_POD_KEY"] = params.pod_key
 
         argv = list(cspec.command) + list(cspec.args)
-        if not argv:
+        uid, gid = cspec.run_as_uid, cspec.run_as_gid
+        working_dir = cspec.working_dir
+        rootfs, chroot_only, mounts = "", False, []
+        if image is not None:
+            # k8s image semantics: command overrides Entrypoint; args
+            # override Cmd; with neither, Entrypoint+Cmd run.
+            if not cspec.command:
+                argv = list(image.config.entrypoint) + list(
+                    cspec.args or image.config.cmd)
+            if not argv:
+                raise RuntimeError(
+                    f"container {cspec.name}: no command and image "
+                    f"{cspec.image} defines no entrypoint/cmd")
+            working_dir = cspec.working_dir or image.config.working_dir
+            if uid < 0 and gid < 0 and image.config.user:
+                uid, gid = self.image_store.resolve_user(
+                    self.image_store.rootfs_for(image), image.config.user)
+            prepared = self._rootfs_mgr.prepare(
+                inst.id, cspec.name, image,
+                params.hostname or params.name,
+                gpu_device_paths=self._gpu_device_paths(inst),
+                working_dir=working_dir)
+            argv[0] = self._rootfs_mgr.resolve_argv0(image, prepared,
+                                                     argv[0])
+            rootfs = prepared.rootfs
+            chroot_only = prepared.chroot_only
+            mounts = prepared.mounts
+        elif not argv:
             argv = [self.podworker_path(), "--hold"]
             if inst.gpu_indices:
                 argv += ["--expect-gpus", str(len(inst.gpu_indices))]
@@ -234,32 +310,69 @@ class ProcessRuntime(Runtime):
                 argv += ["--listen-port", str(port)]
         elif argv[0] in ("podworker", "amdvk-podworker"):
             argv[0] = self.podworker_path()
-        needs_fast_path = (cspec.run_as_uid >= 0 or cspec.run_as_gid >= 0
-                           or self.pod_namespaces)
-        if needs_fast_path and "/" not in argv[0]:
-            # Credential dropping / namespaces happen on the execve fast
-            # path, which does no PATH search — resolve here instead.
-            resolved = shutil.which(argv[0])
-            if resolved:
-                argv[0] = resolved
+        if image is None:
+            needs_fast_path = (uid >= 0 or gid >= 0 or self.pod_namespaces)
+            if needs_fast_path and "/" not in argv[0]:
+                # Credential dropping / namespaces happen on the execve fast
+                # path, which does no PATH search — resolve here instead.
+                resolved = shutil.which(argv[0])
+                if resolved:
+                    argv[0] = resolved
 
         env = dict(base_env)
         env.update(cspec.env)
         envp = [f"{k}={v}" for k, v in env.items()]
         stdout_path = str(self.logs_dir / f"{inst.id}-{cspec.name}.log")
 
-        pid, pidfd, ready_fd, spawn_s, cgroup_s = self._native.launch_process(
-            argv, envp,
-            cspec.working_dir or "",
-            stdout_path, stdout_path,
-            inst.cgroup_dir, True, True,
-            cspec.run_as_uid, cspec.run_as_gid,
-            self.pod_namespaces,
-            (params.hostname or params.name) if self.pod_namespaces else "",
-        )
+        def do_launch():
+            return self._native.launch_process(
+                argv, envp,
+                working_dir or "",
+                stdout_path, stdout_path,
+                inst.cgroup_dir, True, True,
+                uid, gid,
+                self.pod_namespaces,
+                (params.hostname or params.name) if self.pod_namespaces
+                else "",
+                rootfs, chroot_only, mounts,
+            )
+
+        try:
+            pid, pidfd, ready_fd, spawn_s, cgroup_s = do_launch()
+        except RuntimeError as exc:
+            if image is None or "rootfs setup" not in str(exc) or chroot_only:
+                raise
+            # mountns mode failed on this host (overlayfs/caps): downgrade
+            # to the chroot mode once, then retry this launch with a
+            # freshly prepared copied rootfs.
+            log.warning("mountns image launch failed; retrying in chroot "
+                        "mode", extra={"err": str(exc)})
+            self._rootfs_mgr.downgrade_to_chroot()
+            prepared = self._rootfs_mgr.prepare(
+                inst.id, cspec.name, image,
+                params.hostname or params.name,
+                gpu_device_paths=self._gpu_device_paths(inst),
+                working_dir=working_dir)
+            argv[0] = self._rootfs_mgr.resolve_argv0(
+                image, prepared, (cspec.command or image.config.entrypoint
+                                  or argv)[0])
+            rootfs = prepared.rootfs
+            chroot_only = prepared.chroot_only
+            mounts = prepared.mounts
+            pid, pidfd, ready_fd, spawn_s, cgroup_s = do_launch()
         if inst.cgroup_dir:
             metrics.cgroup_migrate_seconds.observe(cgroup_s)
         cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
+        # k8s readiness semantics: a running container WITHOUT a
+        # readinessProbe is Ready as soon as it starts. The AMDVK_READY_FD
+        # pipe protocol (readiness deferred until the workload signals its
+        # GPU context is up) applies only to the in-tree podworker, which
+        # opts in by writing to the pipe — generic/image entrypoints must
+        # not sit NotReady until exit.
+        pipe_gated = bool(argv) and argv[0] == self.podworker_path()
+        if (into is not inst.init_containers and not pipe_gated
+                and cspec.readiness is None):
+            cinfo.ready = True
         into.append(cinfo)
         with self._lock:
             self._pid_to_instance[pid] = inst.id
@@ -473,7 +586,7 @@ class ProcessRuntime(Runtime):
                 cinfo.started_at = fresh.started_at
                 cinfo.finished_at = 0.0
                 cinfo.exit_code = None
-                cinfo.ready = False
+                cinfo.ready = fresh.ready  # started-is-ready when unprobed
                 cinfo.message = ""
                 cinfo.restart_count += 1
                 cinfo.backoff_until = 0
@@ -825,6 +938,8 @@ class ProcessRuntime(Runtime):
             self.binder.unbind(inst.pod_key)
             self._clear_probe_state(instance_id)
             (self.instances_dir / f"{instance_id}.json").unlink(missing_ok=True)
+            if self._rootfs_mgr is not None:
+                self._rootfs_mgr.cleanup(instance_id)
 
     def healthy(self) -> bool:
         return self.ledger.any_schedulable() or self.ledger.total_gpus() == 0
