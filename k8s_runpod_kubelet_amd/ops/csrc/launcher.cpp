@@ -153,6 +153,7 @@ struct ChildCtx {
   const ChildMount* mounts;
   size_t n_mounts;
   bool chroot_only;
+  int setns_fd;                // pidfd of the ns to join (-1 = off)
   const sigset_t* parent_mask;  // restored just before exec
   volatile int* exec_errno;     // shared (CLONE_VM): child reports failure
   volatile int* setup_errno;    // rootfs/mount setup failure (distinct so
@@ -188,6 +189,18 @@ int ChildMain(void* p) {
     }
   }
   if (c->ready_fd >= 0) syscall(SYS_fcntl, c->ready_fd, F_SETFD, 0);
+  // ---- nsenter-style exec into a live container (kubectl exec / exec
+  // probes): join the target's mount+UTS namespaces. After the log fds
+  // are open (host paths). Fail-closed like rootfs setup.
+  if (c->setns_fd >= 0) {
+    if (syscall(SYS_setns, c->setns_fd,
+                CLONE_NEWNS | CLONE_NEWUTS) != 0 &&
+        syscall(SYS_setns, c->setns_fd, CLONE_NEWNS) != 0) {
+      *c->setup_errno = errno ? errno : EPERM;
+      return 125;
+    }
+    syscall(SYS_chdir, "/");
+  }
   // ---- OCI rootfs setup (after the log fds are open on the HOST paths;
   // fds survive pivot_root). Any failure aborts: a pod that asked for a
   // container rootfs must never exec against the host filesystem.
@@ -375,10 +388,22 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
   // (posix_spawn has no mount/pivot_root hook either). Needs an
   // absolute/relative path in argv[0] — execve does no PATH search.
   if ((!spec.cgroup_dir.empty() || spec.uid >= 0 || spec.gid >= 0 ||
-       spec.new_pid_ns || !spec.hostname.empty() || !spec.rootfs.empty()) &&
+       spec.new_pid_ns || !spec.hostname.empty() || !spec.rootfs.empty() ||
+       spec.setns_pid >= 0) &&
       spec.argv[0].find('/') != std::string::npos) {
     volatile int exec_errno = 0;
     volatile int setup_errno = 0;
+    int setns_fd = -1;
+    if (spec.setns_pid >= 0) {
+      setns_fd = PidfdOpen(static_cast<pid_t>(spec.setns_pid));
+      if (setns_fd < 0) {
+        if (ready_pipe[0] >= 0) close(ready_pipe[0]);
+        if (ready_pipe[1] >= 0) close(ready_pipe[1]);
+        res.error = "setns target: " + std::to_string(spec.setns_pid) +
+                    " not alive";
+        return res;
+      }
+    }
     std::string pivot_old = spec.rootfs + "/.amdvk-oldroot";
     std::vector<ChildMount> cmounts;
     cmounts.reserve(spec.mounts.size());
@@ -404,11 +429,13 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
     ctx.mounts = cmounts.data();
     ctx.n_mounts = cmounts.size();
     ctx.chroot_only = spec.chroot_only;
+    ctx.setns_fd = setns_fd;
     ctx.exec_errno = &exec_errno;
     ctx.setup_errno = &setup_errno;
     auto t0 = std::chrono::steady_clock::now();
     int pidfd = -1;
     long rv = SpawnIntoCgroup(spec.cgroup_dir.c_str(), &ctx, &pidfd);
+    if (setns_fd >= 0) close(setns_fd);
     if (rv > 0) {
       res.spawn_ns = std::chrono::duration_cast<std::chrono::nanoseconds>(
                          std::chrono::steady_clock::now() - t0)
@@ -440,10 +467,11 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
     // rv == 0: fast path unavailable here — fall through to
     // posix_spawn + cgroup.procs migration. Never silently run a pod that
     // asked for dropped credentials (or a container rootfs) on the host.
-    if (spec.uid >= 0 || spec.gid >= 0 || !spec.rootfs.empty()) {
+    if (spec.uid >= 0 || spec.gid >= 0 || !spec.rootfs.empty() ||
+        spec.setns_pid >= 0) {
       if (ready_pipe[0] >= 0) close(ready_pipe[0]);
       if (ready_pipe[1] >= 0) close(ready_pipe[1]);
-      res.error = !spec.rootfs.empty()
+      res.error = (!spec.rootfs.empty() || spec.setns_pid >= 0)
                       ? "rootfs setup: clone3 spawn path unavailable"
                       : "runAsUser/runAsGroup requires the clone3 spawn "
                         "path, which is unavailable here";
@@ -451,10 +479,11 @@ LaunchResult LaunchProcess(const LaunchSpec& spec) {
     }
   }
 #endif
-  if (spec.uid >= 0 || spec.gid >= 0 || !spec.rootfs.empty()) {
+  if (spec.uid >= 0 || spec.gid >= 0 || !spec.rootfs.empty() ||
+      spec.setns_pid >= 0) {
     if (ready_pipe[0] >= 0) close(ready_pipe[0]);
     if (ready_pipe[1] >= 0) close(ready_pipe[1]);
-    res.error = !spec.rootfs.empty()
+    res.error = (!spec.rootfs.empty() || spec.setns_pid >= 0)
                     ? "rootfs execution requires an absolute path in argv[0]"
                     : "runAsUser/runAsGroup requires an absolute path in "
                       "argv[0] (no PATH search on the credential-dropping "

@@ -50,6 +50,10 @@ struct LaunchSpec {
   std::string rootfs;
   std::vector<MountSpec> mounts;
   bool chroot_only = false;
+  // kubectl-exec into a live image container: join the mount (+UTS)
+  // namespace of this pid before exec (nsenter semantics). Mutually
+  // exclusive with rootfs/mounts. -1 = off.
+  int64_t setns_pid = -1;
 };
 
 struct LaunchResult {

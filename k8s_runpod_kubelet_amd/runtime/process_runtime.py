@@ -360,6 +360,8 @@ class ProcessRuntime(Runtime):
             chroot_only = prepared.chroot_only
             mounts = prepared.mounts
             pid, pidfd, ready_fd, spawn_s, cgroup_s = do_launch()
+        if image is not None:
+            inst.image_mode = "chroot" if chroot_only else "mountns"
         if inst.cgroup_dir:
             metrics.cgroup_migrate_seconds.observe(cgroup_s)
         cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
@@ -718,15 +720,20 @@ class ProcessRuntime(Runtime):
                                                   self.binder.ledger.inventory))
                             env.update(cspec.env)
                         # confine the probe like the container itself:
-                        # same cgroup (device filter included) and the
-                        # container's runAsUser/runAsGroup credentials
+                        # same cgroup (device filter included), the
+                        # container's runAsUser/runAsGroup credentials,
+                        # and — for image pods — the container's rootfs
                         cgroup_dir = inst.cgroup_dir
                         uid, gid = cspec.run_as_uid, cspec.run_as_gid
+                        setns_pid, rootfs_p = self._container_entry(
+                            inst, cspec)
 
                         def exec_runner(command, penv, timeout_s,
-                                        _cg=cgroup_dir, _uid=uid, _gid=gid):
+                                        _cg=cgroup_dir, _uid=uid, _gid=gid,
+                                        _sp=setns_pid, _rf=rootfs_p):
                             return self._run_confined(
-                                command, penv, _cg, _uid, _gid, timeout_s)
+                                command, penv, _cg, _uid, _gid, timeout_s,
+                                setns_pid=_sp, rootfs=_rf)
                     ok = run_probe(spec, env or {}, exec_runner=exec_runner)
                     outcome = advance(st, spec, ok)
                     if kind == "readiness" and outcome is not None:
@@ -1024,23 +1031,33 @@ class ProcessRuntime(Runtime):
 
     def _run_confined(self, command: List[str], env: Dict[str, str],
                       cgroup_dir: str, uid: int, gid: int,
-                      timeout_s: float, out_path: str = "") -> int:
+                      timeout_s: float, out_path: str = "",
+                      setns_pid: int = -1, rootfs: str = "") -> int:
         """Run one command to completion inside the pod's confinement
-        (cgroup + device filter + credentials) via the native launcher.
-        Returns the exit code (127 = spawn failure, 124 = timeout killed).
-        Used by kubectl-exec and by exec probes — probes must NOT run as
-        the kubelet (root) when the container has runAsUser."""
+        (cgroup + device filter + credentials — and for image pods, the
+        container's own rootfs: setns into the live container's mount ns,
+        or chroot into its rootfs copy) via the native launcher. Returns
+        the exit code (127 = spawn failure, 124 = timeout killed). Used by
+        kubectl-exec and by exec probes — k8s runs both INSIDE the
+        container, never as the kubelet (root) on the host."""
         argv = list(command)
         if "/" not in argv[0]:
-            # credential-drop/exec fast path does no PATH search
-            resolved = shutil.which(argv[0])
-            if resolved:
-                argv[0] = resolved
+            # exec fast path does no PATH search — resolve against the
+            # container's view when entering one, else the host
+            if setns_pid >= 0:
+                argv[0] = _resolve_via_proc_root(setns_pid, argv[0])
+            elif rootfs:
+                argv[0] = _resolve_in_tree(rootfs, argv[0])
+            else:
+                resolved = shutil.which(argv[0])
+                if resolved:
+                    argv[0] = resolved
         sink = out_path or "/dev/null"
         try:
             pid, pidfd, _, _, _ = self._native.launch_process(
                 argv, [f"{k}={v}" for k, v in env.items()],
                 "", sink, sink, cgroup_dir, True, False, uid, gid,
+                False, "", rootfs, bool(rootfs), [], setns_pid,
             )
         except RuntimeError:
             return 127
@@ -1056,6 +1073,26 @@ class ProcessRuntime(Runtime):
             self._native.signal_process(pid, 9, True)
             exit_code = 124
         return exit_code
+
+    def _container_entry(self, inst: Instance, cspec) -> tuple:
+        """(setns_pid, rootfs) for entering the container of an image pod:
+        mountns mode -> join the live container's namespaces; chroot mode
+        -> its private rootfs copy; host pods -> (-1, "")."""
+        if inst.image_mode == "mountns" and cspec is not None:
+            cinfo = next((c for c in inst.containers
+                          if c.name == cspec.name and c.exit_code is None
+                          and c.pid > 0), None)
+            if cinfo is None:
+                cinfo = next((c for c in inst.containers
+                              if c.exit_code is None and c.pid > 0), None)
+            return (cinfo.pid if cinfo else -1), ""
+        if inst.image_mode == "chroot" and cspec is not None \
+                and self._rootfs_mgr is not None:
+            rootfs = (self._rootfs_mgr.containers_dir
+                      / f"{inst.id}-{cspec.name}" / "rootfs")
+            if rootfs.is_dir():
+                return -1, str(rootfs)
+        return -1, ""
 
     def exec_in_instance(self, instance_id: str, command: List[str],
                          timeout_s: float = 30.0) -> tuple:
@@ -1074,10 +1111,13 @@ class ProcessRuntime(Runtime):
         cspec = inst.params.containers[0] if inst.params.containers else None
         uid = cspec.run_as_uid if cspec else -1
         gid = cspec.run_as_gid if cspec else -1
+        setns_pid, rootfs = self._container_entry(inst, cspec)
+        if inst.image_mode == "mountns" and setns_pid < 0:
+            return 126, "no running container to exec into"
         out_path = self.logs_dir / f".exec-{inst.id}-{secrets.token_hex(4)}.log"
         exit_code = self._run_confined(
             list(command), env, inst.cgroup_dir, uid, gid, timeout_s,
-            out_path=str(out_path))
+            out_path=str(out_path), setns_pid=setns_pid, rootfs=rootfs)
         if exit_code == 127 and not out_path.exists():
             return 127, f"exec spawn failed: {command[0]!r}"
         output = out_path.read_text(errors="replace") if out_path.exists() else ""
@@ -1106,6 +1146,7 @@ class ProcessRuntime(Runtime):
             "active_deadline_s": inst.params.active_deadline_s,
             "termination_grace_s": inst.params.termination_grace_s,
             "deadline_exceeded": inst.deadline_exceeded,
+            "image_mode": inst.image_mode,
             "containers": [
                 {
                     "name": c.name,
@@ -1234,6 +1275,7 @@ class ProcessRuntime(Runtime):
                 cost_per_hr=rec.get("cost_per_hr", 0.0),
                 init_index=rec.get("init_index", 0),
                 deadline_exceeded=rec.get("deadline_exceeded", False),
+                image_mode=rec.get("image_mode", ""),
             )
             for c in rec.get("init_containers", []):
                 icinfo = ContainerRuntimeInfo(
@@ -1363,6 +1405,25 @@ class ProcessRuntime(Runtime):
             timer.cancel()
         for timer in self._deadline_timers.values():
             timer.cancel()
+
+
+def _resolve_via_proc_root(pid: int, argv0: str) -> str:
+    """PATH-resolve inside a live container via /proc/<pid>/root (the
+    kernel's view of its mount namespace root)."""
+    for p in ("/usr/local/sbin", "/usr/local/bin", "/usr/sbin", "/usr/bin",
+              "/sbin", "/bin"):
+        if os.path.exists(f"/proc/{pid}/root{p}/{argv0}"):
+            return f"{p}/{argv0}"
+    return argv0
+
+
+def _resolve_in_tree(rootfs: str, argv0: str) -> str:
+    """PATH-resolve against a rootfs directory tree (chroot mode)."""
+    for p in ("/usr/local/sbin", "/usr/local/bin", "/usr/sbin", "/usr/bin",
+              "/sbin", "/bin"):
+        if os.path.exists(f"{rootfs}{p}/{argv0}"):
+            return f"{p}/{argv0}"
+    return argv0
 
 
 def _listening_tcp_inodes() -> dict:

@@ -173,3 +173,6 @@ class Instance:
     cost_per_hr: float = 0.0
     last_error: str = ""
     deadline_exceeded: bool = False  # activeDeadlineSeconds fired
+    # image-backed execution mode: "" (host binaries) | "mountns" | "chroot"
+    # — drives how kubectl-exec/probes enter the container (setns vs chroot)
+    image_mode: str = ""

@@ -322,3 +322,78 @@ def test_image_pod_end_to_end_stack(tmp_state_dir, app_bin, tmp_path):
         assert wait_until(gone, timeout_s=20)
     finally:
         stack.stop()
+
+
+def test_exec_enters_container_rootfs_mountns(image_runtime):
+    """kubectl-exec on an image pod joins the live container's mount
+    namespace (setns): the exec sees the image's filesystem, not the
+    host's."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    st = deploy_image_pod(rt, "exns", command=["/usr/local/bin/app"],
+                          args=["hold"])
+    time.sleep(0.3)  # container up and holding
+    code, out = rt.exec_in_instance(
+        st.id, ["/usr/local/bin/app", "query"], timeout_s=15)
+    assert code == 0, out
+    assert "image-file=present" in out   # sees the image fs
+    assert "host-python=absent" in out   # not the host fs
+    rt.terminate(st.id)
+    wait_status(rt, st.id, PodStatus.TERMINATED)
+
+
+def test_exec_enters_container_rootfs_chroot(synthetic_ledger,
+                                             tmp_state_dir, image_store):
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                        enable_cgroups=False, image_store=image_store,
+                        image_isolation="chroot")
+    try:
+        st = deploy_image_pod(rt, "exch", command=["/usr/local/bin/app"],
+                              args=["hold"])
+        time.sleep(0.3)
+        code, out = rt.exec_in_instance(
+            st.id, ["/usr/local/bin/app", "query"], timeout_s=15)
+        assert code == 0, out
+        assert "image-file=present" in out
+        assert "host-python=absent" in out
+        rt.terminate(st.id)
+        wait_status(rt, st.id, PodStatus.TERMINATED)
+    finally:
+        rt.close()
+
+
+def test_exec_probe_runs_inside_image(image_runtime):
+    """An exec readinessProbe on an image pod runs inside the container
+    (k8s semantics): it can only pass by seeing an image-only file."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+    from k8s_runpod_kubelet_amd.runtime.probes import ProbeSpec
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-iprobe", name="iprobe",
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/usr/local/bin/app"], args=["hold"],
+            readiness=ProbeSpec(
+                kind="exec",
+                # /etc/app-release exists only in the image; probe binary
+                # itself must resolve inside the container too
+                command=["/usr/local/bin/app", "query"],
+                period_s=1.0, timeout_s=10.0))],
+    ))
+    deadline = time.time() + 15
+    ready = False
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        if s.containers and s.containers[0].ready:
+            ready = True
+            break
+        time.sleep(0.1)
+    assert ready, rt.get_logs(st.id)
+    rt.terminate(st.id)
+    wait_status(rt, st.id, PodStatus.TERMINATED)
