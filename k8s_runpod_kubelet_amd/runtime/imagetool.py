@@ -36,6 +36,13 @@ def main(argv=None) -> int:
     p_add.add_argument("dir")
     p_add.add_argument("--ref", required=True)
 
+    p_pull = sub.add_parser("pull", help="pull from an OCI registry")
+    p_pull.add_argument("ref")
+    p_pull.add_argument("--registry", default="",
+                        help="registry base URL override, e.g. http://host:5000")
+    p_pull.add_argument("--token", default="")
+    p_pull.add_argument("--insecure", action="store_true")
+
     p_b = sub.add_parser("build", help="build an image from a rootfs dir")
     p_b.add_argument("rootfs")
     p_b.add_argument("--ref", required=True)
@@ -59,6 +66,17 @@ def main(argv=None) -> int:
     if args.cmd == "add-layout":
         ref = store.add_layout(args.dir, args.ref)
         print(f"added {ref}")
+        return 0
+    if args.cmd == "pull":
+        from .registry import RegistryClient
+
+        client = RegistryClient(base_url=args.registry, token=args.token,
+                                verify=not args.insecure)
+        try:
+            ref = client.pull(args.ref, store)
+        finally:
+            client.close()
+        print(f"pulled {ref}")
         return 0
     if args.cmd == "build":
         with tempfile.TemporaryDirectory(prefix="amdvk-build-") as td:

@@ -1,0 +1,162 @@
+"""OCI Distribution (registry v2) client: pull images into the local store.
+
+The reference's backend pulls `Containers[0].Image` from a registry
+(RunPod's side of runpod_client.go:1304); this node normally has no
+egress, so the store is fed from layouts/archives — but clusters commonly
+run an in-network registry, and `RegistryClient.pull` speaks the standard
+`/v2/` protocol (manifest negotiation incl. manifest lists, blob fetch,
+sha256 verification) into an OCI layout registered in the ImageStore.
+`registry_server.py` is the matching in-repo server, which also lets one
+node serve its store to others."""
+
+from __future__ import annotations
+
+import hashlib
+import json
+import logging
+import tempfile
+from pathlib import Path
+from typing import Dict, Optional, Tuple
+
+import httpx
+
+from .oci import (
+    _INDEX_TYPES,
+    _MANIFEST_TYPES,
+    ImageError,
+    ImageStore,
+    normalize_ref,
+)
+
+log = logging.getLogger("runtime.registry")
+
+_ACCEPT = ", ".join(sorted(_MANIFEST_TYPES | _INDEX_TYPES))
+
+
+def parse_ref(ref: str) -> Tuple[str, str, str]:
+    """normalized ref -> (registry_host, repository, tag_or_digest)."""
+    norm = normalize_ref(ref)
+    host, _, rest = norm.partition("/")
+    if "@" in rest:
+        name, _, digest = rest.partition("@")
+        return host, name, digest
+    name, _, tag = rest.rpartition(":")
+    return host, name, tag or "latest"
+
+
+class RegistryError(ImageError):
+    pass
+
+
+class RegistryClient:
+    def __init__(self, base_url: str = "", token: str = "",
+                 timeout_s: float = 60.0, verify: bool = True):
+        """base_url overrides the registry host parsed from the ref (use
+        for mirrors / plain-HTTP in-cluster registries: 'http://host:port')."""
+        self.base_url = base_url.rstrip("/")
+        headers = {"Accept": _ACCEPT}
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        self._http = httpx.Client(headers=headers, timeout=timeout_s,
+                                  verify=verify, follow_redirects=True)
+
+    def close(self) -> None:
+        self._http.close()
+
+    def _url(self, host: str, path: str) -> str:
+        base = self.base_url or f"https://{host}"
+        return f"{base}{path}"
+
+    def _get(self, url: str, accept: Optional[str] = None) -> httpx.Response:
+        headers = {"Accept": accept} if accept else {}
+        resp = self._http.get(url, headers=headers)
+        if resp.status_code == 401:
+            raise RegistryError(f"unauthorized: {url}")
+        if resp.status_code == 404:
+            raise RegistryError(f"not found: {url}")
+        if resp.status_code >= 400:
+            raise RegistryError(f"registry error {resp.status_code}: {url}")
+        return resp
+
+    def pull(self, ref: str, store: ImageStore) -> str:
+        """Pull ref into the store as an OCI layout; returns the
+        normalized reference. Every blob is sha256-verified against the
+        digest that named it."""
+        host, name, tagish = parse_ref(ref)
+        manifest_url = self._url(host, f"/v2/{name}/manifests/{tagish}")
+        resp = self._get(manifest_url, accept=_ACCEPT)
+        body = resp.content
+        mtype = resp.headers.get("Content-Type", "").split(";")[0]
+        if mtype in _INDEX_TYPES:
+            index = json.loads(body)
+            desc = _pick_platform(index.get("manifests", []))
+            if desc is None:
+                raise RegistryError(f"{ref}: no linux/amd64 manifest")
+            resp = self._get(
+                self._url(host, f"/v2/{name}/manifests/{desc['digest']}"),
+                accept=", ".join(sorted(_MANIFEST_TYPES)))
+            body = resp.content
+            _verify(desc["digest"], body)
+            mtype = resp.headers.get("Content-Type", "").split(";")[0]
+        if mtype not in _MANIFEST_TYPES:
+            raise RegistryError(f"{ref}: unexpected manifest type {mtype!r}")
+        manifest = json.loads(body)
+        man_digest = "sha256:" + hashlib.sha256(body).hexdigest()
+
+        with tempfile.TemporaryDirectory(prefix="amdvk-pull-") as td:
+            layout = Path(td)
+            blobs = layout / "blobs" / "sha256"
+            blobs.mkdir(parents=True)
+            _write_blob(blobs, man_digest, body)
+            cfg_desc = manifest.get("config", {})
+            self._fetch_blob(host, name, cfg_desc.get("digest", ""), blobs)
+            for lay in manifest.get("layers", []):
+                self._fetch_blob(host, name, lay.get("digest", ""), blobs)
+            (layout / "oci-layout").write_text(
+                json.dumps({"imageLayoutVersion": "1.0.0"}))
+            (layout / "index.json").write_text(json.dumps({
+                "schemaVersion": 2,
+                "manifests": [{
+                    "mediaType":
+                        "application/vnd.oci.image.manifest.v1+json",
+                    "digest": man_digest, "size": len(body),
+                    "annotations": {
+                        "org.opencontainers.image.ref.name":
+                            normalize_ref(ref)},
+                }],
+            }))
+            out = store.add_layout(str(layout), ref)
+        log.info("image pulled", extra={"ref": out, "digest": man_digest})
+        return out
+
+    def _fetch_blob(self, host: str, name: str, digest: str,
+                    blobs: Path) -> None:
+        if not digest:
+            raise RegistryError("manifest names a blob with no digest")
+        resp = self._get(self._url(host, f"/v2/{name}/blobs/{digest}"))
+        data = resp.content
+        _verify(digest, data)
+        _write_blob(blobs, digest, data)
+
+
+def _verify(digest: str, data: bytes) -> None:
+    algo, _, hexd = digest.partition(":")
+    if algo != "sha256":
+        raise RegistryError(f"unsupported digest algorithm {algo!r}")
+    actual = hashlib.sha256(data).hexdigest()
+    if actual != hexd:
+        raise RegistryError(
+            f"digest mismatch: expected {digest}, got sha256:{actual}")
+
+
+def _write_blob(blobs: Path, digest: str, data: bytes) -> None:
+    (blobs / digest.partition(":")[2]).write_bytes(data)
+
+
+def _pick_platform(manifests) -> Optional[Dict]:
+    for m in manifests:
+        p = m.get("platform") or {}
+        if (p.get("os") in (None, "linux")
+                and p.get("architecture") in (None, "amd64")):
+            return m
+    return manifests[0] if manifests else None

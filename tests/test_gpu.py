@@ -791,3 +791,115 @@ def test_gpu_image_pod_runs_podworker_in_rootfs(tmp_path):
         assert "ok" in logs, logs
     finally:
         stack.stop()
+
+
+def _ldd_closure(binary):
+    """{container_path: host_path} for a binary's shared-library closure."""
+    import re
+    import subprocess
+
+    out = subprocess.run(["ldd", binary], capture_output=True, text=True,
+                         check=True).stdout
+    deps = {}
+    for line in out.splitlines():
+        m = re.search(r"=>\s+(\S+)\s+\(", line)
+        if m:
+            deps[m.group(1)] = m.group(1)
+        else:
+            m2 = re.search(r"^\s*(/\S*ld-linux\S*)\s+\(", line)
+            if m2:
+                deps[m2.group(1)] = m2.group(1)
+    return deps
+
+
+def test_gpu_image_pod_chroot_selfcontained(tmp_path):
+    """Degraded-isolation GPU image pod: a self-contained image (podworker
+    + its library closure + a KFD topology snapshot) running in CHROOT
+    mode with mknod'd GPU device nodes — the path sandboxes without
+    CAP_SYS_ADMIN take. Skips (with the reason) where the environment
+    denies mknod or HSA cannot enumerate without live sysfs."""
+    _require_gpu()
+    import shutil
+    import stat as statmod
+
+    from k8s_runpod_kubelet_amd.gpu.inventory import Inventory
+    from k8s_runpod_kubelet_amd.gpu.ledger import Ledger
+    from k8s_runpod_kubelet_amd.ops import podworker_binary
+    from k8s_runpod_kubelet_amd.runtime.oci import ImageStore, build_layout
+    from k8s_runpod_kubelet_amd.runtime.process_runtime import ProcessRuntime
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+
+    # can we mknod a char device at all?
+    probe = tmp_path / "mknod-probe"
+    try:
+        os.mknod(probe, 0o600 | statmod.S_IFCHR, os.makedev(1, 3))
+    except (OSError, PermissionError):
+        pytest.skip("environment denies mknod (no CAP_MKNOD)")
+    probe.unlink()
+
+    tree = tmp_path / "tree"
+    (tree / "bin").mkdir(parents=True)
+    shutil.copy2(podworker_binary(), tree / "bin" / "podworker")
+    for cpath, hpath in _ldd_closure(podworker_binary()).items():
+        dst = tree / cpath.lstrip("/")
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        shutil.copy2(os.path.realpath(hpath), dst)
+    # HSA enumerates GPUs from sysfs: snapshot the KFD topology + node
+    # properties into the image. Content-only copy — sysfs permits
+    # neither copystat nor utime (shutil.copy2 EPERMs on it).
+    def snapshot_sysfs(src, dst_root):
+        for root, _dirs, files in os.walk(src):
+            rel = os.path.relpath(root, "/")
+            os.makedirs(dst_root / rel, exist_ok=True)
+            for f in files:
+                try:
+                    with open(os.path.join(root, f), "rb") as fh:
+                        data = fh.read()
+                except OSError:
+                    continue
+                (dst_root / rel / f).write_bytes(data)
+
+    for sysdir in ("/sys/class/kfd/kfd/topology",
+                   "/sys/devices/virtual/kfd/kfd/topology"):
+        if os.path.isdir(sysdir):
+            snapshot_sysfs(sysdir, tree)
+            break
+
+    layout = tmp_path / "layout"
+    layout.mkdir()
+    build_layout(str(layout), "example/gpuchroot:v1", str(tree),
+                 entrypoint=["/bin/podworker"],
+                 cmd=["--expect-gpus", "1", "--run-for", "0.3"],
+                 env=["PATH=/bin"])
+    store = ImageStore(str(tmp_path / "store"))
+    store.add_layout(str(layout), "example/gpuchroot:v1")
+
+    inv = Inventory(sysfs_root="/sys", allow_synthetic=False)
+    if not inv.discover():
+        pytest.skip("no GPUs")
+    ledger = Ledger(inv)
+    ledger.sync_inventory()
+    rt = ProcessRuntime(ledger, str(tmp_path / "state"),
+                        enable_cgroups=False, image_store=store,
+                        image_isolation="chroot")
+    try:
+        st = rt.deploy(DeployParams(
+            pod_key="default-gchroot", name="gchroot", gpu_count=1,
+            containers=[ContainerSpec(
+                name="main", image="example/gpuchroot:v1")],
+        ))
+        deadline = time.time() + 180
+        while time.time() < deadline:
+            s = rt.get_detailed_status(st.id)
+            if s.desired_status == PodStatus.EXITED:
+                break
+            time.sleep(0.1)
+        logs = rt.get_logs(st.id)
+        if s.exit_code == 12 or "expected" in logs:
+            pytest.skip(f"chroot-mode HSA could not see the GPU here "
+                        f"(exit {s.exit_code}): {logs[-300:]}")
+        assert s.exit_code == 0, logs
+        assert "ok" in logs, logs
+    finally:
+        rt.close()
