@@ -866,12 +866,22 @@ def test_gpu_image_pod_chroot_selfcontained(tmp_path):
             snapshot_sysfs(sysdir, tree)
             break
 
+    # podworker's RUNPATH points at the versioned ROCm dir; the ldd
+    # closure landed under the /opt/rocm symlink-resolved paths — provide
+    # both spellings inside the image
+    real_rocm = os.path.realpath("/opt/rocm")
+    if real_rocm != "/opt/rocm" and (tree / "opt").is_dir():
+        link = tree / real_rocm.lstrip("/")
+        if not link.exists():
+            link.symlink_to("rocm")
+
     layout = tmp_path / "layout"
     layout.mkdir()
     build_layout(str(layout), "example/gpuchroot:v1", str(tree),
                  entrypoint=["/bin/podworker"],
                  cmd=["--expect-gpus", "1", "--run-for", "0.3"],
-                 env=["PATH=/bin"])
+                 env=["PATH=/bin",
+                      "LD_LIBRARY_PATH=/opt/rocm/lib:/opt/rocm/lib64"])
     store = ImageStore(str(tmp_path / "store"))
     store.add_layout(str(layout), "example/gpuchroot:v1")
 
