@@ -35,9 +35,74 @@ from ..kube.objects import (
     resource_parse_cpu,
 )
 from ..runtime.probes import ProbeSpec
-from ..runtime.types import ContainerSpec, DeployParams
+from ..runtime.types import (ContainerSpec, DeployParams, VolumeMount,
+                             VolumeSource)
 from . import annotations as ann
 from .envvars import extract_env_vars
+
+
+def _volume_mounts_of(c: Dict[str, Any]) -> List[VolumeMount]:
+    out: List[VolumeMount] = []
+    for m in c.get("volumeMounts", []) or []:
+        if m.get("name") and m.get("mountPath"):
+            out.append(VolumeMount(
+                name=m["name"], mount_path=m["mountPath"],
+                read_only=bool(m.get("readOnly")),
+                sub_path=m.get("subPath", "") or ""))
+    return out
+
+
+def extract_volumes(pod: Dict[str, Any], client: K8sClient) -> Dict[str, VolumeSource]:
+    """spec.volumes -> resolved VolumeSource map (consumed by image-backed
+    containers; host-process pods see the host filesystem, and secret
+    volumes are additionally flattened into env for reference parity —
+    runpod_client.go:949-979). Secret/ConfigMap content is fetched at
+    translation time like env extraction. Unsupported volume types (PVC,
+    projected, downwardAPI, ...) are skipped — mounts referencing them are
+    ignored rather than failing the pod."""
+    from .envvars import ConfigMapCollector, SecretCollector
+
+    namespace = pod.get("metadata", {}).get("namespace", "default")
+    out: Dict[str, VolumeSource] = {}
+    for vol in pod.get("spec", {}).get("volumes", []) or []:
+        name = vol.get("name", "")
+        if not name:
+            continue
+        if "emptyDir" in vol:
+            out[name] = VolumeSource(kind="emptyDir")
+        elif (vol.get("hostPath") or {}).get("path"):
+            out[name] = VolumeSource(kind="hostPath",
+                                     host_path=vol["hostPath"]["path"])
+        elif (vol.get("secret") or {}).get("secretName"):
+            sec = vol["secret"]
+            data = SecretCollector(client, namespace).get(sec["secretName"])
+            files = _project_items(data, sec.get("items"))
+            out[name] = VolumeSource(kind="files", files=files,
+                                     file_mode=int(sec.get("defaultMode",
+                                                           0o644)))
+        elif (vol.get("configMap") or {}).get("name"):
+            cm = vol["configMap"]
+            data = ConfigMapCollector(client, namespace).get(cm["name"])
+            files = _project_items(data, cm.get("items"))
+            out[name] = VolumeSource(kind="files", files=files,
+                                     file_mode=int(cm.get("defaultMode",
+                                                          0o644)))
+    return out
+
+
+def _project_items(data: Dict[str, str], items) -> Dict[str, str]:
+    """k8s items[] projection: each item's key lands at its path (default:
+    key name)."""
+    if not items:
+        return dict(data)
+    files: Dict[str, str] = {}
+    for item in items:
+        key = item.get("key", "")
+        if key in data:
+            files[item.get("path") or key] = data[key]
+    return files
+
+
 from .ports import get_requested_ports, tcp_ports_of_container
 
 log = logging.getLogger("provider.translate")
@@ -259,6 +324,7 @@ def prepare_deploy_params(
                             if gid not in (None, "") else -1),
                 liveness=ProbeSpec.parse(c.get("livenessProbe")),
                 readiness=ProbeSpec.parse(c.get("readinessProbe")),
+                volume_mounts=_volume_mounts_of(c),
             )
         )
     if not containers:
@@ -284,6 +350,7 @@ def prepare_deploy_params(
                             if uid not in (None, "") else -1),
                 run_as_gid=(_int_or_error(gid, "runAsGroup")
                             if gid not in (None, "") else -1),
+                volume_mounts=_volume_mounts_of(c),
             )
         )
 
@@ -306,6 +373,7 @@ def prepare_deploy_params(
         registry_auth_id=registry_auth,
         cpu_limit=cpu_max,
         memory_limit=memory_max,
+        volumes=extract_volumes(pod, client),
         labels=dict(metadata.get("labels", {}) or {}),
         hostname=pod.get("spec", {}).get("hostname")
         or metadata.get("name", ""),

@@ -15,12 +15,16 @@ with:
   pipe, exit from the pidfd — so the provider learns of state changes in
   microseconds instead of on the next 10 s poll tick (kubelet.go:719).
 
-"Image" handling: there is no container-image store on the node, so the
-runtime executes the container's command/args directly (host binaries); the
-image string is recorded as metadata. Synthetic workloads (tests, bench) use
-the in-tree HIP `podworker` binary. A `command`-less container falls back to
-podworker in hold mode sized to the pod's GPU request — mirroring how the
-reference test pod runs a GPU-probe image (runpod_test.go:99).
+"Image" handling (reference contract runpod_client.go:1304 — the image
+actually runs): a pod whose image resolves in the node-local OCI store
+(runtime/oci.py) executes the image's entrypoint inside a per-container
+rootfs (runtime/rootfs.py: overlay + mount-ns + pivot_root, or chroot in
+degraded sandboxes), with k8s command/args vs Entrypoint/Cmd semantics,
+image Env/WorkingDir/User, and pod volumes bound in. Images absent from the
+store keep the legacy host-binary behavior (the reserved `amdvk/*` family,
+bench and test workloads use the in-tree HIP `podworker`); a `command`-less
+unresolved-image container falls back to podworker in hold mode — mirroring
+how the reference test pod runs a GPU-probe image (runpod_test.go:99).
 
 Instance state is journaled under <state_dir>/instances/<id>.json so a
 restarted kubelet re-adopts live pods (reference LoadRunning analogue,
@@ -235,6 +239,41 @@ class ProcessRuntime(Runtime):
             return None
         return self.image_store.resolve(cspec.image)
 
+    def _volume_binds(self, inst: Instance, cspec) -> List[tuple]:
+        """Materialize the pod's volumes and return (src, dst, ro) binds
+        for this container. emptyDir/files volumes live under
+        <state>/volumes/<instance>/ — per-POD, so containers share them and
+        container restarts keep their content (k8s emptyDir semantics)."""
+        out = []
+        if not getattr(cspec, "volume_mounts", None):
+            return out
+        base = self.state_dir / "volumes" / inst.id
+        for vm in cspec.volume_mounts:
+            src_spec = inst.params.volumes.get(vm.name)
+            if src_spec is None:
+                continue  # unsupported/unknown volume type: mount skipped
+            if src_spec.kind == "hostPath":
+                src = src_spec.host_path
+                if not os.path.exists(src):
+                    raise RuntimeError(
+                        f"hostPath volume {vm.name}: {src} does not exist")
+            else:
+                vdir = base / vm.name
+                if not vdir.exists():
+                    vdir.mkdir(parents=True, exist_ok=True)
+                    for fname, content in src_spec.files.items():
+                        fp = vdir / fname.lstrip("/")
+                        fp.parent.mkdir(parents=True, exist_ok=True)
+                        fp.write_text(content)
+                        os.chmod(fp, src_spec.file_mode)
+                src = str(vdir)
+            if vm.sub_path:
+                src = os.path.join(src, vm.sub_path)
+            # secret/configMap projections are read-only like k8s mounts
+            ro = vm.read_only or src_spec.kind == "files"
+            out.append((src, vm.mount_path, ro, src_spec.kind))
+        return out
+
     def _gpu_device_paths(self, inst: Instance) -> List[str]:
         """Host device nodes an image pod needs for its bound GPUs."""
         if not inst.gpu_indices or not os.path.exists("/dev/kfd"):
@@ -296,7 +335,8 @@ class ProcessRuntime(Runtime):
                 inst.id, cspec.name, image,
                 params.hostname or params.name,
                 gpu_device_paths=self._gpu_device_paths(inst),
-                working_dir=working_dir)
+                working_dir=working_dir,
+                volume_binds=self._volume_binds(inst, cspec))
             argv[0] = self._rootfs_mgr.resolve_argv0(image, prepared,
                                                      argv[0])
             rootfs = prepared.rootfs
@@ -352,7 +392,8 @@ class ProcessRuntime(Runtime):
                 inst.id, cspec.name, image,
                 params.hostname or params.name,
                 gpu_device_paths=self._gpu_device_paths(inst),
-                working_dir=working_dir)
+                working_dir=working_dir,
+                volume_binds=self._volume_binds(inst, cspec))
             argv[0] = self._rootfs_mgr.resolve_argv0(
                 image, prepared, (cspec.command or image.config.entrypoint
                                   or argv)[0])
@@ -947,6 +988,8 @@ class ProcessRuntime(Runtime):
             (self.instances_dir / f"{instance_id}.json").unlink(missing_ok=True)
             if self._rootfs_mgr is not None:
                 self._rootfs_mgr.cleanup(instance_id)
+            shutil.rmtree(self.state_dir / "volumes" / instance_id,
+                          ignore_errors=True)
 
     def healthy(self) -> bool:
         return self.ledger.any_schedulable() or self.ledger.total_gpus() == 0

@@ -81,13 +81,19 @@ class RootfsManager:
     def prepare(self, instance_id: str, container_name: str,
                 image: ResolvedImage, hostname: str,
                 gpu_device_paths: Optional[List[str]] = None,
-                working_dir: str = "") -> PreparedRootfs:
+                working_dir: str = "",
+                volume_binds: Optional[List[tuple]] = None) -> PreparedRootfs:
+        """volume_binds: (host_src, container_dst, ro) tuples from the
+        pod's volumes (emptyDir/hostPath/secret/configMap projections,
+        materialized by the runtime)."""
         cdir = self.containers_dir / f"{instance_id}-{container_name}"
         if self.mode() == "mountns":
             return self._prepare_overlay(cdir, image, hostname,
-                                         gpu_device_paths or [], working_dir)
+                                         gpu_device_paths or [], working_dir,
+                                         volume_binds or [])
         return self._prepare_chroot(cdir, image, hostname,
-                                    gpu_device_paths or [], working_dir)
+                                    gpu_device_paths or [], working_dir,
+                                    volume_binds or [])
 
     def _parse_bind(self, entry: str) -> Tuple[str, str, bool]:
         """'src[:dst[:ro|rw]]' -> (src, dst, ro). Default dst=src, ro."""
@@ -99,7 +105,8 @@ class RootfsManager:
 
     def _prepare_overlay(self, cdir: Path, image: ResolvedImage,
                          hostname: str, gpu_devices: List[str],
-                         working_dir: str) -> PreparedRootfs:
+                         working_dir: str,
+                         volume_binds: List[tuple]) -> PreparedRootfs:
         lower = self.store.rootfs_for(image)
         upper = cdir / "upper"
         work = cdir / "work"
@@ -148,12 +155,22 @@ class RootfsManager:
                 target.parent.mkdir(parents=True, exist_ok=True)
                 target.touch()
             mounts.append(mnt.bind(src, str(merged / rel), ro=ro))
+        for src, dst, ro, _kind in volume_binds:
+            rel = dst.lstrip("/")
+            target = upper / rel
+            if os.path.isdir(src):
+                target.mkdir(parents=True, exist_ok=True)
+            else:
+                target.parent.mkdir(parents=True, exist_ok=True)
+                target.touch()
+            mounts.append(mnt.bind(src, str(merged / rel), ro=ro))
         return PreparedRootfs(rootfs=str(merged), mounts=mounts,
                               chroot_only=False, container_dir=str(cdir))
 
     def _prepare_chroot(self, cdir: Path, image: ResolvedImage,
                         hostname: str, gpu_devices: List[str],
-                        working_dir: str) -> PreparedRootfs:
+                        working_dir: str,
+                        volume_binds: List[tuple]) -> PreparedRootfs:
         rootfs = cdir / "rootfs"
         if not (rootfs / ".amdvk-ready").exists():
             cdir.mkdir(parents=True, exist_ok=True)
@@ -173,6 +190,23 @@ class RootfsManager:
                 (rootfs / working_dir.lstrip("/")).mkdir(parents=True,
                                                          exist_ok=True)
             self._write_identity(rootfs, hostname)
+            # chroot mode has no mounts: volumes are materialized INTO the
+            # per-container copy (per-container — cross-container emptyDir
+            # sharing needs the mount-namespace mode; hostPath cannot be
+            # provided at all without bind mounts)
+            for src, dst, ro, kind in volume_binds:
+                if kind == "hostPath":
+                    shutil.rmtree(cdir, ignore_errors=True)
+                    raise RuntimeError(
+                        "hostPath volumes require mount-namespace "
+                        "isolation (unavailable here)")
+                rel = dst.lstrip("/")
+                target = rootfs / rel
+                if os.path.isdir(src):
+                    shutil.copytree(src, target, dirs_exist_ok=True)
+                else:
+                    target.parent.mkdir(parents=True, exist_ok=True)
+                    shutil.copy2(src, target)
             (rootfs / ".amdvk-ready").touch()
         return PreparedRootfs(rootfs=str(rootfs), mounts=[],
                               chroot_only=True, container_dir=str(cdir))

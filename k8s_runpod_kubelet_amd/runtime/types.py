@@ -105,6 +105,31 @@ def is_successful_completion(status: DetailedStatus) -> bool:
 
 
 @dataclass
+class VolumeMount:
+    """containers[].volumeMounts entry."""
+    name: str
+    mount_path: str
+    read_only: bool = False
+    sub_path: str = ""
+
+
+@dataclass
+class VolumeSource:
+    """spec.volumes entry, resolved at translation time.
+
+    kind: "emptyDir" (per-pod scratch dir, shared between the pod's
+    containers, survives container restarts — k8s semantics; medium=Memory
+    is backed by disk here, a documented deviation), "hostPath" (host
+    directory/file bound in; mount-namespace isolation mode only), or
+    "files" (secret/configMap projected to files — content fetched from
+    the API at translation time like env extraction)."""
+    kind: str
+    host_path: str = ""
+    files: Dict[str, str] = field(default_factory=dict)
+    file_mode: int = 0o644
+
+
+@dataclass
 class ContainerSpec:
     name: str
     image: str = ""
@@ -113,6 +138,7 @@ class ContainerSpec:
     env: Dict[str, str] = field(default_factory=dict)
     working_dir: str = ""
     tcp_ports: List[int] = field(default_factory=list)
+    volume_mounts: List[VolumeMount] = field(default_factory=list)
     # securityContext.runAsUser/runAsGroup (container overrides pod; -1 =
     # inherit the kubelet's credentials)
     run_as_uid: int = -1
@@ -151,6 +177,9 @@ class DeployParams:
     registry_auth_id: str = ""
     cpu_limit: str = ""      # cgroup cpu.max, e.g. "200000 100000"
     memory_limit: str = ""   # cgroup memory.max bytes or "max"
+    # spec.volumes resolved (see VolumeSource); consumed by image-backed
+    # containers (host-process pods already see the host filesystem)
+    volumes: Dict[str, VolumeSource] = field(default_factory=dict)
     labels: Dict[str, str] = field(default_factory=dict)
     # k8s pod-hostname semantics (spec.hostname, else pod name); applied in
     # the pod's own UTS namespace when namespace isolation is available

@@ -35,20 +35,34 @@ def closure(binary):
     return deps
 
 
-def snapshot(src, tree, follow=True):
+def snapshot(src, tree, max_files=5000):
+    """Bounded, symlink-free sysfs snapshot (sysfs symlinks form cycles —
+    never follow them) of small text attributes."""
     if not os.path.isdir(src):
         print(f"  [snapshot] missing {src}")
         return
-    for root, _dirs, files in os.walk(src, followlinks=follow):
+    # resolve the top-level dir itself (e.g. /sys/class/kfd -> devices path)
+    # but never follow links below it
+    count = 0
+    for root, _dirs, files in os.walk(src, followlinks=False):
         rel = os.path.relpath(root, "/")
         os.makedirs(tree / rel, exist_ok=True)
         for f in files:
+            fp = os.path.join(root, f)
+            if os.path.islink(fp):
+                continue
             try:
-                with open(os.path.join(root, f), "rb") as fh:
-                    data = fh.read()
+                if os.path.getsize(fp) > 1 << 20:
+                    continue
+                with open(fp, "rb") as fh:
+                    data = fh.read(1 << 20)
             except OSError:
                 continue
             (tree / rel / f).write_bytes(data)
+            count += 1
+            if count >= max_files:
+                print(f"  [snapshot] {src}: truncated at {max_files} files")
+                return
 
 
 def main():
@@ -70,9 +84,14 @@ def main():
         if not link.exists():
             link.symlink_to("rocm")
 
-    # sysfs snapshots HSA/libdrm are known to read
-    snapshot("/sys/class/kfd/kfd/topology", tree)
+    # sysfs snapshots HSA/libdrm are known to read. /sys/class/kfd is a
+    # symlink to devices/virtual — walk the real path, store at BOTH names.
     snapshot("/sys/devices/virtual/kfd/kfd/topology", tree)
+    real = tree / "sys/devices/virtual/kfd"
+    cls = tree / "sys/class/kfd"
+    if real.is_dir() and not cls.exists():
+        cls.parent.mkdir(parents=True, exist_ok=True)
+        cls.symlink_to("../devices/virtual/kfd")
     snapshot("/sys/devices/system/node", tree)
     # /sys/dev/char/<maj:min> device links for kfd + render nodes
     gpus = native.enumerate_gpus("/sys")
@@ -94,22 +113,74 @@ def main():
         except OSError as exc:
             print(f"  [dev] mknod {dev} FAILED: {exc}")
         maj, minr = os.major(st.st_rdev), os.minor(st.st_rdev)
+        # /sys/dev/char/<maj:min> -> device dir: record the link target
+        # shallowly (one level of attributes), not the whole PCI tree
         sysdev = f"/sys/dev/char/{maj}:{minr}"
         if os.path.exists(sysdev):
-            snapshot(os.path.realpath(sysdev), tree)
+            realdev = os.path.realpath(sysdev)
+            reldev = os.path.relpath(realdev, "/")
+            os.makedirs(tree / reldev, exist_ok=True)
+            for f in os.listdir(realdev):
+                fp = os.path.join(realdev, f)
+                if os.path.islink(fp) or os.path.isdir(fp):
+                    continue
+                try:
+                    (tree / reldev / f).write_bytes(open(fp, "rb").read())
+                except OSError:
+                    pass
+            link = tree / "sys/dev/char" / f"{maj}:{minr}"
+            link.parent.mkdir(parents=True, exist_ok=True)
+            if not link.exists():
+                # container-relative target (an absolute host path would
+                # resolve wrongly inside the chroot)
+                link.symlink_to(os.path.relpath(realdev, "/sys/dev/char"))
     # /proc files read-only consumers want (plain files in the rootfs)
-    for pf in ("/proc/cpuinfo", "/proc/meminfo"):
+    for pf in ("/proc/cpuinfo", "/proc/meminfo", "/proc/version"):
         try:
             (tree / pf.lstrip("/")).write_bytes(open(pf, "rb").read())
         except OSError:
             pass
 
+    # LD_PRELOAD syscall spy: log every open/openat/stat failure to stderr
+    # so we can see exactly which path ROCr/ROCt misses inside the chroot
+    spy_c = td / "spy.c"
+    spy_c.write_text(r"""
+#define _GNU_SOURCE
+#include <stdarg.h>
+#include <stdio.h>
+#include <dlfcn.h>
+#include <errno.h>
+#include <fcntl.h>
+static int (*real_openat)(int, const char*, int, ...) = 0;
+static int (*real_open)(const char*, int, ...) = 0;
+int openat(int dirfd, const char* path, int flags, ...) {
+    if (!real_openat) real_openat = dlsym(RTLD_NEXT, "openat");
+    va_list ap; va_start(ap, flags);
+    int mode = va_arg(ap, int); va_end(ap);
+    int rc = real_openat(dirfd, path, flags, mode);
+    if (rc < 0) fprintf(stderr, "[spy] openat(%s) -> %d\n", path, -errno);
+    return rc;
+}
+int open(const char* path, int flags, ...) {
+    if (!real_open) real_open = dlsym(RTLD_NEXT, "open");
+    va_list ap; va_start(ap, flags);
+    int mode = va_arg(ap, int); va_end(ap);
+    int rc = real_open(path, flags, mode);
+    if (rc < 0) fprintf(stderr, "[spy] open(%s) -> %d\n", path, -errno);
+    return rc;
+}
+""")
+    subprocess.run(["gcc", "-shared", "-fPIC", "-O1", "-o",
+                    str(tree / "spy.so"), str(spy_c), "-ldl"], check=True)
+
     log = td / "out.log"
     env = {
         "PATH": "/bin",
         "LD_LIBRARY_PATH": "/opt/rocm/lib:/opt/rocm/lib64",
+        "LD_PRELOAD": "/spy.so",
         "AMD_LOG_LEVEL": "4",
         "HSAKMT_DEBUG_LEVEL": "7",
+        "HSA_ENABLE_IPC_MODE_LEGACY": "0",
         "ROCR_VISIBLE_DEVICES": "0",
     }
     pid, pidfd, _, _, _ = native.launch_process(
@@ -119,7 +190,7 @@ def main():
         False, "", str(tree), True, [])
     loop = native.EventLoop()
     loop.add_process(pid, pidfd, -1, 0)
-    deadline = time.time() + 120
+    deadline = time.time() + 60
     code = None
     while time.time() < deadline and code is None:
         for ev in loop.poll(200):

@@ -80,6 +80,15 @@ def image_store(tmp_path, app_bin):
     import shutil
 
     shutil.copy2(app_bin, tree / "usr" / "local" / "bin" / "app")
+    # a real (dynamically linked) shell inside the image for script tests
+    (tree / "bin").mkdir()
+    (tree / "lib" / "x86_64-linux-gnu").mkdir(parents=True)
+    (tree / "lib64").mkdir()
+    shutil.copy2("/bin/sh", tree / "bin" / "sh")
+    shutil.copy2(os.path.realpath("/lib/x86_64-linux-gnu/libc.so.6"),
+                 tree / "lib" / "x86_64-linux-gnu" / "libc.so.6")
+    shutil.copy2(os.path.realpath("/lib64/ld-linux-x86-64.so.2"),
+                 tree / "lib64" / "ld-linux-x86-64.so.2")
     (tree / "etc" / "app-release").write_text("v1\n")
     (tree / "etc" / "passwd").write_text(
         "root:x:0:0:root:/root:/bin/sh\n"
@@ -397,3 +406,178 @@ def test_exec_probe_runs_inside_image(image_runtime):
     assert ready, rt.get_logs(st.id)
     rt.terminate(st.id)
     wait_status(rt, st.id, PodStatus.TERMINATED)
+
+
+def test_volumes_emptydir_shared_and_secret_files(image_runtime, tmp_path):
+    """Pod volumes for image containers: an emptyDir is shared between the
+    pod's containers (k8s semantics) and a projected files volume
+    (secret/configMap) mounts read-only."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        VolumeMount, VolumeSource)
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    vols = {
+        "scratch": VolumeSource(kind="emptyDir"),
+        "creds": VolumeSource(kind="files",
+                              files={"token": "s3cr3t",
+                                     "nested/extra": "deep"}),
+    }
+    # writer drops a file into the emptyDir and exits; reader (builtins
+    # only — the image carries a bare shell) waits for it, then checks the
+    # shared emptyDir and the secret projection
+    st = rt.deploy(DeployParams(
+        pod_key="default-vols", name="vols", volumes=vols,
+        containers=[
+            ContainerSpec(
+                name="writer", image="example/app:v1",
+                command=["/bin/sh"],
+                args=["-c", "echo shared-data > /scratch/f.txt"],
+                volume_mounts=[VolumeMount("scratch", "/scratch")]),
+            ContainerSpec(
+                name="reader", image="example/app:v1",
+                command=["/bin/sh"],
+                args=["-c",
+                      "until [ -f /scratch/f.txt ]; do :; done; "
+                      "read a < /scratch/f.txt; echo $a; "
+                      "read b < /etc/creds/token; echo $b; "
+                      "read c < /etc/creds/nested/extra; echo $c; "
+                      "( : > /etc/creds/illegal ) 2>/dev/null "
+                      "&& echo RW-LEAK || echo RO-OK"],
+                volume_mounts=[VolumeMount("scratch", "/scratch"),
+                               VolumeMount("creds", "/etc/creds")]),
+        ],
+    ))
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        reader = next((c for c in s.containers if c.name == "reader"), None)
+        if reader is not None and reader.exit_code is not None:
+            break
+        time.sleep(0.1)
+    out = rt.get_logs(st.id, "reader")
+    assert "shared-data" in out, out     # emptyDir shared across containers
+    assert "s3cr3t" in out, out          # secret file projected
+    assert "deep" in out, out            # items with nested path
+    assert "RO-OK" in out, out           # projection is read-only
+    rt.terminate(st.id)
+    wait_status(rt, st.id, PodStatus.TERMINATED)
+
+
+def test_hostpath_volume_and_chroot_refusal(image_runtime, tmp_path,
+                                            synthetic_ledger, tmp_state_dir,
+                                            image_store):
+    from k8s_runpod_kubelet_amd.ops import load_native
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        VolumeMount, VolumeSource)
+
+    hostdir = tmp_path / "hostdata"
+    hostdir.mkdir()
+    (hostdir / "from-host.txt").write_text("host-content")
+    vols = {"hp": VolumeSource(kind="hostPath", host_path=str(hostdir))}
+    mountvm = [VolumeMount("hp", "/mnt/host", read_only=True)]
+
+    if load_native().probe_mount_namespace():
+        rt = image_runtime
+        st = rt.deploy(DeployParams(
+            pod_key="default-hp", name="hp", volumes=vols,
+            containers=[ContainerSpec(
+                name="main", image="example/app:v1",
+                command=["/bin/sh"],
+                args=["-c", "read x < /mnt/host/from-host.txt; echo $x"],
+                volume_mounts=mountvm)],
+        ))
+        s = wait_status(rt, st.id, PodStatus.EXITED)
+        out = rt.get_logs(st.id)
+        assert s.exit_code == 0, out
+        assert "host-content" in out
+
+    # chroot mode cannot provide hostPath (no bind mounts): must refuse
+    # loudly, never run with the volume silently missing
+    rt2 = ProcessRuntime(synthetic_ledger, tmp_state_dir + "-ch",
+                         enable_cgroups=False, image_store=image_store,
+                         image_isolation="chroot")
+    try:
+        with pytest.raises(RuntimeError, match="hostPath"):
+            rt2.deploy(DeployParams(
+                pod_key="default-hp2", name="hp2", volumes=vols,
+                containers=[ContainerSpec(
+                    name="main", image="example/app:v1",
+                    command=["/bin/sh"], args=["-c", "echo hi"],
+                    volume_mounts=mountvm)],
+            ))
+    finally:
+        rt2.close()
+
+
+def test_chroot_mode_emptydir_and_files_volumes(synthetic_ledger,
+                                                tmp_state_dir, image_store):
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        VolumeMount, VolumeSource)
+
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                        enable_cgroups=False, image_store=image_store,
+                        image_isolation="chroot")
+    try:
+        st = rt.deploy(DeployParams(
+            pod_key="default-chv", name="chv",
+            volumes={"scratch": VolumeSource(kind="emptyDir"),
+                     "cfg": VolumeSource(kind="files",
+                                         files={"app.conf": "mode=prod"})},
+            containers=[ContainerSpec(
+                name="main", image="example/app:v1",
+                command=["/bin/sh"],
+                args=["-c",
+                      "echo w > /scratch/x && echo SCRATCH-OK; "
+                      "read v < /etc/cfg/app.conf; echo $v"],
+                volume_mounts=[VolumeMount("scratch", "/scratch"),
+                               VolumeMount("cfg", "/etc/cfg")])],
+        ))
+        s = wait_status(rt, st.id, PodStatus.EXITED)
+        out = rt.get_logs(st.id)
+        assert s.exit_code == 0, out
+        assert "SCRATCH-OK" in out
+        assert "mode=prod" in out
+    finally:
+        rt.close()
+
+
+def test_translate_volumes_from_pod_spec(fake_kube):
+    """spec.volumes/volumeMounts flow through translation: secret content
+    fetched, items projected, emptyDir + hostPath mapped."""
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.provider.translate import (
+        prepare_deploy_params)
+    from tests.conftest import make_pod
+
+    import base64
+
+    def b64(s):
+        return base64.b64encode(s.encode()).decode()
+
+    fake_kube.put_secret("default", {
+        "metadata": {"name": "app-secret"},
+        "data": {"token": b64("t0k3n"), "ignored": b64("x")}})
+    pod = make_pod("volpod", command=["podworker"], args=["--hold"])
+    pod["spec"]["volumes"] = [
+        {"name": "scratch", "emptyDir": {}},
+        {"name": "hostlibs", "hostPath": {"path": "/usr/lib"}},
+        {"name": "creds", "secret": {"secretName": "app-secret",
+                                     "items": [{"key": "token",
+                                                "path": "auth/token"}]}},
+    ]
+    pod["spec"]["containers"][0]["volumeMounts"] = [
+        {"name": "scratch", "mountPath": "/scratch"},
+        {"name": "creds", "mountPath": "/creds", "readOnly": True},
+    ]
+    params = prepare_deploy_params(pod, fake_kube, Config())
+    assert params.volumes["scratch"].kind == "emptyDir"
+    assert params.volumes["hostlibs"].kind == "hostPath"
+    assert params.volumes["hostlibs"].host_path == "/usr/lib"
+    assert params.volumes["creds"].kind == "files"
+    assert params.volumes["creds"].files == {"auth/token": "t0k3n"}
+    vm = params.containers[0].volume_mounts
+    assert [m.name for m in vm] == ["scratch", "creds"]
+    assert vm[1].read_only is True
