@@ -85,7 +85,7 @@ def _container_statuses(
         cs: Dict[str, Any] = {
             "name": name,
             "image": container.get("image", ""),
-            "imageID": "",
+            "imageID": rt.image_id if rt else "",
             "restartCount": rt.restart_count if rt else 0,
             "ready": ready,
         }

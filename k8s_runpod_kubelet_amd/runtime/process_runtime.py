@@ -406,6 +406,8 @@ class ProcessRuntime(Runtime):
         if inst.cgroup_dir:
             metrics.cgroup_migrate_seconds.observe(cgroup_s)
         cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
+        if image is not None:
+            cinfo.image_id = f"{image.ref}@{image.manifest_digest}" 
         # k8s readiness semantics: a running container WITHOUT a
         # readinessProbe is Ready as soon as it starts. The AMDVK_READY_FD
         # pipe protocol (readiness deferred until the workload signals its
@@ -1199,6 +1201,7 @@ class ProcessRuntime(Runtime):
                     "exit_code": c.exit_code,
                     "ready": c.ready,
                     "restart_count": c.restart_count,
+                    "image_id": c.image_id,
                 }
                 for c in inst.containers
             ],
@@ -1349,6 +1352,7 @@ class ProcessRuntime(Runtime):
                     finished_at=c.get("finished_at", 0.0),
                     exit_code=c.get("exit_code"), ready=c.get("ready", False),
                     restart_count=c.get("restart_count", 0),
+                    image_id=c.get("image_id", ""),
                 )
                 if cinfo.exit_code is None:
                     pidfd = self._native.open_pidfd(cinfo.pid)

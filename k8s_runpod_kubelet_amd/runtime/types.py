@@ -42,6 +42,9 @@ class ContainerRuntimeInfo:
     restart_count: int = 0
     backoff_until: float = 0.0
     crash_streak: int = 0
+    # resolved image identity for status.imageID (ref@manifest-digest),
+    # empty for host-binary containers
+    image_id: str = ""
 
 
 @dataclass

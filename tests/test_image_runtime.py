@@ -581,3 +581,63 @@ def test_translate_volumes_from_pod_spec(fake_kube):
     vm = params.containers[0].volume_mounts
     assert [m.name for m in vm] == ["scratch", "creds"]
     assert vm[1].read_only is True
+
+
+def test_image_pod_adoption_across_kubelet_restart(synthetic_ledger,
+                                                   tmp_state_dir,
+                                                   image_store):
+    """A live image pod survives a kubelet restart: re-adopted with its
+    image_mode, exec still enters the container, imageID kept, and a
+    post-restart crash restarts inside the SAME per-container rootfs."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                        enable_cgroups=False, image_store=image_store)
+    st = rt.deploy(DeployParams(
+        pod_key="default-adimg", name="adimg", restart_policy="Always",
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/usr/local/bin/app"], args=["hold"])],
+    ))
+    wait_status(rt, st.id, PodStatus.RUNNING)
+    assert rt.get_detailed_status(st.id).containers[0].image_id.startswith(
+        "docker.io/example/app:v1@sha256:")
+    # kubelet "crash"
+    rt._stop.set(); rt._loop.wake(); rt._watcher.join(timeout=2)
+
+    rt2 = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                         enable_cgroups=False, image_store=image_store)
+    try:
+        rt2.adopt_persisted()
+        s = rt2.get_detailed_status(st.id)
+        assert s.desired_status == PodStatus.RUNNING
+        assert s.containers[0].image_id.startswith("docker.io/example")
+        with rt2._lock:
+            assert rt2._instances[st.id].image_mode == "mountns"
+        # exec enters the adopted container's namespaces
+        code, out = rt2.exec_in_instance(
+            st.id, ["/usr/local/bin/app", "q"], timeout_s=15)
+        assert code == 0, out
+        assert "image-file=present" in out
+        # kill the container: restartPolicy relaunches it in its rootfs
+        pid = s.containers[0].pid
+        import signal as _sig
+
+        os.kill(pid, _sig.SIGKILL)
+        deadline = time.time() + 20
+        ok = False
+        while time.time() < deadline:
+            s2 = rt2.get_detailed_status(st.id)
+            c = s2.containers[0]
+            if c.restart_count >= 1 and c.exit_code is None and c.pid != pid:
+                ok = True
+                break
+            time.sleep(0.05)
+        assert ok, rt2.get_logs(st.id)
+        rt2.terminate(st.id)
+        wait_status(rt2, st.id, PodStatus.TERMINATED)
+    finally:
+        rt2.close()
+        rt.close()
