@@ -356,6 +356,15 @@ class Provider:
         )
         detailed = self.runtime.deploy(params)
         metrics.deploy_seconds.observe(time.monotonic() - t0)
+        # kubelet-style container lifecycle events: Pulled for image-backed
+        # containers (the store resolved it locally), then Started with the
+        # placement info (kubectl describe surface).
+        images_by_name = {c.name: c.image for c in params.containers}
+        for c in detailed.containers:
+            if c.image_id:
+                self._emit(pod, "Normal", "Pulled",
+                           f'Container image "{images_by_name.get(c.name, "")}"'
+                           f" already present on machine ({c.image_id})")
         self._emit(pod, "Normal", "Started",
                    f"instance {detailed.id} on GPUs {detailed.gpu_indices}"
                    if detailed.gpu_indices else f"instance {detailed.id}")
@@ -489,6 +498,15 @@ class Provider:
         if not changed:
             return
 
+        if restart_sig != info.restart_sig:
+            # a container just entered crash backoff: kubectl-visible
+            # BackOff event, like a real kubelet's
+            for c in detailed.containers:
+                if c.backoff_until:
+                    self._emit(pod, "Warning", "BackOff",
+                               f"Back-off restarting failed container "
+                               f"{c.name} (exit {c.exit_code}, "
+                               f"restarts {c.restart_count})")
         info.status = detailed.desired_status
         info.ports_exposed = ports_exposed
         info.restart_sig = restart_sig

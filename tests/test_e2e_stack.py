@@ -630,3 +630,26 @@ def test_bare_pod_defaults_to_always_and_crash_loops(stack):
     assert p["status"]["phase"] != "Succeeded"
     kube.delete_pod("default", "bare")
     assert wait_until(lambda: gone(kube, "bare"), timeout_s=20)
+
+
+def test_backoff_event_emitted(stack):
+    """A crash-looping container produces a kubectl-visible Warning/BackOff
+    event (real-kubelet surface)."""
+    s, kube = stack
+    pod = make_pod("loopev", command=["podworker"],
+                   args=["--run-for", "0.05", "--exit-code", "1"])
+    pod["spec"]["restartPolicy"] = "OnFailure"
+    kube.create_pod("default", pod)
+
+    def backoff_event():
+        evs = [e for e in kube.events.objects.values()
+               if e["involvedObject"]["name"] == "loopev"
+               and e["reason"] == "BackOff"]
+        return evs or None
+
+    evs = wait_until(backoff_event, timeout_s=20)
+    assert evs is not None
+    assert evs[0]["type"] == "Warning"
+    assert "Back-off restarting" in evs[0]["message"]
+    kube.delete_pod("default", "loopev")
+    assert wait_until(lambda: gone(kube, "loopev"), timeout_s=20)

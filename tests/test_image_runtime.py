@@ -319,6 +319,11 @@ def test_image_pod_end_to_end_stack(tmp_state_dir, app_bin, tmp_path):
         logs = stack.provider.get_container_logs("default", "imaged")
         assert "image-file=present" in logs
         assert "host-python=absent" in logs
+        # kubelet-style Pulled event for the locally-resolved image
+        pulled = [e for e in kube.events.objects.values()
+                  if e["involvedObject"]["name"] == "imaged"
+                  and e["reason"] == "Pulled"]
+        assert pulled and "example/holder:v3" in pulled[0]["message"]
         kube.delete_pod("default", "imaged")
 
         def gone():
