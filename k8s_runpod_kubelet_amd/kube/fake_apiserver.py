@@ -49,10 +49,15 @@ class FakeApiServer:
     """Threaded HTTP server delegating every route to a FakeKube."""
 
     def __init__(self, kube: Optional[FakeKube] = None,
-                 host: str = "127.0.0.1", port: int = 0):
+                 host: str = "127.0.0.1", port: int = 0,
+                 required_token: str = ""):
         self.kube = kube if kube is not None else FakeKube()
         self.host = host
         self.port = port
+        # Bearer-token enforcement (mutable at runtime — tests rotate it
+        # to exercise the client's 401-refresh path, like a real apiserver
+        # after a bound SA token expires)
+        self.required_token = required_token
         self._server: Optional[ThreadingHTTPServer] = None
         self._thread: Optional[threading.Thread] = None
 
@@ -68,6 +73,30 @@ class FakeApiServer:
 
             def log_message(self, fmt, *args):
                 log.debug(fmt % args)
+
+            def _authorized(self) -> bool:
+                if not outer.required_token:
+                    return True
+                return (self.headers.get("Authorization", "")
+                        == f"Bearer {outer.required_token}")
+
+            def parse_request(self):
+                # enforce auth uniformly before routing
+                ok = super().parse_request()
+                if ok and not self._authorized():
+                    body = json.dumps({"kind": "Status", "code": 401,
+                                       "message": "Unauthorized"}).encode()
+                    self.send_response(401)
+                    self.send_header("Content-Type", "application/json")
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                    # drain request body if any so keep-alive stays sane
+                    length = int(self.headers.get("Content-Length", 0) or 0)
+                    if length:
+                        self.rfile.read(length)
+                    return False
+                return ok
 
             # -- plumbing --
 

@@ -32,13 +32,18 @@ SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
 
 class ClusterConfig:
     def __init__(self, server: str, token: str = "", ca_path: str = "",
-                 client_cert: str = "", client_key: str = "", verify: bool = True):
+                 client_cert: str = "", client_key: str = "", verify: bool = True,
+                 token_path: str = ""):
         self.server = server.rstrip("/")
         self.token = token
         self.ca_path = ca_path
         self.client_cert = client_cert
         self.client_key = client_key
         self.verify = verify
+        # service-account token file: re-read on 401 (kubelet semantics —
+        # bound SA tokens rotate; a client that caches the startup token
+        # goes Unauthorized after ~1 h on a real cluster)
+        self.token_path = token_path
 
 
 def _materialize(data_b64: str, suffix: str) -> str:
@@ -92,6 +97,7 @@ def load_in_cluster() -> Optional[ClusterConfig]:
         server=f"https://{host}:{port}",
         token=token,
         ca_path=os.path.join(SA_DIR, "ca.crt"),
+        token_path=token_path,
     )
 
 
@@ -156,6 +162,23 @@ class HttpK8sClient(K8sClient):
             return {}
         return resp.json()
 
+    def _refresh_token(self) -> bool:
+        """Re-read a rotated service-account token (in-cluster only).
+        Returns True when the Authorization header changed."""
+        if not self.cc.token_path:
+            return False
+        try:
+            with open(self.cc.token_path, "r", encoding="utf-8") as fh:
+                token = fh.read().strip()
+        except OSError:
+            return False
+        if not token or token == self.cc.token:
+            return False
+        self.cc.token = token
+        self._http.headers["Authorization"] = f"Bearer {token}"
+        log.info("service-account token refreshed")
+        return True
+
     def _request(self, method: str, path: str, *, params=None, json_body=None,
                  content_type: str = "application/json") -> Dict[str, Any]:
         def attempt() -> Dict[str, Any]:
@@ -165,6 +188,13 @@ class HttpK8sClient(K8sClient):
                 content=json.dumps(json_body) if json_body is not None else None,
                 headers=headers,
             )
+            if resp.status_code == 401 and self._refresh_token():
+                resp = self._http.request(
+                    method, path, params=params,
+                    content=json.dumps(json_body)
+                    if json_body is not None else None,
+                    headers=headers,
+                )
             return self._check(resp)
 
         # 404/409 are terminal-valid outcomes (reference treats 200/404 as
@@ -236,6 +266,10 @@ class HttpK8sClient(K8sClient):
             params["resourceVersion"] = resource_version
         with self._http.stream("GET", path, params=params,
                                timeout=timeout_s + 10) as resp:
+            if resp.status_code == 401:
+                # rotated SA token: refresh for the caller's retry (the
+                # informer re-establishes the watch on error)
+                self._refresh_token()
             if resp.status_code >= 400:
                 resp.read()
                 self._check(resp)

@@ -200,3 +200,38 @@ def test_http_watch_rv_continuity_and_410():
     finally:
         client.close()
         srv.stop()
+
+
+def test_service_account_token_rotation(tmp_path):
+    """Bound SA tokens rotate (~1 h on real clusters): a client caching the
+    startup token goes Unauthorized until it re-reads the token file. The
+    client must refresh on 401 and retry transparently."""
+    import pytest as _pytest
+
+    from k8s_runpod_kubelet_amd.kube.client import ApiError
+
+    token_file = tmp_path / "token"
+    token_file.write_text("token-v1\n")
+    srv = FakeApiServer(required_token="token-v1").start()
+    client = HttpK8sClient(ClusterConfig(
+        server=srv.url, token="token-v1", token_path=str(token_file)))
+    try:
+        client.create_pod("default", make_pod("rotpod"))
+        # rotate: apiserver now only accepts v2, the file has v2, but the
+        # client still holds v1
+        srv.required_token = "token-v2"
+        token_file.write_text("token-v2\n")
+        pod = client.get_pod("default", "rotpod")  # 401 -> refresh -> retry
+        assert pod["metadata"]["name"] == "rotpod"
+
+        # rotation where the FILE is stale too: hard 401 surfaces
+        srv.required_token = "token-v3"
+        with _pytest.raises(ApiError) as ei:
+            client.get_pod("default", "rotpod")
+        assert ei.value.status_code == 401
+        # once the file catches up, the next call recovers
+        token_file.write_text("token-v3\n")
+        assert client.get_pod("default", "rotpod")
+    finally:
+        client.close()
+        srv.stop()
