@@ -49,6 +49,13 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--runtime", choices=["process", "fake"], default="process")
     p.add_argument("--workers", type=int, default=4,
                    help="pod controller workers (reference uses 1)")
+    p.add_argument("--image-store-dir", default="",
+                   help="node-local OCI image store (default: "
+                        "<state-dir>/images)")
+    p.add_argument("--image-isolation", choices=["auto", "mountns", "chroot"],
+                   default="auto",
+                   help="image rootfs isolation: overlay+mount-ns+pivot_root "
+                        "(mountns), chroot fallback, or probe (auto)")
     p.add_argument("--gpu-count-override", type=int, default=-1,
                    help="force synthetic GPU inventory (CPU-only dev)")
     p.add_argument("--fake-apiserver", action="store_true",
@@ -90,6 +97,9 @@ def main(argv=None) -> int:
     cfg.runtime = args.runtime
     cfg.pod_controller_workers = args.workers
     cfg.gpu_count_override = args.gpu_count_override
+    if args.image_store_dir:
+        cfg.image_store_dir = args.image_store_dir
+    cfg.image_isolation = args.image_isolation
 
     validate_environment()
     fake_srv = None
