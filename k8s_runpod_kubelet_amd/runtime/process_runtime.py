@@ -331,6 +331,11 @@ class ProcessRuntime(Runtime):
             if uid < 0 and gid < 0 and image.config.user:
                 uid, gid = self.image_store.resolve_user(
                     self.image_store.rootfs_for(image), image.config.user)
+                # record the image-derived identity on the spec so exec
+                # probes / kubectl-exec run as the container's user (k8s
+                # runs both inside the container), and restarts/adoption
+                # keep it (it is persisted with the spec)
+                cspec.run_as_uid, cspec.run_as_gid = uid, gid
             prepared = self._rootfs_mgr.prepare(
                 inst.id, cspec.name, image,
                 params.hostname or params.name,

@@ -646,3 +646,31 @@ def test_image_pod_adoption_across_kubelet_restart(synthetic_ledger,
     finally:
         rt2.close()
         rt.close()
+
+
+def test_exec_runs_as_image_user(image_runtime):
+    """kubectl-exec (and exec probes) on a pod whose IMAGE sets User must
+    run as that user, not as the kubelet (root) — the identity resolved at
+    launch is recorded on the container spec."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    store = rt.image_store
+    img = store.resolve("example/app:v1")
+    tree_root = store.rootfs_for(img)
+    layout = Path(str(tree_root) + "-lexec")
+    layout.mkdir(exist_ok=True)
+    build_layout(str(layout), "example/appexec:v1", str(tree_root),
+                 entrypoint=["/usr/local/bin/app"], user="svc")
+    store.add_layout(str(layout), "example/appexec:v1")
+    st = deploy_image_pod(rt, "execuser", image="example/appexec:v1",
+                          args=["hold"])
+    time.sleep(0.3)
+    code, out = rt.exec_in_instance(
+        st.id, ["/usr/local/bin/app", "q"], timeout_s=15)
+    assert code == 0, out
+    assert "uid=1234 gid=4321" in out, out   # image User=svc, not root
+    rt.terminate(st.id)
+    wait_status(rt, st.id, PodStatus.TERMINATED)
