@@ -979,8 +979,21 @@ class Provider:
                 "conditions": conditions,
                 "addresses": [{"type": "InternalIP", "address": self.config.internal_ip}],
                 "daemonEndpoints": {"kubeletEndpoint": {"Port": self.config.listen_port}},
+                # node.status.images from the local OCI store (kubectl
+                # describe node surface; a real kubelet lists its cache)
+                "images": self._node_images(),
             },
         }
+
+    def _node_images(self) -> List[Dict[str, Any]]:
+        store = getattr(self.runtime, "image_store", None)
+        if store is None:
+            return []
+        try:
+            return [{"names": [ref], "sizeBytes": size}
+                    for ref, size in store.image_sizes()]
+        except Exception:
+            return []
 
     # ------------------------------------------------------------------
     # logs / exec (reference stubs kubelet.go:2027-2066 — real here)

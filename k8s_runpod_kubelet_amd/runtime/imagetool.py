@@ -27,6 +27,12 @@ def main(argv=None) -> int:
 
     sub.add_parser("list", help="list stored image references")
 
+    p_rm = sub.add_parser("rm", help="remove an image reference")
+    p_rm.add_argument("ref")
+
+    sub.add_parser("gc", help="reclaim unpacked-rootfs cache space for "
+                              "removed images")
+
     p_imp = sub.add_parser("import", help="import an oci-archive tar")
     p_imp.add_argument("tar")
     p_imp.add_argument("--ref", default="",
@@ -56,8 +62,19 @@ def main(argv=None) -> int:
     store = ImageStore(args.store or Config().resolved_image_store_dir())
 
     if args.cmd == "list":
-        for ref in store.list_refs():
-            print(ref)
+        for ref, size in store.image_sizes():
+            print(f"{ref}\t{size >> 20} MiB")
+        return 0
+    if args.cmd == "rm":
+        if not store.remove(args.ref):
+            print(f"not found: {args.ref}", file=sys.stderr)
+            return 1
+        removed = store.gc()
+        print(f"removed {args.ref} (+{len(removed)} cache entries)")
+        return 0
+    if args.cmd == "gc":
+        removed = store.gc()
+        print(f"removed {len(removed)} unreferenced cache entries")
         return 0
     if args.cmd == "import":
         ref = store.import_archive(args.tar, args.ref)

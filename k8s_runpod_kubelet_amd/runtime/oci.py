@@ -285,6 +285,59 @@ class ImageStore:
                  extra={"ref": image.ref, "digest": image.manifest_digest})
         return dst
 
+    def image_sizes(self) -> List[Tuple[str, int]]:
+        """(ref, total blob bytes) per stored image — feeds
+        node.status.images (kubectl describe node surface)."""
+        out: List[Tuple[str, int]] = []
+        if not self.layouts_dir.is_dir():
+            return out
+        for d in sorted(self.layouts_dir.iterdir()):
+            ref_file = d / ".amdvk-ref"
+            if not ref_file.exists():
+                continue
+            size = 0
+            blobs = d / "blobs"
+            if blobs.is_dir():
+                for blob in blobs.rglob("*"):
+                    if blob.is_file():
+                        size += blob.stat().st_size
+            out.append((ref_file.read_text().strip(), size))
+        return out
+
+    def remove(self, ref: str) -> bool:
+        """Unregister an image reference (its layout); the shared rootfs
+        cache is reclaimed by gc()."""
+        layout = self._layout_for(ref)
+        if layout is None:
+            return False
+        shutil.rmtree(layout, ignore_errors=True)
+        return True
+
+    def gc(self) -> List[str]:
+        """Remove unpacked-rootfs cache entries whose manifest digest no
+        longer belongs to any stored image (kubelet image-GC analogue).
+        Returns the removed digests. Never touches per-pod container dirs
+        — those belong to the runtime."""
+        live = set()
+        for ref in self.list_refs():
+            try:
+                img = self.resolve(ref)
+            except ImageError:
+                continue
+            if img is not None:
+                live.add(img.manifest_digest.replace(":", "-"))
+        removed = []
+        if self.rootfs_dir.is_dir():
+            for d in self.rootfs_dir.iterdir():
+                if d.name.startswith("."):
+                    continue  # in-flight unpack
+                if d.name not in live:
+                    shutil.rmtree(d, ignore_errors=True)
+                    removed.append(d.name)
+        if removed:
+            log.info("image cache gc", extra={"removed": len(removed)})
+        return removed
+
     def resolve_user(self, rootfs: Path, user: str) -> Tuple[int, int]:
         """OCI config User ('uid', 'uid:gid', 'name', 'name:group') ->
         numeric (uid, gid); (-1, -1) when unset (inherit)."""

@@ -196,3 +196,52 @@ def test_changed_type_replaces(store, tmp_path):
     assert (rootfs / "thing").read_text() == "now-a-file"
     assert (rootfs / "plain").is_dir()
     assert (rootfs / "plain/sub").read_text() == "now-a-dir"
+
+
+def test_image_sizes_remove_gc(store, tmp_path):
+    build_and_add(store, tmp_path, "g1", "example/keep:v1",
+                  {"a.txt": "x" * 1000})
+    build_and_add(store, tmp_path, "g2", "example/drop:v1",
+                  {"b.txt": "y" * 2000})
+    sizes = dict(store.image_sizes())
+    assert sizes["docker.io/example/keep:v1"] > 1000
+    assert sizes["docker.io/example/drop:v1"] > 2000
+    # unpack both so the cache has two entries
+    keep = store.resolve("example/keep:v1")
+    drop = store.resolve("example/drop:v1")
+    store.rootfs_for(keep)
+    store.rootfs_for(drop)
+    assert store.remove("example/drop:v1")
+    assert not store.remove("example/drop:v1")  # already gone
+    removed = store.gc()
+    assert drop.manifest_digest.replace(":", "-") in removed
+    # kept image untouched and still resolvable/unpacked
+    assert store.resolve("example/keep:v1") is not None
+    assert (store.rootfs_for(keep) / "a.txt").exists()
+    assert store.gc() == []  # idempotent
+
+
+def test_node_status_lists_images(tmp_path, store):
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.gpu.inventory import Inventory
+    from k8s_runpod_kubelet_amd.gpu.ledger import Ledger
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.provider.provider import Provider
+    from k8s_runpod_kubelet_amd.runtime.fake import FakeRuntime
+
+    build_and_add(store, tmp_path, "n1", "example/visible:v2",
+                  {"f": "data"})
+    rt = FakeRuntime(gpu_count=2)
+    rt.image_store = store  # duck-typed like ProcessRuntime
+    inv = Inventory(synthetic_count=2)
+    inv.discover()
+    ledger = Ledger(inv)
+    ledger.sync_inventory()
+    prov = Provider(FakeKube(), Config(), rt, ledger=ledger, inventory=inv)
+    try:
+        node = prov.get_node_status()
+        images = node["status"]["images"]
+        assert any("docker.io/example/visible:v2" in im["names"]
+                   and im["sizeBytes"] > 0 for im in images)
+    finally:
+        prov.stop()
