@@ -25,7 +25,7 @@ from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 from typing import Any, Dict, Optional, Tuple
 from urllib.parse import parse_qs, urlparse
 
-from .client import ApiError, ConflictError, GoneError, NotFoundError
+from .client import ConflictError, GoneError, NotFoundError
 from .fake import FakeKube
 
 log = logging.getLogger("kube.fake_apiserver")

@@ -51,7 +51,7 @@ from .instance import InstanceInfo
 from .ports import check_ports_exposed
 from .registration import Registrar
 from .selector import GpuOfferCatalog
-from .status import merge_container_status, translate_status, translate_status_to_phase
+from .status import merge_container_status, translate_status
 from .translate import ValidationError, prepare_deploy_params
 
 log = logging.getLogger("provider")

@@ -31,7 +31,6 @@ import json
 import logging
 import os
 import shutil
-import stat
 import tarfile
 import tempfile
 from dataclasses import dataclass, field

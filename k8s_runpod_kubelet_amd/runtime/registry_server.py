@@ -13,7 +13,6 @@ reference annotations; manifests are also addressable by digest."""
 from __future__ import annotations
 
 import argparse
-import json
 import logging
 import re
 import sys
