@@ -674,3 +674,55 @@ def test_exec_runs_as_image_user(image_runtime):
     assert "uid=1234 gid=4321" in out, out   # image User=svc, not root
     rt.terminate(st.id)
     wait_status(rt, st.id, PodStatus.TERMINATED)
+
+
+def test_image_churn_no_leaks(image_runtime):
+    """Mini-soak: 24 image pods churned through deploy/crash/terminate —
+    no leaked GPU reservations, tracked processes, or per-container rootfs
+    dirs afterwards (the r1 soak methodology applied to the image path)."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    ids = []
+    for i in range(24):
+        kind = i % 3
+        if kind == 0:      # run-to-completion
+            st = deploy_image_pod(rt, f"ch{i}")
+        elif kind == 1:    # holder, terminated mid-flight
+            st = deploy_image_pod(rt, f"ch{i}",
+                                  command=["/usr/local/bin/app"],
+                                  args=["hold"])
+        else:              # failing container
+            st = rt.deploy(DeployParams(
+                pod_key=f"default-ch{i}", name=f"ch{i}",
+                containers=[ContainerSpec(
+                    name="main", image="example/app:v1",
+                    command=["/bin/sh"], args=["-c", "exit 7"])],
+            ))
+        ids.append((st.id, kind))
+    time.sleep(0.5)
+    for iid, kind in ids:
+        if kind == 1:
+            rt.terminate(iid)
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        states = [rt.get_detailed_status(i).desired_status for i, _ in ids]
+        if all(s in (PodStatus.EXITED, PodStatus.TERMINATED)
+               for s in states):
+            break
+        time.sleep(0.1)
+    assert all(rt.get_detailed_status(i).desired_status
+               in (PodStatus.EXITED, PodStatus.TERMINATED)
+               for i, _ in ids), states
+    for iid, _ in ids:
+        rt.remove(iid)
+    # leak checks
+    assert rt.tracked_process_count() == 0
+    assert not rt.ledger.reservations
+    leftovers = list((Path(rt.state_dir) / "containers").glob("amdvk-*"))
+    assert not leftovers, leftovers
+    vol_leftovers = list((Path(rt.state_dir) / "volumes").glob("amdvk-*")) \
+        if (Path(rt.state_dir) / "volumes").exists() else []
+    assert not vol_leftovers, vol_leftovers
