@@ -58,6 +58,9 @@ class PodController:
             t.start()
             self._threads.append(t)
         self.provider.notify_pods(self._notify_from_provider)
+        # deleted pods re-enqueue when their instance turns terminal, so
+        # the API delete completes the moment the containers are dead
+        self.provider.deletion_resync = self.queue.add
         self._ready.set()
         log.info("pod controller started", extra={"workers": self.workers})
 
@@ -120,7 +123,14 @@ class PodController:
             return
 
         if deletion_timestamp(pod):
-            self.provider.delete_pod(pod)
+            self.provider.delete_pod(pod)  # idempotent: record + terminate
+            # k8s semantics: the API object stays Terminating until the
+            # containers are actually dead — a TERM-ignoring container
+            # holds the object through its grace period. The instance's
+            # exit event re-enqueues this key (provider.deletion_resync);
+            # informer resync is the fallback.
+            if not self.provider.deletion_finalized(ns, name):
+                return
             try:
                 self.client.delete_pod(ns, name, grace_period_s=0)
             except Exception as exc:

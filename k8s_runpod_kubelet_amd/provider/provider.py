@@ -98,6 +98,9 @@ class Provider:
         # the reference gets from the virtual-kubelet library's recorder
         # (main.go:172-177).
         self.recorder = None
+        # Set by the PodController: re-enqueue a deleted pod ("ns/name")
+        # when its instance turns terminal, completing the API delete.
+        self.deletion_resync: Optional[Callable[[str], None]] = None
 
         self.registrar = Registrar(
             config.registration_endpoint,
@@ -251,6 +254,24 @@ class Provider:
             self._pods.pop(key, None)
             self._pod_status.pop(key, None)
         metrics.pods_deleted.inc()
+
+    def deletion_finalized(self, namespace: str, name: str) -> bool:
+        """True once the pod's instance is actually dead (terminal or
+        unknown) — the gate the PodController uses before removing the API
+        object. Real k8s keeps a pod Terminating until its containers are
+        gone; deleting the object while a TERM-ignoring container waits out
+        its grace period would lie to kubectl (found by the image-pod
+        churn soak: pid-1-in-namespace entrypoints ignore SIGTERM)."""
+        with self._deleted_lock:
+            instance_id = self._deleted_pods.get(f"{namespace}/{name}")
+        if not instance_id:
+            return True
+        try:
+            status = self.runtime.get_status(instance_id)
+        except Exception:
+            return True
+        return status in (PodStatus.EXITED, PodStatus.TERMINATED,
+                          PodStatus.NOT_FOUND)
 
     def get_pod(self, namespace: str, name: str) -> Optional[Dict[str, Any]]:
         """GetPod (reference kubelet.go:654-667)."""
@@ -431,6 +452,18 @@ class Provider:
             status = PodStatus.NOT_FOUND
         if status in (PodStatus.EXITED, PodStatus.TERMINATED, PodStatus.NOT_FOUND):
             self._nudge_pending()
+            # Deletion finalize: a deleted pod's API object is held until
+            # its instance is dead (deletion_finalized); this event is what
+            # re-enqueues it so the PodController can complete the delete.
+            if self.deletion_resync is not None:
+                with self._deleted_lock:
+                    fkey = next((k for k, iid in self._deleted_pods.items()
+                                 if iid == instance_id), None)
+                if fkey is not None:
+                    try:
+                        self.deletion_resync(fkey)
+                    except Exception:
+                        log.exception("deletion resync failed")
 
     def _emit(self, pod: Dict[str, Any], etype: str, reason: str,
               message: str) -> None:

@@ -35,6 +35,9 @@ def main() -> int:
     ap.add_argument("--restart-kubelet-at", type=float, default=0.0,
                     help="simulate a kubelet crash+restart T seconds in "
                          "(pods untouched; adoption must resume everything)")
+    ap.add_argument("--images", action="store_true",
+                    help="mix OCI image pods (overlay+pivot_root) into the "
+                         "churn (needs mount-ns capability and gcc)")
     args = ap.parse_args()
     rng = random.Random(args.seed)
 
@@ -55,6 +58,40 @@ def main() -> int:
         notify_interval_s=0,
         pod_controller_workers=8,
     )
+    image_ref = ""
+    if args.images:
+        import shutil
+        import subprocess
+
+        from k8s_runpod_kubelet_amd.ops import load_native
+        from k8s_runpod_kubelet_amd.runtime.oci import ImageStore, build_layout
+
+        if not load_native().probe_mount_namespace():
+            print("--images needs mount-namespace capability", file=sys.stderr)
+            return 2
+        bdir = tempfile.mkdtemp(prefix="amdvk-soakimg-")
+        with open(os.path.join(bdir, "app.c"), "w") as fh:
+            fh.write(
+                '#include <unistd.h>\n#include <stdio.h>\n'
+                '#include <string.h>\n'
+                'int main(int c, char** v) {\n'
+                '  printf("img-ready\\n"); fflush(stdout);\n'
+                '  if (c > 1 && !strcmp(v[1], "hold")) pause();\n'
+                '  return 0;\n}\n')
+        subprocess.run(["gcc", "-static", "-O1", "-o",
+                        os.path.join(bdir, "app"),
+                        os.path.join(bdir, "app.c")], check=True)
+        tree = os.path.join(bdir, "tree")
+        os.makedirs(os.path.join(tree, "bin"))
+        shutil.copy2(os.path.join(bdir, "app"),
+                     os.path.join(tree, "bin", "app"))
+        layout = os.path.join(bdir, "layout")
+        os.makedirs(layout)
+        build_layout(layout, "soak/img:v1", tree, entrypoint=["/bin/app"])
+        ImageStore(cfg.resolved_image_store_dir()).add_layout(
+            layout, "soak/img:v1")
+        image_ref = "soak/img:v1"
+
     stack = build_stack(cfg, client=client)
     stack.start(serve_http=False)
 
@@ -62,6 +99,31 @@ def main() -> int:
     pw_args = (["--expect-gpus", "1"] if gpus else [])
 
     def make(name, mode):
+        if mode == "img-ok":
+            # image pod, no command: the image entrypoint runs in its rootfs
+            return {
+                "apiVersion": "v1", "kind": "Pod",
+                "metadata": {"name": name, "namespace": "default"},
+                "spec": {"nodeName": cfg.node_name,
+                         "restartPolicy": "Never",
+                         "containers": [{"name": "main",
+                                         "image": image_ref}]},
+            }
+        if mode == "img-hold":
+            # pid-1-in-namespace ignores default-action TERM: the delete
+            # path exercises the grace->SIGKILL ladder AND the
+            # object-visible-until-dead semantics every time; short grace
+            # keeps the churn brisk
+            return {
+                "apiVersion": "v1", "kind": "Pod",
+                "metadata": {"name": name, "namespace": "default"},
+                "spec": {"nodeName": cfg.node_name,
+                         "restartPolicy": "Never",
+                         "terminationGracePeriodSeconds": 2,
+                         "containers": [{"name": "main", "image": image_ref,
+                                         "command": ["/bin/app"],
+                                         "args": ["hold"]}]},
+            }
         if mode in ("hold", "crash", "probed"):
             a = pw_args + ["--hold"]
         elif mode == "ok":
@@ -139,9 +201,13 @@ def main() -> int:
                 gc.collect()
             # top up population (only while inside the window)
             while now < deadline and len(active) < args.max_active:
-                mode = rng.choices(
-                    ["hold", "ok", "fail", "crash", "probed", "restarting"],
-                    weights=[3, 4, 2, 1, 2, 1])[0]
+                modes = ["hold", "ok", "fail", "crash", "probed",
+                         "restarting"]
+                weights = [3, 4, 2, 1, 2, 1]
+                if image_ref:
+                    modes += ["img-ok", "img-hold"]
+                    weights += [3, 2]
+                mode = rng.choices(modes, weights=weights)[0]
                 name = f"soak-{seq:05d}"
                 seq += 1
                 client.create_pod("default", make(name, mode))
@@ -171,8 +237,9 @@ def main() -> int:
                         pass
                     del active[name]
                     continue
-                if st["mode"] in ("ok", "fail"):
-                    want = "Succeeded" if st["mode"] == "ok" else "Failed"
+                if st["mode"] in ("ok", "fail", "img-ok"):
+                    want = ("Failed" if st["mode"] == "fail"
+                            else "Succeeded")
                     if state == want and not st["deleted"]:
                         counters["succeeded" if want == "Succeeded"
                                  else "failed"] += 1
@@ -197,7 +264,7 @@ def main() -> int:
                         counters["restarted"] += 1
                         st["deleted"] = True
                         client.delete_pod("default", name)
-                else:  # hold / probed
+                else:  # hold / probed / img-hold
                     if state == "Ready" and age > st["dwell"] and not st["deleted"]:
                         counters["deleted_holds"] += 1
                         st["deleted"] = True

@@ -653,3 +653,34 @@ def test_backoff_event_emitted(stack):
     assert "Back-off restarting" in evs[0]["message"]
     kube.delete_pod("default", "loopev")
     assert wait_until(lambda: gone(kube, "loopev"), timeout_s=20)
+
+
+def test_term_ignoring_pod_stays_until_grace_kill(stack):
+    """k8s semantics: a deleted pod whose container ignores SIGTERM stays
+    visible (Terminating) until the grace-period SIGKILL actually ends it —
+    the API object must not vanish while the process lives (found by the
+    image-pod churn soak: pid-1 entrypoints ignore default-action TERM)."""
+    import os
+    import time
+
+    s, kube = stack
+    pod = make_pod("stubborn2", command=["/bin/sh"],
+                   args=["-c", "trap '' TERM; sleep 60"])
+    pod["spec"]["terminationGracePeriodSeconds"] = 1
+    kube.create_pod("default", pod)
+    assert wait_until(lambda: ready(kube, "stubborn2"), timeout_s=15)
+    info = s.provider.instance_info("default", "stubborn2")
+    det = s.runtime.get_detailed_status(info.instance_id)
+    pid = det.containers[0].pid
+    t0 = time.time()
+    kube.delete_pod("default", "stubborn2")
+    # while the process is alive, the pod must still exist in the apiserver
+    time.sleep(0.4)
+    assert os.path.exists(f"/proc/{pid}"), "expected TERM to be ignored"
+    assert not gone(kube, "stubborn2"), \
+        "API object removed while the container was still running"
+    # after the 1 s grace SIGKILL, deletion completes promptly
+    assert wait_until(lambda: gone(kube, "stubborn2"), timeout_s=15)
+    took = time.time() - t0
+    assert took >= 0.9, f"deleted before the grace window ({took:.2f}s)"
+    assert not os.path.exists(f"/proc/{pid}")

@@ -302,6 +302,10 @@ def test_image_pod_end_to_end_stack(tmp_state_dir, app_bin, tmp_path):
         pod = make_pod("imaged", restart_policy=None)
         pod["spec"]["containers"][0] = {
             "name": "main", "image": "example/holder:v3"}
+        # the static entrypoint is pid 1 in its namespace and ignores
+        # default-action TERM; with k8s-correct deletion the object stays
+        # until the grace SIGKILL — keep the test brisk
+        pod["spec"]["terminationGracePeriodSeconds"] = 1
         kube.create_pod("default", pod)
 
         def ready():
