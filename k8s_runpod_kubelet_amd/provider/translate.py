@@ -190,8 +190,9 @@ def extract_gpu_memory_gb(pod: Dict[str, Any], job: Optional[Dict[str, Any]],
             text = text[: -len(suffix)]
             break
     try:
+        # OverflowError covers float('inf')-family strings ("INF", "1e999")
         return max(0, int(float(text)))
-    except ValueError:
+    except (ValueError, OverflowError):
         log.warning("unparseable gpu-memory annotation", extra={"value": raw})
         return default_gb
 
