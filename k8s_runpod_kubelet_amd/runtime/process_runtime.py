@@ -175,6 +175,16 @@ class ProcessRuntime(Runtime):
             else:
                 spawn_s = self._launch_containers(inst)
         except Exception:
+            # Partial deploy failure (e.g. container 2 of 2 fails to
+            # spawn): containers already launched must not leak — kill
+            # them and drop their event-loop registrations, or a --hold
+            # first container would run untracked forever.
+            for c in list(inst.containers) + list(inst.init_containers):
+                if c.pid > 0 and c.exit_code is None:
+                    self._native.signal_process(c.pid, 9, True)
+                    self._loop.remove_process(c.pid)
+                    with self._lock:
+                        self._pid_to_instance.pop(c.pid, None)
             if gpu_indices:
                 self.binder.unbind(params.pod_key)
             self._release_cgroup(inst.cgroup_dir)

@@ -375,3 +375,25 @@ def test_port_exposure_attributed_to_pod_pids(process_runtime):
         wait_status(rt, st.id, PodStatus.TERMINATED)
     finally:
         other.close()
+
+
+def test_partial_deploy_failure_kills_launched_containers(process_runtime):
+    """Multi-container pod where the SECOND container fails to spawn: the
+    already-running first container must be killed and untracked, not leak
+    as an orphan process (found by adversarial review)."""
+    rt = process_runtime
+    with pytest.raises(Exception):
+        rt.deploy(DeployParams(
+            pod_key="default-partial", name="partial",
+            containers=[
+                ContainerSpec(name="good", command=["/bin/sh"],
+                              args=["-c", "sleep 60"]),
+                ContainerSpec(name="bad",
+                              command=["/nonexistent-binary-xyz"]),
+            ],
+        ))
+    # the good container's process must be gone and nothing tracked
+    deadline = time.time() + 5
+    while time.time() < deadline and rt.tracked_process_count() > 0:
+        time.sleep(0.05)
+    assert rt.tracked_process_count() == 0
