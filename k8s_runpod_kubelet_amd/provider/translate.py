@@ -325,6 +325,7 @@ def prepare_deploy_params(
                             if gid not in (None, "") else -1),
                 liveness=ProbeSpec.parse(c.get("livenessProbe")),
                 readiness=ProbeSpec.parse(c.get("readinessProbe")),
+                startup=ProbeSpec.parse(c.get("startupProbe")),
                 volume_mounts=_volume_mounts_of(c),
             )
         )

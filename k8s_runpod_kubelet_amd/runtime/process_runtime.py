@@ -431,7 +431,7 @@ class ProcessRuntime(Runtime):
         # not sit NotReady until exit.
         pipe_gated = bool(argv) and argv[0] == self.podworker_path()
         if (into is not inst.init_containers and not pipe_gated
-                and cspec.readiness is None):
+                and cspec.readiness is None and cspec.startup is None):
             cinfo.ready = True
         into.append(cinfo)
         with self._lock:
@@ -552,7 +552,7 @@ class ProcessRuntime(Runtime):
         # A defined readinessProbe owns the Ready state: the AMDVK_READY_FD
         # pipe / process-start heuristics are ignored for that container.
         probe_gated = not is_init and any(
-            s.readiness is not None
+            s.readiness is not None or s.startup is not None
             for s in inst.params.containers if s.name == cinfo.name)
         if ev.type == "ready":
             if not is_init and not probe_gated:
@@ -570,7 +570,9 @@ class ProcessRuntime(Runtime):
             cinfo.exit_code = ev.exit_code
             cinfo.finished_at = time.time()
             cinfo.ready = False
-            if ev.exit_code:
+            if ev.exit_code and not cinfo.message:
+                # keep a pre-set reason (liveness/startup probe kill,
+                # deadline) over the generic exit-code message
                 cinfo.message = f"exit code {ev.exit_code}"
             if is_init:
                 self._on_init_exit(inst, cinfo)
@@ -752,12 +754,61 @@ class ProcessRuntime(Runtime):
         for inst in insts:
             by_name = {c.name: c for c in inst.containers}
             for cspec in inst.params.containers:
-                if cspec.liveness is None and cspec.readiness is None:
+                if (cspec.liveness is None and cspec.readiness is None
+                        and cspec.startup is None):
                     continue
                 cinfo = by_name.get(cspec.name)
                 if cinfo is None or cinfo.exit_code is not None:
                     continue
                 env = None
+                if cspec.startup is not None:
+                    skey = (inst.id, cspec.name, "startup")
+                    sst = self._probe_states.setdefault(skey, ProbeState())
+                    if sst.result is not True:
+                        # startup owns the container until it passes
+                        # (k8s: not Started yet -> no liveness/readiness)
+                        spec = cspec.startup
+                        if now - cinfo.started_at < spec.initial_delay_s:
+                            continue
+                        if now - sst.last_run < spec.period_s:
+                            continue
+                        sst.last_run = now
+                        if spec.kind == "exec":
+                            env = dict(os.environ)
+                            env.update(inst.params.env)
+                            env.update(device_env(
+                                inst.gpu_indices,
+                                self.binder.ledger.inventory))
+                            env.update(cspec.env)
+                            cg = inst.cgroup_dir
+                            uid, gid = cspec.run_as_uid, cspec.run_as_gid
+                            sp, rf = self._container_entry(inst, cspec)
+
+                            def srunner(command, penv, timeout_s, _cg=cg,
+                                        _uid=uid, _gid=gid, _sp=sp, _rf=rf):
+                                return self._run_confined(
+                                    command, penv, _cg, _uid, _gid,
+                                    timeout_s, setns_pid=_sp, rootfs=_rf)
+                        else:
+                            srunner = None
+                        ok = run_probe(spec, env or {}, exec_runner=srunner)
+                        outcome = advance(sst, spec, ok)
+                        if outcome is False:
+                            log.warning(
+                                "startup probe failed; killing container",
+                                extra={"instance": inst.id,
+                                       "container": cspec.name,
+                                       "failures": sst.failures})
+                            cinfo.message = "startup probe failed"
+                            self._clear_probe_state(inst.id, cspec.name)
+                            self._native.signal_process(cinfo.pid, 9, True)
+                        elif outcome is True and cspec.readiness is None:
+                            # started == ready when no readinessProbe
+                            if not cinfo.ready:
+                                cinfo.ready = True
+                                self._persist(inst)
+                                self._notify(inst.id)
+                        continue
                 for kind, spec in (("liveness", cspec.liveness),
                                    ("readiness", cspec.readiness)):
                     if spec is None:
@@ -1244,6 +1295,8 @@ class ProcessRuntime(Runtime):
                     if c.readiness else None,
                     "liveness": dataclasses.asdict(c.liveness)
                     if c.liveness else None,
+                    "startup": dataclasses.asdict(c.startup)
+                    if c.startup else None,
                     "env": c.env,  # exec probes run in the container env
                     # credentials/cwd must survive restarts: an adopted pod
                     # whose container crash-restarts would otherwise relaunch
@@ -1310,6 +1363,8 @@ class ProcessRuntime(Runtime):
                         if c.get("readiness") else None,
                         liveness=ProbeSpec(**c["liveness"])
                         if c.get("liveness") else None,
+                        startup=ProbeSpec(**c["startup"])
+                        if c.get("startup") else None,
                         run_as_uid=c.get("run_as_uid", -1),
                         run_as_gid=c.get("run_as_gid", -1),
                         working_dir=c.get("working_dir", ""),

@@ -146,9 +146,12 @@ class ContainerSpec:
     # inherit the kubelet's credentials)
     run_as_uid: int = -1
     run_as_gid: int = -1
-    # livenessProbe / readinessProbe (None = absent)
+    # livenessProbe / readinessProbe / startupProbe (None = absent)
     liveness: Optional["ProbeSpec"] = None
     readiness: Optional["ProbeSpec"] = None
+    # startupProbe gates the other two (k8s: the container is not Started
+    # until it passes; failureThreshold exhausted kills the container)
+    startup: Optional["ProbeSpec"] = None
 
 
 @dataclass

@@ -158,3 +158,67 @@ def test_liveness_probe_kills_and_restarts(stack):
     assert wait_until(restarted, timeout_s=30) is not None
     kube.delete_pod("default", "livefail")
     assert wait_until(lambda: not s.ledger.reservations, timeout_s=15)
+
+
+def test_startup_probe_gates_readiness_then_passes(process_runtime,
+                                                   tmp_path):
+    """startupProbe semantics: the container is not Ready (and liveness
+    does not run) until the startup probe passes; with no readinessProbe,
+    startup success makes it Ready."""
+    import time
+
+    from k8s_runpod_kubelet_amd.runtime.probes import ProbeSpec
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+
+    rt = process_runtime
+    gate = tmp_path / "started-marker"
+    st = rt.deploy(DeployParams(
+        pod_key="default-sp1", name="sp1",
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"], args=["-c", "sleep 30"],
+            startup=ProbeSpec(kind="exec",
+                              command=["/usr/bin/test", "-f", str(gate)],
+                              period_s=1.0, timeout_s=5.0,
+                              failure_threshold=30))],
+    ))
+    time.sleep(2.2)  # a couple of startup attempts fail (marker absent)
+    s = rt.get_detailed_status(st.id)
+    assert not s.containers[0].ready, "ready before startup probe passed"
+    gate.write_text("up")
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        if s.containers[0].ready:
+            break
+        time.sleep(0.1)
+    assert s.containers[0].ready, "startup pass did not mark Ready"
+    rt.terminate(st.id)
+
+
+def test_startup_probe_failure_kills_container(process_runtime):
+    """failureThreshold exhausted on the startupProbe kills the container
+    into the restartPolicy machinery (Never here -> pod fails)."""
+    import time
+
+    from k8s_runpod_kubelet_amd.runtime.probes import ProbeSpec
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+
+    rt = process_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-sp2", name="sp2",
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"], args=["-c", "sleep 30"],
+            startup=ProbeSpec(kind="exec", command=["/bin/false"],
+                              period_s=1.0, timeout_s=5.0,
+                              failure_threshold=2))],
+    ))
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        if s.desired_status == PodStatus.EXITED:
+            break
+        time.sleep(0.1)
+    assert s.desired_status == PodStatus.EXITED
+    assert s.containers[0].message == "startup probe failed"
