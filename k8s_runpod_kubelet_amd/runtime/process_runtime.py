@@ -358,6 +358,18 @@ class ProcessRuntime(Runtime):
             chroot_only = prepared.chroot_only
             mounts = prepared.mounts
         elif not argv:
+            if (self._rootfs_mgr is not None and cspec.image
+                    and not cspec.image.startswith("amdvk/")):
+                # The user named a real image with no command and it is not
+                # in the local store: failing loudly (pod stays Pending,
+                # retried) is the honest ErrImagePull analogue — once the
+                # operator imports/pulls the image, the retry succeeds.
+                # Silently running the synthetic podworker instead would
+                # lie about what is executing.
+                raise RuntimeError(
+                    f"image {cspec.image!r} not in the local store "
+                    f"(import it: python -m "
+                    f"k8s_runpod_kubelet_amd.runtime.imagetool)")
             argv = [self.podworker_path(), "--hold"]
             if inst.gpu_indices:
                 argv += ["--expect-gpus", str(len(inst.gpu_indices))]

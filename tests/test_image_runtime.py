@@ -768,3 +768,61 @@ def test_malicious_mount_path_clamped(synthetic_ledger, tmp_state_dir,
         assert not found_outside
     finally:
         rt.close()
+
+
+def test_missing_image_pending_until_imported(synthetic_ledger,
+                                              tmp_state_dir, image_store,
+                                              tmp_path, app_bin):
+    """ErrImagePull analogue: a command-less pod whose image is NOT in the
+    store stays Pending (deploy fails loudly, retried) — and becomes Ready
+    the moment the operator imports the image."""
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from tests.conftest import make_pod, wait_until
+
+    cfg = Config(state_dir=tmp_state_dir + "-pend", gpu_count_override=8,
+                 pending_retry_interval_s=0.3, notify_interval_s=0)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.runtime.enable_cgroups = False
+    stack.start(serve_http=False)
+    try:
+        pod = make_pod("notyet", restart_policy=None)
+        pod["spec"]["containers"][0] = {
+            "name": "main", "image": "example/latecomer:v1"}
+        kube.create_pod("default", pod)
+        time.sleep(1.0)
+        p = kube.get_pod("default", "notyet")
+        assert p.get("status", {}).get("phase", "Pending") in ("Pending", "")
+        # deploy error surfaced as a kubectl event
+        evs = [e for e in kube.events.objects.values()
+               if e["involvedObject"]["name"] == "notyet"]
+        assert any("not in the local store" in e.get("message", "")
+                   for e in evs), [e.get("message") for e in evs]
+
+        # operator imports the image -> next retry runs it
+        import shutil
+
+        tree = tmp_path / "latetree"
+        (tree / "bin").mkdir(parents=True)
+        shutil.copy2(app_bin, tree / "bin" / "app")
+        layout = tmp_path / "latelayout"
+        layout.mkdir()
+        build_layout(str(layout), "example/latecomer:v1", str(tree),
+                     entrypoint=["/bin/app"], cmd=["hold"])
+        ImageStore(cfg.resolved_image_store_dir()).add_layout(
+            str(layout), "example/latecomer:v1")
+
+        def ready():
+            try:
+                p = kube.get_pod("default", "notyet")
+            except Exception:
+                return None
+            conds = {c["type"]: c["status"]
+                     for c in p.get("status", {}).get("conditions", [])}
+            return p if conds.get("Ready") == "True" else None
+
+        assert wait_until(ready, timeout_s=20) is not None
+    finally:
+        stack.stop()
