@@ -376,6 +376,8 @@ def prepare_deploy_params(
         cpu_limit=cpu_max,
         memory_limit=memory_max,
         volumes=extract_volumes(pod, client),
+        fs_group=_int_or_error(pod_sc.get("fsGroup"), "fsGroup")
+        if pod_sc.get("fsGroup") not in (None, "") else -1,
         labels=dict(metadata.get("labels", {}) or {}),
         hostname=pod.get("spec", {}).get("hostname")
         or metadata.get("name", ""),

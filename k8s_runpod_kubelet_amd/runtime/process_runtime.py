@@ -276,6 +276,16 @@ class ProcessRuntime(Runtime):
                         fp.parent.mkdir(parents=True, exist_ok=True)
                         fp.write_text(content)
                         os.chmod(fp, src_spec.file_mode)
+                    fsg = inst.params.fs_group
+                    if fsg >= 0:
+                        # securityContext.fsGroup: volume group-owned and
+                        # group-writable (k8s volume ownership management)
+                        try:
+                            for p_ in [vdir, *vdir.rglob("*")]:
+                                os.chown(p_, -1, fsg)
+                                os.chmod(p_, os.stat(p_).st_mode | 0o070)
+                        except OSError:
+                            pass
                 src = str(vdir)
             if vm.sub_path:
                 src = os.path.join(src, vm.sub_path)

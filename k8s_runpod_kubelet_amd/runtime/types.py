@@ -190,6 +190,9 @@ class DeployParams:
     # k8s pod-hostname semantics (spec.hostname, else pod name); applied in
     # the pod's own UTS namespace when namespace isolation is available
     hostname: str = ""
+    # securityContext.fsGroup: emptyDir/projected volumes are group-owned
+    # and group-writable by this gid (-1 = unset)
+    fs_group: int = -1
 
 
 @dataclass

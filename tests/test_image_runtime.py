@@ -826,3 +826,30 @@ def test_missing_image_pending_until_imported(synthetic_ledger,
         assert wait_until(ready, timeout_s=20) is not None
     finally:
         stack.stop()
+
+
+def test_fsgroup_volume_ownership(image_runtime):
+    """securityContext.fsGroup: emptyDir volumes are group-owned and
+    group-writable, so a non-root container can write its volume."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        VolumeMount, VolumeSource)
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-fsg", name="fsg", fs_group=4321,
+        volumes={"data": VolumeSource(kind="emptyDir")},
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh"],
+            args=["-c", "echo can-write > /data2/f.txt "
+                        "&& echo WRITE-OK"],
+            run_as_uid=1234, run_as_gid=4321,
+            volume_mounts=[VolumeMount("data", "/data2")])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    out = rt.get_logs(st.id)
+    assert s.exit_code == 0, out
+    assert "WRITE-OK" in out, out
