@@ -144,8 +144,9 @@ class KubeletApiServer:
                         self._respond(400, b"missing command")
                         return
                     namespace, pod = parts[1], parts[2]
+                    container = parts[3] if len(parts) > 3 else ""
                     code, output = outer.provider.run_in_container(
-                        namespace, pod, commands
+                        namespace, pod, commands, container=container
                     )
                     body = json.dumps({"exitCode": code, "output": output}).encode()
                     self._respond(200 if code == 0 else 500, body,

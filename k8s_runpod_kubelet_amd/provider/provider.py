@@ -1064,7 +1064,7 @@ class Provider:
                                PodStatus.NOT_FOUND)
 
     def run_in_container(self, namespace: str, name: str, command: List[str],
-                         timeout_s: float = 30.0) -> tuple:
+                         timeout_s: float = 30.0, container: str = "") -> tuple:
         """One-shot exec in the pod's environment (GPU binding included).
 
         The reference returns "not supported by RunPod" (kubelet.go:2027-2047);
@@ -1082,7 +1082,8 @@ class Provider:
         exec_fn = getattr(self.runtime, "exec_in_instance", None)
         if exec_fn is None:
             return 501, "runtime does not support exec"
-        return exec_fn(info.instance_id, command, timeout_s)
+        return exec_fn(info.instance_id, command, timeout_s,
+                       container=container)
 
     def instance_info(self, namespace: str, name: str) -> Optional[InstanceInfo]:
         with self._pods_lock:

@@ -1228,10 +1228,11 @@ class ProcessRuntime(Runtime):
         return -1, ""
 
     def exec_in_instance(self, instance_id: str, command: List[str],
-                         timeout_s: float = 30.0) -> tuple:
+                         timeout_s: float = 30.0, container: str = "") -> tuple:
         """Non-interactive exec with the instance's environment (same GPU
-        binding, same cgroup, same credentials as the first container).
-        Returns (exit_code, combined_output)."""
+        binding, same cgroup, the targeted container's credentials and —
+        for image pods — its rootfs; kubectl exec -c selects `container`,
+        default first). Returns (exit_code, combined_output)."""
         with self._lock:
             inst = self._instances.get(instance_id)
         if inst is None:
@@ -1241,7 +1242,14 @@ class ProcessRuntime(Runtime):
         env.pop("HIP_VISIBLE_DEVICES", None)
         env.update(inst.params.env)
         env.update(device_env(inst.gpu_indices, self.binder.ledger.inventory))
-        cspec = inst.params.containers[0] if inst.params.containers else None
+        cspec = None
+        if container:
+            cspec = next((c for c in inst.params.containers
+                          if c.name == container), None)
+            if cspec is None:
+                return 127, f"container {container!r} not found in pod"
+        elif inst.params.containers:
+            cspec = inst.params.containers[0]
         uid = cspec.run_as_uid if cspec else -1
         gid = cspec.run_as_gid if cspec else -1
         setns_pid, rootfs = self._container_entry(inst, cspec)

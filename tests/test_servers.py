@@ -253,3 +253,58 @@ def test_exec_on_pending_pod_reports_state(provider):
     assert code == 126
     assert "no running instance" in out
     assert "STARTING" in out
+
+
+def test_exec_targets_named_container(provider, process_runtime):
+    """kubectl exec -c <name>: the exec runs with the targeted container's
+    credentials/confinement (multi-container pod)."""
+    import urllib.request
+
+    prov, _, kube = provider
+    prov.runtime = process_runtime
+    pod = make_pod("multi", containers=[
+        {"name": "a", "image": "amdvk/test:latest",
+         "command": ["/bin/sh"], "args": ["-c", "sleep 30"]},
+        {"name": "b", "image": "amdvk/test:latest",
+         "command": ["/bin/sh"], "args": ["-c", "sleep 30"],
+         "securityContext": {"runAsUser": 65534, "runAsGroup": 65534}},
+    ])
+    kube.create_pod("default", pod)
+    prov.create_pod(kube.get_pod("default", "multi"))
+    srv = KubeletApiServer(prov, "127.0.0.1", 0)
+    srv.start()
+    try:
+        base = f"http://127.0.0.1:{srv.port}"
+        # container b runs as nobody; exec -c b must too
+        req = urllib.request.Request(
+            f"{base}/exec/default/multi/b?command=/usr/bin/id&command=-u",
+            method="POST")
+        with urllib.request.urlopen(req, timeout=15) as resp:
+            out = json.loads(resp.read())
+        assert out["exitCode"] == 0, out
+        assert "65534" in out["output"], out
+        # default (container a) runs as the kubelet user (root here)
+        req = urllib.request.Request(
+            f"{base}/exec/default/multi/a?command=/usr/bin/id&command=-u",
+            method="POST")
+        with urllib.request.urlopen(req, timeout=15) as resp:
+            out = json.loads(resp.read())
+        assert out["exitCode"] == 0, out
+        assert out["output"].strip().splitlines()[-1] == "0", out
+        # unknown container -> clear error
+        req = urllib.request.Request(
+            f"{base}/exec/default/multi/ghost?command=/bin/true",
+            method="POST")
+        import urllib.error
+
+        try:
+            urllib.request.urlopen(req, timeout=15)
+            raise AssertionError("exec into unknown container succeeded")
+        except urllib.error.HTTPError as exc:
+            body = json.loads(exc.read())
+            assert "not found" in body["output"]
+    finally:
+        srv.stop()
+        pod_obj = prov.get_pod("default", "multi")
+        if pod_obj:
+            prov.delete_pod(pod_obj)
