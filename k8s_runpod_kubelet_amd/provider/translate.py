@@ -378,6 +378,11 @@ def prepare_deploy_params(
         volumes=extract_volumes(pod, client),
         fs_group=_int_or_error(pod_sc.get("fsGroup"), "fsGroup")
         if pod_sc.get("fsGroup") not in (None, "") else -1,
+        host_aliases=[
+            (ha.get("ip", ""), list(ha.get("hostnames", []) or []))
+            for ha in pod.get("spec", {}).get("hostAliases", []) or []
+            if ha.get("ip") and ha.get("hostnames")
+        ],
         labels=dict(metadata.get("labels", {}) or {}),
         hostname=pod.get("spec", {}).get("hostname")
         or metadata.get("name", ""),

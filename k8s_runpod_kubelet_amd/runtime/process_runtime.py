@@ -361,7 +361,8 @@ class ProcessRuntime(Runtime):
                 params.hostname or params.name,
                 gpu_device_paths=self._gpu_device_paths(inst),
                 working_dir=working_dir,
-                volume_binds=self._volume_binds(inst, cspec))
+                volume_binds=self._volume_binds(inst, cspec),
+                host_aliases=params.host_aliases)
             argv[0] = self._rootfs_mgr.resolve_argv0(image, prepared,
                                                      argv[0])
             rootfs = prepared.rootfs

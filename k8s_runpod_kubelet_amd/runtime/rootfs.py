@@ -101,10 +101,12 @@ class RootfsManager:
                 image: ResolvedImage, hostname: str,
                 gpu_device_paths: Optional[List[str]] = None,
                 working_dir: str = "",
-                volume_binds: Optional[List[tuple]] = None) -> PreparedRootfs:
+                volume_binds: Optional[List[tuple]] = None,
+                host_aliases: Optional[List[tuple]] = None) -> PreparedRootfs:
         """volume_binds: (host_src, container_dst, ro) tuples from the
         pod's volumes (emptyDir/hostPath/secret/configMap projections,
         materialized by the runtime)."""
+        self._host_aliases = list(host_aliases or [])
         cdir = self.containers_dir / f"{instance_id}-{container_name}"
         if self.mode() == "mountns":
             return self._prepare_overlay(cdir, image, hostname,
@@ -245,16 +247,18 @@ class RootfsManager:
         except (OSError, PermissionError):
             log.debug("mknod unavailable", extra={"dev": host_dev})
 
-    @staticmethod
-    def _write_identity(root: Path, hostname: str) -> None:
-        """kubelet-managed identity files (k8s writes these per pod)."""
+    def _write_identity(self, root: Path, hostname: str) -> None:
+        """kubelet-managed identity files (k8s writes these per pod),
+        including spec.hostAliases lines."""
         etc = root / "etc"
         etc.mkdir(parents=True, exist_ok=True)
+        lines = ["127.0.0.1\tlocalhost",
+                 f"127.0.1.1\t{hostname or 'pod'}"]
+        for ip, names in getattr(self, "_host_aliases", []):
+            lines.append(f"{ip}\t" + " ".join(names))
         try:
             (etc / "hostname").write_text((hostname or "pod") + "\n")
-            (etc / "hosts").write_text(
-                "127.0.0.1\tlocalhost\n"
-                f"127.0.1.1\t{hostname or 'pod'}\n")
+            (etc / "hosts").write_text("\n".join(lines) + "\n")
         except OSError:
             pass
 

@@ -193,6 +193,8 @@ class DeployParams:
     # securityContext.fsGroup: emptyDir/projected volumes are group-owned
     # and group-writable by this gid (-1 = unset)
     fs_group: int = -1
+    # spec.hostAliases -> extra /etc/hosts lines in image pods
+    host_aliases: List[tuple] = field(default_factory=list)  # (ip, [names])
 
 
 @dataclass

@@ -853,3 +853,27 @@ def test_fsgroup_volume_ownership(image_runtime):
     out = rt.get_logs(st.id)
     assert s.exit_code == 0, out
     assert "WRITE-OK" in out, out
+
+
+def test_host_aliases_in_etc_hosts(image_runtime):
+    """spec.hostAliases lands in the pod-managed /etc/hosts (k8s kubelet
+    behavior)."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-ha", name="ha",
+        host_aliases=[("10.9.8.7", ["backend", "backend.local"])],
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh"],
+            args=["-c", "read a < /etc/hostname; echo host=$a; "
+                        "while read l; do echo hosts=$l; done < /etc/hosts"])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    out = rt.get_logs(st.id)
+    assert s.exit_code == 0, out
+    assert "host=ha" in out
+    assert "hosts=10.9.8.7" in out and "backend.local" in out, out
