@@ -106,15 +106,15 @@ class RootfsManager:
         """volume_binds: (host_src, container_dst, ro) tuples from the
         pod's volumes (emptyDir/hostPath/secret/configMap projections,
         materialized by the runtime)."""
-        self._host_aliases = list(host_aliases or [])
         cdir = self.containers_dir / f"{instance_id}-{container_name}"
+        aliases = list(host_aliases or [])
         if self.mode() == "mountns":
             return self._prepare_overlay(cdir, image, hostname,
                                          gpu_device_paths or [], working_dir,
-                                         volume_binds or [])
+                                         volume_binds or [], aliases)
         return self._prepare_chroot(cdir, image, hostname,
                                     gpu_device_paths or [], working_dir,
-                                    volume_binds or [])
+                                    volume_binds or [], aliases)
 
     def _parse_bind(self, entry: str) -> Tuple[str, str, bool]:
         """'src[:dst[:ro|rw]]' -> (src, dst, ro). Default dst=src, ro."""
@@ -126,8 +126,8 @@ class RootfsManager:
 
     def _prepare_overlay(self, cdir: Path, image: ResolvedImage,
                          hostname: str, gpu_devices: List[str],
-                         working_dir: str,
-                         volume_binds: List[tuple]) -> PreparedRootfs:
+                         working_dir: str, volume_binds: List[tuple],
+                         host_aliases: List[tuple]) -> PreparedRootfs:
         lower = self.store.rootfs_for(image)
         upper = cdir / "upper"
         work = cdir / "work"
@@ -144,7 +144,7 @@ class RootfsManager:
         if working_dir:
             (upper / _container_rel(working_dir)).mkdir(parents=True,
                                                         exist_ok=True)
-        self._write_identity(upper, hostname)
+        self._write_identity(upper, hostname, host_aliases)
 
         mounts = [mnt.overlay(str(merged), str(lower), str(upper),
                               str(work)),
@@ -190,8 +190,8 @@ class RootfsManager:
 
     def _prepare_chroot(self, cdir: Path, image: ResolvedImage,
                         hostname: str, gpu_devices: List[str],
-                        working_dir: str,
-                        volume_binds: List[tuple]) -> PreparedRootfs:
+                        working_dir: str, volume_binds: List[tuple],
+                        host_aliases: List[tuple]) -> PreparedRootfs:
         rootfs = cdir / "rootfs"
         if not (rootfs / ".amdvk-ready").exists():
             cdir.mkdir(parents=True, exist_ok=True)
@@ -210,7 +210,7 @@ class RootfsManager:
             if working_dir:
                 (rootfs / _container_rel(working_dir)).mkdir(parents=True,
                                                              exist_ok=True)
-            self._write_identity(rootfs, hostname)
+            self._write_identity(rootfs, hostname, host_aliases)
             # chroot mode has no mounts: volumes are materialized INTO the
             # per-container copy (per-container — cross-container emptyDir
             # sharing needs the mount-namespace mode; hostPath cannot be
@@ -247,14 +247,15 @@ class RootfsManager:
         except (OSError, PermissionError):
             log.debug("mknod unavailable", extra={"dev": host_dev})
 
-    def _write_identity(self, root: Path, hostname: str) -> None:
+    def _write_identity(self, root: Path, hostname: str,
+                        host_aliases: List[tuple]) -> None:
         """kubelet-managed identity files (k8s writes these per pod),
         including spec.hostAliases lines."""
         etc = root / "etc"
         etc.mkdir(parents=True, exist_ok=True)
         lines = ["127.0.0.1\tlocalhost",
                  f"127.0.1.1\t{hostname or 'pod'}"]
-        for ip, names in getattr(self, "_host_aliases", []):
+        for ip, names in host_aliases:
             lines.append(f"{ip}\t" + " ".join(names))
         try:
             (etc / "hostname").write_text((hostname or "pod") + "\n")
