@@ -326,6 +326,9 @@ def prepare_deploy_params(
                 liveness=ProbeSpec.parse(c.get("livenessProbe")),
                 readiness=ProbeSpec.parse(c.get("readinessProbe")),
                 startup=ProbeSpec.parse(c.get("startupProbe")),
+                termination_message_path=c.get(
+                    "terminationMessagePath", "/dev/termination-log")
+                or "/dev/termination-log",
                 volume_mounts=_volume_mounts_of(c),
             )
         )

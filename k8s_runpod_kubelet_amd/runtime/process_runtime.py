@@ -593,7 +593,12 @@ class ProcessRuntime(Runtime):
             cinfo.exit_code = ev.exit_code
             cinfo.finished_at = time.time()
             cinfo.ready = False
-            if ev.exit_code and not cinfo.message:
+            tmsg = self._read_termination_message(inst, cinfo.name)
+            if tmsg:
+                # terminationMessagePath: the container's own last words
+                # win (k8s surfaces them in the terminated state)
+                cinfo.message = tmsg
+            elif ev.exit_code and not cinfo.message:
                 # keep a pre-set reason (liveness/startup probe kill,
                 # deadline) over the generic exit-code message
                 cinfo.message = f"exit code {ev.exit_code}"
@@ -602,6 +607,28 @@ class ProcessRuntime(Runtime):
             else:
                 self._on_container_exit(inst, cinfo)
             touched.add(inst_id)
+
+
+    def _read_termination_message(self, inst: Instance,
+                                  container: str) -> str:
+        """spec.containers[].terminationMessagePath for image pods: read
+        the file the container wrote inside its rootfs (overlay upper /
+        chroot copy). Truncated to 4 KiB like the kubelet."""
+        if self._rootfs_mgr is None or not inst.image_mode:
+            return ""
+        cspec = next((c for c in inst.params.containers
+                      if c.name == container), None)
+        if cspec is None:
+            return ""
+        rel = cspec.termination_message_path.lstrip("/")
+        cdir = self._rootfs_mgr.containers_dir / f"{inst.id}-{container}"
+        base = cdir / ("rootfs" if inst.image_mode == "chroot" else "upper")
+        try:
+            with open(base / rel, "r", encoding="utf-8",
+                      errors="replace") as fh:
+                return fh.read(4096).strip()
+        except OSError:
+            return ""
 
     def _on_container_exit(self, inst: Instance, cinfo=None) -> None:
         # spec.restartPolicy (k8s semantics; the reference's cloud instances

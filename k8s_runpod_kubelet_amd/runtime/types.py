@@ -152,6 +152,9 @@ class ContainerSpec:
     # startupProbe gates the other two (k8s: the container is not Started
     # until it passes; failureThreshold exhausted kills the container)
     startup: Optional["ProbeSpec"] = None
+    # spec.containers[].terminationMessagePath: file the container writes
+    # its exit message to, surfaced in terminated status (image pods)
+    termination_message_path: str = "/dev/termination-log"
 
 
 @dataclass

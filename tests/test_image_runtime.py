@@ -877,3 +877,25 @@ def test_host_aliases_in_etc_hosts(image_runtime):
     assert s.exit_code == 0, out
     assert "host=ha" in out
     assert "hosts=10.9.8.7" in out and "backend.local" in out, out
+
+
+def test_termination_message_surfaces(image_runtime):
+    """terminationMessagePath: the container's written message appears as
+    the terminated status message (kubectl describe surface)."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-tmsg", name="tmsg",
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh"],
+            args=["-c", "echo oom-adjacent-sadness > /dev/termination-log; "
+                        "exit 3"])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.exit_code == 3
+    assert s.containers[0].message == "oom-adjacent-sadness", \
+        s.containers[0].message
