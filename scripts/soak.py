@@ -124,7 +124,7 @@ def main() -> int:
                                          "command": ["/bin/app"],
                                          "args": ["hold"]}]},
             }
-        if mode in ("hold", "crash", "probed"):
+        if mode in ("hold", "crash", "probed", "started"):
             a = pw_args + ["--hold"]
         elif mode == "ok":
             a = pw_args + ["--run-for", f"{rng.uniform(0.05, 0.4):.2f}"]
@@ -144,6 +144,11 @@ def main() -> int:
             container["readinessProbe"] = {
                 "exec": {"command": ["/bin/true"]},
                 "periodSeconds": 1, "failureThreshold": 1,
+            }
+        elif mode == "started":
+            container["startupProbe"] = {
+                "exec": {"command": ["/bin/true"]},
+                "periodSeconds": 1, "failureThreshold": 3,
             }
         elif mode == "restarting":
             spec["restartPolicy"] = "OnFailure"
@@ -202,8 +207,8 @@ def main() -> int:
             # top up population (only while inside the window)
             while now < deadline and len(active) < args.max_active:
                 modes = ["hold", "ok", "fail", "crash", "probed",
-                         "restarting"]
-                weights = [3, 4, 2, 1, 2, 1]
+                         "restarting", "started"]
+                weights = [3, 4, 2, 1, 2, 1, 1]
                 if image_ref:
                     modes += ["img-ok", "img-hold"]
                     weights += [3, 2]
@@ -264,7 +269,7 @@ def main() -> int:
                         counters["restarted"] += 1
                         st["deleted"] = True
                         client.delete_pod("default", name)
-                else:  # hold / probed / img-hold
+                else:  # hold / probed / started / img-hold
                     if state == "Ready" and age > st["dwell"] and not st["deleted"]:
                         counters["deleted_holds"] += 1
                         st["deleted"] = True
