@@ -684,3 +684,35 @@ def test_term_ignoring_pod_stays_until_grace_kill(stack):
     took = time.time() - t0
     assert took >= 0.9, f"deleted before the grace window ({took:.2f}s)"
     assert not os.path.exists(f"/proc/{pid}")
+
+
+def test_deletion_finalize_fallback_via_resync(tmp_state_dir):
+    """If the instance's terminal event is lost (deletion_resync disabled
+    here), the informer's periodic resync still completes the API delete —
+    the finalize path must not depend solely on the push event."""
+    import time
+
+    cfg = Config(
+        state_dir=tmp_state_dir,
+        gpu_count_override=8,
+        pending_retry_interval_s=0.2,
+        reconcile_interval_s=0.3,  # informer resync doubles as fallback
+        notify_interval_s=0,
+    )
+    kube = FakeKube()
+    s = build_stack(cfg, client=kube)
+    s.runtime.enable_cgroups = False
+    s.start(serve_http=False)
+    try:
+        s.provider.deletion_resync = None  # simulate a lost event path
+        pod = make_pod("fallback", command=["/bin/sh"],
+                       args=["-c", "trap '' TERM; sleep 60"])
+        pod["spec"]["terminationGracePeriodSeconds"] = 1
+        kube.create_pod("default", pod)
+        assert wait_until(lambda: ready(kube, "fallback"), timeout_s=15)
+        kube.delete_pod("default", "fallback")
+        # grace 1s SIGKILL ends the container; the next resync (0.3 s)
+        # must finalize the delete without any push notification
+        assert wait_until(lambda: gone(kube, "fallback"), timeout_s=20)
+    finally:
+        s.stop()
