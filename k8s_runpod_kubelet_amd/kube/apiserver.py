@@ -112,11 +112,13 @@ class KubeletApiServer:
                     tail = int(query.get("tailLines", ["-1"])[0])
                     follow = query.get(
                         "follow", ["false"])[0].lower() in ("true", "1")
-                    if follow:
+                    previous = query.get(
+                        "previous", ["false"])[0].lower() in ("true", "1")
+                    if follow and not previous:
                         self._stream_logs(namespace, pod, container, tail)
                         return
                     text = outer.provider.get_container_logs(
-                        namespace, pod, container, tail
+                        namespace, pod, container, tail, previous=previous
                     )
                     self._respond(200, text.encode())
                 elif parts and parts[0] == "pods":

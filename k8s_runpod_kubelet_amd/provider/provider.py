@@ -1033,12 +1033,16 @@ class Provider:
     # ------------------------------------------------------------------
 
     def get_container_logs(self, namespace: str, name: str, container: str = "",
-                           tail: int = -1) -> str:
+                           tail: int = -1, previous: bool = False) -> str:
         with self._pods_lock:
             info = self._pod_status.get(f"{namespace}-{name}")
         if info is None or not info.instance_id:
             return ""
-        return self.runtime.get_logs(info.instance_id, container, tail)
+        try:
+            return self.runtime.get_logs(info.instance_id, container, tail,
+                                         previous=previous)
+        except TypeError:
+            return self.runtime.get_logs(info.instance_id, container, tail)
 
     def get_container_log_path(self, namespace: str, name: str,
                                container: str = "") -> Optional[str]:

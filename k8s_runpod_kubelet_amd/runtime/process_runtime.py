@@ -681,6 +681,14 @@ class ProcessRuntime(Runtime):
             cinfo = next((c for c in inst.containers if c.name == name), None)
             if cspec is None or cinfo is None:
                 return
+            # kubectl logs --previous: rotate the crashed run's output
+            logp = self.logs_dir / f"{instance_id}-{name}.log"
+            try:
+                if logp.exists():
+                    logp.rename(self.logs_dir
+                                / f"{instance_id}-{name}.log.prev")
+            except OSError:
+                pass
             old_pid = cinfo.pid
             tmp: List = []
             try:
@@ -1178,11 +1186,14 @@ class ProcessRuntime(Runtime):
             name = names[0] if names else ""
         return str(self.logs_dir / f"{instance_id}-{name}.log")
 
-    def get_logs(self, instance_id: str, container: str = "", tail: int = -1) -> str:
+    def get_logs(self, instance_id: str, container: str = "",
+                 tail: int = -1, previous: bool = False) -> str:
+        """previous=True: the prior (pre-restart) run's log, like
+        `kubectl logs --previous`."""
         path = self.get_log_path(instance_id, container)
         if path is None:
             return ""
-        path = Path(path)
+        path = Path(path + ".prev") if previous else Path(path)
         if not path.exists():
             return ""
         text = path.read_text(errors="replace")

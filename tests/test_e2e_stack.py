@@ -716,3 +716,35 @@ def test_deletion_finalize_fallback_via_resync(tmp_state_dir):
         assert wait_until(lambda: gone(kube, "fallback"), timeout_s=20)
     finally:
         s.stop()
+
+
+def test_logs_previous_after_restart(stack):
+    """kubectl logs --previous: the pre-restart run's output is retained
+    and served separately from the current run's."""
+    s, kube = stack
+    pod = make_pod("prevlog", command=["/bin/sh"],
+                   args=["-c", "echo run-$$; sleep 0.2; exit 1"])
+    pod["spec"]["restartPolicy"] = "OnFailure"
+    kube.create_pod("default", pod)
+
+    def restarted():
+        try:
+            p = kube.get_pod("default", "prevlog")
+        except NotFoundError:
+            return None
+        css = p.get("status", {}).get("containerStatuses", [])
+        return p if css and css[0].get("restartCount", 0) >= 1 else None
+
+    assert wait_until(restarted, timeout_s=20)
+    prev = wait_until(
+        lambda: s.provider.get_container_logs(
+            "default", "prevlog", previous=True) or None, timeout_s=10)
+    cur = wait_until(
+        lambda: s.provider.get_container_logs("default", "prevlog") or None,
+        timeout_s=10)
+    # each run logs exactly one "run-" line; rotation keeps them in
+    # separate files (unrotated they would pile up in one)
+    assert prev and prev.count("run-") == 1, prev
+    assert cur and cur.count("run-") >= 1, cur
+    kube.delete_pod("default", "prevlog")
+    assert wait_until(lambda: gone(kube, "prevlog"), timeout_s=20)
