@@ -227,11 +227,46 @@ class Provider:
             self._emit(pod, "Warning", "DeployError", str(exc))
 
     def update_pod(self, pod: Dict[str, Any]) -> None:
-        """UpdatePod (reference kubelet.go:421-432): refresh the cached copy."""
+        """UpdatePod (reference kubelet.go:421-432): refresh the cached
+        copy — and start any newly added spec.ephemeralContainers
+        (kubectl debug attaches by updating the pod spec)."""
         key = pod_key_of(pod)
         with self._pods_lock:
             if key in self._pods:
                 self._pods[key] = pod
+            info = self._pod_status.get(key)
+        if info is not None and info.instance_id:
+            self._sync_ephemeral_containers(pod, info)
+
+    def _sync_ephemeral_containers(self, pod: Dict[str, Any], info) -> None:
+        fn = getattr(self.runtime, "add_ephemeral_container", None)
+        eph = pod.get("spec", {}).get("ephemeralContainers", []) or []
+        if fn is None or not eph:
+            return
+        from ..runtime.types import ContainerSpec
+
+        det = self.runtime.get_detailed_status(info.instance_id)
+        existing = {c.name for c in det.ephemeral_containers}
+        for ec in eph:
+            name = ec.get("name", "")
+            if not name or name in existing:
+                continue
+            cspec = ContainerSpec(
+                name=name, image=ec.get("image", ""),
+                command=list(ec.get("command", []) or []),
+                args=list(ec.get("args", []) or []),
+                env={e["name"]: str(e.get("value", ""))
+                     for e in ec.get("env", []) or [] if e.get("name")},
+            )
+            try:
+                fn(info.instance_id, cspec)
+                self._emit(pod, "Normal", "Started",
+                           f"ephemeral container {name}")
+            except Exception as exc:
+                log.warning("ephemeral container start failed",
+                            extra={"pod": pod_key_of(pod), "err": str(exc)})
+                self._emit(pod, "Warning", "Failed",
+                           f"ephemeral container {name}: {exc}")
 
     def delete_pod(self, pod: Dict[str, Any]) -> None:
         """DeletePod (reference kubelet.go:621-651): record in deletedPods,
@@ -521,8 +556,15 @@ class Provider:
         # restartPolicy transitions (crash → backoff → restarted) keep the
         # pod RUNNING; the signature makes them visible to change detection
         # so CrashLoopBackOff/restartCount reach the apiserver.
-        restart_sig = sum(c.restart_count for c in detailed.containers) * 2 + \
-            sum(1 for c in detailed.containers if c.backoff_until)
+        restart_sig = (
+            sum(c.restart_count for c in detailed.containers) * 2
+            + sum(1 for c in detailed.containers if c.backoff_until),
+            # ephemeral (kubectl debug) containers appear/terminate without
+            # touching pod phase — their transitions must still patch status
+            len(detailed.ephemeral_containers),
+            sum(1 for c in detailed.ephemeral_containers
+                if c.exit_code is not None),
+        )
         changed = (
             detailed.desired_status != info.status
             or ports_exposed != info.ports_exposed

@@ -136,6 +136,35 @@ def _container_statuses(
     return out
 
 
+def _ephemeral_container_statuses(
+    detailed: Optional[DetailedStatus],
+) -> List[Dict[str, Any]]:
+    """spec.ephemeralContainers status projection (kubectl debug)."""
+    out = []
+    if detailed is None:
+        return out
+    for c in getattr(detailed, "ephemeral_containers", []):
+        cs: Dict[str, Any] = {"name": c.name, "image": "", "imageID": "",
+                              "restartCount": 0, "ready": False}
+        if c.pid > 0:
+            cs["containerID"] = f"amdvk://{c.pid}"
+        if c.exit_code is None:
+            cs["state"] = {"running": {
+                "startedAt": ts_rfc3339(c.started_at) if c.started_at
+                else now_rfc3339()}}
+        else:
+            cs["state"] = {"terminated": {
+                "exitCode": c.exit_code,
+                "reason": "Completed" if c.exit_code == 0 else "Error",
+                "startedAt": ts_rfc3339(c.started_at) if c.started_at
+                else now_rfc3339(),
+                "finishedAt": ts_rfc3339(c.finished_at) if c.finished_at
+                else now_rfc3339(),
+            }}
+        out.append(cs)
+    return out
+
+
 def _init_container_statuses(
     detailed: Optional[DetailedStatus],
 ) -> List[Dict[str, Any]]:
@@ -184,6 +213,9 @@ def translate_status(
     inits = _init_container_statuses(detailed)
     if inits:
         base["initContainerStatuses"] = inits
+    ephs = _ephemeral_container_statuses(detailed)
+    if ephs:
+        base["ephemeralContainerStatuses"] = ephs
     if status == PodStatus.RUNNING:
         ready = info.ports_exposed
         if ready:

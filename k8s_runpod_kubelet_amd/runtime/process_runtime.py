@@ -564,6 +564,16 @@ class ProcessRuntime(Runtime):
                         self._handle_one_event(ev, inst, touched, inst_id)
 
     def _handle_one_event(self, ev, inst, touched, inst_id) -> None:
+        # ephemeral (kubectl-debug) containers: record exits, nothing else —
+        # they never gate readiness, restart, or pod completion
+        eph = next((c for c in inst.ephemeral_containers
+                    if c.pid == ev.pid), None)
+        if eph is not None:
+            if ev.type == "exited":
+                eph.exit_code = ev.exit_code
+                eph.finished_at = time.time()
+                touched.add(inst_id)
+            return
         cinfo = next((c for c in inst.containers if c.pid == ev.pid), None)
         is_init = False
         if cinfo is None:
@@ -793,6 +803,9 @@ class ProcessRuntime(Runtime):
             if iid == inst.id:
                 t.cancel()
                 self._restart_timers.pop((iid, cname), None)
+        for c in inst.ephemeral_containers:
+            if c.exit_code is None and c.pid > 0:
+                self._native.signal_process(c.pid, 9, True)
         # Release GPUs as soon as the workload is gone — HBM headroom returns
         # to the ledger without waiting for pod deletion.
         self.binder.unbind(inst.pod_key)
@@ -1011,6 +1024,8 @@ class ProcessRuntime(Runtime):
             containers=[ContainerRuntimeInfo(**vars(c)) for c in inst.containers],
             init_containers=[ContainerRuntimeInfo(**vars(c))
                              for c in inst.init_containers],
+            ephemeral_containers=[ContainerRuntimeInfo(**vars(c))
+                                  for c in inst.ephemeral_containers],
             gpu_indices=list(inst.gpu_indices),
             cost_per_hr=inst.cost_per_hr,
             last_error=inst.last_error,
@@ -1090,12 +1105,14 @@ class ProcessRuntime(Runtime):
         if inst is None:
             return
         if any(c.exit_code is None
-               for c in list(inst.containers) + list(inst.init_containers)):
+               for c in (list(inst.containers) + list(inst.init_containers)
+                         + list(inst.ephemeral_containers))):
             log.warning("grace period expired; SIGKILL", extra={"instance": instance_id})
             self._signal_all(inst, 9)
 
     def _signal_all(self, inst: Instance, sig: int) -> None:
-        for c in list(inst.containers) + list(inst.init_containers):
+        for c in (list(inst.containers) + list(inst.init_containers)
+                  + list(inst.ephemeral_containers)):
             if c.exit_code is None and c.pid > 0:
                 # Whole process group: the container is a session leader.
                 rc = self._native.signal_process(c.pid, sig, True)
@@ -1106,7 +1123,8 @@ class ProcessRuntime(Runtime):
         with self._lock:
             inst = self._instances.pop(instance_id, None)
             if inst is not None:
-                for c in list(inst.containers) + list(inst.init_containers):
+                for c in (list(inst.containers) + list(inst.init_containers)
+                          + list(inst.ephemeral_containers)):
                     self._pid_to_instance.pop(c.pid, None)
         if inst is not None:
             self.binder.unbind(inst.pod_key)
@@ -1179,11 +1197,14 @@ class ProcessRuntime(Runtime):
             inst = self._instances.get(instance_id)
         if inst is None:
             return None
-        names = [c.name for c in inst.params.containers]
+        names = ([c.name for c in inst.params.containers]
+                 + [c.name for c in inst.params.init_containers]
+                 + [c.name for c in inst.ephemeral_specs])
         if container and container in names:
             name = container
         else:
-            name = names[0] if names else ""
+            name = (inst.params.containers[0].name
+                    if inst.params.containers else "")
         return str(self.logs_dir / f"{instance_id}-{name}.log")
 
     def get_logs(self, instance_id: str, container: str = "",
@@ -1304,6 +1325,67 @@ class ProcessRuntime(Runtime):
         out_path.unlink(missing_ok=True)
         return exit_code, output
 
+
+    def add_ephemeral_container(self, instance_id: str, cspec) -> None:
+        """kubectl-debug ephemeral container: runs alongside the pod's
+        containers (same cgroup, GPU env; for image pods it joins a live
+        container's namespaces / rootfs copy) — never restarted, never
+        gating readiness or completion. Raises on a dead pod or a
+        duplicate name."""
+        with self._lock:
+            inst = self._instances.get(instance_id)
+            if inst is None or inst.desired_status != PodStatus.RUNNING:
+                raise RuntimeError(
+                    f"instance {instance_id} is not running")
+            all_names = ([c.name for c in inst.params.containers]
+                         + [c.name for c in inst.params.init_containers]
+                         + [c.name for c in inst.ephemeral_specs])
+            if cspec.name in all_names:
+                raise RuntimeError(
+                    f"container name {cspec.name!r} already in use")
+        params = inst.params
+        env = dict(os.environ)
+        env.pop("ROCR_VISIBLE_DEVICES", None)
+        env.pop("HIP_VISIBLE_DEVICES", None)
+        env.update(params.env)
+        env.update(device_env(inst.gpu_indices, self.binder.ledger.inventory))
+        env.update(cspec.env)
+        env["AMDVK_INSTANCE_ID"] = inst.id
+        argv = list(cspec.command) + list(cspec.args)
+        if not argv:
+            raise RuntimeError("ephemeral container needs a command")
+        target = (inst.params.containers[0]
+                  if inst.params.containers else None)
+        setns_pid, rootfs = self._container_entry(inst, target)
+        if "/" not in argv[0]:
+            if setns_pid >= 0:
+                argv[0] = _resolve_via_proc_root(setns_pid, argv[0])
+            elif rootfs:
+                argv[0] = _resolve_in_tree(rootfs, argv[0])
+            else:
+                resolved = shutil.which(argv[0])
+                if resolved:
+                    argv[0] = resolved
+        stdout_path = str(self.logs_dir / f"{inst.id}-{cspec.name}.log")
+        pid, pidfd, _, _, _ = self._native.launch_process(
+            argv, [f"{k}={v}" for k, v in env.items()],
+            cspec.working_dir or "", stdout_path, stdout_path,
+            inst.cgroup_dir, True, False,
+            cspec.run_as_uid, cspec.run_as_gid,
+            False, "", rootfs, bool(rootfs), [], setns_pid,
+        )
+        cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid,
+                                     started_at=time.time())
+        with self._lock:
+            inst.ephemeral_containers.append(cinfo)
+            inst.ephemeral_specs.append(cspec)
+            self._pid_to_instance[pid] = inst.id
+        self._loop.add_process(pid, pidfd, -1, pid)
+        self._persist(inst)
+        self._notify(instance_id)
+        log.info("ephemeral container started",
+                 extra={"instance": instance_id, "container": cspec.name})
+
     # ------------- persistence / adoption -------------
 
     def _persist(self, inst: Instance) -> None:
@@ -1339,6 +1421,22 @@ class ProcessRuntime(Runtime):
                     "image_id": c.image_id,
                 }
                 for c in inst.containers
+            ],
+            "ephemeral_containers": [
+                {
+                    "name": c.name,
+                    "pid": c.pid,
+                    "started_at": c.started_at,
+                    "finished_at": c.finished_at,
+                    "exit_code": c.exit_code,
+                }
+                for c in inst.ephemeral_containers
+            ],
+            "ephemeral_specs": [
+                {"name": c.name, "image": c.image, "command": c.command,
+                 "args": c.args, "env": c.env,
+                 "run_as_uid": c.run_as_uid, "run_as_gid": c.run_as_gid}
+                for c in inst.ephemeral_specs
             ],
             "init_containers": [
                 {
@@ -1484,6 +1582,30 @@ class ProcessRuntime(Runtime):
                 inst.init_containers.append(icinfo)
             if any(c.exit_code not in (None, 0) for c in inst.init_containers):
                 inst.desired_status = PodStatus.EXITED
+            for c in rec.get("ephemeral_specs", []):
+                inst.ephemeral_specs.append(ContainerSpec(
+                    name=c["name"], image=c.get("image", ""),
+                    command=c.get("command", []), args=c.get("args", []),
+                    env=c.get("env", {}) or {},
+                    run_as_uid=c.get("run_as_uid", -1),
+                    run_as_gid=c.get("run_as_gid", -1)))
+            for c in rec.get("ephemeral_containers", []):
+                einfo = ContainerRuntimeInfo(
+                    name=c["name"], pid=c["pid"],
+                    started_at=c.get("started_at", 0.0),
+                    finished_at=c.get("finished_at", 0.0),
+                    exit_code=c.get("exit_code"))
+                if einfo.exit_code is None:
+                    pidfd = self._native.open_pidfd(einfo.pid)
+                    if pidfd >= 0:
+                        self._loop.add_process(einfo.pid, pidfd, -1,
+                                               einfo.pid)
+                        with self._lock:
+                            self._pid_to_instance[einfo.pid] = inst.id
+                    else:
+                        einfo.exit_code = -1
+                        einfo.finished_at = time.time()
+                inst.ephemeral_containers.append(einfo)
             all_alive = True
             for c in rec.get("containers", []):
                 cinfo = ContainerRuntimeInfo(

@@ -60,6 +60,8 @@ class DetailedStatus:
     name: str = ""
     containers: List[ContainerRuntimeInfo] = field(default_factory=list)
     init_containers: List[ContainerRuntimeInfo] = field(default_factory=list)
+    ephemeral_containers: List[ContainerRuntimeInfo] = field(
+        default_factory=list)
     gpu_indices: List[int] = field(default_factory=list)
     cost_per_hr: float = 0.0
     last_error: str = ""
@@ -210,6 +212,11 @@ class Instance:
     desired_status: str = PodStatus.STARTING
     containers: List[ContainerRuntimeInfo] = field(default_factory=list)
     init_containers: List[ContainerRuntimeInfo] = field(default_factory=list)
+    # kubectl-debug ephemeral containers: run alongside, never restarted,
+    # never gate readiness or completion
+    ephemeral_containers: List[ContainerRuntimeInfo] = field(
+        default_factory=list)
+    ephemeral_specs: List["ContainerSpec"] = field(default_factory=list)
     init_index: int = 0  # next init container to run
     cgroup_dir: str = ""
     created_at: float = field(default_factory=time.time)

@@ -748,3 +748,45 @@ def test_logs_previous_after_restart(stack):
     assert cur and cur.count("run-") >= 1, cur
     kube.delete_pod("default", "prevlog")
     assert wait_until(lambda: gone(kube, "prevlog"), timeout_s=20)
+
+
+def test_ephemeral_container_kubectl_debug(stack):
+    """kubectl debug: adding spec.ephemeralContainers to a running pod
+    starts a debug container alongside (same cgroup/GPU env), visible in
+    ephemeralContainerStatuses with its own logs — never gating readiness
+    or pod completion."""
+    s, kube = stack
+    kube.create_pod("default", make_pod(
+        "debugee", gpus=1, command=["podworker"], args=["--hold"]))
+    assert wait_until(lambda: ready(kube, "debugee"), timeout_s=15)
+
+    pod = kube.get_pod("default", "debugee")
+    pod["spec"]["ephemeralContainers"] = [{
+        "name": "debugger",
+        "image": "amdvk/debug:latest",
+        "command": ["/bin/sh"],
+        "args": ["-c", "echo debugging $AMDVK_INSTANCE_ID; sleep 0.3"],
+    }]
+    kube.update_pod("default", pod)
+
+    def eph_done():
+        try:
+            p = kube.get_pod("default", "debugee")
+        except NotFoundError:
+            return None
+        ecs = p.get("status", {}).get("ephemeralContainerStatuses", [])
+        for c in ecs:
+            if c["name"] == "debugger" and "terminated" in c.get("state", {}):
+                return c
+        return None
+
+    ec = wait_until(eph_done, timeout_s=15)
+    assert ec is not None
+    assert ec["state"]["terminated"]["exitCode"] == 0
+    # debug logs are real and carry the instance env
+    logs = s.provider.get_container_logs("default", "debugee", "debugger")
+    assert "debugging amdvk-" in logs
+    # the pod itself is unaffected
+    assert ready(kube, "debugee") is not None
+    kube.delete_pod("default", "debugee")
+    assert wait_until(lambda: gone(kube, "debugee"), timeout_s=20)
