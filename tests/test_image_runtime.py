@@ -899,3 +899,37 @@ def test_termination_message_surfaces(image_runtime):
     assert s.exit_code == 3
     assert s.containers[0].message == "oom-adjacent-sadness", \
         s.containers[0].message
+
+
+def test_ephemeral_container_joins_image_rootfs(image_runtime):
+    """kubectl debug on an IMAGE pod: the ephemeral container joins the
+    live container's mount namespace and sees the image filesystem."""
+    from k8s_runpod_kubelet_amd.ops import load_native
+
+    if not load_native().probe_mount_namespace():
+        pytest.skip("no mount-namespace capability")
+    rt = image_runtime
+    st = deploy_image_pod(rt, "dbgimg", command=["/usr/local/bin/app"],
+                          args=["hold"])
+    time.sleep(0.3)
+    rt.add_ephemeral_container(st.id, ContainerSpec(
+        name="dbg", command=["/bin/sh"],
+        args=["-c", "read v < /etc/app-release; echo img=$v; "
+                    "[ -e /usr/bin/python3 ] && echo HOST-LEAK "
+                    "|| echo CONTAINED"]))
+    deadline = time.time() + 10
+    done = None
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        eph = next((c for c in s.ephemeral_containers
+                    if c.name == "dbg"), None)
+        if eph is not None and eph.exit_code is not None:
+            done = eph
+            break
+        time.sleep(0.1)
+    assert done is not None and done.exit_code == 0
+    out = rt.get_logs(st.id, "dbg")
+    assert "img=v1" in out, out        # image fs visible
+    assert "CONTAINED" in out, out     # host fs not
+    rt.terminate(st.id)
+    wait_status(rt, st.id, PodStatus.TERMINATED)
