@@ -326,6 +326,10 @@ def prepare_deploy_params(
                 liveness=ProbeSpec.parse(c.get("livenessProbe")),
                 readiness=ProbeSpec.parse(c.get("readinessProbe")),
                 startup=ProbeSpec.parse(c.get("startupProbe")),
+                post_start=ProbeSpec.parse_hook(
+                    (c.get("lifecycle") or {}).get("postStart")),
+                pre_stop=ProbeSpec.parse_hook(
+                    (c.get("lifecycle") or {}).get("preStop")),
                 termination_message_path=c.get(
                     "terminationMessagePath", "/dev/termination-log")
                 or "/dev/termination-log",

@@ -74,7 +74,28 @@ class ProbeSpec:
             cmd = list(raw["exec"].get("command", []) or [])
             if cmd:
                 return ProbeSpec(kind="exec", command=cmd, **common)
+        if raw.get("sleep"):
+            try:
+                secs = float(raw["sleep"].get("seconds", 0))
+            except (TypeError, ValueError):
+                return None
+            # lifecycle-hook sleep handler (k8s 1.29+): modeled as a probe
+            # whose "timeout" is the sleep duration
+            return ProbeSpec(kind="sleep", timeout_s=max(0.0, secs),
+                             **{k: v for k, v in common.items()
+                                if k != "timeout_s"})
         return None
+
+    @staticmethod
+    def parse_hook(raw) -> Optional["ProbeSpec"]:
+        """Parse a lifecycle handler (postStart/preStop): same handler
+        shapes as probes plus sleep; hook executions default to a 30 s
+        bound instead of the probe's 1 s."""
+        spec = ProbeSpec.parse(raw)
+        if spec is not None and spec.kind != "sleep" \
+                and "timeoutSeconds" not in (raw or {}):
+            spec.timeout_s = 30.0
+        return spec
 
 
 @dataclass
@@ -109,6 +130,11 @@ def run_probe(spec: ProbeSpec, env: Dict[str, str],
                 return 200 <= status < 400
             finally:
                 conn.close()
+        if spec.kind == "sleep":
+            import time as _time
+
+            _time.sleep(spec.timeout_s)
+            return True
         if spec.kind == "exec":
             if exec_runner is not None:
                 return exec_runner(spec.command, env, spec.timeout_s) == 0

@@ -457,6 +457,11 @@ class ProcessRuntime(Runtime):
                 and cspec.readiness is None and cspec.startup is None):
             cinfo.ready = True
         into.append(cinfo)
+        if (getattr(cspec, "post_start", None) is not None
+                and into is not inst.init_containers):
+            threading.Thread(target=self._post_start_hook,
+                             args=(inst, cspec, pid),
+                             name="post-start", daemon=True).start()
         with self._lock:
             self._pid_to_instance[pid] = inst.id
         self._loop.add_process(pid, pidfd, ready_fd, pid)
@@ -812,6 +817,63 @@ class ProcessRuntime(Runtime):
         self._release_cgroup(inst.cgroup_dir)
         inst.cgroup_dir = ""
 
+    def _run_hook(self, inst: Instance, cspec, spec) -> bool:
+        """Execute one lifecycle hook (postStart/preStop) confined like an
+        exec probe: container credentials, cgroup, and (image pods) its
+        rootfs."""
+        from .probes import run_probe
+
+        env = dict(os.environ)
+        env.update(inst.params.env)
+        env.update(device_env(inst.gpu_indices,
+                              self.binder.ledger.inventory))
+        env.update(cspec.env)
+        setns_pid, rootfs = self._container_entry(inst, cspec)
+
+        def runner(command, penv, timeout_s):
+            return self._run_confined(
+                command, penv, inst.cgroup_dir,
+                cspec.run_as_uid, cspec.run_as_gid, timeout_s,
+                setns_pid=setns_pid, rootfs=rootfs)
+
+        return run_probe(spec, env, exec_runner=runner)
+
+    def _post_start_hook(self, inst: Instance, cspec, pid: int) -> None:
+        """k8s postStart: runs right after the container starts; failure
+        kills the container into the restartPolicy machinery."""
+        ok = False
+        try:
+            ok = self._run_hook(inst, cspec, cspec.post_start)
+        except Exception:
+            log.exception("postStart hook error")
+        if ok:
+            return
+        log.warning("postStart hook failed; killing container",
+                    extra={"instance": inst.id, "container": cspec.name})
+        with self._lock:
+            cinfo = next((c for c in inst.containers
+                          if c.pid == pid and c.exit_code is None), None)
+            if cinfo is not None:
+                cinfo.message = "postStart hook failed"
+        if cinfo is not None:
+            self._native.signal_process(pid, 9, True)
+
+    def _pre_stop_then_term(self, instance_id: str, inst: Instance,
+                            hooks) -> None:
+        """k8s preStop: hooks run inside the grace window, then SIGTERM.
+        The force-kill timer was armed at terminate() time, so a slow hook
+        cannot extend the pod's life beyond its grace period."""
+        for cspec, cinfo in hooks:
+            if cinfo.exit_code is not None:
+                continue
+            try:
+                self._run_hook(inst, cspec, cspec.pre_stop)
+            except Exception:
+                log.exception("preStop hook error")
+        with self._lock:
+            if inst.desired_status == PodStatus.TERMINATING:
+                self._signal_all(inst, 15)
+
     def _run_probes(self) -> None:
         """One probe-scheduler tick: run due liveness/readiness probes on
         RUNNING containers (see runtime/probes.py; the reference has no
@@ -1086,15 +1148,29 @@ class ProcessRuntime(Runtime):
                 self._teardown_resources(inst)
                 notify_done = True
             else:
-                self._signal_all(inst, 15)  # SIGTERM
                 timer = threading.Timer(
-                    # spec.terminationGracePeriodSeconds (default 10 s here;
-                    # our pods are processes, not images pulling state)
+                    # spec.terminationGracePeriodSeconds — the window
+                    # covers preStop hooks AND the TERM->KILL ladder
                     max(0.1, inst.params.termination_grace_s),
                     self._force_kill, args=(instance_id,))
                 timer.daemon = True
                 self._kill_timers[instance_id] = timer
                 timer.start()
+                hooks = []
+                for cspec in inst.params.containers:
+                    if getattr(cspec, "pre_stop", None) is None:
+                        continue
+                    cinfo = next((c for c in inst.containers
+                                  if c.name == cspec.name
+                                  and c.exit_code is None), None)
+                    if cinfo is not None:
+                        hooks.append((cspec, cinfo))
+                if hooks:
+                    threading.Thread(target=self._pre_stop_then_term,
+                                     args=(instance_id, inst, hooks),
+                                     name="pre-stop", daemon=True).start()
+                else:
+                    self._signal_all(inst, 15)  # SIGTERM
         self._persist(inst)
         if notify_done:
             self._notify(instance_id)
@@ -1464,6 +1540,10 @@ class ProcessRuntime(Runtime):
                     if c.liveness else None,
                     "startup": dataclasses.asdict(c.startup)
                     if c.startup else None,
+                    "post_start": dataclasses.asdict(c.post_start)
+                    if c.post_start else None,
+                    "pre_stop": dataclasses.asdict(c.pre_stop)
+                    if c.pre_stop else None,
                     "env": c.env,  # exec probes run in the container env
                     # credentials/cwd must survive restarts: an adopted pod
                     # whose container crash-restarts would otherwise relaunch
@@ -1532,6 +1612,10 @@ class ProcessRuntime(Runtime):
                         if c.get("liveness") else None,
                         startup=ProbeSpec(**c["startup"])
                         if c.get("startup") else None,
+                        post_start=ProbeSpec(**c["post_start"])
+                        if c.get("post_start") else None,
+                        pre_stop=ProbeSpec(**c["pre_stop"])
+                        if c.get("pre_stop") else None,
                         run_as_uid=c.get("run_as_uid", -1),
                         run_as_gid=c.get("run_as_gid", -1),
                         working_dir=c.get("working_dir", ""),

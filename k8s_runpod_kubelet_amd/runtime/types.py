@@ -157,6 +157,11 @@ class ContainerSpec:
     # spec.containers[].terminationMessagePath: file the container writes
     # its exit message to, surfaced in terminated status (image pods)
     termination_message_path: str = "/dev/termination-log"
+    # lifecycle hooks (exec/httpGet/sleep handlers): postStart runs right
+    # after the container starts (failure kills it into restartPolicy);
+    # preStop runs before SIGTERM, inside the grace window
+    post_start: Optional["ProbeSpec"] = None
+    pre_stop: Optional["ProbeSpec"] = None
 
 
 @dataclass

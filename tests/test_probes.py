@@ -222,3 +222,85 @@ def test_startup_probe_failure_kills_container(process_runtime):
         time.sleep(0.1)
     assert s.desired_status == PodStatus.EXITED
     assert s.containers[0].message == "startup probe failed"
+
+
+def test_pre_stop_hook_runs_before_sigterm(process_runtime, tmp_path):
+    """preStop lifecycle hook runs inside the grace window before SIGTERM
+    reaches the container."""
+    from k8s_runpod_kubelet_amd.runtime.probes import ProbeSpec
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+    import time
+
+    rt = process_runtime
+    marker = tmp_path / "prestop-ran"
+    st = rt.deploy(DeployParams(
+        pod_key="default-ps", name="ps", termination_grace_s=8.0,
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"],
+            args=["-c", f"trap 'echo got-term; [ -f {marker} ] && "
+                        f"echo HOOK-FIRST; exit 0' TERM; "
+                        f"while :; do sleep 0.05; done"],
+            pre_stop=ProbeSpec.parse_hook(
+                {"exec": {"command": ["/bin/sh", "-c",
+                                      f"touch {marker}"]}}))],
+    ))
+    time.sleep(0.3)
+    rt.terminate(st.id)
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        if s.desired_status == PodStatus.TERMINATED:
+            break
+        time.sleep(0.05)
+    assert s.desired_status == PodStatus.TERMINATED
+    assert marker.exists(), "preStop hook never ran"
+    out = rt.get_logs(st.id)
+    assert "HOOK-FIRST" in out, out  # marker existed when TERM arrived
+
+
+def test_post_start_hook_failure_kills(process_runtime):
+    """postStart failure kills the container (k8s semantics)."""
+    from k8s_runpod_kubelet_amd.runtime.probes import ProbeSpec
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+    import time
+
+    rt = process_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-psf", name="psf",
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"], args=["-c", "sleep 30"],
+            post_start=ProbeSpec.parse_hook(
+                {"exec": {"command": ["/bin/false"]}}))],
+    ))
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        s = rt.get_detailed_status(st.id)
+        if s.desired_status == PodStatus.EXITED:
+            break
+        time.sleep(0.05)
+    assert s.desired_status == PodStatus.EXITED
+    assert s.containers[0].message == "postStart hook failed"
+
+
+def test_post_start_sleep_hook_ok(process_runtime):
+    """postStart sleep handler (k8s 1.29 sleep action) succeeds and leaves
+    the container running."""
+    from k8s_runpod_kubelet_amd.runtime.probes import ProbeSpec
+    from k8s_runpod_kubelet_amd.runtime.types import (
+        ContainerSpec, DeployParams, PodStatus)
+    import time
+
+    rt = process_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-pss", name="pss",
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"], args=["-c", "sleep 30"],
+            post_start=ProbeSpec.parse_hook({"sleep": {"seconds": 0.2}}))],
+    ))
+    time.sleep(0.8)
+    s = rt.get_detailed_status(st.id)
+    assert s.desired_status == PodStatus.RUNNING
+    assert s.containers[0].exit_code is None
+    rt.terminate(st.id)
