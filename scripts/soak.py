@@ -124,7 +124,7 @@ def main() -> int:
                                          "command": ["/bin/app"],
                                          "args": ["hold"]}]},
             }
-        if mode in ("hold", "crash", "probed", "started"):
+        if mode in ("hold", "crash", "probed", "started", "hooked"):
             a = pw_args + ["--hold"]
         elif mode == "ok":
             a = pw_args + ["--run-for", f"{rng.uniform(0.05, 0.4):.2f}"]
@@ -149,6 +149,11 @@ def main() -> int:
             container["startupProbe"] = {
                 "exec": {"command": ["/bin/true"]},
                 "periodSeconds": 1, "failureThreshold": 3,
+            }
+        elif mode == "hooked":
+            container["lifecycle"] = {
+                "postStart": {"exec": {"command": ["/bin/true"]}},
+                "preStop": {"sleep": {"seconds": 0.1}},
             }
         elif mode == "restarting":
             spec["restartPolicy"] = "OnFailure"
@@ -207,8 +212,8 @@ def main() -> int:
             # top up population (only while inside the window)
             while now < deadline and len(active) < args.max_active:
                 modes = ["hold", "ok", "fail", "crash", "probed",
-                         "restarting", "started"]
-                weights = [3, 4, 2, 1, 2, 1, 1]
+                         "restarting", "started", "hooked"]
+                weights = [3, 4, 2, 1, 2, 1, 1, 1]
                 if image_ref:
                     modes += ["img-ok", "img-hold"]
                     weights += [3, 2]
@@ -269,7 +274,24 @@ def main() -> int:
                         counters["restarted"] += 1
                         st["deleted"] = True
                         client.delete_pod("default", name)
-                else:  # hold / probed / started / img-hold
+                else:  # hold / probed / started / hooked / img-hold
+                    if (st["mode"] == "hooked" and state == "Ready"
+                            and not st.get("debugged")):
+                        # kubectl-debug under churn: attach an ephemeral
+                        # container to a fraction of hooked pods
+                        st["debugged"] = True
+                        if rng.random() < 0.5:
+                            try:
+                                p2 = client.get_pod("default", name)
+                                p2["spec"]["ephemeralContainers"] = [{
+                                    "name": "dbg",
+                                    "image": "amdvk/debug:soak",
+                                    "command": ["/bin/sh"],
+                                    "args": ["-c", "echo dbg; exit 0"],
+                                }]
+                                client.update_pod("default", p2)
+                            except NotFoundError:
+                                pass
                     if state == "Ready" and age > st["dwell"] and not st["deleted"]:
                         counters["deleted_holds"] += 1
                         st["deleted"] = True
