@@ -87,7 +87,37 @@ def extract_volumes(pod: Dict[str, Any], client: K8sClient) -> Dict[str, VolumeS
             out[name] = VolumeSource(kind="files", files=files,
                                      file_mode=int(cm.get("defaultMode",
                                                           0o644)))
+        elif vol.get("downwardAPI") is not None:
+            da = vol["downwardAPI"] or {}
+            files = {}
+            for item in da.get("items", []) or []:
+                path = item.get("path", "")
+                fref = (item.get("fieldRef") or {}).get("fieldPath", "")
+                if path and fref:
+                    files[path] = _downward_field(pod, fref)
+            out[name] = VolumeSource(kind="files", files=files,
+                                     file_mode=int(da.get("defaultMode",
+                                                          0o644)))
     return out
+
+
+def _downward_field(pod: Dict[str, Any], path: str) -> str:
+    """Downward-API fieldRef subset for volume projection (same fields as
+    the env fieldRef support in envvars.py, plus labels/annotations)."""
+    md = pod.get("metadata", {})
+    if path == "metadata.labels":
+        return "\n".join(f'{k}="{v}"'
+                          for k, v in sorted((md.get("labels") or {}).items()))
+    if path == "metadata.annotations":
+        return "\n".join(f'{k}="{v}"'
+                          for k, v in sorted((md.get("annotations")
+                                              or {}).items()))
+    return {
+        "metadata.name": md.get("name", ""),
+        "metadata.namespace": md.get("namespace", "default"),
+        "metadata.uid": md.get("uid", ""),
+        "spec.nodeName": pod.get("spec", {}).get("nodeName", ""),
+    }.get(path, "")
 
 
 def _project_items(data: Dict[str, str], items) -> Dict[str, str]:

@@ -933,3 +933,30 @@ def test_ephemeral_container_joins_image_rootfs(image_runtime):
     assert "CONTAINED" in out, out     # host fs not
     rt.terminate(st.id)
     wait_status(rt, st.id, PodStatus.TERMINATED)
+
+
+def test_downward_api_volume(image_runtime):
+    """downwardAPI volume: pod metadata projected as files."""
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.provider.translate import (
+        prepare_deploy_params)
+    from tests.conftest import make_pod
+
+    pod = make_pod("dapod", command=["podworker"], args=["--hold"],
+                   labels={"team": "ml"})
+    pod["metadata"]["uid"] = "uid-123"
+    pod["spec"]["volumes"] = [{
+        "name": "podinfo",
+        "downwardAPI": {"items": [
+            {"path": "name", "fieldRef": {"fieldPath": "metadata.name"}},
+            {"path": "labels", "fieldRef": {"fieldPath": "metadata.labels"}},
+        ]},
+    }]
+    pod["spec"]["containers"][0]["volumeMounts"] = [
+        {"name": "podinfo", "mountPath": "/etc/podinfo"}]
+    params = prepare_deploy_params(pod, FakeKube(), Config())
+    v = params.volumes["podinfo"]
+    assert v.kind == "files"
+    assert v.files["name"] == "dapod"
+    assert v.files["labels"] == 'team="ml"'
