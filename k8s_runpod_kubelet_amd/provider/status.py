@@ -136,6 +136,28 @@ def _container_statuses(
     return out
 
 
+def qos_class(pod: Dict[str, Any]) -> str:
+    """k8s QoS classification (kubectl shows status.qosClass): Guaranteed
+    when every container has requests==limits for both cpu and memory;
+    BestEffort when no container has any request/limit; else Burstable."""
+    containers = pod.get("spec", {}).get("containers", []) or []
+    any_set = False
+    guaranteed = bool(containers)
+    for c in containers:
+        res = c.get("resources", {}) or {}
+        req = res.get("requests", {}) or {}
+        lim = res.get("limits", {}) or {}
+        if req or lim:
+            any_set = True
+        for key in ("cpu", "memory"):
+            limv = lim.get(key)
+            if limv is None or req.get(key, limv) != limv:
+                guaranteed = False
+    if guaranteed and any_set:
+        return "Guaranteed"
+    return "Burstable" if any_set else "BestEffort"
+
+
 def _ephemeral_container_statuses(
     detailed: Optional[DetailedStatus],
 ) -> List[Dict[str, Any]]:
@@ -209,6 +231,7 @@ def translate_status(
         "hostIP": node_ip,
         "podIP": node_ip,
         "startTime": start_time,
+        "qosClass": qos_class(pod),
     }
     inits = _init_container_statuses(detailed)
     if inits:

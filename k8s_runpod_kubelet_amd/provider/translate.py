@@ -57,9 +57,10 @@ def extract_volumes(pod: Dict[str, Any], client: K8sClient) -> Dict[str, VolumeS
     containers; host-process pods see the host filesystem, and secret
     volumes are additionally flattened into env for reference parity —
     runpod_client.go:949-979). Secret/ConfigMap content is fetched at
-    translation time like env extraction. Unsupported volume types (PVC,
-    projected, downwardAPI, ...) are skipped — mounts referencing them are
-    ignored rather than failing the pod."""
+    translation time like env extraction; downwardAPI items project pod
+    metadata. Unsupported volume types (PVC, projected, CSI, ...) are
+    skipped — mounts referencing them are ignored rather than failing the
+    pod."""
     from .envvars import ConfigMapCollector, SecretCollector
 
     namespace = pod.get("metadata", {}).get("namespace", "default")

@@ -141,3 +141,25 @@ def test_merge_container_status_preserves_identity():
     assert merged[0]["imageID"] == "img@sha"
     assert merged[0]["restartCount"] == 3
     assert merged[0]["started"] is True
+
+
+def test_qos_class():
+    from k8s_runpod_kubelet_amd.provider.status import qos_class
+
+    def pod(resources_list):
+        return {"spec": {"containers": [
+            {"name": f"c{i}", "resources": r}
+            for i, r in enumerate(resources_list)]}}
+
+    assert qos_class(pod([{}])) == "BestEffort"
+    assert qos_class(pod([{"requests": {"cpu": "1"}}])) == "Burstable"
+    assert qos_class(pod([
+        {"limits": {"cpu": "1", "memory": "1Gi"},
+         "requests": {"cpu": "1", "memory": "1Gi"}}])) == "Guaranteed"
+    # limits-only counts as Guaranteed (requests default to limits)
+    assert qos_class(pod([
+        {"limits": {"cpu": "1", "memory": "1Gi"}}])) == "Guaranteed"
+    # mixed containers -> Burstable
+    assert qos_class(pod([
+        {"limits": {"cpu": "1", "memory": "1Gi"}},
+        {"requests": {"cpu": "1"}}])) == "Burstable"
