@@ -118,6 +118,7 @@ def build_stack(
                 cgroup_root=config.cgroup_root,
                 cgroup_parent=config.cgroup_parent,
                 pod_namespaces=config.pod_namespaces,
+                log_max_bytes=config.pod_log_max_bytes,
                 image_store=ImageStore(config.resolved_image_store_dir()),
                 image_isolation=config.image_isolation,
                 image_gpu_binds=config.image_gpu_binds,

@@ -82,6 +82,9 @@ class Config:
     pod_namespaces: bool = True
     runtime: str = "process"  # process | fake
     pod_log_dir: str = ""  # defaults to <state_dir>/logs
+    # per-container log cap (copytruncate rotation into the --previous
+    # slot; kubelet default is 10 MB — we default higher for GPU jobs)
+    pod_log_max_bytes: int = 50 * 1024 * 1024
     # OCI image execution: pods whose image is present in the local store
     # run inside it (mount-ns overlay + pivot_root, or chroot fallback).
     # The store holds OCI image layouts/archives imported via

@@ -70,6 +70,7 @@ class ProcessRuntime(Runtime):
         podworker: Optional[str] = None,
         enable_cgroups: bool = True,
         pod_namespaces: bool = True,
+        log_max_bytes: int = 50 * 1024 * 1024,
         image_store=None,
         image_isolation: str = "auto",
         image_gpu_binds: Optional[List[str]] = None,
@@ -91,6 +92,7 @@ class ProcessRuntime(Runtime):
         # descendants die with it) + UTS namespace with the pod's hostname.
         # Degrades automatically when CAP_SYS_ADMIN is absent.
         self.pod_namespaces = pod_namespaces
+        self.log_max_bytes = log_max_bytes
         self._podworker = podworker
         # OCI-image execution (reference contract: Containers[0].Image
         # actually runs, runpod_client.go:1304). image_store=None keeps the
@@ -134,6 +136,8 @@ class ProcessRuntime(Runtime):
 
         self._probe_states: Dict[tuple, object] = {}
         self._probe_ticker = Ticker(1.0, self._run_probes, "probes").start()
+        self._logrotate_ticker = Ticker(5.0, self._rotate_big_logs,
+                                        "logrotate").start()
 
     # ------------- deploy -------------
 
@@ -993,6 +997,33 @@ class ProcessRuntime(Runtime):
                         self._native.signal_process(cinfo.pid, 9, True)
                         # exit event drives restartPolicy from here
 
+    def _rotate_big_logs(self) -> None:
+        """Cap per-container log files (copytruncate into the --previous
+        slot): the child keeps its O_APPEND fd, so rename-style rotation
+        would chase the fd — copy+truncate is the standard answer. A
+        chatty container must not fill the node's disk."""
+        if self.log_max_bytes <= 0:
+            return
+        with self._lock:
+            insts = [i for i in self._instances.values()
+                     if i.desired_status in (PodStatus.RUNNING,
+                                             PodStatus.STARTING)]
+        for inst in insts:
+            names = ([c.name for c in inst.params.containers]
+                     + [c.name for c in inst.ephemeral_specs])
+            for name in names:
+                path = self.logs_dir / f"{inst.id}-{name}.log"
+                try:
+                    if path.stat().st_size <= self.log_max_bytes:
+                        continue
+                    shutil.copyfile(path, str(path) + ".prev")
+                    with open(path, "r+b") as fh:
+                        fh.truncate(0)
+                    log.info("container log rotated",
+                             extra={"instance": inst.id, "container": name})
+                except OSError:
+                    continue
+
     def _clear_probe_state(self, instance_id: str, container: str = "") -> None:
         for key in list(self._probe_states):
             if key[0] == instance_id and (not container or key[1] == container):
@@ -1789,6 +1820,7 @@ class ProcessRuntime(Runtime):
     def close(self) -> None:
         self._stop.set()
         self._probe_ticker.stop()
+        self._logrotate_ticker.stop()
         self._loop.wake()
         self._watcher.join(timeout=2.0)
         for timer in self._kill_timers.values():

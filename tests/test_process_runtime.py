@@ -397,3 +397,37 @@ def test_partial_deploy_failure_kills_launched_containers(process_runtime):
     while time.time() < deadline and rt.tracked_process_count() > 0:
         time.sleep(0.05)
     assert rt.tracked_process_count() == 0
+
+
+def test_log_rotation_caps_chatty_containers(synthetic_ledger,
+                                             tmp_state_dir):
+    """A container writing unbounded output gets copytruncate-rotated at
+    the cap (content preserved in the --previous slot; disk bounded)."""
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                        enable_cgroups=False, log_max_bytes=64 * 1024)
+    try:
+        st = rt.deploy(DeployParams(
+            pod_key="default-chatty", name="chatty",
+            containers=[ContainerSpec(
+                name="main", command=["/bin/sh"],
+                args=["-c", "while :; do printf "
+                            "'xxxxxxxxxxxxxxxxxxxxxxxxxxxxxxxx%.0s' "
+                            "$(seq 200); sleep 0.02; done"])],
+        ))
+        from pathlib import Path
+
+        logp = Path(tmp_state_dir) / "logs" / f"{st.id}-main.log"
+        deadline = time.time() + 30
+        rotated = False
+        while time.time() < deadline:
+            prev = Path(str(logp) + ".prev")
+            if prev.exists() and prev.stat().st_size >= 64 * 1024:
+                rotated = True
+                break
+            time.sleep(0.2)
+        assert rotated, "log never rotated"
+        # live file stays bounded (cap + a few seconds of writes)
+        assert logp.stat().st_size < 2 * 64 * 1024 + 512 * 1024
+        rt.terminate(st.id)
+    finally:
+        rt.close()
