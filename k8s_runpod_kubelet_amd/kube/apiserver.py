@@ -86,6 +86,10 @@ class KubeletApiServer:
                             size = os.path.getsize(path)
                         except OSError:
                             size = offset
+                        if size < offset:
+                            # copytruncate rotation shrank the file:
+                            # resume from the new beginning
+                            offset = 0
                         if size > offset:
                             with open(path, "rb") as fh:
                                 fh.seek(offset)
