@@ -431,3 +431,41 @@ def test_log_rotation_caps_chatty_containers(synthetic_ledger,
         rt.terminate(st.id)
     finally:
         rt.close()
+
+
+def test_ephemeral_container_adoption(synthetic_ledger, tmp_state_dir):
+    """A live ephemeral (kubectl-debug) container survives a kubelet
+    restart: re-adopted, watched, and killed with the pod."""
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                        enable_cgroups=False)
+    st = rt.deploy(params(pod_key="default-ephad", args=["--hold"]))
+    wait_ready(rt, st.id)
+    rt.add_ephemeral_container(st.id, ContainerSpec(
+        name="dbg", command=["/bin/sh"], args=["-c", "sleep 60"]))
+    time.sleep(0.2)
+    eph_pid = rt.get_detailed_status(st.id).ephemeral_containers[0].pid
+    assert eph_pid > 0
+    rt._stop.set(); rt._loop.wake(); rt._watcher.join(timeout=2)
+
+    rt2 = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                         enable_cgroups=False)
+    try:
+        rt2.adopt_persisted()
+        s = rt2.get_detailed_status(st.id)
+        eph = s.ephemeral_containers[0]
+        assert eph.name == "dbg" and eph.pid == eph_pid
+        assert eph.exit_code is None
+        import os as _os
+
+        assert _os.path.exists(f"/proc/{eph_pid}")
+        rt2.terminate(st.id)
+        s = wait_status(rt2, st.id, PodStatus.TERMINATED)
+        assert s.desired_status == PodStatus.TERMINATED
+        deadline = time.time() + 5
+        while time.time() < deadline and _os.path.exists(f"/proc/{eph_pid}"):
+            time.sleep(0.05)
+        assert not _os.path.exists(f"/proc/{eph_pid}"), \
+            "ephemeral survived pod termination"
+    finally:
+        rt2.close()
+        rt.close()
