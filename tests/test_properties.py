@@ -311,3 +311,51 @@ def test_translate_status_invariants(status, ports_exposed, exit_code,
     ready_cond = {c["type"]: c["status"] for c in out["conditions"]}["Ready"]
     if ready_cond == "True":
         assert status == "RUNNING" and ports_exposed
+
+
+@given(st.text(alphabet="abcdefghij./-_:@", min_size=0, max_size=60))
+@settings(max_examples=300, deadline=None)
+def test_normalize_ref_idempotent_and_parseable(ref):
+    """normalize_ref is idempotent, and parse_ref round-trips every
+    normalized non-empty reference without raising."""
+    from k8s_runpod_kubelet_amd.runtime.oci import normalize_ref
+    from k8s_runpod_kubelet_amd.runtime.registry import parse_ref
+
+    norm = normalize_ref(ref)
+    assert normalize_ref(norm) == norm
+    if norm:
+        host, name, tagish = parse_ref(norm)
+        assert host
+        assert isinstance(name, str)
+        assert isinstance(tagish, str)
+
+
+@given(st.lists(
+    st.sampled_from(["a", "b", "..", ".", "", "c.d", "..."]),
+    min_size=0, max_size=8).map("/".join))
+@settings(max_examples=300, deadline=None)
+def test_safe_join_never_escapes(path):
+    """_safe_join stays inside the root for arbitrary member paths (the
+    unpacker-escape guard, lexical half; the symlink half is covered by
+    the attack regression tests in test_oci.py)."""
+    import tempfile
+    from pathlib import Path
+
+    from k8s_runpod_kubelet_amd.runtime.oci import _safe_join
+
+    root = Path(tempfile.gettempdir()) / "amdvk-prop-root"
+    out = _safe_join(root, path)
+    # result is root or strictly under it
+    assert out == root or root in out.parents, (path, out)
+
+
+@given(st.text(alphabet="ab/.", min_size=0, max_size=40))
+@settings(max_examples=300, deadline=None)
+def test_container_rel_never_escapes(path):
+    """_container_rel (pod-spec mountPath/workingDir clamp) never yields
+    a path with '..' or an absolute component."""
+    from k8s_runpod_kubelet_amd.runtime.rootfs import _container_rel
+
+    rel = _container_rel(path)
+    assert not rel.startswith("/")
+    assert ".." not in rel.split("/")
