@@ -22,4 +22,6 @@ class InstanceInfo:
     last_error: str = ""
     ready_time: Optional[float] = None  # first time the pod went Ready (metrics)
     deploying: bool = False  # claim flag: a deploy for this pod is in flight
-    restart_sig: int = 0  # last seen restart/backoff signature (change detection)
+    # last seen restart/backoff/ephemeral signature (change detection);
+    # a tuple since ephemeral-container transitions joined the key
+    restart_sig: object = 0

@@ -39,6 +39,7 @@ from ..runtime.types import (ContainerSpec, DeployParams, VolumeMount,
                              VolumeSource)
 from . import annotations as ann
 from .envvars import extract_env_vars
+from .ports import get_requested_ports, tcp_ports_of_container
 
 
 def _volume_mounts_of(c: Dict[str, Any]) -> List[VolumeMount]:
@@ -133,8 +134,6 @@ def _project_items(data: Dict[str, str], items) -> Dict[str, str]:
             files[item.get("path") or key] = data[key]
     return files
 
-
-from .ports import get_requested_ports, tcp_ports_of_container
 
 log = logging.getLogger("provider.translate")
 
