@@ -141,3 +141,35 @@ def test_informer_no_relist_on_watch_timeout():
         assert kube.list_calls == 1
     finally:
         inf.stop()
+
+
+def test_fake_watch_replays_deletes(fake_kube):
+    """Events that happened while a watcher was away — including DELETEs —
+    replay in order on RV resume (grace-0 delete emits DELETED)."""
+    fake_kube.create_pod("default", make_pod("a"))
+    _, rv = fake_kube.list_pods_with_rv()
+    fake_kube.create_pod("default", make_pod("b"))
+    fake_kube.delete_pod("default", "b", grace_period_s=0)
+    events = [(t, p["metadata"]["name"])
+              for t, p in fake_kube.watch_pods(resource_version=rv,
+                                               timeout_s=0.2)]
+    assert events == [("ADDED", "b"), ("DELETED", "b")]
+
+
+def test_informer_cache_consistent_after_missed_delete():
+    """A pod created AND deleted while the informer's watch was down (but
+    within the retained window) must not linger in the cache."""
+    kube = CountingKube()
+    inf = PodInformer(kube, "virtual-runpod", resync_interval_s=300,
+                      watch_timeout_s=0.15)
+    inf.add_handler(lambda t, p: None)
+    inf.start()
+    try:
+        assert inf.wait_for_sync(5)
+        kube.create_pod("default", make_pod("flash"))
+        kube.delete_pod("default", "flash", grace_period_s=0)
+        time.sleep(0.8)  # several watch cycles with replay
+        assert inf.get("default", "flash") is None
+        assert kube.list_calls == 1  # no relist needed
+    finally:
+        inf.stop()
