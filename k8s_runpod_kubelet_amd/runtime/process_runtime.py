@@ -435,7 +435,8 @@ class ProcessRuntime(Runtime):
                 params.hostname or params.name,
                 gpu_device_paths=self._gpu_device_paths(inst),
                 working_dir=working_dir,
-                volume_binds=self._volume_binds(inst, cspec))
+                volume_binds=self._volume_binds(inst, cspec),
+                host_aliases=params.host_aliases)
             argv[0] = self._rootfs_mgr.resolve_argv0(
                 image, prepared, (cspec.command or image.config.entrypoint
                                   or argv)[0])
@@ -449,7 +450,7 @@ class ProcessRuntime(Runtime):
             metrics.cgroup_migrate_seconds.observe(cgroup_s)
         cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
         if image is not None:
-            cinfo.image_id = f"{image.ref}@{image.manifest_digest}" 
+            cinfo.image_id = f"{image.ref}@{image.manifest_digest}"
         # k8s readiness semantics: a running container WITHOUT a
         # readinessProbe is Ready as soon as it starts. The AMDVK_READY_FD
         # pipe protocol (readiness deferred until the workload signals its
