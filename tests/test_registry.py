@@ -132,3 +132,36 @@ def test_imagetool_pull_cli(served_store, tmp_path):
         assert dst.resolve("example/pullme:v1") is not None
     finally:
         srv.stop()
+
+
+def test_imagetool_build_list_rm_gc_cli(tmp_path):
+    """Full imagetool CLI lifecycle: build from a rootfs dir, list with
+    sizes, rm, gc."""
+    import io
+    from contextlib import redirect_stdout
+
+    from k8s_runpod_kubelet_amd.runtime.imagetool import main
+
+    tree = tmp_path / "clitree"
+    (tree / "bin").mkdir(parents=True)
+    (tree / "bin" / "x").write_text("#!/bin/true\n")
+    store_dir = str(tmp_path / "clistore")
+
+    assert main(["--store", store_dir, "build", str(tree),
+                 "--ref", "cli/app:v1", "--entrypoint", "/bin/x"]) == 0
+    out = io.StringIO()
+    with redirect_stdout(out):
+        assert main(["--store", store_dir, "list"]) == 0
+    assert "docker.io/cli/app:v1" in out.getvalue()
+
+    store = ImageStore(store_dir)
+    img = store.resolve("cli/app:v1")
+    assert img.config.entrypoint == ["/bin/x"]
+    store.rootfs_for(img)  # populate cache so gc has work after rm
+
+    assert main(["--store", store_dir, "rm", "cli/app:v1"]) == 0
+    assert store.resolve("cli/app:v1") is None
+    assert main(["--store", store_dir, "rm", "cli/app:v1"]) == 1  # gone
+    assert main(["--store", store_dir, "gc"]) == 0
+    assert not any(store.rootfs_dir.iterdir()) \
+        if store.rootfs_dir.is_dir() else True
