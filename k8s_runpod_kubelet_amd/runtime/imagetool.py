@@ -53,7 +53,8 @@ def main(argv=None) -> int:
     p_b.add_argument("rootfs")
     p_b.add_argument("--ref", required=True)
     p_b.add_argument("--entrypoint", nargs="+", default=[])
-    p_b.add_argument("--cmd", nargs="+", default=[])
+    p_b.add_argument("--cmd", dest="image_cmd", nargs="+",
+                     default=[])
     p_b.add_argument("--env", nargs="*", default=[])
     p_b.add_argument("--workdir", default="")
     p_b.add_argument("--user", default="")
@@ -98,7 +99,7 @@ def main(argv=None) -> int:
     if args.cmd == "build":
         with tempfile.TemporaryDirectory(prefix="amdvk-build-") as td:
             build_layout(td, args.ref, args.rootfs,
-                         entrypoint=args.entrypoint, cmd=args.cmd,
+                         entrypoint=args.entrypoint, cmd=args.image_cmd,
                          env=args.env or None, working_dir=args.workdir,
                          user=args.user)
             ref = store.add_layout(td, args.ref)
