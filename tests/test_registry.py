@@ -165,3 +165,47 @@ def test_imagetool_build_list_rm_gc_cli(tmp_path):
     assert main(["--store", store_dir, "gc"]) == 0
     assert not any(store.rootfs_dir.iterdir()) \
         if store.rootfs_dir.is_dir() else True
+
+
+def test_registry_server_head_and_blob_404(served_store):
+    """Distribution API details: HEAD mirrors GET headers (clients probe
+    with HEAD), unknown blobs 404, /v2/ pings."""
+    import urllib.error
+    import urllib.request
+
+    srv = RegistryServer(served_store).start()
+    try:
+        base = srv.url
+        img = served_store.resolve("example/pullme:v1")
+        # ping
+        with urllib.request.urlopen(f"{base}/v2/", timeout=5) as r:
+            assert r.status == 200
+        # HEAD manifest
+        req = urllib.request.Request(
+            f"{base}/v2/example/pullme/manifests/v1", method="HEAD")
+        with urllib.request.urlopen(req, timeout=5) as r:
+            assert r.status == 200
+            assert int(r.headers["Content-Length"]) > 0
+            assert r.read() == b""  # HEAD: headers only
+        # GET manifest by digest
+        with urllib.request.urlopen(
+                f"{base}/v2/example/pullme/manifests/"
+                f"{img.manifest_digest}", timeout=5) as r:
+            assert r.status == 200
+        # unknown blob 404s
+        try:
+            urllib.request.urlopen(
+                f"{base}/v2/example/pullme/blobs/sha256:" + "0" * 64,
+                timeout=5)
+            raise AssertionError("unknown blob served")
+        except urllib.error.HTTPError as exc:
+            assert exc.code == 404
+        # unknown tag 404s
+        try:
+            urllib.request.urlopen(
+                f"{base}/v2/example/pullme/manifests/nope", timeout=5)
+            raise AssertionError("unknown tag served")
+        except urllib.error.HTTPError as exc:
+            assert exc.code == 404
+    finally:
+        srv.stop()
