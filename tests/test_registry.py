@@ -209,3 +209,27 @@ def test_registry_server_head_and_blob_404(served_store):
             assert exc.code == 404
     finally:
         srv.stop()
+
+
+def test_imagetool_import_cli(tmp_path):
+    import tarfile
+
+    from k8s_runpod_kubelet_amd.runtime.imagetool import main
+    from k8s_runpod_kubelet_amd.runtime.oci import build_layout
+
+    tree = tmp_path / "itree"
+    (tree / "f").parent.mkdir(parents=True, exist_ok=True)
+    (tree / "f").write_text("data")
+    layout = tmp_path / "ilayout"
+    layout.mkdir()
+    build_layout(str(layout), "cli/imported:v3", str(tree))
+    tar_path = tmp_path / "img.tar"
+    with tarfile.open(tar_path, "w") as tf:
+        for p in sorted(layout.rglob("*")):
+            tf.add(p, arcname=str(p.relative_to(layout)), recursive=False)
+    store_dir = str(tmp_path / "istore")
+    assert main(["--store", store_dir, "import", str(tar_path)]) == 0
+    store = ImageStore(store_dir)
+    img = store.resolve("cli/imported:v3")
+    assert img is not None
+    assert (store.rootfs_for(img) / "f").read_text() == "data"
