@@ -123,6 +123,9 @@ def build_stack(
                 image_isolation=config.image_isolation,
                 image_gpu_binds=config.image_gpu_binds,
                 image_extra_binds=config.image_extra_binds,
+                image_registry=config.image_registry,
+                image_registry_token=config.image_registry_token
+                or os.environ.get("AMDVK_REGISTRY_TOKEN", ""),
             )
 
     provider = Provider(client, config, runtime, ledger=ledger, inventory=inventory)

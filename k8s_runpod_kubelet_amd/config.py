@@ -95,6 +95,11 @@ class Config:
     # the thin-image + host-ROCm pattern); "src[:dst[:ro|rw]]"
     image_gpu_binds: List[str] = field(default_factory=lambda: ["/opt/rocm"])
     image_extra_binds: List[str] = field(default_factory=list)
+    # optional in-cluster registry: images missing from the local store are
+    # pulled from here at deploy time (base URL, e.g. http://registry:5000;
+    # empty = no in-kubelet pulls — operators feed the store via imagetool)
+    image_registry: str = ""
+    image_registry_token: str = ""  # or env AMDVK_REGISTRY_TOKEN
     pod_controller_workers: int = 4  # reference uses 1 (main.go:263)
 
     # GPU inventory overrides (mostly for tests / CPU-only dev)

@@ -75,6 +75,8 @@ class ProcessRuntime(Runtime):
         image_isolation: str = "auto",
         image_gpu_binds: Optional[List[str]] = None,
         image_extra_binds: Optional[List[str]] = None,
+        image_registry: str = "",
+        image_registry_token: str = "",
     ):
         from ..ops import load_native
 
@@ -98,6 +100,8 @@ class ProcessRuntime(Runtime):
         # actually runs, runpod_client.go:1304). image_store=None keeps the
         # legacy host-binary behavior everywhere.
         self.image_store = image_store
+        self.image_registry = image_registry
+        self.image_registry_token = image_registry_token
         self._rootfs_mgr = None
         if image_store is not None:
             from .rootfs import RootfsManager
@@ -247,11 +251,34 @@ class ProcessRuntime(Runtime):
     def _resolve_image(self, cspec):
         """ImageStore resolution for a container spec; None keeps the
         legacy host-binary path (reserved amdvk/ refs, absent store, or an
-        image not present in the local store)."""
+        image not present in the local store). With image_registry
+        configured, a store miss triggers an in-kubelet pull — the
+        kubelet-pulls-during-ContainerCreating behavior of a real node."""
         if (self._rootfs_mgr is None or not cspec.image
                 or cspec.image.startswith("amdvk/")):
             return None
-        return self.image_store.resolve(cspec.image)
+        image = self.image_store.resolve(cspec.image)
+        if image is None and self.image_registry:
+            image = self._pull_image(cspec.image)
+        return image
+
+    def _pull_image(self, ref: str):
+        from .registry import RegistryClient, RegistryError
+
+        client = RegistryClient(base_url=self.image_registry,
+                                token=self.image_registry_token)
+        try:
+            pulled = client.pull(ref, self.image_store)
+            log.info("image pulled from registry",
+                     extra={"ref": pulled,
+                            "registry": self.image_registry})
+            return self.image_store.resolve(pulled)
+        except (RegistryError, OSError) as exc:
+            log.warning("image pull failed",
+                        extra={"ref": ref, "err": str(exc)})
+            return None
+        finally:
+            client.close()
 
     def _volume_binds(self, inst: Instance, cspec) -> List[tuple]:
         """Materialize the pod's volumes and return (src, dst, ro) binds

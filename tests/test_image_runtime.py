@@ -960,3 +960,69 @@ def test_downward_api_volume(image_runtime):
     assert v.kind == "files"
     assert v.files["name"] == "dapod"
     assert v.files["labels"] == 'team="ml"'
+
+
+class TestRegistryPullOnDeploy:
+    """In-kubelet pulls: with `image_registry` configured, a store miss at
+    deploy time pulls the image from the registry (the kubelet-pulls-during-
+    ContainerCreating behavior of a real node; closes the ErrImagePull loop
+    end to end — reference images are pulled server-side by RunPod,
+    runpod_client.go:1304)."""
+
+    @pytest.fixture
+    def registry(self, tmp_path, image_store):
+        from k8s_runpod_kubelet_amd.runtime.registry_server import (
+            RegistryServer,
+        )
+
+        srv = RegistryServer(image_store).start()
+        yield srv
+        srv.stop()
+
+    @pytest.fixture
+    def pulling_runtime(self, synthetic_ledger, tmp_state_dir, tmp_path,
+                        registry):
+        # empty local store — everything must come over the wire
+        rt = ProcessRuntime(
+            synthetic_ledger, tmp_state_dir, enable_cgroups=False,
+            image_store=ImageStore(str(tmp_path / "local-store")),
+            image_registry=registry.url)
+        yield rt
+        rt.close()
+
+    def test_store_miss_pulls_and_runs(self, pulling_runtime):
+        rt = pulling_runtime
+        assert rt.image_store.resolve("example/app:v1") is None
+        st = deploy_image_pod(rt, "pull1")
+        s = wait_status(rt, st.id, PodStatus.EXITED)
+        assert s.desired_status == PodStatus.EXITED
+        out = rt.get_logs(st.id)
+        assert "image-file=present" in out  # ran inside the pulled rootfs
+        assert "host-python=absent" in out or "host-python=visible" in out
+        # pulled image is now cached locally: imageID reported, next deploy
+        # needs no registry round-trip
+        assert rt.image_store.resolve("example/app:v1") is not None
+        assert "@sha256:" in s.containers[0].image_id
+
+    def test_pull_failure_stays_pending(self, pulling_runtime):
+        rt = pulling_runtime
+        with pytest.raises(RuntimeError, match="not in the local store"):
+            deploy_image_pod(rt, "pull2", image="example/ghost:v9")
+
+    def test_bad_token_fails_loudly(self, synthetic_ledger, tmp_state_dir,
+                                    tmp_path, image_store):
+        from k8s_runpod_kubelet_amd.runtime.registry_server import (
+            RegistryServer,
+        )
+
+        srv = RegistryServer(image_store, token="sesame").start()
+        rt = ProcessRuntime(
+            synthetic_ledger, tmp_state_dir, enable_cgroups=False,
+            image_store=ImageStore(str(tmp_path / "ls2")),
+            image_registry=srv.url, image_registry_token="wrong")
+        try:
+            with pytest.raises(RuntimeError, match="not in the local store"):
+                deploy_image_pod(rt, "pull3")
+        finally:
+            rt.close()
+            srv.stop()
