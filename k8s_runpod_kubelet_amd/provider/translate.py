@@ -59,7 +59,9 @@ def extract_volumes(pod: Dict[str, Any], client: K8sClient) -> Dict[str, VolumeS
     volumes are additionally flattened into env for reference parity —
     runpod_client.go:949-979). Secret/ConfigMap content is fetched at
     translation time like env extraction; downwardAPI items project pod
-    metadata. Unsupported volume types (PVC, projected, CSI, ...) are
+    metadata; projected volumes merge those three source kinds (plus a
+    best-effort serviceAccountToken) into one directory.
+    Unsupported volume types (PVC, CSI, ...) are
     skipped — mounts referencing them are ignored rather than failing the
     pod."""
     from .envvars import ConfigMapCollector, SecretCollector
@@ -100,7 +102,45 @@ def extract_volumes(pod: Dict[str, Any], client: K8sClient) -> Dict[str, VolumeS
             out[name] = VolumeSource(kind="files", files=files,
                                      file_mode=int(da.get("defaultMode",
                                                           0o644)))
+        elif vol.get("projected") is not None:
+            pr = vol["projected"] or {}
+            files: Dict[str, str] = {}
+            for src in pr.get("sources", []) or []:
+                if (src.get("secret") or {}).get("name"):
+                    s = src["secret"]
+                    data = SecretCollector(client, namespace).get(s["name"])
+                    files.update(_project_items(data, s.get("items")))
+                elif (src.get("configMap") or {}).get("name"):
+                    c = src["configMap"]
+                    data = ConfigMapCollector(client, namespace).get(c["name"])
+                    files.update(_project_items(data, c.get("items")))
+                elif src.get("downwardAPI") is not None:
+                    for item in (src["downwardAPI"] or {}).get("items",
+                                                               []) or []:
+                        path = item.get("path", "")
+                        fref = (item.get("fieldRef") or {}).get("fieldPath",
+                                                                "")
+                        if path and fref:
+                            files[path] = _downward_field(pod, fref)
+                elif (src.get("serviceAccountToken") or {}).get("path"):
+                    # no token issuer offline: project the kubelet's own SA
+                    # token if one is mounted, else an empty placeholder so
+                    # workloads see the expected file
+                    files[src["serviceAccountToken"]["path"]] = \
+                        _local_sa_token()
+            out[name] = VolumeSource(kind="files", files=files,
+                                     file_mode=int(pr.get("defaultMode",
+                                                          0o644)))
     return out
+
+
+def _local_sa_token() -> str:
+    try:
+        with open("/var/run/secrets/kubernetes.io/serviceaccount/token",
+                  "r", encoding="utf-8") as fh:
+            return fh.read()
+    except OSError:
+        return ""
 
 
 def _downward_field(pod: Dict[str, Any], path: str) -> str:

@@ -313,3 +313,37 @@ def test_configmap_and_fieldref_env(fake_kube, pod_factory):
     }])
     pod_env2, _ = extract_env_vars(pod2, fake_kube)
     assert "MODE" not in pod_env2
+
+
+def test_projected_volume_merges_sources(fake_kube):
+    """projected volume: secret + configMap + downwardAPI sources land in
+    one files map; items[] renames apply per source; serviceAccountToken
+    projects a (possibly empty, offline) token file."""
+    from k8s_runpod_kubelet_amd.provider.translate import extract_volumes
+
+    fake_kube.put_secret("default", {"metadata": {"name": "ps"},
+                                     "data": {"tok": b64("s3cr3t")}})
+    fake_kube.put_configmap("default", {"metadata": {"name": "pc"},
+                                         "data": {"conf": "cfg-val"}})
+    pod = make_pod()
+    pod["metadata"]["labels"] = {"app": "demo"}
+    pod["spec"]["volumes"] = [{"name": "proj", "projected": {
+        "defaultMode": 0o600,
+        "sources": [
+            {"secret": {"name": "ps",
+                        "items": [{"key": "tok", "path": "token.txt"}]}},
+            {"configMap": {"name": "pc"}},
+            {"downwardAPI": {"items": [
+                {"path": "labels", "fieldRef": {"fieldPath":
+                                                "metadata.labels"}}]}},
+            {"serviceAccountToken": {"path": "sa-token",
+                                     "audience": "api"}},
+        ]}}]
+    vols = extract_volumes(pod, fake_kube)
+    v = vols["proj"]
+    assert v.kind == "files"
+    assert v.file_mode == 0o600
+    assert v.files["token.txt"] == "s3cr3t"
+    assert v.files["conf"] == "cfg-val"
+    assert 'app="demo"' in v.files["labels"]
+    assert "sa-token" in v.files  # best-effort: empty offline
