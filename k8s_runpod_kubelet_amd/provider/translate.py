@@ -403,6 +403,9 @@ def prepare_deploy_params(
                 termination_message_path=c.get(
                     "terminationMessagePath", "/dev/termination-log")
                 or "/dev/termination-log",
+                termination_message_policy=c.get(
+                    "terminationMessagePolicy", "File") or "File",
+                image_pull_policy=c.get("imagePullPolicy", "") or "",
                 volume_mounts=_volume_mounts_of(c),
             )
         )

@@ -258,8 +258,19 @@ class ProcessRuntime(Runtime):
                 or cspec.image.startswith("amdvk/")):
             return None
         image = self.image_store.resolve(cspec.image)
-        if image is None and self.image_registry:
-            image = self._pull_image(cspec.image)
+        policy = cspec.image_pull_policy or (
+            "Always" if (":" not in cspec.image.rsplit("/", 1)[-1]
+                         or cspec.image.endswith(":latest"))
+            and "@sha256:" not in cspec.image else "IfNotPresent")
+        if policy == "Never" or not self.image_registry:
+            return image
+        if image is None or policy == "Always":
+            pulled = self._pull_image(cspec.image)
+            if pulled is not None:
+                return pulled
+            # Always-policy pull failure: fall back to the cached copy
+            # (availability over the kubelet's strict ErrImagePull — an
+            # offline-first node should not brick on a registry blip)
         return image
 
     def _pull_image(self, ref: str):
@@ -641,6 +652,8 @@ class ProcessRuntime(Runtime):
             cinfo.finished_at = time.time()
             cinfo.ready = False
             tmsg = self._read_termination_message(inst, cinfo.name)
+            if not tmsg and ev.exit_code:
+                tmsg = self._termination_log_fallback(inst, cinfo.name)
             if tmsg:
                 # terminationMessagePath: the container's own last words
                 # win (k8s surfaces them in the terminated state)
@@ -676,6 +689,28 @@ class ProcessRuntime(Runtime):
                 return fh.read(4096).strip()
         except OSError:
             return ""
+
+    def _termination_log_fallback(self, inst: Instance,
+                                  container: str) -> str:
+        """terminationMessagePolicy=FallbackToLogsOnError: a failed
+        container with no termination-message file surfaces the tail of
+        its log (kubelet caps this at 2048 bytes / 80 lines)."""
+        cspec = next((c for c in inst.params.containers
+                      if c.name == container), None)
+        if cspec is None or                 cspec.termination_message_policy != "FallbackToLogsOnError":
+            return ""
+        path = self.get_log_path(inst.id, container)
+        if not path:
+            return ""
+        try:
+            with open(path, "rb") as fh:
+                fh.seek(0, os.SEEK_END)
+                size = fh.tell()
+                fh.seek(max(0, size - 2048))
+                raw = fh.read().decode("utf-8", errors="replace")
+        except OSError:
+            return ""
+        return "\n".join(raw.splitlines()[-80:]).strip()
 
     def _on_container_exit(self, inst: Instance, cinfo=None) -> None:
         # spec.restartPolicy (k8s semantics; the reference's cloud instances

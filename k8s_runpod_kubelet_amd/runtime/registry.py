@@ -69,7 +69,12 @@ class RegistryClient:
 
     def _get(self, url: str, accept: Optional[str] = None) -> httpx.Response:
         headers = {"Accept": accept} if accept else {}
-        resp = self._http.get(url, headers=headers)
+        try:
+            resp = self._http.get(url, headers=headers)
+        except httpx.HTTPError as exc:
+            # uniform error surface: transport failures (refused, DNS,
+            # timeout, TLS) become RegistryError like protocol failures
+            raise RegistryError(f"registry unreachable: {url}: {exc}")
         if resp.status_code == 401:
             raise RegistryError(f"unauthorized: {url}")
         if resp.status_code == 404:

@@ -157,6 +157,12 @@ class ContainerSpec:
     # spec.containers[].terminationMessagePath: file the container writes
     # its exit message to, surfaced in terminated status (image pods)
     termination_message_path: str = "/dev/termination-log"
+    # terminationMessagePolicy: File | FallbackToLogsOnError (tail of the
+    # container log used when the file is empty and the container failed)
+    termination_message_policy: str = "File"
+    # imagePullPolicy: IfNotPresent | Always | Never ("" = k8s default:
+    # Always for :latest/untagged refs, IfNotPresent otherwise)
+    image_pull_policy: str = ""
     # lifecycle hooks (exec/httpGet/sleep handlers): postStart runs right
     # after the container starts (failure kills it into restartPolicy);
     # preStop runs before SIGTERM, inside the grace window

@@ -1026,3 +1026,121 @@ class TestRegistryPullOnDeploy:
         finally:
             rt.close()
             srv.stop()
+
+
+def test_termination_message_fallback_to_logs(image_runtime):
+    """terminationMessagePolicy=FallbackToLogsOnError: a failed container
+    with no termination-message file surfaces its log tail as the
+    terminated message."""
+    rt = image_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-tmf", name="tmf",
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c", "echo boom-reason; exit 3"],
+            termination_message_policy="FallbackToLogsOnError")],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.containers[0].exit_code == 3
+    assert "boom-reason" in (s.containers[0].message or "")
+
+
+def test_termination_message_file_wins_over_log_fallback(image_runtime):
+    """Even with FallbackToLogsOnError, a written termination-message file
+    wins (k8s: the file is used when present)."""
+    rt = image_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-tmw", name="tmw",
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c",
+                     "echo noise; echo real-reason > /dev/termination-log;"
+                     " exit 5"],
+            termination_message_policy="FallbackToLogsOnError")],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.containers[0].message == "real-reason"
+
+
+class TestImagePullPolicy:
+    def test_never_blocks_pull(self, synthetic_ledger, tmp_state_dir,
+                               tmp_path, image_store):
+        from k8s_runpod_kubelet_amd.runtime.registry_server import (
+            RegistryServer,
+        )
+
+        srv = RegistryServer(image_store).start()
+        rt = ProcessRuntime(
+            synthetic_ledger, tmp_state_dir, enable_cgroups=False,
+            image_store=ImageStore(str(tmp_path / "ls-never")),
+            image_registry=srv.url)
+        try:
+            with pytest.raises(RuntimeError, match="not in the local store"):
+                deploy_image_pod(rt, "nv1", image_pull_policy="Never")
+            assert rt.image_store.resolve("example/app:v1") is None
+        finally:
+            rt.close()
+            srv.stop()
+
+    def test_always_refreshes_cached_image(self, synthetic_ledger,
+                                           tmp_state_dir, tmp_path, app_bin):
+        """imagePullPolicy=Always re-pulls a tag even when cached: the
+        registry's newer content under the same tag replaces the local
+        resolve target."""
+        import shutil
+
+        from k8s_runpod_kubelet_amd.runtime.registry_server import (
+            RegistryServer,
+        )
+
+        def mk_tree(marker):
+            tree = tmp_path / f"tree-{marker}"
+            (tree / "usr" / "local" / "bin").mkdir(parents=True)
+            (tree / "etc").mkdir()
+            (tree / "data").mkdir()
+            shutil.copy2(app_bin, tree / "usr" / "local" / "bin" / "app")
+            (tree / "etc" / "app-release").write_text(marker + "\n")
+            return tree
+
+        def mk_store(root, marker):
+            store = ImageStore(str(tmp_path / root))
+            layout = tmp_path / f"layout-{root}"
+            layout.mkdir()
+            build_layout(str(layout), "example/app:v1", str(mk_tree(marker)),
+                         entrypoint=["app"], env=["PATH=/usr/local/bin"])
+            store.add_layout(str(layout), "example/app:v1")
+            return store
+
+        local = mk_store("local-v1", "old")
+        remote = mk_store("remote-v2", "new")
+        old_digest = local.resolve("example/app:v1").manifest_digest
+        new_digest = remote.resolve("example/app:v1").manifest_digest
+        assert old_digest != new_digest
+
+        srv = RegistryServer(remote).start()
+        rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                            enable_cgroups=False, image_store=local,
+                            image_registry=srv.url)
+        try:
+            st = deploy_image_pod(rt, "al1", image_pull_policy="Always")
+            s = wait_status(rt, st.id, PodStatus.EXITED)
+            assert s.desired_status == PodStatus.EXITED
+            assert local.resolve(
+                "example/app:v1").manifest_digest == new_digest
+        finally:
+            rt.close()
+            srv.stop()
+
+    def test_always_falls_back_to_cache_when_registry_down(
+            self, synthetic_ledger, tmp_state_dir, image_store):
+        rt = ProcessRuntime(
+            synthetic_ledger, tmp_state_dir, enable_cgroups=False,
+            image_store=image_store,
+            image_registry="http://127.0.0.1:1")  # nothing listens
+        try:
+            st = deploy_image_pod(rt, "al2", image_pull_policy="Always")
+            s = wait_status(rt, st.id, PodStatus.EXITED)
+            assert s.desired_status == PodStatus.EXITED
+            assert "image-file=present" in rt.get_logs(st.id)
+        finally:
+            rt.close()

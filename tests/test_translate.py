@@ -347,3 +347,15 @@ def test_projected_volume_merges_sources(fake_kube):
     assert v.files["conf"] == "cfg-val"
     assert 'app="demo"' in v.files["labels"]
     assert "sa-token" in v.files  # best-effort: empty offline
+
+
+def test_container_pull_and_termination_policies_parsed(fake_kube):
+    pod = make_pod(containers=[{
+        "name": "a", "image": "example/x:latest",
+        "imagePullPolicy": "Never",
+        "terminationMessagePolicy": "FallbackToLogsOnError",
+    }])
+    params = prepare_deploy_params(pod, fake_kube, Config())
+    c = params.containers[0]
+    assert c.image_pull_policy == "Never"
+    assert c.termination_message_policy == "FallbackToLogsOnError"
