@@ -380,6 +380,8 @@ def prepare_deploy_params(
         c_sc = c.get("securityContext", {}) or {}
         uid = c_sc.get("runAsUser", pod_sc.get("runAsUser"))
         gid = c_sc.get("runAsGroup", pod_sc.get("runAsGroup"))
+        non_root = bool(c_sc.get("runAsNonRoot",
+                                 pod_sc.get("runAsNonRoot", False)))
         containers.append(
             ContainerSpec(
                 name=c.get("name", f"c{i}"),
@@ -406,6 +408,9 @@ def prepare_deploy_params(
                 termination_message_policy=c.get(
                     "terminationMessagePolicy", "File") or "File",
                 image_pull_policy=c.get("imagePullPolicy", "") or "",
+                run_as_non_root=non_root,
+                read_only_root_fs=bool(
+                    c_sc.get("readOnlyRootFilesystem", False)),
                 volume_mounts=_volume_mounts_of(c),
             )
         )
@@ -420,6 +425,8 @@ def prepare_deploy_params(
         c_sc = c.get("securityContext", {}) or {}
         uid = c_sc.get("runAsUser", pod_sc.get("runAsUser"))
         gid = c_sc.get("runAsGroup", pod_sc.get("runAsGroup"))
+        non_root = bool(c_sc.get("runAsNonRoot",
+                                 pod_sc.get("runAsNonRoot", False)))
         init_containers.append(
             ContainerSpec(
                 name=c.get("name", f"init{i}"),

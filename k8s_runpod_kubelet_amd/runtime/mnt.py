@@ -26,6 +26,7 @@ def tmpfs(dst: str, opts: str = "mode=1777") -> Mount:
     return ("tmpfs", dst, "tmpfs", opts, MS_NOSUID | MS_NODEV, False)
 
 
-def overlay(dst: str, lower: str, upper: str, work: str) -> Mount:
+def overlay(dst: str, lower: str, upper: str, work: str,
+            ro: bool = False) -> Mount:
     data = f"lowerdir={lower},upperdir={upper},workdir={work}"
-    return ("overlay", dst, "overlay", data, 0, False)
+    return ("overlay", dst, "overlay", data, 0, ro)

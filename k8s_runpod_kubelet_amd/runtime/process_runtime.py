@@ -404,7 +404,8 @@ class ProcessRuntime(Runtime):
                 gpu_device_paths=self._gpu_device_paths(inst),
                 working_dir=working_dir,
                 volume_binds=self._volume_binds(inst, cspec),
-                host_aliases=params.host_aliases)
+                host_aliases=params.host_aliases,
+                read_only=cspec.read_only_root_fs)
             argv[0] = self._rootfs_mgr.resolve_argv0(image, prepared,
                                                      argv[0])
             rootfs = prepared.rootfs
@@ -438,6 +439,16 @@ class ProcessRuntime(Runtime):
                 resolved = shutil.which(argv[0])
                 if resolved:
                     argv[0] = resolved
+
+        if cspec.run_as_non_root and uid <= 0:
+            # securityContext.runAsNonRoot: k8s refuses to start a
+            # container whose effective user is (or defaults to) root —
+            # CreateContainerConfigError; pod stays Pending until fixed
+            raise RuntimeError(
+                f"container {cspec.name}: runAsNonRoot is set but the "
+                f"container would run as "
+                f"{'root' if uid == 0 else 'the kubelet user (root)'}"
+                " — set runAsUser or an image USER")
 
         env = dict(base_env)
         env.update(cspec.env)
@@ -1645,6 +1656,8 @@ class ProcessRuntime(Runtime):
                     # securityContext.runAsUser
                     "run_as_uid": c.run_as_uid,
                     "run_as_gid": c.run_as_gid,
+                    "run_as_non_root": c.run_as_non_root,
+                    "read_only_root_fs": c.read_only_root_fs,
                     "working_dir": c.working_dir,
                 }
                 for c in inst.params.containers
@@ -1712,6 +1725,9 @@ class ProcessRuntime(Runtime):
                         if c.get("pre_stop") else None,
                         run_as_uid=c.get("run_as_uid", -1),
                         run_as_gid=c.get("run_as_gid", -1),
+                        run_as_non_root=c.get("run_as_non_root", False),
+                        read_only_root_fs=c.get("read_only_root_fs",
+                                                False),
                         working_dir=c.get("working_dir", ""),
                     )
                     for c in rec.get("container_specs", [])

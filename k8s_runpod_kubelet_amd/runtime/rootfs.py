@@ -102,16 +102,24 @@ class RootfsManager:
                 gpu_device_paths: Optional[List[str]] = None,
                 working_dir: str = "",
                 volume_binds: Optional[List[tuple]] = None,
-                host_aliases: Optional[List[tuple]] = None) -> PreparedRootfs:
+                host_aliases: Optional[List[tuple]] = None,
+                read_only: bool = False) -> PreparedRootfs:
         """volume_binds: (host_src, container_dst, ro) tuples from the
         pod's volumes (emptyDir/hostPath/secret/configMap projections,
-        materialized by the runtime)."""
+        materialized by the runtime). read_only = securityContext.
+        readOnlyRootFilesystem: the overlay mountpoint is remounted
+        MS_RDONLY (volume/tmpfs sub-mounts stay writable — k8s
+        semantics); unenforceable in chroot mode (warned, degraded)."""
         cdir = self.containers_dir / f"{instance_id}-{container_name}"
         aliases = list(host_aliases or [])
         if self.mode() == "mountns":
             return self._prepare_overlay(cdir, image, hostname,
                                          gpu_device_paths or [], working_dir,
-                                         volume_binds or [], aliases)
+                                         volume_binds or [], aliases,
+                                         read_only)
+        if read_only:
+            log.warning("readOnlyRootFilesystem not enforceable in chroot "
+                        "fallback mode; continuing writable")
         return self._prepare_chroot(cdir, image, hostname,
                                     gpu_device_paths or [], working_dir,
                                     volume_binds or [], aliases)
@@ -127,7 +135,8 @@ class RootfsManager:
     def _prepare_overlay(self, cdir: Path, image: ResolvedImage,
                          hostname: str, gpu_devices: List[str],
                          working_dir: str, volume_binds: List[tuple],
-                         host_aliases: List[tuple]) -> PreparedRootfs:
+                         host_aliases: List[tuple],
+                         read_only: bool = False) -> PreparedRootfs:
         lower = self.store.rootfs_for(image)
         upper = cdir / "upper"
         work = cdir / "work"
@@ -147,7 +156,7 @@ class RootfsManager:
         self._write_identity(upper, hostname, host_aliases)
 
         mounts = [mnt.overlay(str(merged), str(lower), str(upper),
-                              str(work)),
+                              str(work), ro=read_only),
                   mnt.proc(str(merged / "proc")),
                   # ro sysfs view (ROCm userspace reads KFD topology from
                   # /sys/class/kfd; standard for non-netns containers)

@@ -163,6 +163,12 @@ class ContainerSpec:
     # imagePullPolicy: IfNotPresent | Always | Never ("" = k8s default:
     # Always for :latest/untagged refs, IfNotPresent otherwise)
     image_pull_policy: str = ""
+    # securityContext.runAsNonRoot: refuse to start if the effective uid
+    # resolves to root (CreateContainerConfigError analogue)
+    run_as_non_root: bool = False
+    # securityContext.readOnlyRootFilesystem: rootfs mounted read-only
+    # (volume mounts stay writable); mountns/overlay mode only
+    read_only_root_fs: bool = False
     # lifecycle hooks (exec/httpGet/sleep handlers): postStart runs right
     # after the container starts (failure kills it into restartPolicy);
     # preStop runs before SIGTERM, inside the grace window

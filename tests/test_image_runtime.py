@@ -1144,3 +1144,46 @@ class TestImagePullPolicy:
             assert "image-file=present" in rt.get_logs(st.id)
         finally:
             rt.close()
+
+
+def test_read_only_root_filesystem(image_runtime):
+    """securityContext.readOnlyRootFilesystem: writes to the rootfs fail,
+    writes to volume mounts still succeed (k8s semantics)."""
+    rt = image_runtime
+    if rt._rootfs_mgr.mode() != "mountns":
+        pytest.skip("overlay (mountns) mode required")
+    from k8s_runpod_kubelet_amd.runtime.types import (VolumeMount,
+                                                      VolumeSource)
+
+    st = rt.deploy(DeployParams(
+        pod_key="default-rofs", name="rofs",
+        volumes={"scratch": VolumeSource(kind="emptyDir")},
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c",
+                     "if echo x > /rootfs-write 2>/dev/null; then "
+                     "echo rootfs=writable; else echo rootfs=readonly; fi; "
+                     "if echo y > /scratch/f 2>/dev/null; then "
+                     "echo volume=writable; else echo volume=readonly; fi"],
+            read_only_root_fs=True,
+            volume_mounts=[VolumeMount(name="scratch",
+                                       mount_path="/scratch")])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    out = rt.get_logs(st.id)
+    assert s.exit_code == 0, out
+    assert "rootfs=readonly" in out
+    assert "volume=writable" in out
+
+
+def test_run_as_non_root_refuses_root(image_runtime):
+    """runAsNonRoot + root identity => refused (CreateContainerConfigError
+    analogue); with a non-root runAsUser it starts."""
+    rt = image_runtime
+    with pytest.raises(RuntimeError, match="runAsNonRoot"):
+        deploy_image_pod(rt, "nr1", run_as_non_root=True)  # image user root
+    st = deploy_image_pod(rt, "nr2", run_as_non_root=True,
+                          run_as_uid=1234, run_as_gid=4321)
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.exit_code == 0
+    assert "uid=1234" in rt.get_logs(st.id)
