@@ -313,6 +313,11 @@ class ProcessRuntime(Runtime):
                 vdir = base / vm.name
                 if not vdir.exists():
                     vdir.mkdir(parents=True, exist_ok=True)
+                    if src_spec.kind == "emptyDir":
+                        # kubelet makes emptyDir world-writable (0777): a
+                        # runAsUser container must be able to write its
+                        # scratch volume without fsGroup plumbing
+                        os.chmod(vdir, 0o777)
                     for fname, content in src_spec.files.items():
                         fp = vdir / fname.lstrip("/")
                         fp.parent.mkdir(parents=True, exist_ok=True)

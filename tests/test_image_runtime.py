@@ -1187,3 +1187,27 @@ def test_run_as_non_root_refuses_root(image_runtime):
     s = wait_status(rt, st.id, PodStatus.EXITED)
     assert s.exit_code == 0
     assert "uid=1234" in rt.get_logs(st.id)
+
+
+def test_empty_dir_writable_by_non_root_user(image_runtime):
+    """kubelet creates emptyDir world-writable (0777): a runAsUser
+    container writes its scratch volume without fsGroup (found by
+    scripts/gpu_quickcheck.py — the materialized dir was root 0755)."""
+    rt = image_runtime
+    from k8s_runpod_kubelet_amd.runtime.types import (VolumeMount,
+                                                      VolumeSource)
+
+    st = rt.deploy(DeployParams(
+        pod_key="default-edw", name="edw",
+        volumes={"scratch": VolumeSource(kind="emptyDir")},
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c",
+                     "echo data > /scratch/f && echo scratch=ok"],
+            run_as_uid=1234, run_as_gid=4321,
+            volume_mounts=[VolumeMount(name="scratch",
+                                       mount_path="/scratch")])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.exit_code == 0, rt.get_logs(st.id)
+    assert "scratch=ok" in rt.get_logs(st.id)
