@@ -1211,3 +1211,94 @@ def test_empty_dir_writable_by_non_root_user(image_runtime):
     s = wait_status(rt, st.id, PodStatus.EXITED)
     assert s.exit_code == 0, rt.get_logs(st.id)
     assert "scratch=ok" in rt.get_logs(st.id)
+
+
+def test_secured_image_pod_pulled_through_stack(tmp_state_dir, app_bin,
+                                                tmp_path):
+    """Full stack, kubectl-shaped: an image pod whose image is NOT local is
+    pulled from the configured registry at deploy time and runs with
+    runAsNonRoot + readOnlyRootFilesystem + an emptyDir scratch volume —
+    the scripts/gpu_quickcheck.py scenario as a permanent test."""
+    import shutil
+    import subprocess as sp
+
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.runtime.registry_server import RegistryServer
+    from tests.conftest import wait_until
+
+    src = tmp_path / "sec.c"
+    src.write_text(r'''
+#include <stdio.h>
+#include <unistd.h>
+int main(void) {
+    printf("uid=%d\n", (int)getuid());
+    FILE* w = fopen("/rootfs-write", "w");
+    printf("rootfs=%s\n", w ? "writable" : "readonly");
+    if (w) fclose(w);
+    FILE* v = fopen("/scratch/f", "w");
+    printf("volume=%s\n", v ? "writable" : "readonly");
+    if (v) fclose(v);
+    fflush(stdout);
+    return 0;
+}
+''')
+    binp = tmp_path / "sec"
+    sp.run(["gcc", "-static", "-O1", "-o", str(binp), str(src)], check=True)
+    tree = tmp_path / "sectree"
+    (tree / "bin").mkdir(parents=True)
+    (tree / "etc").mkdir()
+    shutil.copy2(binp, tree / "bin" / "sec")
+    (tree / "etc" / "passwd").write_text("qc:x:1500:1500::/:/bin/sec\n")
+    (tree / "etc" / "group").write_text("qc:x:1500:\n")
+    layout = tmp_path / "seclayout"
+    layout.mkdir()
+    build_layout(str(layout), "example/secured:v1", str(tree),
+                 entrypoint=["/bin/sec"])
+    remote = ImageStore(str(tmp_path / "remote-store"))
+    remote.add_layout(str(layout), "example/secured:v1")
+    srv = RegistryServer(remote).start()
+
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=8,
+                 pending_retry_interval_s=0.2, notify_interval_s=0,
+                 image_registry=srv.url)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.runtime.enable_cgroups = False
+    stack.start(serve_http=False)
+    try:
+        assert stack.runtime.image_store.resolve("example/secured:v1") is None
+        pod = {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": "secured", "namespace": "default"},
+            "spec": {
+                "nodeName": cfg.node_name, "restartPolicy": "Never",
+                "volumes": [{"name": "scratch", "emptyDir": {}}],
+                "containers": [{
+                    "name": "main", "image": "example/secured:v1",
+                    "securityContext": {"runAsUser": 1500,
+                                        "runAsGroup": 1500,
+                                        "runAsNonRoot": True,
+                                        "readOnlyRootFilesystem": True},
+                    "volumeMounts": [{"name": "scratch",
+                                      "mountPath": "/scratch"}],
+                }],
+            },
+        }
+        kube.create_pod("default", pod)
+        assert wait_until(
+            lambda: (kube.get_pod("default", "secured").get("status") or
+                     {}).get("phase") == "Succeeded", timeout_s=20)
+        logs = stack.provider.get_container_logs(
+            "default", "secured", "main", 100)
+        assert "uid=1500" in logs
+        assert "volume=writable" in logs
+        if stack.runtime._rootfs_mgr.mode() == "mountns":
+            assert "rootfs=readonly" in logs
+        # pulled into the stack's local store
+        assert stack.runtime.image_store.resolve(
+            "example/secured:v1") is not None
+    finally:
+        stack.stop()
+        srv.stop()
