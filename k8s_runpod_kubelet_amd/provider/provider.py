@@ -417,7 +417,12 @@ class Provider:
         # placement info (kubectl describe surface).
         images_by_name = {c.name: c.image for c in params.containers}
         for c in detailed.containers:
-            if c.image_id:
+            if c.image_id and c.pull_seconds > 0:
+                self._emit(pod, "Normal", "Pulled",
+                           f'Successfully pulled image '
+                           f'"{images_by_name.get(c.name, "")}" in '
+                           f"{c.pull_seconds:.3f}s ({c.image_id})")
+            elif c.image_id:
                 self._emit(pod, "Normal", "Pulled",
                            f'Container image "{images_by_name.get(c.name, "")}"'
                            f" already present on machine ({c.image_id})")

@@ -102,6 +102,8 @@ class ProcessRuntime(Runtime):
         self.image_store = image_store
         self.image_registry = image_registry
         self.image_registry_token = image_registry_token
+        # ref -> seconds of the most recent registry pull (event surface)
+        self._last_pull_s: dict = {}
         self._rootfs_mgr = None
         if image_store is not None:
             from .rootfs import RootfsManager
@@ -265,8 +267,10 @@ class ProcessRuntime(Runtime):
         if policy == "Never" or not self.image_registry:
             return image
         if image is None or policy == "Always":
+            t0 = time.monotonic()
             pulled = self._pull_image(cspec.image)
             if pulled is not None:
+                self._last_pull_s[cspec.image] = time.monotonic() - t0
                 return pulled
             # Always-policy pull failure: fall back to the cached copy
             # (availability over the kubelet's strict ErrImagePull — an
@@ -505,6 +509,7 @@ class ProcessRuntime(Runtime):
         cinfo = ContainerRuntimeInfo(name=cspec.name, pid=pid, started_at=time.time())
         if image is not None:
             cinfo.image_id = f"{image.ref}@{image.manifest_digest}"
+            cinfo.pull_seconds = self._last_pull_s.pop(cspec.image, 0.0)
         # k8s readiness semantics: a running container WITHOUT a
         # readinessProbe is Ready as soon as it starts. The AMDVK_READY_FD
         # pipe protocol (readiness deferred until the workload signals its

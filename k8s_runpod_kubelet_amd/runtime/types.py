@@ -45,6 +45,8 @@ class ContainerRuntimeInfo:
     # resolved image identity for status.imageID (ref@manifest-digest),
     # empty for host-binary containers
     image_id: str = ""
+    # in-kubelet registry pull time for this launch (0 = local cache hit)
+    pull_seconds: float = 0.0
 
 
 @dataclass

@@ -1302,3 +1302,62 @@ int main(void) {
     finally:
         stack.stop()
         srv.stop()
+
+
+def test_pulled_event_distinguishes_pull_from_cache(tmp_state_dir, app_bin,
+                                                    tmp_path):
+    """kubelet event parity: first deploy of a registry image emits
+    'Successfully pulled image ... in Xs'; a second pod of the same image
+    emits 'already present on machine'."""
+    import shutil
+
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+    from k8s_runpod_kubelet_amd.runtime.registry_server import RegistryServer
+    from tests.conftest import wait_until
+
+    tree = tmp_path / "evtree"
+    (tree / "bin").mkdir(parents=True)
+    shutil.copy2(app_bin, tree / "bin" / "app")
+    layout = tmp_path / "evlayout"
+    layout.mkdir()
+    build_layout(str(layout), "example/ev:v1", str(tree),
+                 entrypoint=["/bin/app"])
+    remote = ImageStore(str(tmp_path / "ev-remote"))
+    remote.add_layout(str(layout), "example/ev:v1")
+    srv = RegistryServer(remote).start()
+
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=8,
+                 pending_retry_interval_s=0.2, notify_interval_s=0,
+                 image_registry=srv.url)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.runtime.enable_cgroups = False
+    stack.start(serve_http=False)
+    try:
+        def mk(name):
+            return {"apiVersion": "v1", "kind": "Pod",
+                    "metadata": {"name": name, "namespace": "default"},
+                    "spec": {"nodeName": cfg.node_name,
+                             "restartPolicy": "Never",
+                             "containers": [{"name": "main",
+                                             "image": "example/ev:v1"}]}}
+
+        def msgs():
+            with kube._lock:
+                return [e.get("message") or ""
+                        for e in kube.events.objects.values()
+                        if e.get("reason") == "Pulled"]
+
+        kube.create_pod("default", mk("ev1"))
+        assert wait_until(
+            lambda: any("Successfully pulled" in m for m in msgs()),
+            timeout_s=10)
+        kube.create_pod("default", mk("ev2"))
+        assert wait_until(
+            lambda: any("already present" in m for m in msgs()),
+            timeout_s=10)
+    finally:
+        stack.stop()
+        srv.stop()
