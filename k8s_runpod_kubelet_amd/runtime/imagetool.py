@@ -49,6 +49,14 @@ def main(argv=None) -> int:
     p_pull.add_argument("--token", default="")
     p_pull.add_argument("--insecure", action="store_true")
 
+    p_push = sub.add_parser("push", help="push a stored image to a "
+                            "registry (server needs --allow-push)")
+    p_push.add_argument("ref")
+    p_push.add_argument("--registry", default="",
+                        help="registry base URL override, e.g. http://host:5000")
+    p_push.add_argument("--token", default="")
+    p_push.add_argument("--insecure", action="store_true")
+
     p_b = sub.add_parser("build", help="build an image from a rootfs dir")
     p_b.add_argument("rootfs")
     p_b.add_argument("--ref", required=True)
@@ -95,6 +103,17 @@ def main(argv=None) -> int:
         finally:
             client.close()
         print(f"pulled {ref}")
+        return 0
+    if args.cmd == "push":
+        from .registry import RegistryClient
+
+        client = RegistryClient(base_url=args.registry, token=args.token,
+                                verify=not args.insecure)
+        try:
+            digest = client.push(args.ref, store)
+        finally:
+            client.close()
+        print(f"pushed {args.ref} ({digest})")
         return 0
     if args.cmd == "build":
         with tempfile.TemporaryDirectory(prefix="amdvk-build-") as td:

@@ -134,6 +134,72 @@ class RegistryClient:
         log.info("image pulled", extra={"ref": out, "digest": man_digest})
         return out
 
+    def push(self, ref: str, store: ImageStore) -> str:
+        """Push a locally stored image to the registry (standard upload
+        flow: blob existence HEAD, POST /blobs/uploads/ + monolithic PUT,
+        manifest PUT by tag). Returns the manifest digest."""
+        img = store.resolve(ref)
+        if img is None:
+            raise RegistryError(f"{ref}: not in the local store")
+        host, name, tagish = parse_ref(ref)
+        if tagish.startswith("sha256:"):
+            raise RegistryError("push needs a tag reference")
+        man_file = (img.layout_dir / "blobs" /
+                    img.manifest_digest.replace(":", "/"))
+        body = man_file.read_bytes()
+        manifest = json.loads(body)
+        for desc in [manifest.get("config") or {}] + list(
+                manifest.get("layers") or []):
+            digest = desc.get("digest", "")
+            if not digest:
+                raise RegistryError("manifest names a blob with no digest")
+            blob = img.layout_dir / "blobs" / digest.replace(":", "/")
+            self._push_blob(host, name, digest, blob.read_bytes())
+        url = self._url(host, f"/v2/{name}/manifests/{tagish}")
+        try:
+            resp = self._http.put(url, content=body, headers={
+                "Content-Type": manifest.get(
+                    "mediaType",
+                    "application/vnd.oci.image.manifest.v1+json")})
+        except httpx.HTTPError as exc:
+            raise RegistryError(f"registry unreachable: {url}: {exc}")
+        if resp.status_code == 401:
+            raise RegistryError(f"unauthorized: {url}")
+        if resp.status_code not in (200, 201):
+            raise RegistryError(
+                f"manifest push failed {resp.status_code}: {resp.text}")
+        log.info("image pushed", extra={"ref": normalize_ref(ref),
+                                        "digest": img.manifest_digest})
+        return img.manifest_digest
+
+    def _push_blob(self, host: str, name: str, digest: str,
+                   data: bytes) -> None:
+        try:
+            head = self._http.head(
+                self._url(host, f"/v2/{name}/blobs/{digest}"))
+            if head.status_code == 200:
+                return  # registry already has it (content-addressed)
+            start = self._http.post(
+                self._url(host, f"/v2/{name}/blobs/uploads/"))
+            if start.status_code == 401:
+                raise RegistryError("unauthorized: blob upload")
+            if start.status_code != 202:
+                raise RegistryError(
+                    f"upload start failed {start.status_code} "
+                    f"(registry read-only? start it with --allow-push)")
+            loc = start.headers.get("Location", "")
+            if loc.startswith("/"):
+                base = self.base_url or f"https://{host}"
+                loc = f"{base}{loc}"
+            sep = "&" if "?" in loc else "?"
+            fin = self._http.put(f"{loc}{sep}digest={digest}",
+                                 content=data)
+            if fin.status_code not in (200, 201):
+                raise RegistryError(
+                    f"blob upload failed {fin.status_code}: {fin.text}")
+        except httpx.HTTPError as exc:
+            raise RegistryError(f"registry unreachable: {exc}")
+
     def _fetch_blob(self, host: str, name: str, digest: str,
                     blobs: Path) -> None:
         if not digest:

@@ -233,3 +233,93 @@ def test_imagetool_import_cli(tmp_path):
     img = store.resolve("cli/imported:v3")
     assert img is not None
     assert (store.rootfs_for(img) / "f").read_text() == "data"
+
+
+def test_push_roundtrip(served_store, tmp_path):
+    """Push flow: a node publishes a local image to a push-enabled
+    registry; another store pulls it back bit-exact."""
+    hub_store = ImageStore(str(tmp_path / "hub"))
+    hub = RegistryServer(hub_store, allow_push=True).start()
+    client = RegistryClient(base_url=hub.url)
+    try:
+        digest = client.push("example/pullme:v1", served_store)
+        assert digest == served_store.resolve(
+            "example/pullme:v1").manifest_digest
+        # the hub can serve it now
+        assert "docker.io/example/pullme:v1" in hub_store.list_refs()
+        dst = ImageStore(str(tmp_path / "dst"))
+        ref = client.pull("example/pullme:v1", dst)
+        img = dst.resolve(ref)
+        assert img.manifest_digest == digest
+        rootfs = dst.rootfs_for(img)
+        assert (rootfs / "etc" / "release").read_text() == "pulled\n"
+        # idempotent re-push (blobs deduped via HEAD)
+        assert client.push("example/pullme:v1", served_store) == digest
+    finally:
+        client.close()
+        hub.stop()
+
+
+def test_push_refused_without_allow_push(served_store, tmp_path):
+    hub = RegistryServer(ImageStore(str(tmp_path / "ro-hub"))).start()
+    client = RegistryClient(base_url=hub.url)
+    try:
+        with pytest.raises(RegistryError, match="read-only|405"):
+            client.push("example/pullme:v1", served_store)
+    finally:
+        client.close()
+        hub.stop()
+
+
+def test_push_requires_token(served_store, tmp_path):
+    hub = RegistryServer(ImageStore(str(tmp_path / "t-hub")),
+                         token="sesame", allow_push=True).start()
+    bad = RegistryClient(base_url=hub.url, token="wrong")
+    good = RegistryClient(base_url=hub.url, token="sesame")
+    try:
+        with pytest.raises(RegistryError, match="unauthorized"):
+            bad.push("example/pullme:v1", served_store)
+        assert good.push("example/pullme:v1", served_store)
+    finally:
+        bad.close()
+        good.close()
+        hub.stop()
+
+
+def test_push_manifest_missing_blob_rejected(served_store, tmp_path):
+    """Manifest referencing an un-uploaded blob -> 400 BLOB_UNKNOWN (the
+    server refuses to register a partial image)."""
+    import hashlib
+    import json
+
+    import httpx
+
+    hub = RegistryServer(ImageStore(str(tmp_path / "p-hub")),
+                         allow_push=True).start()
+    try:
+        manifest = {"schemaVersion": 2,
+                    "mediaType": "application/vnd.oci.image.manifest.v1+json",
+                    "config": {"digest": "sha256:" + "0" * 64, "size": 2},
+                    "layers": []}
+        r = httpx.put(f"{hub.url}/v2/example/ghost/manifests/v1",
+                      content=json.dumps(manifest).encode())
+        assert r.status_code == 400
+        assert "BLOB_UNKNOWN" in r.text
+    finally:
+        hub.stop()
+
+
+def test_imagetool_push_cli(served_store, tmp_path, capsys):
+    from k8s_runpod_kubelet_amd.runtime.imagetool import main as itool
+
+    hub_store_dir = str(tmp_path / "cli-hub")
+    hub = RegistryServer(ImageStore(hub_store_dir), allow_push=True).start()
+    try:
+        rc = itool(["--store", str(served_store.root), "push",
+                    "example/pullme:v1", "--registry", hub.url])
+        assert rc == 0
+        assert "pushed" in capsys.readouterr().out
+        assert ImageStore(hub_store_dir).resolve(
+            "example/pullme:v1") is not None
+    finally:
+        hub.stop()
