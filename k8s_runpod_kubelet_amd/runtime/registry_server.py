@@ -187,6 +187,11 @@ class RegistryServer:
                     return self._respond(401, b'{"errors":[]}')
                 if self.path == "/v2/" or self.path == "/v2":
                     return self._respond(200, b"{}")
+                if self.path.split("?", 1)[0] == "/v2/_catalog":
+                    repos = sorted({outer._repo_tag(r)[0]
+                                    for r in outer.store.list_refs()})
+                    return self._respond(200, json.dumps(
+                        {"repositories": repos}).encode())
                 if m := _MANIFEST_RE.match(self.path):
                     name, ref = m.group(1), m.group(2)
                     layout, desc = outer._find_layout(name, ref)

@@ -323,3 +323,50 @@ def test_imagetool_push_cli(served_store, tmp_path, capsys):
             "example/pullme:v1") is not None
     finally:
         hub.stop()
+
+
+def test_catalog_endpoint(served_store):
+    import httpx
+
+    srv = RegistryServer(served_store).start()
+    try:
+        r = httpx.get(f"{srv.url}/v2/_catalog")
+        assert r.status_code == 200
+        assert r.json() == {"repositories": ["example/pullme"]}
+    finally:
+        srv.stop()
+
+
+def test_chunked_blob_upload(tmp_path):
+    """Docker-style chunked upload: POST -> PATCH chunks -> PUT?digest=
+    finalize; blob then serves."""
+    import hashlib
+
+    import httpx
+
+    hub = RegistryServer(ImageStore(str(tmp_path / "c-hub")),
+                         allow_push=True).start()
+    try:
+        data = b"x" * 10_000 + b"y" * 10_000
+        digest = "sha256:" + hashlib.sha256(data).hexdigest()
+        start = httpx.post(f"{hub.url}/v2/example/chunky/blobs/uploads/")
+        assert start.status_code == 202
+        loc = hub.url + start.headers["Location"]
+        r1 = httpx.patch(loc, content=data[:10_000])
+        assert r1.status_code == 202
+        assert r1.headers["Range"] == "0-9999"
+        r2 = httpx.patch(loc, content=data[10_000:])
+        assert r2.status_code == 202
+        fin = httpx.put(f"{loc}?digest={digest}")
+        assert fin.status_code == 201
+        assert fin.headers["Docker-Content-Digest"] == digest
+        # server finds it content-addressed for manifest validation
+        assert hub._find_blob_file(digest).read_bytes() == data
+        # wrong-digest finalize is refused
+        start2 = httpx.post(f"{hub.url}/v2/example/chunky/blobs/uploads/")
+        loc2 = hub.url + start2.headers["Location"]
+        httpx.patch(loc2, content=b"garbage")
+        bad = httpx.put(f"{loc2}?digest={digest}")
+        assert bad.status_code == 400
+    finally:
+        hub.stop()
