@@ -56,6 +56,10 @@ def build_parser() -> argparse.ArgumentParser:
                    default="auto",
                    help="image rootfs isolation: overlay+mount-ns+pivot_root "
                         "(mountns), chroot fallback, or probe (auto)")
+    p.add_argument("--image-registry", default="",
+                   help="in-cluster registry base URL; images missing from "
+                        "the local store are pulled from here at deploy "
+                        "time (token via AMDVK_REGISTRY_TOKEN)")
     p.add_argument("--gpu-count-override", type=int, default=-1,
                    help="force synthetic GPU inventory (CPU-only dev)")
     p.add_argument("--fake-apiserver", action="store_true",
@@ -100,6 +104,8 @@ def main(argv=None) -> int:
     if args.image_store_dir:
         cfg.image_store_dir = args.image_store_dir
     cfg.image_isolation = args.image_isolation
+    if args.image_registry:
+        cfg.image_registry = args.image_registry
 
     validate_environment()
     fake_srv = None
