@@ -73,6 +73,23 @@ class RegistryServer:
                     return blob
         return None
 
+    def _purge_stale_uploads(self, staging: Path,
+                             max_age_s: float = 3600.0) -> None:
+        """Abandoned uploads (client died mid-push) are reaped after an
+        hour so the staging dir cannot grow without bound."""
+        import time as _time
+
+        now = _time.time()
+        with self._push_lock:
+            for f in staging.iterdir():
+                try:
+                    if now - f.stat().st_mtime > max_age_s:
+                        f.unlink()
+                        self._uploads = {k: v for k, v in
+                                         self._uploads.items() if v != f}
+                except OSError:
+                    pass
+
     def _commit_blob(self, digest: str, data: bytes) -> bool:
         if ("sha256:" + hashlib.sha256(data).hexdigest()) != digest:
             return False
@@ -257,6 +274,7 @@ class RegistryServer:
                 uid = str(uuidlib.uuid4())
                 staging = Path(outer.store.root) / "_push" / "uploads"
                 staging.mkdir(parents=True, exist_ok=True)
+                outer._purge_stale_uploads(staging)
                 with outer._push_lock:
                     outer._uploads[uid] = staging / uid
                 (staging / uid).write_bytes(b"")

@@ -370,3 +370,60 @@ def test_chunked_blob_upload(tmp_path):
         assert bad.status_code == 400
     finally:
         hub.stop()
+
+
+def test_concurrent_pushes_and_pulls(served_store, tmp_path):
+    """Thread-safety: N clients pushing distinct tags while others pull —
+    the hub ends with every tag resolvable and bit-exact manifests."""
+    import threading
+
+    hub_store = ImageStore(str(tmp_path / "mt-hub"))
+    hub = RegistryServer(hub_store, allow_push=True).start()
+    src_digest = served_store.resolve("example/pullme:v1").manifest_digest
+    errors = []
+
+    def push_one(i):
+        c = RegistryClient(base_url=hub.url)
+        try:
+            # same content under a unique tag (content-addressed dedup)
+            layout = served_store._layout_for("example/pullme:v1")
+            dst = ImageStore(str(tmp_path / f"src-{i}"))
+            dst.add_layout(str(layout), f"example/mt-{i}:v1")
+            c.push(f"example/mt-{i}:v1", dst)
+        except Exception as exc:  # noqa: BLE001
+            errors.append(f"push{i}: {exc}")
+        finally:
+            c.close()
+
+    threads = [threading.Thread(target=push_one, args=(i,))
+               for i in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=30)
+    assert not errors, errors
+    for i in range(6):
+        img = hub_store.resolve(f"example/mt-{i}:v1")
+        assert img is not None
+        assert img.manifest_digest == src_digest
+
+    def pull_one(i):
+        c = RegistryClient(base_url=hub.url)
+        try:
+            dst = ImageStore(str(tmp_path / f"pl-{i}"))
+            c.pull(f"example/mt-{i}:v1", dst)
+            assert dst.resolve(
+                f"example/mt-{i}:v1").manifest_digest == src_digest
+        except Exception as exc:  # noqa: BLE001
+            errors.append(f"pull{i}: {exc}")
+        finally:
+            c.close()
+
+    threads = [threading.Thread(target=pull_one, args=(i,))
+               for i in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=30)
+    assert not errors, errors
+    hub.stop()
