@@ -158,11 +158,24 @@ class RegistryServer:
             if not idx_file.exists():
                 continue
             index = _read_json(idx_file)
+            # the store's registration ref (.amdvk-ref) is authoritative:
+            # a layout re-registered under a new tag serves under THAT tag
+            # even though its internal annotation still names the original
+            stored_ref = ""
+            try:
+                stored_ref = (d / ".amdvk-ref").read_text().strip()
+            except OSError:
+                pass
+            stored_match = bool(
+                stored_ref
+                and self._repo_tag(normalize_ref(stored_ref)) == (name, ref))
             for m in index.get("manifests", []):
                 if ref.startswith("sha256:"):
                     if m.get("digest") == ref:
                         return d, m
                     continue
+                if stored_match:
+                    return d, m
                 ann = (m.get("annotations") or {}).get(
                     "org.opencontainers.image.ref.name", "")
                 if ann and self._repo_tag(normalize_ref(ann)) == (name, ref):

@@ -91,6 +91,18 @@ def main() -> int:
         ImageStore(cfg.resolved_image_store_dir()).add_layout(
             layout, "soak/img:v1")
         image_ref = "soak/img:v1"
+        # registry pull-on-miss under churn: a hub serves the same content
+        # under many unique tags; img-pull pods each force a live pull
+        # into the kubelet's store (concurrent deploy workers pulling)
+        from k8s_runpod_kubelet_amd.runtime.registry_server import (
+            RegistryServer,
+        )
+
+        hub_store = ImageStore(os.path.join(bdir, "hub-store"))
+        for i in range(64):
+            hub_store.add_layout(layout, f"soak/pull-{i}:v1")
+        hub = RegistryServer(hub_store).start()
+        cfg.image_registry = hub.url
 
     stack = build_stack(cfg, client=client)
     stack.start(serve_http=False)
@@ -99,6 +111,16 @@ def main() -> int:
     pw_args = (["--expect-gpus", "1"] if gpus else [])
 
     def make(name, mode):
+        if mode == "img-pull":
+            # unique tag -> guaranteed store miss -> in-kubelet pull
+            tag = f"soak/pull-{rng.randrange(64)}:v1"
+            return {
+                "apiVersion": "v1", "kind": "Pod",
+                "metadata": {"name": name, "namespace": "default"},
+                "spec": {"nodeName": cfg.node_name,
+                         "restartPolicy": "Never",
+                         "containers": [{"name": "main", "image": tag}]},
+            }
         if mode == "img-ok":
             # image pod, no command: the image entrypoint runs in its rootfs
             return {
@@ -215,8 +237,8 @@ def main() -> int:
                          "restarting", "started", "hooked"]
                 weights = [3, 4, 2, 1, 2, 1, 1, 1]
                 if image_ref:
-                    modes += ["img-ok", "img-hold"]
-                    weights += [3, 2]
+                    modes += ["img-ok", "img-hold", "img-pull"]
+                    weights += [3, 2, 2]
                 mode = rng.choices(modes, weights=weights)[0]
                 name = f"soak-{seq:05d}"
                 seq += 1
@@ -247,7 +269,7 @@ def main() -> int:
                         pass
                     del active[name]
                     continue
-                if st["mode"] in ("ok", "fail", "img-ok"):
+                if st["mode"] in ("ok", "fail", "img-ok", "img-pull"):
                     want = ("Failed" if st["mode"] == "fail"
                             else "Succeeded")
                     if state == want and not st["deleted"]:

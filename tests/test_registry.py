@@ -427,3 +427,22 @@ def test_concurrent_pushes_and_pulls(served_store, tmp_path):
         t.join(timeout=30)
     assert not errors, errors
     hub.stop()
+
+
+def test_retagged_layout_served_under_store_ref(served_store, tmp_path):
+    """A layout re-registered under a new tag (add_layout) serves under
+    that tag: the store's .amdvk-ref is authoritative over the layout's
+    internal annotation (bug found by the img-pull soak mode: every
+    re-tagged pull 404'd)."""
+    layout = served_store._layout_for("example/pullme:v1")
+    served_store.add_layout(str(layout), "example/retag:v7")
+    srv = RegistryServer(served_store).start()
+    client = RegistryClient(base_url=srv.url)
+    try:
+        dst = ImageStore(str(tmp_path / "rt-dst"))
+        ref = client.pull("example/retag:v7", dst)
+        assert ref == "docker.io/example/retag:v7"
+        assert dst.resolve("example/retag:v7") is not None
+    finally:
+        client.close()
+        srv.stop()
