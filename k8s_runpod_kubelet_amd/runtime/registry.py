@@ -154,7 +154,7 @@ class RegistryClient:
             if not digest:
                 raise RegistryError("manifest names a blob with no digest")
             blob = img.layout_dir / "blobs" / digest.replace(":", "/")
-            self._push_blob(host, name, digest, blob.read_bytes())
+            self._push_blob(host, name, digest, blob)
         url = self._url(host, f"/v2/{name}/manifests/{tagish}")
         try:
             resp = self._http.put(url, content=body, headers={
@@ -173,7 +173,7 @@ class RegistryClient:
         return img.manifest_digest
 
     def _push_blob(self, host: str, name: str, digest: str,
-                   data: bytes) -> None:
+                   blob_file: Path) -> None:
         try:
             head = self._http.head(
                 self._url(host, f"/v2/{name}/blobs/{digest}"))
@@ -192,8 +192,10 @@ class RegistryClient:
                 base = self.base_url or f"https://{host}"
                 loc = f"{base}{loc}"
             sep = "&" if "?" in loc else "?"
-            fin = self._http.put(f"{loc}{sep}digest={digest}",
-                                 content=data)
+            with open(blob_file, "rb") as fh:
+                # file-object content: httpx streams it chunked from disk
+                fin = self._http.put(f"{loc}{sep}digest={digest}",
+                                     content=fh)
             if fin.status_code not in (200, 201):
                 raise RegistryError(
                     f"blob upload failed {fin.status_code}: {fin.text}")
@@ -204,10 +206,38 @@ class RegistryClient:
                     blobs: Path) -> None:
         if not digest:
             raise RegistryError("manifest names a blob with no digest")
-        resp = self._get(self._url(host, f"/v2/{name}/blobs/{digest}"))
-        data = resp.content
-        _verify(digest, data)
-        _write_blob(blobs, digest, data)
+        algo, _, hexd = digest.partition(":")
+        if algo != "sha256":
+            raise RegistryError(f"unsupported digest algorithm {algo!r}")
+        url = self._url(host, f"/v2/{name}/blobs/{digest}")
+        dst = blobs / hexd
+        tmp = dst.with_name(f".{hexd}.part")
+        h = hashlib.sha256()
+        try:
+            with self._http.stream("GET", url) as resp:
+                if resp.status_code == 401:
+                    raise RegistryError(f"unauthorized: {url}")
+                if resp.status_code == 404:
+                    raise RegistryError(f"not found: {url}")
+                if resp.status_code >= 400:
+                    raise RegistryError(
+                        f"registry error {resp.status_code}: {url}")
+                # stream to disk hashing on the fly: layer blobs are
+                # GB-scale for ROCm images, never buffered whole
+                with open(tmp, "wb") as fh:
+                    for chunk in resp.iter_bytes(1 << 20):
+                        h.update(chunk)
+                        fh.write(chunk)
+        except httpx.HTTPError as exc:
+            tmp.unlink(missing_ok=True)
+            raise RegistryError(f"registry unreachable: {url}: {exc}")
+        except BaseException:
+            tmp.unlink(missing_ok=True)
+            raise
+        if h.hexdigest() != hexd:
+            tmp.unlink(missing_ok=True)
+            raise RegistryError(f"digest mismatch for {digest}")
+        tmp.rename(dst)
 
 
 def _verify(digest: str, data: bytes) -> None:

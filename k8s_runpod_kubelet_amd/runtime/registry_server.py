@@ -138,7 +138,8 @@ class RegistryServer:
                             normalize_ref(ref)},
                 }],
             }))
-            self.store.add_layout(str(layout), ref)
+            with self._push_lock:
+                self.store.add_layout(str(layout), ref)
         log.info("image pushed", extra={"ref": ref, "digest": man_digest})
         return man_digest
 
@@ -236,15 +237,23 @@ class RegistryServer:
                                  "application/vnd.oci.image.manifest.v1+json"))
                 if m := _BLOB_RE.match(self.path):
                     name, digest = m.group(1), m.group(2)
-                    # search every layout for the blob (content-addressed)
-                    if outer.store.layouts_dir.is_dir():
-                        for d in outer.store.layouts_dir.iterdir():
-                            blob = d / "blobs" / digest.replace(":", "/")
-                            if blob.exists():
-                                return self._respond(
-                                    200, blob.read_bytes(),
-                                    "application/octet-stream")
-                    return self._respond(404, b'{"errors":[]}')
+                    blob = outer._find_blob_file(digest)
+                    if blob is None or not blob.exists():
+                        return self._respond(404, b'{"errors":[]}')
+                    # stream: layer blobs are GB-scale for ROCm images
+                    size = blob.stat().st_size
+                    self.send_response(200)
+                    self.send_header("Content-Type",
+                                     "application/octet-stream")
+                    self.send_header("Content-Length", str(size))
+                    self.send_header("Docker-Content-Digest", digest)
+                    self.end_headers()
+                    if self.command == "HEAD":
+                        return
+                    with open(blob, "rb") as fh:
+                        while chunk := fh.read(1 << 20):
+                            self.wfile.write(chunk)
+                    return
                 return self._respond(404, b'{"errors":[]}')
 
             do_HEAD = do_GET  # noqa: N815

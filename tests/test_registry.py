@@ -446,3 +446,40 @@ def test_retagged_layout_served_under_store_ref(served_store, tmp_path):
     finally:
         client.close()
         srv.stop()
+
+
+def test_large_blob_streams_roundtrip(tmp_path):
+    """GB-scale layer posture: a 48 MiB layer pushes and pulls bit-exact
+    through the streaming paths (server chunked GET, client stream-to-disk
+    with on-the-fly hashing, file-object upload)."""
+    import hashlib
+    import io
+    import tarfile
+
+    # build an image whose single layer contains a 48 MiB file
+    tree = tmp_path / "big-tree"
+    tree.mkdir()
+    payload = bytes(range(256)) * (48 * 1024 * 4)  # 48 MiB patterned
+    (tree / "blob.bin").write_bytes(payload)
+    layout = tmp_path / "big-layout"
+    layout.mkdir()
+    build_layout(str(layout), "example/big:v1", str(tree))
+    src = ImageStore(str(tmp_path / "big-src"))
+    src.add_layout(str(layout), "example/big:v1")
+
+    hub_store = ImageStore(str(tmp_path / "big-hub"))
+    hub = RegistryServer(hub_store, allow_push=True).start()
+    client = RegistryClient(base_url=hub.url)
+    try:
+        digest = client.push("example/big:v1", src)
+        dst = ImageStore(str(tmp_path / "big-dst"))
+        client.pull("example/big:v1", dst)
+        img = dst.resolve("example/big:v1")
+        assert img.manifest_digest == digest
+        rootfs = dst.rootfs_for(img)
+        assert hashlib.sha256(
+            (rootfs / "blob.bin").read_bytes()).digest() == \
+            hashlib.sha256(payload).digest()
+    finally:
+        client.close()
+        hub.stop()
