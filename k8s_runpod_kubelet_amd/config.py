@@ -102,6 +102,13 @@ class Config:
     image_registry_token: str = ""  # or env AMDVK_REGISTRY_TOKEN
     pod_controller_workers: int = 4  # reference uses 1 (main.go:263)
 
+    # Node-pressure eviction (kubelet memory.available hard-eviction
+    # analogue; 0 disables — this node may share memory with non-pod
+    # workloads, so the operator opts in with a threshold, e.g. 512 for
+    # the kubelet's classic memory.available<...Mi signal)
+    eviction_memory_threshold_mb: int = 0
+    eviction_interval_s: float = 10.0
+
     # GPU inventory overrides (mostly for tests / CPU-only dev)
     gpu_count_override: int = -1
     gpu_vram_gb_override: int = -1

@@ -98,6 +98,8 @@ class Provider:
         # the reference gets from the virtual-kubelet library's recorder
         # (main.go:172-177).
         self.recorder = None
+        self.meminfo_reader = None  # injectable memory.available source
+        self._eviction = None
         # Set by the PodController: re-enqueue a deleted pod ("ns/name")
         # when its instance turns terminal, completing the API delete.
         self.deletion_resync: Optional[Callable[[str], None]] = None
@@ -141,6 +143,16 @@ class Provider:
             Ticker(self.config.pending_retry_interval_s, self.process_pending_pods,
                    "pending-pods").start(),
         ]
+        if self.config.eviction_memory_threshold_mb > 0:
+            from .eviction import EvictionManager
+
+            self._eviction = EvictionManager(
+                self,
+                self.config.eviction_memory_threshold_mb * 1024 * 1024,
+                meminfo_reader=self.meminfo_reader)
+            self._tickers.append(
+                Ticker(self.config.eviction_interval_s,
+                       self._eviction.check, "eviction").start())
 
     def stop(self) -> None:
         for t in self._tickers:
@@ -289,6 +301,56 @@ class Provider:
             self._pods.pop(key, None)
             self._pod_status.pop(key, None)
         metrics.pods_deleted.inc()
+
+    def eviction_candidates(self):
+        """(pod_key, pod_obj, memory_usage_bytes) for every live local pod
+        (eviction.py ranks them). Pending pods are skipped — they hold no
+        node memory yet."""
+        stats_fn = getattr(self.runtime, "get_stats", None)
+        with self._pods_lock:
+            items = [(k, dict(p), self._pod_status.get(k))
+                     for k, p in self._pods.items()]
+        out = []
+        for key, pod, info in items:
+            if info is None or not info.instance_id:
+                continue
+            if phase_of(pod) in ("Succeeded", "Failed"):
+                continue
+            usage = 0
+            if callable(stats_fn):
+                try:
+                    usage = int(stats_fn(info.instance_id).get(
+                        "memoryUsageBytes", 0) or 0)
+                except Exception:
+                    usage = 0
+            out.append((key, pod, usage))
+        return out
+
+    def evict_pod(self, key: str, message: str) -> None:
+        """Node-pressure eviction of one pod: kill its instance now (hard
+        eviction has no grace) and mark the API object Failed/Evicted —
+        the object is NOT deleted (kubectl keeps showing Evicted pods,
+        k8s semantics)."""
+        with self._pods_lock:
+            pod = self._pods.get(key)
+            info = self._pod_status.get(key)
+        if pod is None or info is None or not info.instance_id:
+            return
+        self._emit(pod, "Warning", "Evicted", message)
+        try:
+            self.runtime.terminate(info.instance_id)
+        except Exception:
+            log.exception("eviction terminate failed",
+                          extra={"instance": info.instance_id})
+        info.status = PodStatus.TERMINATED
+        status = {
+            "phase": "Failed",
+            "reason": "Evicted",
+            "message": message,
+            "startTime": ts_rfc3339(info.creation_time),
+        }
+        self._push_status(pod, status)
+        metrics.pods_evicted.inc()
 
     def deletion_finalized(self, namespace: str, name: str) -> bool:
         """True once the pod's instance is actually dead (terminal or

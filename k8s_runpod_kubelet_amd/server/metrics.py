@@ -21,6 +21,9 @@ registry = CollectorRegistry()
 
 pods_created = Counter("amdvk_pods_created_total", "Pods accepted by CreatePod",
                        registry=registry)
+pods_evicted = Counter("amdvk_pods_evicted_total",
+                       "Pods evicted under node memory pressure",
+                       registry=registry)
 pods_deleted = Counter("amdvk_pods_deleted_total", "Pods removed by DeletePod",
                        registry=registry)
 
