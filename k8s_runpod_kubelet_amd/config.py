@@ -108,6 +108,10 @@ class Config:
     # the kubelet's classic memory.available<...Mi signal)
     eviction_memory_threshold_mb: int = 0
     eviction_interval_s: float = 10.0
+    # fail pods bound to a GPU that goes unhealthy (RAS uncorrectable)
+    # mid-run, so controllers reschedule them instead of wedging on a
+    # dead device; the GPU itself is cordoned by the health re-probe
+    evict_on_gpu_failure: bool = True
 
     # GPU inventory overrides (mostly for tests / CPU-only dev)
     gpu_count_override: int = -1

@@ -171,6 +171,11 @@ class Provider:
                     metrics.observe_gpus(self.ledger.snapshot())
             except Exception:
                 log.exception("inventory refresh failed")
+            if self.config.evict_on_gpu_failure:
+                try:
+                    self._evict_gpu_failure_pods()
+                except Exception:
+                    log.exception("gpu-failure eviction failed")
         self.update_all_pod_statuses()
         self.check_backend_health()
 
@@ -326,17 +331,19 @@ class Provider:
             out.append((key, pod, usage))
         return out
 
-    def evict_pod(self, key: str, message: str) -> None:
+    def evict_pod(self, key: str, message: str,
+                  reason: str = "Evicted") -> None:
         """Node-pressure eviction of one pod: kill its instance now (hard
         eviction has no grace) and mark the API object Failed/Evicted —
         the object is NOT deleted (kubectl keeps showing Evicted pods,
-        k8s semantics)."""
+        k8s semantics). GPU-failure eviction reuses this with
+        reason=GPUFailure."""
         with self._pods_lock:
             pod = self._pods.get(key)
             info = self._pod_status.get(key)
         if pod is None or info is None or not info.instance_id:
             return
-        self._emit(pod, "Warning", "Evicted", message)
+        self._emit(pod, "Warning", reason, message)
         try:
             self.runtime.terminate(info.instance_id)
         except Exception:
@@ -345,12 +352,39 @@ class Provider:
         info.status = PodStatus.TERMINATED
         status = {
             "phase": "Failed",
-            "reason": "Evicted",
+            "reason": reason,
             "message": message,
             "startTime": ts_rfc3339(info.creation_time),
         }
         self._push_status(pod, status)
         metrics.pods_evicted.inc()
+
+    def _evict_gpu_failure_pods(self) -> None:
+        """Pods bound to a GPU that lost health (RAS uncorrectable seen by
+        the per-tick re-probe) fail fast with reason GPUFailure, so a Job/
+        Deployment controller reschedules the work — leaving them running
+        against a dead device would wedge them silently. The GPU stays
+        cordoned by the ledger until an operator intervenes."""
+        if self.inventory is None:
+            return
+        bad = {g.index for g in self.inventory.gpus if not g.healthy}
+        if not bad:
+            return
+        with self._pods_lock:
+            items = [(k, dict(p), self._pod_status.get(k))
+                     for k, p in self._pods.items()]
+        for key, pod, info in items:
+            if info is None or not info.instance_id:
+                continue
+            if phase_of(pod) in ("Succeeded", "Failed"):
+                continue
+            hit = sorted(bad.intersection(info.gpu_indices))
+            if hit:
+                self.evict_pod(
+                    key,
+                    f"GPU {hit} became unhealthy (RAS uncorrectable "
+                    f"errors); pod failed so it can be rescheduled",
+                    reason="GPUFailure")
 
     def deletion_finalized(self, namespace: str, name: str) -> bool:
         """True once the pod's instance is actually dead (terminal or

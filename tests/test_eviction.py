@@ -130,3 +130,57 @@ def test_eviction_end_to_end_stack(tmp_state_dir):
         assert evs and "memory" in evs[0]["message"]
     finally:
         stack.stop()
+
+
+def test_gpu_failure_evicts_bound_pods(tmp_state_dir):
+    """A GPU that loses health mid-run (RAS) fails its bound pods fast
+    (reason GPUFailure) so controllers reschedule; pods on healthy GPUs
+    survive; the evicted object stays visible."""
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=8,
+                 runtime="fake", pending_retry_interval_s=0.2,
+                 notify_interval_s=0)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.start(serve_http=False)
+    try:
+        def gpu_pod(name):
+            pod = make_pod(name)
+            pod["spec"]["containers"][0]["resources"] = {
+                "limits": {"amd.com/gpu": "1"}}
+            return pod
+
+        kube.create_pod("default", gpu_pod("g1"))
+        kube.create_pod("default", gpu_pod("g2"))
+        assert wait_until(
+            lambda: all(
+                (kube.get_pod("default", n).get("status") or {}).get(
+                    "phase") == "Running" for n in ("g1", "g2")),
+            timeout_s=10)
+        bound = {}
+        with stack.provider._pods_lock:
+            for k, info in stack.provider._pod_status.items():
+                bound[k.split("-", 1)[1]] = list(info.gpu_indices)
+        assert bound["g1"] and bound["g2"]
+        assert bound["g1"] != bound["g2"]
+
+        # GPU of g1 goes unhealthy; reconcile tick notices
+        dead = bound["g1"][0]
+        for g in stack.inventory.gpus:
+            if g.index == dead:
+                g.healthy = False
+        stack.provider._periodic_reconcile()
+
+        v = kube.get_pod("default", "g1")
+        assert (v.get("status") or {}).get("reason") == "GPUFailure"
+        assert v["status"]["phase"] == "Failed"
+        s = kube.get_pod("default", "g2")
+        assert (s.get("status") or {}).get("phase") == "Running"
+        with kube._lock:
+            assert any(e.get("reason") == "GPUFailure"
+                       for e in kube.events.objects.values())
+    finally:
+        stack.stop()
