@@ -359,3 +359,42 @@ def test_container_rel_never_escapes(path):
     rel = _container_rel(path)
     assert not rel.startswith("/")
     assert ".." not in rel.split("/")
+
+
+@given(st.lists(
+    st.tuples(
+        st.sampled_from(["besteffort", "burstable", "guaranteed"]),
+        st.integers(min_value=-10, max_value=2_100_000_000),
+        st.integers(min_value=0, max_value=1 << 40),
+    ), max_size=12))
+@settings(max_examples=80, deadline=None)
+def test_eviction_ranking_invariants(specs):
+    """rank_victims: never returns critical-priority pods; output is a
+    subset of input; ordering is stable w.r.t. the documented key (qos
+    class rank, then priority, then -usage)."""
+    from k8s_runpod_kubelet_amd.provider.eviction import (
+        CRITICAL_PRIORITY,
+        rank_victims,
+    )
+
+    def mk(i, qos, prio):
+        c = {"name": "c"}
+        if qos == "guaranteed":
+            c["resources"] = {"requests": {"cpu": "1", "memory": "1Gi"},
+                              "limits": {"cpu": "1", "memory": "1Gi"}}
+        elif qos == "burstable":
+            c["resources"] = {"requests": {"memory": "1Mi"}}
+        return {"metadata": {"name": f"p{i}", "namespace": "default"},
+                "spec": {"priority": prio, "containers": [c]}}
+
+    cands = [(f"k{i}", mk(i, qos, prio), usage)
+             for i, (qos, prio, usage) in enumerate(specs)]
+    ranked = rank_victims(cands)
+    keys_in = {k for k, _, _ in cands}
+    assert all(k in keys_in for k, _, _ in ranked)
+    assert all(int(p["spec"]["priority"]) < CRITICAL_PRIORITY
+               for _, p, _ in ranked)
+    rank_of = {"besteffort": 0, "burstable": 1, "guaranteed": 2}
+    seq = [(rank_of[specs[int(k[1:])][0]], specs[int(k[1:])][1],
+            -specs[int(k[1:])][2]) for k, _, _ in ranked]
+    assert seq == sorted(seq)
