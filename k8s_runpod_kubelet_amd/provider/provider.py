@@ -345,7 +345,9 @@ class Provider:
             return
         self._emit(pod, "Warning", reason, message)
         try:
-            self.runtime.terminate(info.instance_id)
+            # hard eviction: no grace (kubelet memory hard-eviction
+            # semantics) — TERM immediately, KILL right behind it
+            self.runtime.terminate(info.instance_id, grace_override_s=0.0)
         except Exception:
             log.exception("eviction terminate failed",
                           extra={"instance": info.instance_id})

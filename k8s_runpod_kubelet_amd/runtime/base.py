@@ -31,7 +31,8 @@ class Runtime(ABC):
         (id + cost). Raises on immediate failure (no GPUs free, bad spec)."""
 
     @abstractmethod
-    def terminate(self, instance_id: str) -> None:
+    def terminate(self, instance_id: str,
+                  grace_override_s: float = -1.0) -> None:
         """Request instance stop (SIGTERM, escalating to SIGKILL)."""
 
     @abstractmethod

@@ -115,7 +115,8 @@ class FakeRuntime(Runtime):
                 c.ready = True
         self._notify(instance_id)
 
-    def terminate(self, instance_id: str) -> None:
+    def terminate(self, instance_id: str,
+                  grace_override_s: float = -1.0) -> None:
         with self._lock:
             self.terminate_calls += 1
             inst = self.instances.get(instance_id)

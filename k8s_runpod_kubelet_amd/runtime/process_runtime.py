@@ -1235,7 +1235,10 @@ class ProcessRuntime(Runtime):
 
     # ------------- terminate / GC -------------
 
-    def terminate(self, instance_id: str) -> None:
+    def terminate(self, instance_id: str,
+                  grace_override_s: float = -1.0) -> None:
+        """grace_override_s >= 0 replaces spec.terminationGracePeriodSeconds
+        for this call (hard eviction passes 0: TERM now, KILL ~now)."""
         notify_done = False
         # Transition under the runtime lock — atomic against the exit-event
         # handler and restart timers (a terminate interleaving with a
@@ -1263,10 +1266,12 @@ class ProcessRuntime(Runtime):
                 self._teardown_resources(inst)
                 notify_done = True
             else:
+                grace = (grace_override_s if grace_override_s >= 0
+                         else inst.params.termination_grace_s)
                 timer = threading.Timer(
                     # spec.terminationGracePeriodSeconds — the window
                     # covers preStop hooks AND the TERM->KILL ladder
-                    max(0.1, inst.params.termination_grace_s),
+                    max(0.1, grace),
                     self._force_kill, args=(instance_id,))
                 timer.daemon = True
                 self._kill_timers[instance_id] = timer

@@ -184,3 +184,67 @@ def test_gpu_failure_evicts_bound_pods(tmp_state_dir):
                        for e in kube.events.objects.values())
     finally:
         stack.stop()
+
+
+@pytest.mark.skipif(__import__("os").geteuid() != 0,
+                    reason="process runtime tests need root")
+def test_eviction_kills_real_process_and_frees_gpu(tmp_state_dir):
+    """Hard eviction against the real ProcessRuntime: the workload process
+    dies, the GPU reservation frees (re-placeable), nothing leaks."""
+    import os
+
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=1,
+                 pending_retry_interval_s=0.2, notify_interval_s=0,
+                 eviction_memory_threshold_mb=100,
+                 eviction_interval_s=0.1)
+    kube = FakeKube()
+    stack = build_stack(cfg, client=kube)
+    stack.runtime.enable_cgroups = False
+    pressure = [1 << 40]
+    stack.provider.meminfo_reader = lambda: pressure[0]
+    stack.start(serve_http=False)
+    try:
+        pod = make_pod("evictme")
+        pod["spec"]["containers"][0]["command"] = ["sleep"]
+        pod["spec"]["containers"][0]["args"] = ["300"]
+        pod["spec"]["containers"][0]["resources"] = {
+            "limits": {"amd.com/gpu": "1"}}
+        kube.create_pod("default", pod)
+        assert wait_until(
+            lambda: (kube.get_pod("default", "evictme").get("status") or
+                     {}).get("phase") == "Running", timeout_s=10)
+        with stack.provider._pods_lock:
+            info = next(iter(stack.provider._pod_status.values()))
+            iid = info.instance_id
+        pid = stack.runtime.get_detailed_status(iid).containers[0].pid
+        assert pid and os.path.exists(f"/proc/{pid}")
+
+        pressure[0] = 1 << 20
+        assert wait_until(
+            lambda: (kube.get_pod("default", "evictme").get("status") or
+                     {}).get("reason") == "Evicted", timeout_s=10)
+        pressure[0] = 1 << 40
+        # process actually dies and the GPU frees for the next pod
+        assert wait_until(lambda: not os.path.exists(f"/proc/{pid}")
+                          or open(f"/proc/{pid}/stat").read()
+                          .rsplit(") ", 1)[-1].startswith("Z"),
+                          timeout_s=10)
+        assert wait_until(
+            lambda: all(not s.pod_keys for s in stack.ledger.snapshot()),
+            timeout_s=10)
+        # node can place a new GPU pod after the eviction
+        pod2 = make_pod("next")
+        pod2["spec"]["containers"][0]["command"] = ["sleep"]
+        pod2["spec"]["containers"][0]["args"] = ["30"]
+        pod2["spec"]["containers"][0]["resources"] = {
+            "limits": {"amd.com/gpu": "1"}}
+        kube.create_pod("default", pod2)
+        assert wait_until(
+            lambda: (kube.get_pod("default", "next").get("status") or
+                     {}).get("phase") == "Running", timeout_s=10)
+    finally:
+        stack.stop()
