@@ -59,6 +59,11 @@ class SecretCollector:
         self.client = client
         self.namespace = namespace
         self._cache: Dict[str, Dict[str, str]] = {}
+        self._missing: set = set()
+
+    def exists(self, name: str) -> bool:
+        self.get(name)
+        return name not in self._missing
 
     def get(self, name: str) -> Dict[str, str]:
         if name not in self._cache:
@@ -68,6 +73,7 @@ class SecretCollector:
             except Exception as exc:
                 if is_not_found(exc):
                     log.warning("secret not found", extra={"secret": name, "ns": self.namespace})
+                    self._missing.add(name)
                     self._cache[name] = {}
                 else:
                     raise
@@ -83,6 +89,11 @@ class ConfigMapCollector:
         self.client = client
         self.namespace = namespace
         self._cache: Dict[str, Dict[str, str]] = {}
+        self._missing: set = set()
+
+    def exists(self, name: str) -> bool:
+        self.get(name)
+        return name not in self._missing
 
     def get(self, name: str) -> Dict[str, str]:
         if name not in self._cache:
@@ -95,6 +106,7 @@ class ConfigMapCollector:
                 if is_not_found(exc):
                     log.warning("configmap not found",
                                 extra={"configmap": name, "ns": self.namespace})
+                    self._missing.add(name)
                     self._cache[name] = {}
                 else:
                     raise

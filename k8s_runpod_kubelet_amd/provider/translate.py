@@ -73,20 +73,39 @@ def extract_volumes(pod: Dict[str, Any], client: K8sClient) -> Dict[str, VolumeS
         if not name:
             continue
         if "emptyDir" in vol:
-            out[name] = VolumeSource(kind="emptyDir")
+            ed = vol.get("emptyDir") or {}
+            size = 0
+            if ed.get("sizeLimit"):
+                size = int(resource_parse_bytes(str(ed["sizeLimit"])))
+            out[name] = VolumeSource(kind="emptyDir",
+                                     medium=str(ed.get("medium", "") or ""),
+                                     size_limit_bytes=size)
         elif (vol.get("hostPath") or {}).get("path"):
             out[name] = VolumeSource(kind="hostPath",
                                      host_path=vol["hostPath"]["path"])
         elif (vol.get("secret") or {}).get("secretName"):
             sec = vol["secret"]
-            data = SecretCollector(client, namespace).get(sec["secretName"])
+            col = SecretCollector(client, namespace)
+            data = col.get(sec["secretName"])
+            if not col.exists(sec["secretName"]) and \
+                    not bool(sec.get("optional")):
+                # k8s: a non-optional missing secret blocks the pod
+                # (ContainerCreating); Pending-with-retry is the local
+                # equivalent — the pod starts once the secret appears
+                raise ValidationError(
+                    f'volume "{name}": secret '
+                    f'{sec["secretName"]!r} not found')
             files = _project_items(data, sec.get("items"))
             out[name] = VolumeSource(kind="files", files=files,
                                      file_mode=int(sec.get("defaultMode",
                                                            0o644)))
         elif (vol.get("configMap") or {}).get("name"):
             cm = vol["configMap"]
-            data = ConfigMapCollector(client, namespace).get(cm["name"])
+            col = ConfigMapCollector(client, namespace)
+            data = col.get(cm["name"])
+            if not col.exists(cm["name"]) and not bool(cm.get("optional")):
+                raise ValidationError(
+                    f'volume "{name}": configMap {cm["name"]!r} not found')
             files = _project_items(data, cm.get("items"))
             out[name] = VolumeSource(kind="files", files=files,
                                      file_mode=int(cm.get("defaultMode",

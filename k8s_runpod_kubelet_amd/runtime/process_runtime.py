@@ -313,6 +313,15 @@ class ProcessRuntime(Runtime):
                 if not os.path.exists(src):
                     raise RuntimeError(
                         f"hostPath volume {vm.name}: {src} does not exist")
+            elif (src_spec.kind == "emptyDir"
+                    and src_spec.medium == "Memory"
+                    and self._rootfs_mgr is not None
+                    and self._rootfs_mgr.mode() == "mountns"):
+                # emptyDir.medium=Memory: tmpfs mounted by the child
+                # (size capped by sizeLimit); nothing materialized on disk
+                out.append((str(src_spec.size_limit_bytes), vm.mount_path,
+                            False, "tmpfs"))
+                continue
             else:
                 vdir = base / vm.name
                 if not vdir.exists():

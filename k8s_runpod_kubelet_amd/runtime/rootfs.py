@@ -185,9 +185,17 @@ class RootfsManager:
                 target.parent.mkdir(parents=True, exist_ok=True)
                 target.touch()
             mounts.append(mnt.bind(src, str(merged / rel), ro=ro))
-        for src, dst, ro, _kind in volume_binds:
+        for src, dst, ro, kind in volume_binds:
             rel = _container_rel(dst)
             target = upper / rel
+            if kind == "tmpfs":
+                # emptyDir.medium=Memory: src carries sizeLimit bytes
+                target.mkdir(parents=True, exist_ok=True)
+                opts = "mode=1777"
+                if src and src != "0":
+                    opts += f",size={src}"
+                mounts.append(mnt.tmpfs(str(merged / rel), opts))
+                continue
             if os.path.isdir(src):
                 target.mkdir(parents=True, exist_ok=True)
             else:

@@ -126,7 +126,8 @@ class VolumeSource:
 
     kind: "emptyDir" (per-pod scratch dir, shared between the pod's
     containers, survives container restarts — k8s semantics; medium=Memory
-    is backed by disk here, a documented deviation), "hostPath" (host
+    mounts a tmpfs sized by sizeLimit in mount-ns mode, plain dir in the
+    chroot fallback), "hostPath" (host
     directory/file bound in; mount-namespace isolation mode only), or
     "files" (secret/configMap projected to files — content fetched from
     the API at translation time like env extraction)."""
@@ -134,6 +135,10 @@ class VolumeSource:
     host_path: str = ""
     files: Dict[str, str] = field(default_factory=dict)
     file_mode: int = 0o644
+    # emptyDir.medium ("Memory" -> tmpfs in mountns mode; plain dir in
+    # chroot fallback) and emptyDir.sizeLimit (tmpfs size= cap)
+    medium: str = ""
+    size_limit_bytes: int = 0
 
 
 @dataclass

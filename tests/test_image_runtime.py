@@ -1361,3 +1361,42 @@ def test_pulled_event_distinguishes_pull_from_cache(tmp_state_dir, app_bin,
     finally:
         stack.stop()
         srv.stop()
+
+
+def test_empty_dir_memory_is_tmpfs_with_size_limit(image_runtime):
+    """emptyDir.medium=Memory mounts a tmpfs (visible in /proc/mounts)
+    capped by sizeLimit: a write past the cap fails, a small write
+    succeeds."""
+    rt = image_runtime
+    if rt._rootfs_mgr.mode() != "mountns":
+        pytest.skip("tmpfs emptyDir needs mount-ns mode")
+    from k8s_runpod_kubelet_amd.runtime.types import (VolumeMount,
+                                                      VolumeSource)
+
+    st = rt.deploy(DeployParams(
+        pod_key="default-mem", name="mem",
+        volumes={"m": VolumeSource(kind="emptyDir", medium="Memory",
+                                   size_limit_bytes=1 << 20)},
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c",
+                     # shell-builtins only: the test image has no coreutils
+                     'while read -r l; do case "$l" in '
+                     '*" /m tmpfs"*) echo mount=tmpfs ;; esac; '
+                     "done < /proc/mounts; "
+                     "printf hello > /m/small && echo small=ok; "
+                     'x=0123456789; while [ ${#x} -lt 60000 ]; '
+                     'do x="$x$x"; done; '
+                     "i=0; ok=1; while [ $i -lt 40 ]; do "
+                     'printf %s "$x" >> /m/big 2>/dev/null || '
+                     "{ ok=0; break; }; i=$((i+1)); done; "
+                     "if [ $ok = 1 ]; then echo cap=missing; "
+                     "else echo cap=enforced; fi"],
+            volume_mounts=[VolumeMount(name="m", mount_path="/m")])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    out = rt.get_logs(st.id)
+    assert s.exit_code == 0, out
+    assert "mount=tmpfs" in out
+    assert "small=ok" in out
+    assert "cap=enforced" in out

@@ -359,3 +359,37 @@ def test_container_pull_and_termination_policies_parsed(fake_kube):
     c = params.containers[0]
     assert c.image_pull_policy == "Never"
     assert c.termination_message_policy == "FallbackToLogsOnError"
+
+
+def test_volume_secret_optional_and_missing(fake_kube):
+    from k8s_runpod_kubelet_amd.provider.translate import (ValidationError,
+                                                           extract_volumes)
+
+    pod = make_pod()
+    pod["spec"]["volumes"] = [{"name": "v", "secret": {
+        "secretName": "ghost"}}]
+    with pytest.raises(ValidationError, match="not found"):
+        extract_volumes(pod, fake_kube)
+    pod["spec"]["volumes"] = [{"name": "v", "secret": {
+        "secretName": "ghost", "optional": True}}]
+    vols = extract_volumes(pod, fake_kube)
+    assert vols["v"].kind == "files" and vols["v"].files == {}
+    # configMap analogue
+    pod["spec"]["volumes"] = [{"name": "v", "configMap": {"name": "nope"}}]
+    with pytest.raises(ValidationError, match="not found"):
+        extract_volumes(pod, fake_kube)
+
+
+def test_empty_dir_medium_and_size_parsed(fake_kube):
+    from k8s_runpod_kubelet_amd.provider.translate import extract_volumes
+
+    pod = make_pod()
+    pod["spec"]["volumes"] = [
+        {"name": "mem", "emptyDir": {"medium": "Memory",
+                                     "sizeLimit": "64Mi"}},
+        {"name": "disk", "emptyDir": {}},
+    ]
+    vols = extract_volumes(pod, fake_kube)
+    assert vols["mem"].medium == "Memory"
+    assert vols["mem"].size_limit_bytes == 64 << 20
+    assert vols["disk"].medium == ""
