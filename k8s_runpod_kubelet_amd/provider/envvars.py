@@ -3,7 +3,9 @@
 Counterpart of the reference's ExtractEnvVars pipeline
 (reference pkg/virtual_kubelet/runpod_client.go:866-1054):
 - literal container env,
-- ``valueFrom.secretKeyRef`` single keys,
+- ``valueFrom.secretKeyRef`` single keys (``optional:`` honored — a
+  non-optional missing secret/key blocks the pod like
+  CreateContainerConfigError; the reference silently tolerates),
 - ``envFrom.secretRef`` whole-secret imports,
 - secret *volumes* flattened into env vars (items→path/key names,
   runpod_client.go:949-979),
@@ -124,6 +126,20 @@ def _field_ref(pod: Dict[str, Any], path: str) -> str:
     }.get(path, "")
 
 
+def _missing_ref_error(kind: str, name: str, key: str = ""):
+    from .translate import ValidationError
+
+    what = f"{kind} {name!r}" + (f" key {key!r}" if key else "")
+    return ValidationError(f"env references missing {what}")
+
+
+def _require(collector, ref: Dict[str, Any], kind: str, name: str) -> None:
+    """envFrom k8s semantics: the referenced object must exist unless
+    optional: true."""
+    if not collector.exists(name) and not bool(ref.get("optional")):
+        raise _missing_ref_error(kind, name)
+
+
 def _container_env(container: Dict[str, Any], collector: SecretCollector,
                    cm_collector: ConfigMapCollector,
                    pod: Dict[str, Any]) -> Dict[str, str]:
@@ -133,11 +149,15 @@ def _container_env(container: Dict[str, Any], collector: SecretCollector,
     for ef in container.get("envFrom", []) or []:
         ref = ef.get("secretRef")
         if ref and ref.get("name"):
-            for key, value in collector.get(ref["name"]).items():
+            data = collector.get(ref["name"])
+            _require(collector, ref, "secret", ref["name"])
+            for key, value in data.items():
                 env[key] = _escape(value)
         cref = ef.get("configMapRef")
         if cref and cref.get("name"):
-            for key, value in cm_collector.get(cref["name"]).items():
+            data = cm_collector.get(cref["name"])
+            _require(cm_collector, cref, "configMap", cref["name"])
+            for key, value in data.items():
                 env[key] = _escape(value)
     # explicit env entries (literal + secretKeyRef + configMapKeyRef +
     # fieldRef), later wins
@@ -155,6 +175,11 @@ def _container_env(container: Dict[str, Any], collector: SecretCollector,
             key = ref.get("key", "")
             if key in data:
                 env[name] = _escape(data[key])
+            elif not bool(ref.get("optional")):
+                # k8s: a non-optional missing secret/key blocks the pod
+                # (CreateContainerConfigError); Pending-with-retry is the
+                # local analogue
+                raise _missing_ref_error("secret", ref["name"], key)
             continue
         cref = vf.get("configMapKeyRef")
         if cref and cref.get("name"):
@@ -162,6 +187,8 @@ def _container_env(container: Dict[str, Any], collector: SecretCollector,
             key = cref.get("key", "")
             if key in data:
                 env[name] = _escape(data[key])
+            elif not bool(cref.get("optional")):
+                raise _missing_ref_error("configMap", cref["name"], key)
             continue
         fref = vf.get("fieldRef")
         if fref and fref.get("fieldPath"):

@@ -227,11 +227,33 @@ def test_env_all_containers_parity_plus(fake_kube):
     assert per[1]["X"] == "from-b"
 
 
-def test_missing_secret_tolerated(fake_kube):
+def test_missing_secret_blocks_unless_optional(fake_kube):
+    """k8s envFrom semantics: a non-optional missing secret blocks the pod
+    (CreateContainerConfigError analogue: ValidationError -> Pending,
+    retried); optional: true degrades to empty."""
     pod = make_pod(containers=[{"name": "a",
                                 "envFrom": [{"secretRef": {"name": "nope"}}]}])
+    with pytest.raises(ValidationError, match="missing secret"):
+        extract_env_vars(pod, fake_kube)
+    pod = make_pod(containers=[{
+        "name": "a",
+        "envFrom": [{"secretRef": {"name": "nope", "optional": True}}]}])
     env, _ = extract_env_vars(pod, fake_kube)
     assert env == {}
+    # single-key refs: same rule
+    pod = make_pod(containers=[{
+        "name": "a",
+        "env": [{"name": "X", "valueFrom": {
+            "secretKeyRef": {"name": "nope", "key": "k"}}}]}])
+    with pytest.raises(ValidationError, match="missing secret"):
+        extract_env_vars(pod, fake_kube)
+    pod = make_pod(containers=[{
+        "name": "a",
+        "env": [{"name": "X", "valueFrom": {
+            "secretKeyRef": {"name": "nope", "key": "k",
+                             "optional": True}}}]}])
+    env, _ = extract_env_vars(pod, fake_kube)
+    assert "X" not in env
 
 
 def test_auto_injected_patterns():
@@ -306,13 +328,19 @@ def test_configmap_and_fieldref_env(fake_kube, pod_factory):
     assert pod_env["ONE_KEY"] == "prod"
     assert pod_env["MY_POD"] == "cmenv"
     assert pod_env["MY_NODE"] == "virtual-runpod"
-    # missing configmap degrades to empty, not a crash
+    # missing configmap: optional degrades to empty, non-optional blocks
     pod2 = pod_factory("cm2", containers=[{
         "name": "main", "image": "x",
-        "envFrom": [{"configMapRef": {"name": "nope"}}],
+        "envFrom": [{"configMapRef": {"name": "nope", "optional": True}}],
     }])
     pod_env2, _ = extract_env_vars(pod2, fake_kube)
     assert "MODE" not in pod_env2
+    pod3 = pod_factory("cm3", containers=[{
+        "name": "main", "image": "x",
+        "envFrom": [{"configMapRef": {"name": "nope"}}],
+    }])
+    with pytest.raises(ValidationError, match="missing configMap"):
+        extract_env_vars(pod3, fake_kube)
 
 
 def test_projected_volume_merges_sources(fake_kube):
