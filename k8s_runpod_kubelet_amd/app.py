@@ -60,7 +60,10 @@ class Stack:
             )
             self.health.start()
             self.api_server = KubeletApiServer(
-                self.provider, self.config.internal_ip, self.config.listen_port
+                self.provider, self.config.internal_ip,
+                self.config.listen_port,
+                token=self.config.kubelet_api_token
+                or os.environ.get("AMDVK_KUBELET_TOKEN", ""),
             )
             self.api_server.start()
         # Startup reconciliation (reference main.go:426 → LoadRunning).

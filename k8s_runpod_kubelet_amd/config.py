@@ -69,6 +69,11 @@ class Config:
     # when called from non-loopback peers; empty = loopback-only. Env
     # AMDVK_ADMIN_TOKEN overrides when this is unset.
     admin_token: str = ""
+    # Bearer token for the kubelet API (:10250 — logs/exec/stats) from
+    # non-loopback peers; empty = loopback-only (the real kubelet
+    # authenticates this surface via webhook/x509). Env
+    # AMDVK_KUBELET_TOKEN overrides when unset.
+    kubelet_api_token: str = ""
     log_level: str = "info"
     heartbeat_interval_s: float = 300.0  # 0 disables (reference kubelet.go:73)
     registration_endpoint: str = ""  # optional registration hook, off by default

@@ -23,10 +23,16 @@ log = logging.getLogger("kube.apiserver")
 
 
 class KubeletApiServer:
-    def __init__(self, provider, internal_ip: str = "127.0.0.1", port: int = 10250):
+    def __init__(self, provider, internal_ip: str = "127.0.0.1",
+                 port: int = 10250, token: str = ""):
+        """token: bearer token required from non-loopback peers (the real
+        kubelet authenticates :10250 via webhook/x509 — a static token +
+        loopback allowance is the offline equivalent; empty token =
+        loopback-only for every request)."""
         self.provider = provider
         self.internal_ip = internal_ip
         self.port = port
+        self.token = token
         self._server: Optional[ThreadingHTTPServer] = None
         self._thread: Optional[threading.Thread] = None
 
@@ -106,8 +112,21 @@ class KubeletApiServer:
                 except (BrokenPipeError, ConnectionResetError):
                     pass  # client hung up (Ctrl-C on kubectl logs -f)
 
+            def _authorized(self) -> bool:
+                peer = self.client_address[0] if self.client_address else ""
+                if peer in ("127.0.0.1", "::1", "::ffff:127.0.0.1"):
+                    return True
+                if outer.token:
+                    return (self.headers.get("Authorization", "")
+                            == f"Bearer {outer.token}")
+                return False
+
             def do_GET(self):
                 parsed = urlparse(self.path)
+                if parsed.path not in ("/healthz",) and not self._authorized():
+                    self._respond(401, b"unauthorized: kubelet API needs "
+                                       b"loopback or bearer token")
+                    return
                 parts = [unquote(p) for p in parsed.path.strip("/").split("/") if p]
                 query = parse_qs(parsed.query)
                 if parts and parts[0] == "containerLogs" and len(parts) >= 3:
@@ -141,6 +160,10 @@ class KubeletApiServer:
                     self._respond(404, b"not found")
 
             def do_POST(self):
+                if not self._authorized():
+                    self._respond(401, b"unauthorized: kubelet API needs "
+                                       b"loopback or bearer token")
+                    return
                 parsed = urlparse(self.path)
                 parts = [unquote(p) for p in parsed.path.strip("/").split("/") if p]
                 query = parse_qs(parsed.query)

@@ -308,3 +308,49 @@ def test_exec_targets_named_container(provider, process_runtime):
         pod_obj = prov.get_pod("default", "multi")
         if pod_obj:
             prov.delete_pod(pod_obj)
+
+
+def test_kubelet_api_requires_auth_from_non_loopback():
+    """:10250 parity: loopback is allowed, non-loopback peers need the
+    bearer token (the real kubelet authenticates this surface); /healthz
+    stays open for probes."""
+    import socket
+    import urllib.error
+    import urllib.request
+
+    ip = ""
+    try:
+        probe = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        probe.connect(("192.0.2.1", 9))  # no packets sent (UDP)
+        ip = probe.getsockname()[0]
+        probe.close()
+    except OSError:
+        pass
+    if not ip or ip.startswith("127."):
+        pytest.skip("no non-loopback interface in this environment")
+
+    class _Stub:
+        def get_pods(self):
+            return []
+
+    srv = KubeletApiServer(_Stub(), "0.0.0.0", 0, token="kubelet-secret")
+    srv.start()
+    try:
+        base = f"http://{ip}:{srv.port}"
+        # non-loopback without token: 401
+        with pytest.raises(urllib.error.HTTPError) as exc:
+            urllib.request.urlopen(f"{base}/pods", timeout=5)
+        assert exc.value.code == 401
+        # with the bearer token: allowed
+        req = urllib.request.Request(
+            f"{base}/pods",
+            headers={"Authorization": "Bearer kubelet-secret"})
+        assert urllib.request.urlopen(req, timeout=5).status == 200
+        # healthz open for probes
+        assert urllib.request.urlopen(
+            f"{base}/healthz", timeout=5).status == 200
+        # loopback needs no token
+        assert urllib.request.urlopen(
+            f"http://127.0.0.1:{srv.port}/pods", timeout=5).status == 200
+    finally:
+        srv.stop()
