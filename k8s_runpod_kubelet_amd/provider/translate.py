@@ -317,6 +317,32 @@ def _float_or(value, default: float) -> float:
         return default
 
 
+def _render_dns_config(pod: Dict[str, Any]) -> str:
+    """spec.dnsConfig -> resolv.conf text ("" = inherit the node's file,
+    dnsPolicy Default semantics — there is no cluster DNS on a
+    single-node deployment, so ClusterFirst falls back to the node too).
+    nameservers/searches/options follow the kubelet's rendering."""
+    dc = pod.get("spec", {}).get("dnsConfig") or {}
+    if not dc:
+        return ""
+    lines = []
+    for ns in dc.get("nameservers", []) or []:
+        lines.append(f"nameserver {ns}")
+    searches = [s for s in (dc.get("searches", []) or []) if s]
+    if searches:
+        lines.append("search " + " ".join(searches))
+    opts = []
+    for o in dc.get("options", []) or []:
+        name = o.get("name", "")
+        if not name:
+            continue
+        val = o.get("value")
+        opts.append(f"{name}:{val}" if val not in (None, "") else name)
+    if opts:
+        lines.append("options " + " ".join(opts))
+    return ("\n".join(lines) + "\n") if lines else ""
+
+
 def _cgroup_limits(pod: Dict[str, Any]) -> (str, str):
     """Aggregate container CPU/memory limits into cgroup v2 strings; a
     malformed quantity is a ValidationError (the apiserver would normally
@@ -463,6 +489,7 @@ def prepare_deploy_params(
         )
 
     cpu_max, memory_max = _cgroup_limits(pod)
+    resolv = _render_dns_config(pod)
 
     return DeployParams(
         pod_key=pod_key(pod),
@@ -480,6 +507,7 @@ def prepare_deploy_params(
         template_id=template_id,
         registry_auth_id=registry_auth,
         cpu_limit=cpu_max,
+        resolv_conf=resolv,
         memory_limit=memory_max,
         volumes=extract_volumes(pod, client),
         fs_group=_int_or_error(pod_sc.get("fsGroup"), "fsGroup")

@@ -423,7 +423,8 @@ class ProcessRuntime(Runtime):
                 working_dir=working_dir,
                 volume_binds=self._volume_binds(inst, cspec),
                 host_aliases=params.host_aliases,
-                read_only=cspec.read_only_root_fs)
+                read_only=cspec.read_only_root_fs,
+                resolv_conf=params.resolv_conf)
             argv[0] = self._rootfs_mgr.resolve_argv0(image, prepared,
                                                      argv[0])
             rootfs = prepared.rootfs
@@ -503,7 +504,8 @@ class ProcessRuntime(Runtime):
                 gpu_device_paths=self._gpu_device_paths(inst),
                 working_dir=working_dir,
                 volume_binds=self._volume_binds(inst, cspec),
-                host_aliases=params.host_aliases)
+                host_aliases=params.host_aliases,
+                resolv_conf=params.resolv_conf)
             argv[0] = self._rootfs_mgr.resolve_argv0(
                 image, prepared, (cspec.command or image.config.entrypoint
                                   or argv)[0])
@@ -1612,6 +1614,7 @@ class ProcessRuntime(Runtime):
             # remainder in adopt_persisted) and its grace window
             "active_deadline_s": inst.params.active_deadline_s,
             "termination_grace_s": inst.params.termination_grace_s,
+            "resolv_conf": inst.params.resolv_conf,
             "deadline_exceeded": inst.deadline_exceeded,
             "image_mode": inst.image_mode,
             "containers": [
@@ -1729,6 +1732,7 @@ class ProcessRuntime(Runtime):
                 gpu_memory_bytes=rec.get("gpu_memory_bytes", 0),
                 restart_policy=rec.get("restart_policy", "Never"),
                 active_deadline_s=rec.get("active_deadline_s", 0.0),
+                resolv_conf=rec.get("resolv_conf", ""),
                 termination_grace_s=rec.get("termination_grace_s",
                                             TERM_GRACE_S),
                 containers=[

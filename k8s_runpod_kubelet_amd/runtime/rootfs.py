@@ -103,7 +103,8 @@ class RootfsManager:
                 working_dir: str = "",
                 volume_binds: Optional[List[tuple]] = None,
                 host_aliases: Optional[List[tuple]] = None,
-                read_only: bool = False) -> PreparedRootfs:
+                read_only: bool = False,
+                resolv_conf: str = "") -> PreparedRootfs:
         """volume_binds: (host_src, container_dst, ro) tuples from the
         pod's volumes (emptyDir/hostPath/secret/configMap projections,
         materialized by the runtime). read_only = securityContext.
@@ -116,13 +117,13 @@ class RootfsManager:
             return self._prepare_overlay(cdir, image, hostname,
                                          gpu_device_paths or [], working_dir,
                                          volume_binds or [], aliases,
-                                         read_only)
+                                         read_only, resolv_conf)
         if read_only:
             log.warning("readOnlyRootFilesystem not enforceable in chroot "
                         "fallback mode; continuing writable")
         return self._prepare_chroot(cdir, image, hostname,
                                     gpu_device_paths or [], working_dir,
-                                    volume_binds or [], aliases)
+                                    volume_binds or [], aliases, resolv_conf)
 
     def _parse_bind(self, entry: str) -> Tuple[str, str, bool]:
         """'src[:dst[:ro|rw]]' -> (src, dst, ro). Default dst=src, ro."""
@@ -136,7 +137,8 @@ class RootfsManager:
                          hostname: str, gpu_devices: List[str],
                          working_dir: str, volume_binds: List[tuple],
                          host_aliases: List[tuple],
-                         read_only: bool = False) -> PreparedRootfs:
+                         read_only: bool = False,
+                         resolv_conf: str = "") -> PreparedRootfs:
         lower = self.store.rootfs_for(image)
         upper = cdir / "upper"
         work = cdir / "work"
@@ -153,7 +155,7 @@ class RootfsManager:
         if working_dir:
             (upper / _container_rel(working_dir)).mkdir(parents=True,
                                                         exist_ok=True)
-        self._write_identity(upper, hostname, host_aliases)
+        self._write_identity(upper, hostname, host_aliases, resolv_conf)
 
         mounts = [mnt.overlay(str(merged), str(lower), str(upper),
                               str(work), ro=read_only),
@@ -208,7 +210,8 @@ class RootfsManager:
     def _prepare_chroot(self, cdir: Path, image: ResolvedImage,
                         hostname: str, gpu_devices: List[str],
                         working_dir: str, volume_binds: List[tuple],
-                        host_aliases: List[tuple]) -> PreparedRootfs:
+                        host_aliases: List[tuple],
+                        resolv_conf: str = "") -> PreparedRootfs:
         rootfs = cdir / "rootfs"
         if not (rootfs / ".amdvk-ready").exists():
             cdir.mkdir(parents=True, exist_ok=True)
@@ -227,7 +230,8 @@ class RootfsManager:
             if working_dir:
                 (rootfs / _container_rel(working_dir)).mkdir(parents=True,
                                                              exist_ok=True)
-            self._write_identity(rootfs, hostname, host_aliases)
+            self._write_identity(rootfs, hostname, host_aliases,
+                                 resolv_conf)
             # chroot mode has no mounts: volumes are materialized INTO the
             # per-container copy (per-container — cross-container emptyDir
             # sharing needs the mount-namespace mode; hostPath cannot be
@@ -265,18 +269,27 @@ class RootfsManager:
             log.debug("mknod unavailable", extra={"dev": host_dev})
 
     def _write_identity(self, root: Path, hostname: str,
-                        host_aliases: List[tuple]) -> None:
+                        host_aliases: List[tuple],
+                        resolv_conf: str = "") -> None:
         """kubelet-managed identity files (k8s writes these per pod),
-        including spec.hostAliases lines."""
+        including spec.hostAliases lines and resolv.conf (dnsPolicy
+        Default = the node's resolver; spec.dnsConfig overrides)."""
         etc = root / "etc"
         etc.mkdir(parents=True, exist_ok=True)
         lines = ["127.0.0.1\tlocalhost",
                  f"127.0.1.1\t{hostname or 'pod'}"]
         for ip, names in host_aliases:
             lines.append(f"{ip}\t" + " ".join(names))
+        if not resolv_conf:
+            try:
+                resolv_conf = Path("/etc/resolv.conf").read_text()
+            except OSError:
+                resolv_conf = ""
         try:
             (etc / "hostname").write_text((hostname or "pod") + "\n")
             (etc / "hosts").write_text("\n".join(lines) + "\n")
+            if resolv_conf:
+                (etc / "resolv.conf").write_text(resolv_conf)
         except OSError:
             pass
 

@@ -191,6 +191,9 @@ class DeployParams:
     name: str
     namespace: str = "default"
     containers: List[ContainerSpec] = field(default_factory=list)
+    # spec.dnsConfig rendered to a resolv.conf ("" = node default; image
+    # pods get it written into /etc/resolv.conf like the kubelet)
+    resolv_conf: str = ""
     # spec.initContainers: run sequentially to completion before the main
     # containers start; any nonzero exit fails the pod
     init_containers: List[ContainerSpec] = field(default_factory=list)

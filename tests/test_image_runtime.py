@@ -1400,3 +1400,38 @@ def test_empty_dir_memory_is_tmpfs_with_size_limit(image_runtime):
     assert "mount=tmpfs" in out
     assert "small=ok" in out
     assert "cap=enforced" in out
+
+
+def test_resolv_conf_written_into_rootfs(image_runtime):
+    """Image pods get /etc/resolv.conf: the node's file by default
+    (dnsPolicy Default), spec.dnsConfig rendering when present."""
+    rt = image_runtime
+    st = rt.deploy(DeployParams(
+        pod_key="default-dns", name="dns",
+        resolv_conf="nameserver 10.9.9.9\nsearch custom.local\n",
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c",
+                     "while read -r l; do echo RC:$l; done "
+                     "< /etc/resolv.conf"])],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    out = rt.get_logs(st.id)
+    assert s.exit_code == 0, out
+    assert "RC:nameserver 10.9.9.9" in out
+    assert "RC:search custom.local" in out
+
+    # default: node resolv.conf copied in (if the node has one)
+    import os as _os
+
+    if _os.path.exists("/etc/resolv.conf") and \
+            open("/etc/resolv.conf").read().strip():
+        st2 = rt.deploy(DeployParams(
+            pod_key="default-dns2", name="dns2",
+            containers=[ContainerSpec(
+                name="main", image="example/app:v1",
+                command=["/bin/sh", "-c",
+                         "test -s /etc/resolv.conf && echo rc=present"])],
+        ))
+        wait_status(rt, st2.id, PodStatus.EXITED)
+        assert "rc=present" in rt.get_logs(st2.id)

@@ -421,3 +421,20 @@ def test_empty_dir_medium_and_size_parsed(fake_kube):
     assert vols["mem"].medium == "Memory"
     assert vols["mem"].size_limit_bytes == 64 << 20
     assert vols["disk"].medium == ""
+
+
+def test_dns_config_rendered(fake_kube):
+    pod = make_pod()
+    pod["spec"]["dnsConfig"] = {
+        "nameservers": ["10.0.0.10", "10.0.0.11"],
+        "searches": ["ns.svc.cluster.local", "example.com"],
+        "options": [{"name": "ndots", "value": "5"}, {"name": "edns0"}],
+    }
+    params = prepare_deploy_params(pod, fake_kube, Config())
+    assert params.resolv_conf == (
+        "nameserver 10.0.0.10\nnameserver 10.0.0.11\n"
+        "search ns.svc.cluster.local example.com\n"
+        "options ndots:5 edns0\n")
+    # no dnsConfig -> node default (empty marker)
+    assert prepare_deploy_params(make_pod(), fake_kube,
+                                 Config()).resolv_conf == ""
