@@ -307,6 +307,19 @@ class Provider:
             self._pod_status.pop(key, None)
         metrics.pods_deleted.inc()
 
+    def _memory_pressure(self) -> bool:
+        """node.status MemoryPressure condition: the eviction signal is
+        below threshold right now (kubelet sets the condition from the
+        same memory.available signal that drives eviction)."""
+        if self.config.eviction_memory_threshold_mb <= 0:
+            return False
+        from .eviction import read_available_memory_bytes
+
+        reader = self.meminfo_reader or read_available_memory_bytes
+        avail = reader()
+        return 0 <= avail < (
+            self.config.eviction_memory_threshold_mb * 1024 * 1024)
+
     def eviction_candidates(self):
         """(pod_key, pod_obj, memory_usage_bytes) for every live local pod
         (eviction.py ranks them). Pending pods are skipped — they hold no
@@ -1100,8 +1113,11 @@ class Provider:
              "lastTransitionTime": now_rfc3339()},
             {"type": "OutOfDisk", "status": "False", "reason": "KubeletHasSufficientDisk",
              "lastHeartbeatTime": now_rfc3339(), "lastTransitionTime": now_rfc3339()},
-            {"type": "MemoryPressure", "status": "False",
-             "reason": "KubeletHasSufficientMemory",
+            {"type": "MemoryPressure",
+             "status": "True" if self._memory_pressure() else "False",
+             "reason": ("KubeletHasInsufficientMemory"
+                        if self._memory_pressure()
+                        else "KubeletHasSufficientMemory"),
              "lastHeartbeatTime": now_rfc3339(), "lastTransitionTime": now_rfc3339()},
             {"type": "DiskPressure", "status": "False", "reason": "KubeletHasNoDiskPressure",
              "lastHeartbeatTime": now_rfc3339(), "lastTransitionTime": now_rfc3339()},

@@ -248,3 +248,29 @@ def test_eviction_kills_real_process_and_frees_gpu(tmp_state_dir):
                      {}).get("phase") == "Running", timeout_s=10)
     finally:
         stack.stop()
+
+
+def test_memory_pressure_node_condition(tmp_state_dir):
+    """node.status MemoryPressure flips True while the eviction signal is
+    below threshold (same signal the kubelet uses)."""
+    from k8s_runpod_kubelet_amd.app import build_stack
+    from k8s_runpod_kubelet_amd.config import Config
+    from k8s_runpod_kubelet_amd.kube.fake import FakeKube
+
+    cfg = Config(state_dir=tmp_state_dir, gpu_count_override=2,
+                 runtime="fake", eviction_memory_threshold_mb=100)
+    stack = build_stack(cfg, client=FakeKube())
+    pressure = [1 << 40]
+    stack.provider.meminfo_reader = lambda: pressure[0]
+
+    def cond(node, typ):
+        return next(c for c in node["status"]["conditions"]
+                    if c["type"] == typ)
+
+    node = stack.provider.get_node_status()
+    assert cond(node, "MemoryPressure")["status"] == "False"
+    pressure[0] = 1 << 20
+    node = stack.provider.get_node_status()
+    assert cond(node, "MemoryPressure")["status"] == "True"
+    assert cond(node, "MemoryPressure")["reason"] == \
+        "KubeletHasInsufficientMemory"
