@@ -17,6 +17,7 @@ import argparse
 import json
 import os
 import random
+import signal
 import sys
 import tempfile
 import time
@@ -35,11 +36,38 @@ def main() -> int:
     ap.add_argument("--restart-kubelet-at", type=float, default=0.0,
                     help="simulate a kubelet crash+restart T seconds in "
                          "(pods untouched; adoption must resume everything)")
+    ap.add_argument("--reap-orphans", action="store_true",
+                    help="at startup, SIGKILL podworker processes leaked "
+                         "by previous soaks that were killed outright "
+                         "(exact exe match only)")
     ap.add_argument("--images", action="store_true",
                     help="mix OCI image pods (overlay+pivot_root) into the "
                          "churn (needs mount-ns capability and gcc)")
     args = ap.parse_args()
     rng = random.Random(args.seed)
+
+    if args.reap_orphans:
+        from k8s_runpod_kubelet_amd.ops import podworker_binary
+
+    target = None
+    if args.reap_orphans:
+        target = os.path.realpath(podworker_binary())
+        reaped = 0
+        for pid in os.listdir("/proc"):
+            if not pid.isdigit() or int(pid) == os.getpid():
+                continue
+            try:
+                exe = os.readlink(f"/proc/{pid}/exe")
+            except OSError:
+                continue
+            if exe in (target, target + " (deleted)"):
+                try:
+                    os.kill(int(pid), signal.SIGKILL)
+                    reaped += 1
+                except OSError:
+                    pass
+        if reaped:
+            print(f"reaped {reaped} leaked podworker(s)", file=sys.stderr)
 
     from k8s_runpod_kubelet_amd.app import build_stack
     from k8s_runpod_kubelet_amd.config import Config
@@ -48,6 +76,14 @@ def main() -> int:
     from k8s_runpod_kubelet_amd.kube.real import ClusterConfig, HttpK8sClient
     from k8s_runpod_kubelet_amd.provider import annotations as ann
     from k8s_runpod_kubelet_amd.server import metrics as m
+
+    # an outer `timeout` kills with SIGTERM: a flag (not an exception —
+    # raising from a handler can interrupt lock-held teardown and
+    # deadlock) makes the main loop fall through to the finally-block
+    # reap instead of orphaning pod processes
+    stop_requested = []
+    signal.signal(signal.SIGTERM,
+                  lambda *_: stop_requested.append(True))
 
     srv = FakeApiServer().start()
     client = HttpK8sClient(ClusterConfig(server=srv.url))
@@ -209,7 +245,7 @@ def main() -> int:
     deadline = start_ts + args.duration
     kubelet_restarts = 0
     try:
-        while time.time() < deadline or active:
+        while (time.time() < deadline or active) and not stop_requested:
             now = time.time()
             if (args.restart_kubelet_at > 0 and kubelet_restarts == 0
                     and now - start_ts >= args.restart_kubelet_at):
@@ -347,6 +383,18 @@ def main() -> int:
         print(json.dumps(summary))
         return 0 if ok else 1
     finally:
+        # Reap every pod process before exiting: pods are DESIGNED to
+        # outlive a kubelet (adoption), but a finished soak must not
+        # leave its workloads running on the host (leaked holds from
+        # timeout-killed soaks once stole a fixed test port).
+        try:
+            for inst in stack.runtime.list_instances():
+                try:
+                    stack.runtime.terminate(inst.id, grace_override_s=0.0)
+                except Exception:
+                    pass
+        except Exception:
+            pass
         stack.stop()
         client.close()
         srv.stop()
