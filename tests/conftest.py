@@ -127,3 +127,15 @@ def wait_until(fn, timeout_s=5.0, interval_s=0.01):
 @pytest.fixture
 def waiter():
     return wait_until
+
+
+def free_port() -> int:
+    """OS-assigned free TCP port (bound then released — a fixed port in a
+    test can collide with leftover listeners or parallel workers)."""
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port

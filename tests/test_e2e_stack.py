@@ -107,9 +107,12 @@ def test_failing_pod_reports_failed(stack):
 
 
 def test_port_gated_readiness(stack):
+    from tests.conftest import free_port
+
     s, kube = stack
-    pod = make_pod("srv", ports=[39131], command=["podworker"],
-                   args=["--listen-port", "39131", "--hold"])
+    port = free_port()
+    pod = make_pod("srv", ports=[port], command=["podworker"],
+                   args=["--listen-port", str(port), "--hold"])
     kube.create_pod("default", pod)
     got = wait_until(lambda: ready(kube, "srv"), timeout_s=10)
     assert got is not None  # listening socket detected via /proc/net/tcp

@@ -114,7 +114,9 @@ def test_readiness_probe_gates_ready(stack):
     """A tcpSocket readinessProbe owns Ready: the pod only becomes Ready
     once the port actually accepts, regardless of the process being up."""
     s, kube = stack
-    port = 19123
+    from tests.conftest import free_port
+
+    port = free_port()
     pod = make_pod("probed", command=["podworker"],
                    args=["--startup-delay", "1.5",
                          "--listen-port", str(port), "--hold"])
