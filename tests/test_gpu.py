@@ -232,10 +232,13 @@ def test_gpu_port_readiness_and_exec(tmp_path):
     stack = build_stack(cfg, client=kube)
     stack.start(serve_http=False)
     try:
+        from tests.conftest import free_port
+
+        port = free_port()
         pod = make_pod("gpuport", gpus=1, command=["podworker"],
                        args=["--expect-gpus", "1",
-                             "--listen-port", "18081", "--hold"],
-                       ports=[18081])
+                             "--listen-port", str(port), "--hold"],
+                       ports=[port])
         kube.create_pod("default", pod)
 
         def ready():
@@ -250,7 +253,7 @@ def test_gpu_port_readiness_and_exec(tmp_path):
         assert wait_until(ready, timeout_s=90) is not None
         info = stack.provider.instance_info("default", "gpuport")
         detailed = stack.runtime.get_detailed_status(info.instance_id)
-        assert detailed.port_mappings.get(18081) == 18081
+        assert detailed.port_mappings.get(port) == port
 
         code, out = stack.provider.run_in_container(
             "default", "gpuport", ["/usr/bin/env"])
