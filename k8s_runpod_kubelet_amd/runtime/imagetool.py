@@ -49,6 +49,10 @@ def main(argv=None) -> int:
     p_pull.add_argument("--token", default="")
     p_pull.add_argument("--insecure", action="store_true")
 
+    p_i = sub.add_parser("inspect", help="print an image's config and "
+                         "layer digests as JSON")
+    p_i.add_argument("ref")
+
     p_push = sub.add_parser("push", help="push a stored image to a "
                             "registry (server needs --allow-push)")
     p_push.add_argument("ref")
@@ -103,6 +107,29 @@ def main(argv=None) -> int:
         finally:
             client.close()
         print(f"pulled {ref}")
+        return 0
+    if args.cmd == "inspect":
+        import json as _json
+
+        img = store.resolve(args.ref)
+        if img is None:
+            print(f"not found: {args.ref}", file=sys.stderr)
+            return 1
+        manifest = _json.loads(
+            (img.layout_dir / "blobs" /
+             img.manifest_digest.replace(":", "/")).read_bytes())
+        print(_json.dumps({
+            "ref": img.ref,
+            "digest": img.manifest_digest,
+            "entrypoint": img.config.entrypoint,
+            "cmd": img.config.cmd,
+            "env": img.config.env,
+            "user": img.config.user,
+            "workingDir": img.config.working_dir,
+            "layers": [{"digest": l.get("digest", ""),
+                        "size": l.get("size", 0)}
+                       for l in manifest.get("layers", [])],
+        }, indent=2))
         return 0
     if args.cmd == "push":
         from .registry import RegistryClient

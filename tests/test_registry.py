@@ -483,3 +483,19 @@ def test_large_blob_streams_roundtrip(tmp_path):
     finally:
         client.close()
         hub.stop()
+
+
+def test_imagetool_inspect_cli(served_store, capsys):
+    import json as _json
+
+    from k8s_runpod_kubelet_amd.runtime.imagetool import main as itool
+
+    rc = itool(["--store", str(served_store.root), "inspect",
+                "example/pullme:v1"])
+    assert rc == 0
+    out = _json.loads(capsys.readouterr().out)
+    assert out["entrypoint"] == ["/bin/tool"]
+    assert out["digest"].startswith("sha256:")
+    assert out["layers"] and out["layers"][0]["digest"].startswith("sha256:")
+    assert itool(["--store", str(served_store.root), "inspect",
+                  "nope:v9"]) == 1
