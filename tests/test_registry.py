@@ -499,3 +499,30 @@ def test_imagetool_inspect_cli(served_store, capsys):
     assert out["layers"] and out["layers"][0]["digest"].startswith("sha256:")
     assert itool(["--store", str(served_store.root), "inspect",
                   "nope:v9"]) == 1
+
+
+def test_server_survives_hostile_paths(served_store):
+    """Robustness: arbitrary request paths never crash the server or leak
+    files — everything unexpected is a clean 4xx."""
+    import httpx
+
+    srv = RegistryServer(served_store).start()
+    try:
+        hostile = [
+            "/", "/v2/../../etc/passwd", "/v2//blobs/sha256:zz",
+            "/v2/a/manifests/", "/v2/a/blobs/sha256:" + "g" * 64,
+            "/v2/a/blobs/uploads/../../x", "/v2/%2e%2e/manifests/v1",
+            "/v2/a/manifests/" + "A" * 5000,
+        ]
+        for path in hostile:
+            r = httpx.get(srv.url + path)
+            assert 400 <= r.status_code < 500, (path, r.status_code)
+        # push endpoints refused cleanly when push disabled
+        assert httpx.post(
+            srv.url + "/v2/a/blobs/uploads/").status_code == 405
+        assert httpx.put(
+            srv.url + "/v2/a/manifests/v1", content=b"{}").status_code == 405
+        # server still serves normally afterwards
+        assert httpx.get(srv.url + "/v2/").status_code == 200
+    finally:
+        srv.stop()
