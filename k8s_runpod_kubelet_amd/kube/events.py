@@ -15,18 +15,48 @@ log = logging.getLogger("kube.events")
 
 
 class EventRecorder:
+    """client-go-style correlation: a repeat of the same
+    (object, type, reason, message) within the aggregation window bumps
+    the existing Event's count/lastTimestamp instead of creating a new
+    object — a CrashLoopBackOff pod otherwise floods the apiserver with
+    one BackOff Event per restart attempt."""
+
+    AGGREGATION_WINDOW_S = 600.0
+    _CACHE_MAX = 4096
+
     def __init__(self, client: K8sClient, component: str = "amd-virtual-kubelet"):
         self.client = client
         self.component = component
+        # (ns, name, uid, type, reason, message) -> [event_name, count,
+        # first_ts, last_seen_monotonic]
+        self._seen: Dict[tuple, list] = {}
 
     def event(self, obj: Dict[str, Any], event_type: str, reason: str,
               message: str) -> None:
         namespace = namespace_of(obj)
+        key = (namespace, name_of(obj), uid_of(obj), event_type, reason,
+               message)
+        now_mono = time.monotonic()
+        hit = self._seen.get(key)
+        if hit is not None and now_mono - hit[3] < self.AGGREGATION_WINDOW_S:
+            hit[1] += 1
+            hit[3] = now_mono
+            ev_name, count, first_ts = hit[0], hit[1], hit[2]
+        else:
+            ev_name = f"{name_of(obj)}.{int(time.time() * 1e6):x}"
+            first_ts = now_rfc3339()
+            count = 1
+            if len(self._seen) >= self._CACHE_MAX:
+                # drop the stalest half; bounded memory under pod churn
+                by_age = sorted(self._seen.items(), key=lambda kv: kv[1][3])
+                for k, _ in by_age[: self._CACHE_MAX // 2]:
+                    self._seen.pop(k, None)
+            self._seen[key] = [ev_name, 1, first_ts, now_mono]
         event = {
             "apiVersion": "v1",
             "kind": "Event",
             "metadata": {
-                "name": f"{name_of(obj)}.{int(time.time() * 1e6):x}",
+                "name": ev_name,
                 "namespace": namespace,
             },
             "involvedObject": {
@@ -39,11 +69,16 @@ class EventRecorder:
             "message": message,
             "type": event_type,
             "source": {"component": self.component},
-            "firstTimestamp": now_rfc3339(),
+            "firstTimestamp": first_ts,
             "lastTimestamp": now_rfc3339(),
-            "count": 1,
+            "count": count,
         }
         try:
+            if count > 1:
+                update = getattr(self.client, "update_event", None)
+                if callable(update):
+                    update(namespace, event)
+                    return
             self.client.create_event(namespace, event)
         except Exception as exc:
             log.debug("event create failed", extra={"err": str(exc)})

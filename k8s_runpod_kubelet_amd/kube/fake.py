@@ -403,6 +403,9 @@ class FakeKube(K8sClient):
 
     # ---- events / auth ----
 
+    def update_event(self, namespace: str, event: Dict[str, Any]) -> Dict[str, Any]:
+        return self.create_event(namespace, event)  # name-keyed overwrite
+
     def create_event(self, namespace: str, event: Dict[str, Any]) -> Dict[str, Any]:
         with self._lock:
             event = copy.deepcopy(event)

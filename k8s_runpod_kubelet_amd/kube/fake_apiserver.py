@@ -43,6 +43,7 @@ _SECRET_RE = re.compile(r"^/api/v1/namespaces/([^/]+)/secrets/([^/]+)$")
 _CONFIGMAP_RE = re.compile(r"^/api/v1/namespaces/([^/]+)/configmaps/([^/]+)$")
 _JOB_RE = re.compile(r"^/apis/batch/v1/namespaces/([^/]+)/jobs/([^/]+)$")
 _EVENTS_NS_RE = re.compile(r"^/api/v1/namespaces/([^/]+)/events$")
+_EVENT_RE = re.compile(r"^/api/v1/namespaces/([^/]+)/events/([^/]+)$")
 
 
 class FakeApiServer:
@@ -238,6 +239,9 @@ class FakeApiServer:
                         return self._json(200, kube.update_node(body))
                     if m := _LEASE_RE.match(path):
                         return self._json(200, kube.update_lease(m.group(1), body))
+                    if m := _EVENT_RE.match(path):
+                        return self._json(200,
+                                          kube.update_event(m.group(1), body))
                     raise NotFoundError(f"no route {path}")
                 except Exception as exc:  # noqa: BLE001
                     self._err(exc)

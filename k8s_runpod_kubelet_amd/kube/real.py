@@ -359,6 +359,12 @@ class HttpK8sClient(K8sClient):
 
     # ---- events / auth ----
 
+    def update_event(self, namespace: str, event: Dict[str, Any]) -> Dict[str, Any]:
+        name = event.get("metadata", {}).get("name", "")
+        return self._request(
+            "PUT", f"/api/v1/namespaces/{namespace}/events/{name}",
+            json_body=event)
+
     def create_event(self, namespace: str, event: Dict[str, Any]) -> Dict[str, Any]:
         return self._request("POST", f"/api/v1/namespaces/{namespace}/events",
                              json_body=event)

@@ -235,3 +235,27 @@ def test_service_account_token_rotation(tmp_path):
     finally:
         client.close()
         srv.stop()
+
+
+def test_event_aggregation_over_the_wire(http_stack):
+    """Repeated identical events aggregate client-go-style: one Event
+    object whose count climbs (a crash-looping pod must not flood the
+    apiserver with one BackOff object per attempt)."""
+    from k8s_runpod_kubelet_amd.kube.events import EventRecorder
+
+    _stack, srv, client, _cfg = http_stack
+    rec = EventRecorder(client)
+    obj = {"kind": "Pod", "metadata": {"name": "looper",
+                                       "namespace": "default",
+                                       "uid": "u-1"}}
+    for _ in range(5):
+        rec.event(obj, "Warning", "BackOff", "restarting failed container")
+    rec.event(obj, "Normal", "Started", "instance i-1")  # distinct event
+
+    with srv.kube._lock:
+        evs = list(srv.kube.events.objects.values())
+    backoffs = [e for e in evs if e.get("reason") == "BackOff"]
+    assert len(backoffs) == 1
+    assert backoffs[0]["count"] == 5
+    assert backoffs[0]["firstTimestamp"] <= backoffs[0]["lastTimestamp"]
+    assert sum(1 for e in evs if e.get("reason") == "Started") == 1
