@@ -5,6 +5,7 @@ virtual-kubelet lib emits through it)."""
 from __future__ import annotations
 
 import logging
+import threading
 import time
 from typing import Any, Dict
 
@@ -30,6 +31,7 @@ class EventRecorder:
         # (ns, name, uid, type, reason, message) -> [event_name, count,
         # first_ts, last_seen_monotonic]
         self._seen: Dict[tuple, list] = {}
+        self._lock = threading.Lock()
 
     def event(self, obj: Dict[str, Any], event_type: str, reason: str,
               message: str) -> None:
@@ -37,21 +39,24 @@ class EventRecorder:
         key = (namespace, name_of(obj), uid_of(obj), event_type, reason,
                message)
         now_mono = time.monotonic()
-        hit = self._seen.get(key)
-        if hit is not None and now_mono - hit[3] < self.AGGREGATION_WINDOW_S:
-            hit[1] += 1
-            hit[3] = now_mono
-            ev_name, count, first_ts = hit[0], hit[1], hit[2]
-        else:
-            ev_name = f"{name_of(obj)}.{int(time.time() * 1e6):x}"
-            first_ts = now_rfc3339()
-            count = 1
-            if len(self._seen) >= self._CACHE_MAX:
-                # drop the stalest half; bounded memory under pod churn
-                by_age = sorted(self._seen.items(), key=lambda kv: kv[1][3])
-                for k, _ in by_age[: self._CACHE_MAX // 2]:
-                    self._seen.pop(k, None)
-            self._seen[key] = [ev_name, 1, first_ts, now_mono]
+        with self._lock:  # events fire from deploy/reconcile/event threads
+            hit = self._seen.get(key)
+            if hit is not None and \
+                    now_mono - hit[3] < self.AGGREGATION_WINDOW_S:
+                hit[1] += 1
+                hit[3] = now_mono
+                ev_name, count, first_ts = hit[0], hit[1], hit[2]
+            else:
+                ev_name = f"{name_of(obj)}.{int(time.time() * 1e6):x}"
+                first_ts = now_rfc3339()
+                count = 1
+                if len(self._seen) >= self._CACHE_MAX:
+                    # drop the stalest half; bounded memory under churn
+                    by_age = sorted(self._seen.items(),
+                                    key=lambda kv: kv[1][3])
+                    for k, _ in by_age[: self._CACHE_MAX // 2]:
+                        self._seen.pop(k, None)
+                self._seen[key] = [ev_name, 1, first_ts, now_mono]
         event = {
             "apiVersion": "v1",
             "kind": "Event",
