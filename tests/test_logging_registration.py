@@ -82,3 +82,25 @@ def test_registrar_failure_not_fatal(monkeypatch):
     r = Registrar("http://127.0.0.1:1", "n", "ns", 8, heartbeat_interval_s=0)
     assert r.register() is False
     r.stop_heartbeat()
+
+
+def test_event_recorder_window_expiry(fake_kube, monkeypatch):
+    """Aggregation window: repeats inside the window bump one Event;
+    after the window a fresh Event object starts at count 1."""
+    import k8s_runpod_kubelet_amd.kube.events as ev_mod
+
+    clock = [1000.0]
+    monkeypatch.setattr(ev_mod.time, "monotonic", lambda: clock[0])
+    rec = ev_mod.EventRecorder(fake_kube)
+    obj = {"kind": "Pod", "metadata": {"name": "w", "namespace": "default",
+                                       "uid": "u"}}
+    rec.event(obj, "Warning", "BackOff", "m")
+    clock[0] += 100
+    rec.event(obj, "Warning", "BackOff", "m")
+    clock[0] += rec.AGGREGATION_WINDOW_S + 1
+    rec.event(obj, "Warning", "BackOff", "m")
+    with fake_kube._lock:
+        evs = [e for e in fake_kube.events.objects.values()
+               if e.get("reason") == "BackOff"]
+    assert sorted(e["count"] for e in evs) == [1, 2]
+    assert len(evs) == 2
