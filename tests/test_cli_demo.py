@@ -17,6 +17,7 @@ def test_cli_fake_apiserver_demo():
     p = subprocess.Popen(
         [sys.executable, "-m", "k8s_runpod_kubelet_amd.cli",
          "--gpu-count-override", "8", "--fake-apiserver",
+         "--listen-port", "0",
          "--health-server-address", "127.0.0.1:0"],
         stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, cwd=REPO,
     )
