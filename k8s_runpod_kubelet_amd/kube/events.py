@@ -82,8 +82,24 @@ class EventRecorder:
             if count > 1:
                 update = getattr(self.client, "update_event", None)
                 if callable(update):
-                    update(namespace, event)
-                    return
+                    try:
+                        update(namespace, event)
+                        return
+                    except Exception:
+                        # apiserver TTL-GCs Events (~1 h): the aggregated
+                        # object may be gone — restart the series under a
+                        # fresh name rather than losing the signal (a
+                        # fresh name also survives non-404 update
+                        # failures without clobbering a live object)
+                        ev_name = (f"{name_of(obj)}."
+                                   f"{int(time.time() * 1e6):x}")
+                        first_ts = now_rfc3339()
+                        with self._lock:
+                            self._seen[key] = [ev_name, 1, first_ts,
+                                               now_mono]
+                        event["metadata"]["name"] = ev_name
+                        event["count"] = 1
+                        event["firstTimestamp"] = first_ts
             self.client.create_event(namespace, event)
         except Exception as exc:
             log.debug("event create failed", extra={"err": str(exc)})

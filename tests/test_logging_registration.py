@@ -104,3 +104,25 @@ def test_event_recorder_window_expiry(fake_kube, monkeypatch):
                if e.get("reason") == "BackOff"]
     assert sorted(e["count"] for e in evs) == [1, 2]
     assert len(evs) == 2
+
+
+def test_event_update_falls_back_to_create(fake_kube):
+    """If the aggregated Event was TTL-GC'd by the apiserver, the update
+    404s — the recorder recreates instead of dropping the signal."""
+    from k8s_runpod_kubelet_amd.kube.events import EventRecorder
+
+    rec = EventRecorder(fake_kube)
+    obj = {"kind": "Pod", "metadata": {"name": "g", "namespace": "default",
+                                       "uid": "u"}}
+    rec.event(obj, "Warning", "BackOff", "m")
+
+    def gone(_ns, _ev):
+        raise RuntimeError("404 not found")
+
+    fake_kube.update_event = gone
+    rec.event(obj, "Warning", "BackOff", "m")  # update fails -> recreate
+    with fake_kube._lock:
+        evs = [e for e in fake_kube.events.objects.values()
+               if e.get("reason") == "BackOff"]
+    assert len(evs) == 2  # old + recreated series
+    assert all(e["count"] == 1 for e in evs)
