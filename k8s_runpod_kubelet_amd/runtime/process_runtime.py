@@ -295,6 +295,23 @@ class ProcessRuntime(Runtime):
         finally:
             client.close()
 
+    def _binds_for(self, inst: Instance, cspec) -> List[tuple]:
+        """Volume binds plus kubelet-managed rw files: with
+        readOnlyRootFilesystem the kubelet still keeps
+        terminationMessagePath writable by bind-mounting a host file."""
+        binds = self._volume_binds(inst, cspec)
+        if getattr(cspec, "read_only_root_fs", False) and \
+                cspec.termination_message_path:
+            host = (self.state_dir / "volumes" / inst.id /
+                    f".tmsg-{cspec.name}")
+            host.parent.mkdir(parents=True, exist_ok=True)
+            if not host.exists():
+                host.touch()
+                os.chmod(host, 0o666)
+            binds.append((str(host), cspec.termination_message_path,
+                          False, "file"))
+        return binds
+
     def _volume_binds(self, inst: Instance, cspec) -> List[tuple]:
         """Materialize the pod's volumes and return (src, dst, ro) binds
         for this container. emptyDir/files volumes live under
@@ -421,7 +438,7 @@ class ProcessRuntime(Runtime):
                 params.hostname or params.name,
                 gpu_device_paths=self._gpu_device_paths(inst),
                 working_dir=working_dir,
-                volume_binds=self._volume_binds(inst, cspec),
+                volume_binds=self._binds_for(inst, cspec),
                 host_aliases=params.host_aliases,
                 read_only=cspec.read_only_root_fs,
                 resolv_conf=params.resolv_conf)
@@ -503,7 +520,7 @@ class ProcessRuntime(Runtime):
                 params.hostname or params.name,
                 gpu_device_paths=self._gpu_device_paths(inst),
                 working_dir=working_dir,
-                volume_binds=self._volume_binds(inst, cspec),
+                volume_binds=self._binds_for(inst, cspec),
                 host_aliases=params.host_aliases,
                 resolv_conf=params.resolv_conf)
             argv[0] = self._rootfs_mgr.resolve_argv0(
@@ -712,6 +729,16 @@ class ProcessRuntime(Runtime):
                       if c.name == container), None)
         if cspec is None:
             return ""
+        host = (self.state_dir / "volumes" / inst.id /
+                f".tmsg-{container}")
+        if host.exists():
+            try:
+                text = host.read_text(encoding="utf-8",
+                                      errors="replace")[:4096].strip()
+                if text:
+                    return text
+            except OSError:
+                pass
         rel = cspec.termination_message_path.lstrip("/")
         cdir = self._rootfs_mgr.containers_dir / f"{inst.id}-{container}"
         base = cdir / ("rootfs" if inst.image_mode == "chroot" else "upper")

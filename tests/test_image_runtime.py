@@ -1435,3 +1435,26 @@ def test_resolv_conf_written_into_rootfs(image_runtime):
         ))
         wait_status(rt, st2.id, PodStatus.EXITED)
         assert "rc=present" in rt.get_logs(st2.id)
+
+
+def test_termination_message_writable_with_ro_rootfs(image_runtime):
+    """readOnlyRootFilesystem + terminationMessagePath: the kubelet keeps
+    the message path writable (host file bind) even though the rootfs is
+    read-only."""
+    rt = image_runtime
+    if rt._rootfs_mgr.mode() != "mountns":
+        pytest.skip("ro rootfs needs mount-ns mode")
+    st = rt.deploy(DeployParams(
+        pod_key="default-rotm", name="rotm",
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c",
+                     "if echo x > /probe 2>/dev/null; then echo rw; fi; "
+                     "echo died-of-reasons > /dev/termination-log; exit 7"],
+            read_only_root_fs=True)],
+    ))
+    s = wait_status(rt, st.id, PodStatus.EXITED)
+    assert s.containers[0].exit_code == 7
+    out = rt.get_logs(st.id)
+    assert "rw" not in out.split()  # rootfs really read-only
+    assert s.containers[0].message == "died-of-reasons"
