@@ -1712,6 +1712,10 @@ class ProcessRuntime(Runtime):
                     "run_as_gid": c.run_as_gid,
                     "run_as_non_root": c.run_as_non_root,
                     "read_only_root_fs": c.read_only_root_fs,
+                    "termination_message_path": c.termination_message_path,
+                    "termination_message_policy":
+                        c.termination_message_policy,
+                    "image_pull_policy": c.image_pull_policy,
                     "working_dir": c.working_dir,
                 }
                 for c in inst.params.containers
@@ -1783,6 +1787,12 @@ class ProcessRuntime(Runtime):
                         run_as_non_root=c.get("run_as_non_root", False),
                         read_only_root_fs=c.get("read_only_root_fs",
                                                 False),
+                        termination_message_path=c.get(
+                            "termination_message_path",
+                            "/dev/termination-log"),
+                        termination_message_policy=c.get(
+                            "termination_message_policy", "File"),
+                        image_pull_policy=c.get("image_pull_policy", ""),
                         working_dir=c.get("working_dir", ""),
                     )
                     for c in rec.get("container_specs", [])
