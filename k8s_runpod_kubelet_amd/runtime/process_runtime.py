@@ -1642,6 +1642,15 @@ class ProcessRuntime(Runtime):
             "active_deadline_s": inst.params.active_deadline_s,
             "termination_grace_s": inst.params.termination_grace_s,
             "resolv_conf": inst.params.resolv_conf,
+            # pod-level launch context: a crash-restart after kubelet
+            # restart must rebuild the same mounts/identity/env
+            "hostname": inst.params.hostname,
+            "pod_env": inst.params.env,
+            "fs_group": inst.params.fs_group,
+            "host_aliases": [[ip, list(names)]
+                             for ip, names in inst.params.host_aliases],
+            "volumes": {n: dataclasses.asdict(v)
+                        for n, v in inst.params.volumes.items()},
             "deadline_exceeded": inst.deadline_exceeded,
             "image_mode": inst.image_mode,
             "containers": [
@@ -1716,6 +1725,8 @@ class ProcessRuntime(Runtime):
                     "termination_message_policy":
                         c.termination_message_policy,
                     "image_pull_policy": c.image_pull_policy,
+                    "volume_mounts": [dataclasses.asdict(vm)
+                                      for vm in c.volume_mounts],
                     "working_dir": c.working_dir,
                 }
                 for c in inst.params.containers
@@ -1753,7 +1764,7 @@ class ProcessRuntime(Runtime):
             except (OSError, json.JSONDecodeError):
                 continue
             from .probes import ProbeSpec
-            from .types import ContainerSpec
+            from .types import ContainerSpec, VolumeMount, VolumeSource
 
             params = DeployParams(
                 pod_key=rec["pod_key"],
@@ -1764,6 +1775,13 @@ class ProcessRuntime(Runtime):
                 restart_policy=rec.get("restart_policy", "Never"),
                 active_deadline_s=rec.get("active_deadline_s", 0.0),
                 resolv_conf=rec.get("resolv_conf", ""),
+                hostname=rec.get("hostname", ""),
+                env=rec.get("pod_env", {}) or {},
+                fs_group=rec.get("fs_group", -1),
+                host_aliases=[(ip, list(names)) for ip, names in
+                              rec.get("host_aliases", []) or []],
+                volumes={n: VolumeSource(**v) for n, v in
+                         (rec.get("volumes", {}) or {}).items()},
                 termination_grace_s=rec.get("termination_grace_s",
                                             TERM_GRACE_S),
                 containers=[
@@ -1793,6 +1811,8 @@ class ProcessRuntime(Runtime):
                         termination_message_policy=c.get(
                             "termination_message_policy", "File"),
                         image_pull_policy=c.get("image_pull_policy", ""),
+                        volume_mounts=[VolumeMount(**vm) for vm in
+                                       c.get("volume_mounts", []) or []],
                         working_dir=c.get("working_dir", ""),
                     )
                     for c in rec.get("container_specs", [])

@@ -1458,3 +1458,64 @@ def test_termination_message_writable_with_ro_rootfs(image_runtime):
     out = rt.get_logs(st.id)
     assert "rw" not in out.split()  # rootfs really read-only
     assert s.containers[0].message == "died-of-reasons"
+
+
+def test_adopted_crash_restart_keeps_volumes(synthetic_ledger,
+                                             tmp_state_dir, image_store):
+    """Kubelet restart + crash-restarting image pod: the relaunched
+    container still gets its volume mounts (the pod-level launch context
+    — volumes, volumeMounts, hostname — must survive persistence)."""
+    from k8s_runpod_kubelet_amd.runtime.types import (VolumeMount,
+                                                      VolumeSource)
+
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                        enable_cgroups=False, image_store=image_store)
+    st = rt.deploy(DeployParams(
+        pod_key="default-adv", name="adv", restart_policy="Always",
+        hostname="adv-host",
+        volumes={"scratch": VolumeSource(kind="emptyDir")},
+        containers=[ContainerSpec(
+            name="main", image="example/app:v1",
+            command=["/bin/sh", "-c",
+                     "echo run >> /scratch/runs; "
+                     "while read -r l; do n=$((${n:-0}+1)); done "
+                     "< /scratch/runs; echo runs=$n; exit 1"],
+            volume_mounts=[VolumeMount(name="scratch",
+                                       mount_path="/scratch")])],
+    ))
+    iid = st.id
+    import time as _t
+
+    deadline = _t.time() + 10
+    while _t.time() < deadline:
+        if "runs=1" in rt.get_logs(iid):
+            break
+        _t.sleep(0.05)
+    assert "runs=1" in rt.get_logs(iid)
+    rt.close()  # kubelet "crashes" (pods live on / restart pending)
+
+    rt2 = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                         enable_cgroups=False, image_store=image_store)
+    try:
+        rt2.adopt_persisted()
+        inst = rt2._instances[iid]
+        assert inst.params.volumes["scratch"].kind == "emptyDir"
+        assert inst.params.containers[0].volume_mounts[0].name == "scratch"
+        assert inst.params.hostname == "adv-host"
+        # the next crash-restart runs with the volume mounted and the
+        # SAME per-pod volume content (runs file accumulates)
+        deadline = _t.time() + 15
+        ok = False
+        while _t.time() < deadline:
+            logs = rt2.get_logs(iid)
+            if "runs=2" in logs or "runs=3" in logs:
+                ok = True
+                break
+            _t.sleep(0.1)
+        assert ok, rt2.get_logs(iid)
+    finally:
+        rt2.terminate(iid, grace_override_s=0.0)
+        import time as _t2
+
+        _t2.sleep(0.3)
+        rt2.close()
