@@ -129,13 +129,34 @@ def waiter():
     return wait_until
 
 
+_port_counter = [0]
+
+
 def free_port() -> int:
-    """OS-assigned free TCP port (bound then released — a fixed port in a
-    test can collide with leftover listeners or parallel workers)."""
+    """A free TCP port from a per-xdist-worker range. A plain
+    bind(0)+close is racy across parallel workers (another worker can
+    grab the port before the test's process binds it — seen as a
+    port-gated-readiness flake); disjoint per-worker ranges remove the
+    cross-worker race, and the bind probe skips anything else living
+    in the range."""
+    import os
     import socket
 
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    port = s.getsockname()[1]
-    s.close()
-    return port
+    worker = os.environ.get("PYTEST_XDIST_WORKER", "gw0")
+    try:
+        widx = int(worker.lstrip("gw") or 0)
+    except ValueError:
+        widx = 0
+    base = 21000 + 1000 * (widx % 40)
+    for _ in range(1000):
+        _port_counter[0] += 1
+        port = base + (_port_counter[0] % 1000)
+        probe = socket.socket()
+        try:
+            probe.bind(("127.0.0.1", port))
+        except OSError:
+            continue
+        finally:
+            probe.close()
+        return port
+    raise RuntimeError("no free port in worker range")
