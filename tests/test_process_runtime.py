@@ -99,8 +99,12 @@ def test_subscriber_push(process_runtime):
     process_runtime.subscribe(lambda iid: events.append(iid))
     st = process_runtime.deploy(params(args=["--run-for", "0.05"]))
     wait_status(process_runtime, st.id, PodStatus.EXITED)
-    time.sleep(0.05)
-    assert events.count(st.id) >= 2  # ready + exit
+    # the exit notification is delivered by the event thread slightly
+    # after the status flips — wait, don't sleep a fixed beat
+    from tests.conftest import wait_until
+
+    assert wait_until(lambda: events.count(st.id) >= 2, timeout_s=5), \
+        events  # ready + exit
 
 
 def test_not_found(process_runtime):
