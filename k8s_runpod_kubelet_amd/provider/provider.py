@@ -99,6 +99,7 @@ class Provider:
         # (main.go:172-177).
         self.recorder = None
         self.meminfo_reader = None  # injectable memory.available source
+        self._port_rechecks: Dict[str, int] = {}
         self._eviction = None
         # Set by the PodController: re-enqueue a deleted pod ("ns/name")
         # when its instance turns terminal, completing the API delete.
@@ -645,6 +646,17 @@ class Provider:
             except Exception:
                 log.exception("status sync failed", extra={"pod": key})
 
+    def _schedule_port_recheck(self, key: str, delay_s: float = 0.25,
+                               max_rechecks: int = 40) -> None:
+        with self._pods_lock:
+            n = self._port_rechecks.get(key, 0)
+            if n >= max_rechecks:
+                return
+            self._port_rechecks[key] = n + 1
+        t = threading.Timer(delay_s, self._sync_pod_status, args=(key,))
+        t.daemon = True
+        t.start()
+
     def _sync_pod_status(self, key: str) -> None:
         with self._pods_lock:
             pod = self._pods.get(key)
@@ -662,6 +674,16 @@ class Provider:
             return
 
         ports_exposed = check_ports_exposed(info.requested_ports, detailed.port_mappings)
+        if ports_exposed:
+            with self._pods_lock:
+                self._port_rechecks.pop(key, None)
+        if (not ports_exposed and info.requested_ports
+                and detailed.desired_status == PodStatus.RUNNING):
+            # Port-gated pod whose listen socket is not visible yet: the
+            # push event fires once (at container-ready), and /proc/net
+            # scans can transiently miss entries — re-check shortly
+            # instead of waiting for the 30 s reconcile tick.
+            self._schedule_port_recheck(key)
         # Readiness also requires container readiness (our event signal).
         containers_ready = bool(detailed.containers) and all(
             c.ready for c in detailed.containers

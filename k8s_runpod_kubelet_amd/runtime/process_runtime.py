@@ -1976,6 +1976,18 @@ class ProcessRuntime(Runtime):
                     t.daemon = True
                     self._deadline_timers[inst.id] = t
                     t.start()
+                if inst.desired_status == PodStatus.TERMINATING and any(
+                        c.exit_code is None for c in inst.containers):
+                    # the previous kubelet died mid-grace: its SIGKILL
+                    # timer died with it — re-arm with the full grace
+                    # (elapsed time is unknown; a TERM-immune pid-1 must
+                    # not outlive the window forever)
+                    t = threading.Timer(
+                        max(0.1, params.termination_grace_s),
+                        self._force_kill, args=(inst.id,))
+                    t.daemon = True
+                    self._kill_timers[inst.id] = t
+                    t.start()
             adopted.append(inst.id)
         if adopted:
             log.info("adopted persisted instances", extra={"count": len(adopted)})
@@ -2016,9 +2028,13 @@ def _resolve_in_tree(rootfs: str, argv0: str) -> str:
 
 def _listening_tcp_inodes() -> dict:
     """socket-inode → local port for LISTEN-state TCP sockets from
-    /proc/net/tcp{,6} (state 0A, inode field 9)."""
+    /proc/net/tcp{,6} (state 0A, inode field 9). Two passes, unioned:
+    the kernel's seq_file iteration can skip entries while the socket
+    table mutates under load, and a missed entry here reads as
+    "pod port not exposed"."""
     by_inode = {}
-    for path in ("/proc/net/tcp", "/proc/net/tcp6"):
+    for path in ("/proc/net/tcp", "/proc/net/tcp6",
+                 "/proc/net/tcp", "/proc/net/tcp6"):
         try:
             with open(path, "r", encoding="ascii") as fh:
                 next(fh, None)

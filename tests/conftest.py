@@ -70,6 +70,7 @@ def process_runtime(synthetic_ledger, tmp_state_dir):
 
     rt = ProcessRuntime(synthetic_ledger, tmp_state_dir, enable_cgroups=False)
     yield rt
+    drain_runtime(rt)
     rt.close()
 
 
@@ -130,6 +131,32 @@ def waiter():
 
 
 _port_counter = [0]
+
+
+def drain_runtime(rt, timeout_s: float = 5.0) -> None:
+    """Kill every live instance (grace 0) and wait for the processes to
+    die. Test teardown MUST do this before close(): close() cancels kill
+    timers (kubelet-crash semantics — the next kubelet re-arms them), so
+    a bare close() after terminate() leaks TERM-immune pid-1 workloads
+    (this once leaked 58 infinite log writers totalling ~115 GB of
+    deleted-but-open files)."""
+    import time as _time
+
+    try:
+        for inst in rt.list_instances():
+            try:
+                rt.terminate(inst.id, grace_override_s=0.0)
+            except Exception:
+                pass
+        deadline = _time.time() + timeout_s
+        while _time.time() < deadline:
+            alive = [i for i in rt.list_instances()
+                     if any(c.exit_code is None for c in i.containers)]
+            if not alive:
+                break
+            _time.sleep(0.05)
+    except Exception:
+        pass
 
 
 def free_port() -> int:

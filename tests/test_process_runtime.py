@@ -432,8 +432,17 @@ def test_log_rotation_caps_chatty_containers(synthetic_ledger,
         assert rotated, "log never rotated"
         # live file stays bounded (cap + a few seconds of writes)
         assert logp.stat().st_size < 2 * 64 * 1024 + 512 * 1024
-        rt.terminate(st.id)
+        rt.terminate(st.id, grace_override_s=0.0)
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            if all(c.exit_code is not None
+                   for c in rt.get_detailed_status(st.id).containers):
+                break
+            time.sleep(0.05)
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(rt)
         rt.close()
 
 
@@ -473,3 +482,48 @@ def test_ephemeral_container_adoption(synthetic_ledger, tmp_state_dir):
     finally:
         rt2.close()
         rt.close()
+
+
+def test_adoption_rearms_kill_timer_for_terminating(synthetic_ledger,
+                                                    tmp_state_dir):
+    """Kubelet dies mid-grace while a TERM-immune pid-1 ignores SIGTERM:
+    the next kubelet re-arms the SIGKILL timer on adoption — the
+    container must not outlive the grace window forever."""
+    import os
+
+    rt = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                        enable_cgroups=False)
+    st = rt.deploy(DeployParams(
+        pod_key="default-immune", name="immune",
+        termination_grace_s=1.0,
+        containers=[ContainerSpec(
+            name="main", command=["/bin/sh"],
+            args=["-c", "trap '' TERM; while :; do sleep 1; done"])],
+    ))
+    wait_status(rt, st.id, PodStatus.RUNNING)
+    pid = rt.get_detailed_status(st.id).containers[0].pid
+    rt.terminate(st.id)  # TERM ignored; grace timer armed
+    rt.close()           # kubelet "dies": timer cancelled
+    assert os.path.exists(f"/proc/{pid}")
+
+    rt2 = ProcessRuntime(synthetic_ledger, tmp_state_dir,
+                         enable_cgroups=False)
+    try:
+        rt2.adopt_persisted()
+        deadline = time.time() + 10
+        gone = False
+        while time.time() < deadline:
+            if not os.path.exists(f"/proc/{pid}") or \
+                    open(f"/proc/{pid}/stat").read().rsplit(
+                        ") ", 1)[-1].startswith("Z"):
+                gone = True
+                break
+            time.sleep(0.1)
+        assert gone, "TERM-immune container survived re-armed grace"
+        s = wait_status(rt2, st.id, PodStatus.TERMINATED)
+        assert s.desired_status == PodStatus.TERMINATED
+    finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(rt2)
+        rt2.close()
