@@ -7,6 +7,7 @@ import re
 import signal
 import subprocess
 import sys
+import tempfile
 import time
 import urllib.request
 
@@ -14,10 +15,15 @@ REPO = __file__.rsplit("/tests/", 1)[0]
 
 
 def test_cli_fake_apiserver_demo():
+    # isolated state dir: the CLI default (/var/lib/amd-virtual-kubelet)
+    # accumulates adopted instances across runs (kubelet semantics) —
+    # at ~100 leftovers the adoption/import work ate this test's startup
+    # budget
+    state = tempfile.mkdtemp(prefix="amdvk-clidemo-")
     p = subprocess.Popen(
         [sys.executable, "-m", "k8s_runpod_kubelet_amd.cli",
          "--gpu-count-override", "8", "--fake-apiserver",
-         "--listen-port", "0",
+         "--listen-port", "0", "--state-dir", state,
          "--health-server-address", "127.0.0.1:0"],
         stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, cwd=REPO,
     )

@@ -31,6 +31,9 @@ def stack(tmp_state_dir):
     s.runtime.enable_cgroups = False
     s.start(serve_http=False)
     yield s, kube
+    from tests.conftest import drain_runtime
+
+    drain_runtime(s.runtime)
     s.stop()
 
 

@@ -129,6 +129,9 @@ def test_eviction_end_to_end_stack(tmp_state_dir):
                    if e.get("reason") == "Evicted"]
         assert evs and "memory" in evs[0]["message"]
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(stack.runtime)
         stack.stop()
 
 
@@ -183,6 +186,9 @@ def test_gpu_failure_evicts_bound_pods(tmp_state_dir):
             assert any(e.get("reason") == "GPUFailure"
                        for e in kube.events.objects.values())
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(stack.runtime)
         stack.stop()
 
 
@@ -247,6 +253,9 @@ def test_eviction_kills_real_process_and_frees_gpu(tmp_state_dir):
             lambda: (kube.get_pod("default", "next").get("status") or
                      {}).get("phase") == "Running", timeout_s=10)
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(stack.runtime)
         stack.stop()
 
 

@@ -339,6 +339,9 @@ def test_image_pod_end_to_end_stack(tmp_state_dir, app_bin, tmp_path):
 
         assert wait_until(gone, timeout_s=20)
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(stack.runtime)
         stack.stop()
 
 
@@ -825,6 +828,9 @@ def test_missing_image_pending_until_imported(synthetic_ledger,
 
         assert wait_until(ready, timeout_s=20) is not None
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(stack.runtime)
         stack.stop()
 
 
@@ -1300,6 +1306,9 @@ int main(void) {
         assert stack.runtime.image_store.resolve(
             "example/secured:v1") is not None
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(stack.runtime)
         stack.stop()
         srv.stop()
 
@@ -1359,6 +1368,9 @@ def test_pulled_event_distinguishes_pull_from_cache(tmp_state_dir, app_bin,
             lambda: any("already present" in m for m in msgs()),
             timeout_s=10)
     finally:
+        from tests.conftest import drain_runtime
+
+        drain_runtime(stack.runtime)
         stack.stop()
         srv.stop()
 
