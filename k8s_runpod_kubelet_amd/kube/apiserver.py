@@ -16,6 +16,8 @@ import logging
 import os
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+from ..utils.httpserver import QuietThreadingHTTPServer
 from typing import Optional
 from urllib.parse import parse_qs, unquote, urlparse
 
@@ -183,7 +185,7 @@ class KubeletApiServer:
                 else:
                     self._respond(404, b"not found")
 
-        self._server = ThreadingHTTPServer((self.internal_ip, self.port), Handler)
+        self._server = QuietThreadingHTTPServer((self.internal_ip, self.port), Handler)
         self.port = self._server.server_address[1]
         self._thread = threading.Thread(
             target=self._server.serve_forever, name="kubelet-api", daemon=True

@@ -22,6 +22,8 @@ import logging
 import re
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+from ..utils.httpserver import QuietThreadingHTTPServer
 from typing import Any, Dict, Optional, Tuple
 from urllib.parse import parse_qs, urlparse
 
@@ -280,7 +282,7 @@ class FakeApiServer:
                 except Exception as exc:  # noqa: BLE001
                     self._err(exc)
 
-        self._server = ThreadingHTTPServer((self.host, self.port), Handler)
+        self._server = QuietThreadingHTTPServer((self.host, self.port), Handler)
         self.port = self._server.server_address[1]
         self._thread = threading.Thread(
             target=self._server.serve_forever, name="fake-apiserver", daemon=True)

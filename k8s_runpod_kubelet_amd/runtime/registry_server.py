@@ -27,6 +27,8 @@ import threading
 import uuid as uuidlib
 from pathlib import Path
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+from ..utils.httpserver import QuietThreadingHTTPServer
 from typing import Optional
 
 from .oci import ImageStore, _read_json, normalize_ref
@@ -376,7 +378,7 @@ class RegistryServer:
                     return
                 return self._respond(404, b'{"errors":[]}')
 
-        self._server = ThreadingHTTPServer((self.host, self.port), Handler)
+        self._server = QuietThreadingHTTPServer((self.host, self.port), Handler)
         self.port = self._server.server_address[1]
         self._thread = threading.Thread(target=self._server.serve_forever,
                                         name="registry-server", daemon=True)

@@ -12,6 +12,8 @@ import sys
 import threading
 import traceback
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+from ..utils.httpserver import QuietThreadingHTTPServer
 from typing import Callable, Optional
 
 from . import metrics
@@ -126,7 +128,7 @@ class HealthServer:
                 else:
                     self._respond(404, b"not found")
 
-        self._server = ThreadingHTTPServer((self.host, self.port), Handler)
+        self._server = QuietThreadingHTTPServer((self.host, self.port), Handler)
         self.port = self._server.server_address[1]
         self._thread = threading.Thread(
             target=self._server.serve_forever, name="health-server", daemon=True
