@@ -32,6 +32,9 @@ def http_stack(tmp_state_dir):
     stack.runtime.enable_cgroups = False
     stack.start(serve_http=False)
     yield stack, srv, client, cfg
+    from tests.conftest import drain_runtime
+
+    drain_runtime(stack.runtime)
     stack.stop()
     client.close()
     srv.stop()
