@@ -354,3 +354,29 @@ def test_kubelet_api_requires_auth_from_non_loopback():
             f"http://127.0.0.1:{srv.port}/pods", timeout=5).status == 200
     finally:
         srv.stop()
+
+
+def test_quiet_server_swallows_client_disconnects(capsys):
+    """Client hang-ups (kubectl Ctrl-C mid-stream) must not dump
+    socketserver tracebacks to stderr; real errors still do."""
+    from k8s_runpod_kubelet_amd.utils.httpserver import (
+        QuietThreadingHTTPServer,
+    )
+
+    srv = QuietThreadingHTTPServer(("127.0.0.1", 0),
+                                   __import__("http.server",
+                                              fromlist=["x"]
+                                              ).BaseHTTPRequestHandler)
+    try:
+        try:
+            raise ConnectionResetError(104, "peer reset")
+        except ConnectionResetError:
+            srv.handle_error(None, ("127.0.0.1", 9999))
+        assert "Traceback" not in capsys.readouterr().err
+        try:
+            raise ValueError("real bug")
+        except ValueError:
+            srv.handle_error(None, ("127.0.0.1", 9999))
+        assert "ValueError" in capsys.readouterr().err
+    finally:
+        srv.server_close()
