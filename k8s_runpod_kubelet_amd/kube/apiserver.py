@@ -135,6 +135,7 @@ class KubeletApiServer:
                     namespace, pod = parts[1], parts[2]
                     container = parts[3] if len(parts) > 3 else ""
                     tail = int(query.get("tailLines", ["-1"])[0])
+                    limit_bytes = int(query.get("limitBytes", ["0"])[0])
                     follow = query.get(
                         "follow", ["false"])[0].lower() in ("true", "1")
                     previous = query.get(
@@ -145,7 +146,10 @@ class KubeletApiServer:
                     text = outer.provider.get_container_logs(
                         namespace, pod, container, tail, previous=previous
                     )
-                    self._respond(200, text.encode())
+                    body = text.encode()
+                    if limit_bytes > 0:
+                        body = body[:limit_bytes]  # kubectl logs --limit-bytes
+                    self._respond(200, body)
                 elif parts and parts[0] == "pods":
                     pods = outer.provider.get_pods()
                     body = json.dumps(

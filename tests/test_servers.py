@@ -380,3 +380,34 @@ def test_quiet_server_swallows_client_disconnects(capsys):
         assert "ValueError" in capsys.readouterr().err
     finally:
         srv.server_close()
+
+
+def test_container_logs_limit_bytes(provider, process_runtime):
+    prov, _, kube = provider
+    prov.runtime = process_runtime
+    pod = make_pod("lim1", command=["/bin/sh"],
+                   args=["-c", "printf 'abcdefghij%.0s' $(seq 100)"])
+    kube.create_pod("default", pod)
+    prov.create_pod(kube.get_pod("default", "lim1"))
+    srv = KubeletApiServer(prov, "127.0.0.1", 0)
+    srv.start()
+    try:
+        import time
+
+        base = f"http://127.0.0.1:{srv.port}"
+        deadline = time.time() + 10
+        full = b""
+        while time.time() < deadline:
+            full = urllib.request.urlopen(
+                f"{base}/containerLogs/default/lim1/main",
+                timeout=5).read()
+            if len(full) >= 1000:
+                break
+            time.sleep(0.1)
+        assert len(full) == 1000
+        cut = urllib.request.urlopen(
+            f"{base}/containerLogs/default/lim1/main?limitBytes=64",
+            timeout=5).read()
+        assert cut == full[:64]
+    finally:
+        srv.stop()
