@@ -150,6 +150,15 @@ class KubeletApiServer:
                     if limit_bytes > 0:
                         body = body[:limit_bytes]  # kubectl logs --limit-bytes
                     self._respond(200, body)
+                elif parts and parts[0] == "runningpods":
+                    # kubelet debug endpoint: pods with live containers
+                    # only (kubectl get --raw /api/v1/nodes/<n>/proxy/
+                    # runningpods/ parity)
+                    pods = outer.provider.get_running_pods()
+                    body = json.dumps(
+                        {"kind": "PodList", "apiVersion": "v1",
+                         "items": pods}).encode()
+                    self._respond(200, body, "application/json")
                 elif parts and parts[0] == "pods":
                     pods = outer.provider.get_pods()
                     body = json.dumps(

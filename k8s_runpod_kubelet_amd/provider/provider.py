@@ -430,6 +430,17 @@ class Provider:
         with self._pods_lock:
             return list(self._pods.values())
 
+    def get_running_pods(self) -> List[Dict[str, Any]]:
+        """Pods whose instance is live right now (the kubelet's
+        /runningpods debug endpoint reports runtime truth, not the
+        apiserver's possibly-stale phase)."""
+        with self._pods_lock:
+            items = [(dict(p), self._pod_status.get(k))
+                     for k, p in self._pods.items()]
+        return [p for p, info in items
+                if info is not None and info.instance_id
+                and info.status == PodStatus.RUNNING]
+
     def get_pod_status(self, namespace: str, name: str) -> Optional[Dict[str, Any]]:
         """GetPodStatus (reference kubelet.go:670-696): live port re-check for
         RUNNING pods with requested ports."""

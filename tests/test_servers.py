@@ -411,3 +411,30 @@ def test_container_logs_limit_bytes(provider, process_runtime):
         assert cut == full[:64]
     finally:
         srv.stop()
+
+
+def test_runningpods_endpoint(provider, process_runtime):
+    prov, _, kube = provider
+    prov.runtime = process_runtime
+    pod = make_pod("rp1", command=["podworker"], args=["--hold"])
+    kube.create_pod("default", pod)
+    prov.create_pod(kube.get_pod("default", "rp1"))
+    srv = KubeletApiServer(prov, "127.0.0.1", 0)
+    srv.start()
+    try:
+        import time
+
+        base = f"http://127.0.0.1:{srv.port}"
+        deadline = time.time() + 10
+        names = []
+        while time.time() < deadline:
+            body = json.loads(urllib.request.urlopen(
+                f"{base}/runningpods/", timeout=5).read())
+            names = [i["metadata"]["name"] for i in body["items"]]
+            if "rp1" in names:
+                break
+            time.sleep(0.1)
+        assert "rp1" in names
+        assert body["kind"] == "PodList"
+    finally:
+        srv.stop()
