@@ -169,6 +169,19 @@ class KubeletApiServer:
                     summary = outer.provider.get_stats_summary()
                     self._respond(200, json.dumps(summary).encode(),
                                   "application/json")
+                elif parts and parts[0] == "configz":
+                    # kubelet /configz parity: the live provider config
+                    # (tokens redacted)
+                    import dataclasses as _dc
+
+                    cfg = outer.provider.config
+                    data = _dc.asdict(cfg)
+                    for k in list(data):
+                        if "token" in k:
+                            data[k] = "***" if data[k] else ""
+                    self._respond(200, json.dumps(
+                        {"kubeletconfig": data}).encode(),
+                        "application/json")
                 elif parts and parts[0] == "healthz":
                     self._respond(200, b"ok")
                 else:

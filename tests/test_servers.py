@@ -438,3 +438,19 @@ def test_runningpods_endpoint(provider, process_runtime):
         assert body["kind"] == "PodList"
     finally:
         srv.stop()
+
+
+def test_configz_endpoint_redacts_tokens(provider):
+    prov, _, _ = provider
+    prov.config.admin_token = "sekrit"
+    srv = KubeletApiServer(prov, "127.0.0.1", 0)
+    srv.start()
+    try:
+        body = json.loads(urllib.request.urlopen(
+            f"http://127.0.0.1:{srv.port}/configz", timeout=5).read())
+        cfg = body["kubeletconfig"]
+        assert cfg["node_name"] == prov.config.node_name
+        assert cfg["admin_token"] == "***"
+        assert "sekrit" not in json.dumps(body)
+    finally:
+        srv.stop()
